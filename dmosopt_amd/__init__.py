@@ -1,0 +1,30 @@
+"""dmosopt_amd — MI355X-native multi-objective adaptive surrogate optimization.
+
+A from-scratch rebuild of the capabilities of dmosopt (MO-ASMO) designed for
+AMD Instinct MI355X: PyTorch-ROCm tensors, hand-written gfx950 HIP kernels
+for the hot population/GP math, RCCL collectives over xGMI for distribution.
+"""
+
+__version__ = "0.1.0"
+
+from dmosopt_amd.datatypes import ParameterSpace  # noqa: F401
+
+sopt_dict = {}
+
+
+def run(dopt_params, time_limit=None, feasible=True, return_features=False,
+        return_constraints=False, spawn_workers=False, sequential_spawn=False,
+        spawn_startup_wait=None, spawn_executable=None, spawn_args=[],
+        nprocs_per_worker=1, collective_mode="gather", verbose=True,
+        worker_debug=False):
+    """Run a distributed optimization. See dmosopt_amd.api.run."""
+    from dmosopt_amd.api import run as _run
+
+    return _run(
+        dopt_params,
+        time_limit=time_limit,
+        feasible=feasible,
+        return_features=return_features,
+        return_constraints=return_constraints,
+        verbose=verbose,
+    )

@@ -1,0 +1,179 @@
+"""Quality indicators: IGD, Hypervolume, HVI selection, PopulationDiversity.
+
+Parity with reference indicators.py:208-335. Numpy at the interface (these
+feed host-side termination logic); the heavy metric math (crowding, pareto
+rank) dispatches through dmosopt_amd.ops so it runs on-device when the
+inputs live there.
+"""
+
+from __future__ import annotations
+
+from abc import abstractmethod
+from typing import Optional
+
+import numpy as np
+import torch
+
+from dmosopt_amd import ops
+from dmosopt_amd.hv.exact import HyperVolumeBoxDecomposition
+from dmosopt_amd.normalization import PreNormalization
+
+
+def _to_numpy(a):
+    if isinstance(a, torch.Tensor):
+        return a.detach().cpu().numpy()
+    return np.asarray(a)
+
+
+def crowding_distance_metric(Y) -> np.ndarray:
+    t = Y if isinstance(Y, torch.Tensor) else torch.as_tensor(np.asarray(Y, dtype=np.float64))
+    return _to_numpy(ops.crowding_distance(t))
+
+
+def euclidean_distance_metric(Y) -> np.ndarray:
+    t = Y if isinstance(Y, torch.Tensor) else torch.as_tensor(np.asarray(Y, dtype=np.float64))
+    return _to_numpy(ops.euclidean_distance_metric(t))
+
+
+def pareto_rank_np(Y) -> np.ndarray:
+    t = Y if isinstance(Y, torch.Tensor) else torch.as_tensor(np.asarray(Y, dtype=np.float64))
+    return _to_numpy(ops.pareto_rank(t))
+
+
+def at_least_2d_array(x, extend_as="row"):
+    if x is None:
+        return x
+    x = np.asarray(x)
+    if not isinstance(x, np.ndarray):
+        x = np.array([x])
+    if x.ndim == 1:
+        x = x[None, :] if extend_as == "row" else x[:, None]
+    return x
+
+
+def derive_ideal_and_nadir_from_pf(pf, ideal=None, nadir=None):
+    if pf is not None:
+        if ideal is None:
+            ideal = np.min(pf, axis=0)
+        if nadir is None:
+            nadir = np.max(pf, axis=0)
+    return ideal, nadir
+
+
+class SlidingWindow(list):
+    def __init__(self, size=None) -> None:
+        super().__init__()
+        self.size = size
+
+    def append(self, entry):
+        super().append(entry)
+        if self.size is not None:
+            while len(self) > self.size:
+                self.pop(0)
+
+    def is_full(self):
+        return self.size == len(self)
+
+
+class Indicator(PreNormalization):
+    def __init__(self, **kwargs):
+        super().__init__(**kwargs)
+        self.default_if_empty = 0.0
+
+    def do(self, F, *args, **kwargs):
+        F = _to_numpy(F)
+        if F.ndim == 1:
+            F = F[None, :]
+        if len(F) == 0:
+            return self.default_if_empty
+        F = self.normalization.forward(F)
+        return self._do(F, *args, **kwargs)
+
+    @abstractmethod
+    def _do(self, F, *args, **kwargs):
+        ...
+
+
+class IGD(Indicator):
+    """Inverted generational distance to a known pareto front."""
+
+    def __init__(self, pf, zero_to_one=False, ideal=None, nadir=None, norm_by_dist=False, **kwargs):
+        pf = at_least_2d_array(pf, extend_as="row")
+        ideal, nadir = derive_ideal_and_nadir_from_pf(pf, ideal=ideal, nadir=nadir)
+        super().__init__(zero_to_one=zero_to_one, ideal=ideal, nadir=nadir, **kwargs)
+        self.norm_by_dist = norm_by_dist
+        self.pf = self.normalization.forward(pf)
+
+    def _do(self, F):
+        norm = 1.0
+        if self.norm_by_dist:
+            assert self.ideal is not None and self.nadir is not None
+            norm = self.nadir - self.ideal
+        diff = (self.pf[:, None, :] - F[None, :, :]) / norm
+        D = np.sqrt((diff**2).sum(axis=2))
+        return float(np.mean(np.min(D, axis=1)))
+
+
+class Hypervolume(Indicator):
+    def __init__(self, ref_point=None, pf=None, nds=False, norm_ref_point=True,
+                 ideal=None, nadir=None, **kwargs):
+        pf = at_least_2d_array(pf, extend_as="row")
+        ideal, nadir = derive_ideal_and_nadir_from_pf(pf, ideal=ideal, nadir=nadir)
+        super().__init__(ideal=ideal, nadir=nadir, **kwargs)
+        self.nds = nds
+        if ref_point is None and pf is not None:
+            ref_point = pf.max(axis=0)
+        if norm_ref_point:
+            ref_point = self.normalization.forward(ref_point)
+        self.ref_point = ref_point
+        assert self.ref_point is not None, "Hypervolume needs a reference point"
+
+    def _do(self, F):
+        if self.nds:
+            rank = pareto_rank_np(F)
+            F = np.copy(F[rank == 0, :])
+        return HyperVolumeBoxDecomposition(self.ref_point).compute_hypervolume(F)
+
+
+class HypervolumeImprovement(Indicator):
+    def __init__(self, ref_point=None, pf=None, nds=False, norm_ref_point=True,
+                 ideal=None, nadir=None, **kwargs):
+        pf = at_least_2d_array(pf, extend_as="row")
+        ideal, nadir = derive_ideal_and_nadir_from_pf(pf, ideal=ideal, nadir=nadir)
+        super().__init__(ideal=ideal, nadir=nadir, **kwargs)
+        self.default_if_empty = []
+        self.nds = nds
+        if ref_point is None and pf is not None:
+            ref_point = pf.max(axis=0)
+        if norm_ref_point:
+            ref_point = self.normalization.forward(ref_point)
+        self.ref_point = ref_point
+        assert self.ref_point is not None
+
+    def _do(self, F, means, variances, k):
+        assert k > 0 and len(F) > 0
+        if self.nds:
+            rank = pareto_rank_np(F)
+            nd = np.argwhere(rank == 0).ravel()
+            if len(nd) > 0:
+                F = np.copy(F[nd, :])
+        hv = HyperVolumeBoxDecomposition(self.ref_point)
+        selection, _ = hv.select_candidates(F, _to_numpy(means), _to_numpy(variances), k)
+        assert len(selection) > 0
+        return np.asarray(selection, dtype=int)
+
+
+class PopulationDiversity(Indicator):
+    """Front-0 fraction + crowding-distance CV (indicators.py:316-335)."""
+
+    def _do(self, F, Y):
+        front_0 = np.argwhere(F.flat == 0)
+        diversity = len(front_0) / len(F[0])
+        D = crowding_distance_metric(_to_numpy(Y))
+        if len(front_0) > 1:
+            cd = D[front_0.flat]
+            mean = np.mean(cd)
+            cd_spread = np.std(cd) / mean if mean != 0 else 0.0
+        else:
+            cd_spread = 0
+        return diversity, cd_spread

@@ -1,0 +1,161 @@
+"""Op dispatch layer.
+
+Every op has a pure-PyTorch reference implementation (``torch_ref``) used on
+CPU and as the numerics oracle. On CUDA (= HIP/ROCm) tensors the hand-written
+gfx950 kernels from the in-tree extension ``dmosopt_amd._hipops`` are used.
+Per the framework contract, running on a GPU without the native extension is
+an ERROR (no silent eager fallback): build it with
+``python setup.py build_ext --inplace`` or ``__graft_entry__.build()``.
+
+Set ``DMOSOPT_AMD_FORCE_TORCH=1`` to force the torch path (debugging only).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+from . import torch_ref
+from .torch_ref import (  # re-exported torch-native helpers
+    anyclose,
+    filter_samples,
+    lexsort,
+    sbx_crossover_batch,
+    polynomial_mutation_batch,
+    tournament_selection,
+    tournament_prob_vector,
+)
+
+_FORCE_TORCH = os.environ.get("DMOSOPT_AMD_FORCE_TORCH", "0") == "1"
+
+_native = None
+_native_err: Optional[str] = None
+
+
+def _load_native():
+    global _native, _native_err
+    if _native is not None or _native_err is not None:
+        return _native
+    try:
+        from dmosopt_amd import _hipops  # built in-tree by setup.py
+
+        _native = _hipops
+    except ImportError as e:  # remember why, for the loud failure path
+        _native_err = str(e)
+    return _native
+
+
+def native_available() -> bool:
+    return _load_native() is not None
+
+
+def _use_native(*tensors: torch.Tensor) -> bool:
+    if _FORCE_TORCH:
+        return False
+    if not any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor)):
+        return False
+    native = _load_native()
+    if native is None:
+        raise RuntimeError(
+            "dmosopt_amd: tensors are on GPU but the native HIP extension "
+            "dmosopt_amd._hipops is not built (import error: "
+            f"{_native_err}). Build it with `python setup.py build_ext "
+            "--inplace` (PYTORCH_ROCM_ARCH=gfx950); refusing to fall back "
+            "to eager PyTorch on the GPU."
+        )
+    return True
+
+
+# ------------------------------------------------------------------ ranking
+def pareto_rank(Y: torch.Tensor) -> torch.Tensor:
+    if _use_native(Y):
+        return _native.pareto_rank(Y.contiguous().float())
+    return torch_ref.pareto_rank(Y)
+
+
+def dominance_degree_matrix(Y: torch.Tensor) -> torch.Tensor:
+    if _use_native(Y):
+        return _native.dominance_degree_matrix(Y.contiguous().float())
+    return torch_ref.dominance_degree_matrix(Y)
+
+
+def crowding_distance(Y: torch.Tensor) -> torch.Tensor:
+    if _use_native(Y):
+        return _native.crowding_distance(Y.contiguous().float()).to(Y.dtype)
+    return torch_ref.crowding_distance(Y)
+
+
+def euclidean_distance_metric(Y: torch.Tensor) -> torch.Tensor:
+    return torch_ref.euclidean_distance_metric(Y)
+
+
+def order_mo(
+    x: torch.Tensor,
+    y: torch.Tensor,
+    x_dists: Optional[List[torch.Tensor]] = None,
+    y_distance_metrics: Optional[List] = None,
+):
+    # rank+crowding use dispatched kernels internally via the metric calls
+    rank = pareto_rank(y)
+    y_dist_vals: List[torch.Tensor] = []
+    if y_distance_metrics:
+        for metric in y_distance_metrics:
+            if callable(metric):
+                y_dist_vals.append(metric(y))
+            elif metric == "crowding":
+                y_dist_vals.append(crowding_distance(y))
+            elif metric == "euclidean":
+                y_dist_vals.append(euclidean_distance_metric(y))
+            else:
+                raise RuntimeError(f"order_mo: unknown distance metric {metric}")
+    x_dist_vals = list(x_dists) if x_dists else []
+    keys = [-d for d in x_dist_vals] + [-d for d in y_dist_vals] + [rank.to(y.dtype)]
+    perm = lexsort(keys)
+    return perm, rank[perm], tuple(d[perm] for d in y_dist_vals)
+
+
+def remove_worst(
+    population_parm: torch.Tensor,
+    population_obj: torch.Tensor,
+    pop: int,
+    x_dists: Optional[List[torch.Tensor]] = None,
+    y_distance_metrics: Optional[List] = None,
+):
+    perm, rank, _ = order_mo(
+        population_parm, population_obj, x_dists=x_dists, y_distance_metrics=y_distance_metrics
+    )
+    perm = perm[:pop]
+    return population_parm[perm], population_obj[perm], rank[:pop], perm
+
+
+def get_duplicates(X: torch.Tensor, eps: float = 1e-16) -> torch.Tensor:
+    if _use_native(X):
+        return _native.get_duplicates(X.contiguous().float(), eps)
+    return torch_ref.get_duplicates(X, eps)
+
+
+def remove_duplicates(x: torch.Tensor, y: torch.Tensor, eps: float = 1e-16):
+    dup = get_duplicates(x, eps)
+    return x[~dup], y[~dup]
+
+
+__all__ = [
+    "pareto_rank",
+    "dominance_degree_matrix",
+    "crowding_distance",
+    "euclidean_distance_metric",
+    "order_mo",
+    "remove_worst",
+    "get_duplicates",
+    "remove_duplicates",
+    "lexsort",
+    "sbx_crossover_batch",
+    "polynomial_mutation_batch",
+    "tournament_selection",
+    "tournament_prob_vector",
+    "anyclose",
+    "filter_samples",
+    "native_available",
+]
