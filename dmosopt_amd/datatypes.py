@@ -37,6 +37,12 @@ class Struct:
     def __getitem__(self, key):
         return self.__dict__[key]
 
+    def __setitem__(self, key, val):
+        self.__dict__[key] = val
+
+    def items(self):
+        return self.__dict__.items()
+
     def __contains__(self, key):
         return key in self.__dict__
 
