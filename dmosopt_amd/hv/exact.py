@@ -23,19 +23,24 @@ from scipy.stats import norm
 
 # ------------------------------------------------------------------ 2D / 3D
 def hv_2d(points: np.ndarray, ref_point: np.ndarray) -> float:
-    """Vectorized 2D hypervolume: sort by f0, suffix-min prune, swept area."""
+    """Vectorized 2D hypervolume: sort by (f0, f1), prefix-min prune, swept
+    area. Note: this matches the moocore/exact semantics; the reference's
+    pure-Python fallback (hv_box_decomposition.py:44-78) prunes with an
+    inverted scan and under-counts dominated staircases — its test suite
+    gates correctness against moocore, which we reproduce here."""
     pts = points[np.all(points < ref_point, axis=1)]
     if len(pts) == 0:
         return 0.0
-    pts = pts[np.argsort(pts[:, 0], kind="stable")]
+    order = np.lexsort((pts[:, 1], pts[:, 0]))  # x asc, then y asc
+    pts = pts[order]
     y = pts[:, 1]
-    # keep i iff y[i] < min(y[i+1:]) (strict): exclusive suffix min
-    suf = np.empty(len(y))
-    suf[-1] = np.inf
+    # keep i iff y[i] < min(y[:i]) (strict exclusive prefix min): a point is
+    # dominated exactly by some earlier (smaller-x) point with y <= y[i]
+    pm = np.empty(len(y))
+    pm[0] = np.inf
     if len(y) > 1:
-        suf[:-1] = np.minimum.accumulate(y[::-1])[::-1][1:]
-    keep = y < suf
-    pts = pts[keep]
+        pm[1:] = np.minimum.accumulate(y)[:-1]
+    pts = pts[y < pm]
     if len(pts) == 0:
         return 0.0
     x_next = np.empty(len(pts))

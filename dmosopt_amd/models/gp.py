@@ -158,21 +158,33 @@ class _GPRBase:
         init[:, -1] = math.log(1e-6)
         theta = init.clone().requires_grad_(True)
         opt = torch.optim.Adam([theta], lr=lr)
+        best_theta = init.clone()
+        best_loss = torch.full((m,), float("inf"), dtype=self.dtype, device=self.device)
         for _ in range(iters):
             opt.zero_grad(set_to_none=True)
-            loss_total = 0.0
             losses = []
             for s in range(m):
                 nm = batched_nmll(
-                    X, Yn[:, s], theta[s : s + 1], nu=self.nu, anisotropic=self.anisotropic
+                    X, Yn[:, s], theta[s : s + 1], nu=self.nu,
+                    anisotropic=self.anisotropic, jitter=1e-8,
                 )
                 losses.append(nm)
-            loss = torch.cat(losses).sum()
-            loss.backward()
+            loss_vec = torch.cat(losses)
+            finite = torch.isfinite(loss_vec)
+            if not bool(finite.any()):
+                break
+            with torch.no_grad():
+                better = finite & (loss_vec < best_loss)
+                best_loss = torch.where(better, loss_vec.detach(), best_loss)
+                best_theta[better] = theta.detach()[better]
+            loss_vec.masked_fill(~finite, 0.0).sum().backward()
+            if not torch.isfinite(theta.grad).all():
+                theta.grad.zero_()
             opt.step()
             with torch.no_grad():
                 theta.clamp_(bl_t, bu_t)
-        return theta.detach()
+                theta.nan_to_num_(nan=0.0)
+        return best_theta
 
     # ---------------------------------------------------------------- API
     def predict(self, xin):

@@ -1,0 +1,70 @@
+"""End-to-end MO-ASMO runs on analytic problems (quality gates)."""
+
+import numpy as np
+import pytest
+
+import dmosopt_amd
+from dmosopt_amd.benchmarks.problems import zdt1_pareto
+from dmosopt_amd.hv.exact import hv_2d
+
+
+def _zdt1_objfun(pp):
+    names = sorted(pp.keys(), key=lambda s: int(s[1:]))
+    x = np.array([pp[k] for k in names])
+    f1 = x[0]
+    g = 1 + 9 * x[1:].mean()
+    f2 = g * (1 - np.sqrt(f1 / g))
+    return np.array([f1, f2])
+
+
+def _run(opt_id, **overrides):
+    params = {
+        "opt_id": opt_id,
+        "obj_fun": _zdt1_objfun,
+        "problem_parameters": {},
+        "space": {f"x{i + 1}": [0.0, 1.0] for i in range(10)},
+        "objective_names": ["y1", "y2"],
+        "population_size": 60,
+        "num_generations": 40,
+        "initial_maxiter": 2,
+        "optimizer": "nsga2",
+        "n_initial": 3,
+        "n_epochs": 2,
+        "random_seed": 21,
+    }
+    params.update(overrides)
+    return dmosopt_amd.run(params, verbose=False)
+
+
+def test_zdt1_gpr_epochs_quality():
+    best = _run("t_gpr")
+    bestx, besty = best
+    y = np.column_stack([v for _, v in besty])
+    hv = hv_2d(y, np.array([11.0, 11.0]))
+    hv_true = hv_2d(zdt1_pareto(200), np.array([11.0, 11.0]))
+    assert hv > 0.9 * hv_true
+    # evaluated archive is accessible through sopt_dict
+    x, yev = dmosopt_amd.sopt_dict["t_gpr"].optimizer_dict[0].get_evals()
+    assert x.shape[0] == yev.shape[0] > 0
+
+
+def test_zdt1_no_surrogate():
+    best = _run("t_nosurr", surrogate_method_name=None, num_generations=15)
+    bestx, besty = best
+    y = np.column_stack([v for _, v in besty])
+    assert y.shape[1] == 2 and y.shape[0] > 0
+
+
+def test_optimizer_cycling():
+    best = _run(
+        "t_cycle",
+        optimizer_name=["nsga2", "nsga2"],
+        optimizer_kwargs=[{"mutation_prob": 0.1, "crossover_prob": 0.9}] * 2,
+        num_generations=10,
+    )
+    assert best is not None
+
+
+def test_termination_conditions_true():
+    best = _run("t_term", termination_conditions=True, num_generations=30)
+    assert best is not None

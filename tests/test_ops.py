@@ -1,0 +1,181 @@
+"""Population-op tests against independent numpy oracles.
+
+The oracles re-derive the reference semantics (dda.py, indicators.py,
+MOEA.py) directly in this file, so the library implementation is tested
+against an INDEPENDENT formulation, not against itself.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from dmosopt_amd import ops
+from dmosopt_amd.ops import torch_ref
+
+
+def brute_force_fronts(Y: np.ndarray) -> np.ndarray:
+    """O(n^2 m) straightforward non-dominated sorting oracle.
+
+    dominates(i, j): all(Y[i] <= Y[j]) and any(Y[i] < Y[j]).
+    Identical rows do not dominate each other.
+    """
+    n = Y.shape[0]
+    dominated_by = [
+        [
+            j
+            for j in range(n)
+            if j != i
+            and np.all(Y[j] <= Y[i])
+            and np.any(Y[j] < Y[i])
+        ]
+        for i in range(n)
+    ]
+    rank = np.full(n, -1)
+    k = 0
+    remaining = set(range(n))
+    while remaining:
+        front = [
+            i for i in remaining if not any(j in remaining for j in dominated_by[i])
+        ]
+        for i in front:
+            rank[i] = k
+        remaining -= set(front)
+        k += 1
+    return rank
+
+
+@pytest.mark.parametrize("n,m", [(30, 2), (60, 3), (100, 5), (17, 2)])
+def test_pareto_rank_matches_bruteforce(n, m, rng):
+    Y = rng.random((n, m))
+    r = ops.pareto_rank(torch.as_tensor(Y)).numpy()
+    expected = brute_force_fronts(Y)
+    assert np.array_equal(r, expected)
+
+
+def test_pareto_rank_with_duplicates(rng):
+    Y = rng.random((20, 3))
+    Y = np.vstack([Y, Y[:5]])  # duplicate rows
+    r = ops.pareto_rank(torch.as_tensor(Y)).numpy()
+    expected = brute_force_fronts(Y)
+    assert np.array_equal(r, expected)
+    # duplicated rows land in the same front as their originals
+    assert np.array_equal(r[:5], r[20:])
+
+
+def crowding_oracle(Y):
+    """Reference crowding semantics (indicators.py:12-51) re-derived."""
+    n, d = Y.shape
+    if n == 1:
+        return np.array([1.0])
+    lb, ub = Y.min(0), Y.max(0)
+    span = np.where(ub - lb == 0, 1.0, ub - lb)
+    U = (Y - lb) / span
+    D = np.zeros(n)
+    for j in range(d):
+        idx = np.argsort(U[:, j])
+        us = U[idx, j]
+        ds = np.empty(n)
+        ds[0] = ds[-1] = 1.0
+        for i in range(1, n - 1):
+            ds[i] = us[i + 1] - us[i - 1]
+        for i in range(n):
+            D[idx[i]] += ds[i]
+    return D
+
+
+@pytest.mark.parametrize("n,m", [(2, 2), (50, 2), (80, 4)])
+def test_crowding_distance(n, m, rng):
+    Y = rng.random((n, m))
+    d = ops.crowding_distance(torch.as_tensor(Y)).numpy()
+    assert np.allclose(d, crowding_oracle(Y), atol=1e-12)
+
+
+def test_lexsort_matches_numpy(rng):
+    a = rng.integers(0, 5, 100).astype(float)
+    b = rng.random(100)
+    c = rng.integers(0, 3, 100).astype(float)
+    ours = ops.lexsort([torch.as_tensor(b), torch.as_tensor(-a), torch.as_tensor(c)]).numpy()
+    theirs = np.lexsort((b, -a, c))
+    assert np.array_equal(ours, theirs)
+
+
+def test_order_mo_sorts_by_rank_then_crowding(rng):
+    Y = rng.random((40, 2))
+    X = rng.random((40, 5))
+    perm, rank, dists = ops.order_mo(
+        torch.as_tensor(X), torch.as_tensor(Y), y_distance_metrics=["crowding"]
+    )
+    assert np.all(np.diff(rank.numpy()) >= 0)  # ranks nondecreasing
+    # within a front, crowding distance is nonincreasing
+    r = rank.numpy()
+    d = dists[0].numpy()
+    for k in np.unique(r):
+        dk = d[r == k]
+        assert np.all(np.diff(dk) <= 1e-12)
+
+
+def test_sbx_crossover_bounds_and_mean(rng):
+    d = 10
+    p1 = torch.rand(500, d, dtype=torch.float64)
+    p2 = torch.rand(500, d, dtype=torch.float64)
+    di = torch.full((d,), 1.0, dtype=torch.float64)
+    lo = torch.zeros(d, dtype=torch.float64)
+    hi = torch.ones(d, dtype=torch.float64)
+    g = torch.Generator().manual_seed(0)
+    c1, c2 = ops.sbx_crossover_batch(p1, p2, di, lo, hi, generator=g)
+    assert (c1 >= 0).all() and (c1 <= 1).all()
+    assert (c2 >= 0).all() and (c2 <= 1).all()
+    # SBX preserves the parent midpoint before clipping: c1+c2 == p1+p2
+    mid_parent = (p1 + p2).mean()
+    mid_child = (c1 + c2).mean()
+    assert abs(mid_parent - mid_child) < 0.02
+
+
+def test_polynomial_mutation_bounds(rng):
+    d = 8
+    p = torch.rand(300, d, dtype=torch.float64)
+    di = torch.full((d,), 20.0, dtype=torch.float64)
+    lo = torch.zeros(d, dtype=torch.float64)
+    hi = torch.ones(d, dtype=torch.float64)
+    g = torch.Generator().manual_seed(3)
+    child = ops.polynomial_mutation_batch(p, di, lo, hi, mutation_rate=0.5, generator=g)
+    assert (child >= 0).all() and (child <= 1).all()
+    # high di keeps children near parents
+    assert (child - p).abs().mean() < 0.1
+
+
+def test_get_duplicates_marks_earlier_row(rng):
+    X = rng.random((10, 4))
+    X[7] = X[2]  # pair (2, 7): later index marked (triu-mask semantics)
+    dup = ops.get_duplicates(torch.as_tensor(X)).numpy()
+    assert dup[7] and not dup[2]
+    assert dup.sum() == 1
+
+
+def test_remove_worst_keeps_best_front(rng):
+    X = torch.rand(60, 4, dtype=torch.float64)
+    Y = torch.rand(60, 2, dtype=torch.float64)
+    x2, y2, rank, perm = ops.remove_worst(X, Y, 20, y_distance_metrics=["crowding"])
+    assert x2.shape == (20, 4)
+    full_rank = ops.pareto_rank(Y).numpy()
+    kept_ranks = full_rank[perm.numpy()]
+    dropped = np.setdiff1d(np.arange(60), perm.numpy())
+    assert kept_ranks.max() <= full_rank[dropped].min() or len(dropped) == 0
+
+
+def test_filter_samples_nan_remove():
+    y = torch.tensor([[1.0, 2.0], [float("nan"), 1.0], [3.0, 4.0]])
+    x = torch.arange(3).double()[:, None]
+    y2, x2 = ops.filter_samples(y, x, nan="remove")
+    assert y2.shape[0] == 2 and x2.shape[0] == 2
+
+
+def test_tournament_selection_distribution(rng):
+    # best-ranked individuals must be selected more often
+    pop = 20
+    metrics = [torch.arange(pop, dtype=torch.float64)]  # identity rank
+    counts = np.zeros(pop)
+    for _ in range(200):
+        idx = ops.tournament_selection(pop, 5, metrics, rng).numpy()
+        counts[idx] += 1
+    assert counts[:5].sum() > counts[15:].sum()

@@ -1,0 +1,56 @@
+import numpy as np
+import pytest
+
+from dmosopt_amd import sampling
+
+
+@pytest.mark.parametrize("method", ["mc", "lh", "slh", "glp", "sobol"])
+def test_designs_in_unit_cube(method, rng):
+    fn = getattr(sampling, method)
+    x = fn(50, 4, rng)
+    assert x.shape == (50, 4)
+    assert (x >= 0).all() and (x <= 1).all()
+
+
+def test_lh_stratification(rng):
+    n, s = 40, 3
+    x = sampling.lh(n, s, rng)
+    for j in range(s):
+        counts, _ = np.histogram(x[:, j], bins=n, range=(0, 1))
+        assert (counts == 1).all()
+
+
+def test_slh_symmetry(rng):
+    n, s = 20, 4
+    x = sampling.slh(n, s, rng)
+    # symmetric LH: row i and row n-1-i sum to 1 per column
+    assert np.allclose(x + x[::-1], 1.0)
+    # and it is a latin hypercube
+    for j in range(s):
+        counts, _ = np.histogram(x[:, j], bins=n, range=(0, 1))
+        assert (counts == 1).all()
+
+
+def test_glp_lattice_structure(rng):
+    x = sampling.glp(20, 3, rng)
+    assert x.shape == (20, 3)
+    # lattice points are distinct per column
+    for j in range(3):
+        assert len(np.unique(np.round(x[:, j], 12))) == 20
+
+
+def test_decorr_reduces_correlation(rng):
+    x = sampling.lh(64, 6, rng, maxiter=0)
+    xd = sampling.lh(64, 6, np.random.default_rng(12345), maxiter=3)
+    def max_abs_offdiag_corr(a):
+        c = np.corrcoef(a.T)
+        np.fill_diagonal(c, 0)
+        return np.abs(c).max()
+    assert max_abs_offdiag_corr(xd) <= max_abs_offdiag_corr(x) + 0.05
+
+
+def test_cd2_known_property(rng):
+    # a latin hypercube should have lower CD2 than random MC of same size
+    mc = sampling.mc(64, 4, np.random.default_rng(1))
+    lh = sampling.lh(64, 4, np.random.default_rng(1))
+    assert sampling.cd2(lh) < sampling.cd2(mc)
