@@ -1,0 +1,199 @@
+"""Flagship benchmark: MO-ASMO epoch time (GP fit + 200-gen NSGA2), ZDT1 d=30.
+
+Measures BASELINE.json's headline metric on its named config: one timed
+"step" = one full MO-ASMO epoch — GP surrogate fit (Matern-5/2, SCE-UA
+batched MLL search) on a fixed 300-point archive + 200 NSGA-II generations
+against the surrogate + resample selection + real (synthetic, on-device)
+ZDT1 evaluation of the resample batch. Weak scaling: per-GPU population is
+fixed at 200, global population = 200 * n_gpus; every rank runs the
+replicated NSGA2 control flow (identical seeds) while surrogate prediction
+is rank-sharded and all-gathered over RCCL/xGMI each generation.
+
+Run: python bench.py [--gpus N] [--steps K] [--warmup W]
+Distributed: torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N ...
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from dmosopt_amd.benchmarks.problems import zdt1
+from dmosopt_amd.core import engine
+from dmosopt_amd.hv.exact import hv_2d
+from dmosopt_amd.models.gp import GPRMatern
+from dmosopt_amd.models.model import Model
+from dmosopt_amd.parallel import comm
+
+D_IN = 30
+N_OBJ = 2
+POP_PER_GPU = 200
+N_GEN = 200
+ARCHIVE_N = 300
+
+
+class ShardedGPObjective:
+    """GP posterior evaluate() with rank-sharded prediction.
+
+    Queries arrive replicated on every rank; each rank predicts rows
+    [rank::world] on its GPU and the full result is re-assembled with ONE
+    all_gather per generation (a single large collective over xGMI, per the
+    comm design of SURVEY.md section 2.10).
+    """
+
+    def __init__(self, gp: GPRMatern, rank: int, world: int, device):
+        self.gp = gp
+        self.rank = rank
+        self.world = world
+        self.device = device
+
+    def evaluate(self, x):
+        import torch.distributed as dist
+
+        x = np.asarray(x, dtype=np.float64)
+        if self.world == 1:
+            mean, _ = self.gp.predict(x)
+            return mean
+        P = x.shape[0]
+        pad = (self.world - P % self.world) % self.world
+        if pad:
+            x = np.vstack([x, np.repeat(x[-1:], pad, axis=0)])
+        shard = x[self.rank :: self.world]
+        mean, _ = self.gp.predict(shard)  # (P/world, m)
+        mean_t = torch.as_tensor(mean, dtype=torch.float32, device=self.device)
+        out = [torch.empty_like(mean_t) for _ in range(self.world)]
+        dist.all_gather(out, mean_t)
+        full = torch.stack(out, dim=1).reshape(-1, mean_t.shape[1])  # interleave
+        return full[:P].cpu().numpy()
+
+
+def make_archive(seed: int) -> tuple:
+    rng = np.random.default_rng(seed)
+    X = rng.random((ARCHIVE_N, D_IN))
+    Y = zdt1(X).numpy()
+    return X, Y
+
+
+def one_epoch(X, Y, pop, rank, world, device, seed, n_gen=N_GEN):
+    """One MO-ASMO epoch; returns (resample_x, predicted_y, hv)."""
+    gp = GPRMatern(
+        X, Y, D_IN, N_OBJ, np.zeros(D_IN), np.ones(D_IN),
+        optimizer="sceua", seed=seed, device=device,
+    )
+    mdl = Model(objective=ShardedGPObjective(gp, rank, world, device))
+
+    from dmosopt_amd.moea.nsga2 import NSGA2Optimizer
+
+    optimizer = NSGA2Optimizer(
+        popsize=pop, nInput=D_IN, nOutput=N_OBJ, model=mdl,
+        distance_metric=None, sampling_method="slh", mutation_rate=None, nchildren=1,
+    )
+    if device.type == "cuda":
+        optimizer.set_device(device)
+    local_random = np.random.default_rng(seed + 1)
+    res = engine.optimize_loop(
+        n_gen, optimizer, mdl, D_IN, N_OBJ, np.zeros(D_IN), np.ones(D_IN),
+        popsize=pop, initial=(X.astype(np.float32), Y.astype(np.float32)),
+        local_random=local_random,
+    )
+    # resample selection (25% of pop) + real evaluation of the batch
+    from dmosopt_amd import ops
+
+    best_x = torch.as_tensor(res.best_x, dtype=torch.float32, device=device)
+    best_y = torch.as_tensor(res.best_y, dtype=torch.float32, device=device)
+    Dc = ops.crowding_distance(best_y)
+    n_res = max(1, int(0.25 * pop))
+    idx = torch.argsort(Dc, descending=True)[:n_res]
+    x_res = best_x[idx]
+    shard = x_res[rank::world] if world > 1 else x_res
+    y_res = zdt1(shard.double())  # on-device synthetic objective
+    hv = hv_2d(res.best_y, np.array([11.0, 11.0]))
+    return x_res, y_res, hv
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--gens", type=int, default=N_GEN)
+    args = ap.parse_args()
+
+    rank, world = comm.init_from_env()
+    if torch.cuda.is_available():
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0)) % torch.cuda.device_count())
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    pop = POP_PER_GPU * world  # weak scaling: global population grows
+    X, Y = make_archive(seed=1234)
+
+    import torch.distributed as dist
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
+    hv = None
+    for w in range(args.warmup):
+        one_epoch(X, Y, pop, rank, world, device, seed=100 + w, n_gen=args.gens)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        _, _, hv = one_epoch(X, Y, pop, rank, world, device, seed=200 + s, n_gen=args.gens)
+    barrier_sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if device.type == "cuda":
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = 1000.0 * elapsed / args.steps
+    if rank == 0:
+        out = {
+            "metric": "MO-ASMO epoch time (GP fit + 200-gen NSGA2), ZDT1 d=30",
+            "value": ms_per_step,
+            "unit": "ms_per_epoch",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic ZDT1 d=30, random-init archive of 300 evals",
+            "config": {
+                "model": "GPR-Matern52 surrogate + NSGA2",
+                "global_batch": pop,
+                "seq_len": None,
+                "parallelism": f"replicated-moea+sharded-gp dp{world}",
+                "population_size": pop,
+                "num_generations": args.gens,
+                "archive_size": ARCHIVE_N,
+                "final_hypervolume_ref11": hv,
+            },
+        }
+        print(json.dumps(out))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -167,6 +167,7 @@ class _GPRBase:
                 nm = batched_nmll(
                     X, Yn[:, s], theta[s : s + 1], nu=self.nu,
                     anisotropic=self.anisotropic, jitter=1e-8,
+                    differentiable=True,
                 )
                 losses.append(nm)
             loss_vec = torch.cat(losses)

@@ -52,7 +52,7 @@ def matern_from_d2(d2: torch.Tensor, nu: float) -> torch.Tensor:
     raise ValueError(f"Unsupported Matern nu={nu}")
 
 
-def build_kernel(
+def build_kernel_torch(
     X1: torch.Tensor,
     X2: Optional[torch.Tensor],
     theta: torch.Tensor,
@@ -88,6 +88,23 @@ def build_kernel(
     return K
 
 
+def build_kernel(
+    X1: torch.Tensor,
+    X2: Optional[torch.Tensor],
+    theta: torch.Tensor,
+    nu: float = 2.5,
+    anisotropic: bool = False,
+    jitter: float = 0.0,
+) -> torch.Tensor:
+    """Dispatching kernel build: gfx950 fused-assembly kernel on GPU,
+    torch on CPU."""
+    from dmosopt_amd import ops
+
+    if X2 is None:
+        return ops.matern_train_kernel(X1, theta, nu, anisotropic, jitter)
+    return ops.matern_cross_kernel(X1, X2, theta, nu, anisotropic)
+
+
 def batched_nmll(
     X: torch.Tensor,
     y: torch.Tensor,
@@ -95,23 +112,38 @@ def batched_nmll(
     nu: float = 2.5,
     anisotropic: bool = False,
     jitter: float = 1e-10,
+    differentiable: bool = False,
 ) -> torch.Tensor:
     """Negative log marginal likelihood for a batch of theta.
 
     X: (N, d), y: (N,) standardized. theta: (B, p). Returns (B,).
     Failed factorizations get +inf (SCE-UA treats them as bad points).
+    ``differentiable=True`` forces the autograd-capable torch path (Adam
+    optimizer); the default dispatches to the fused gfx950 kernels.
     """
     N = X.shape[0]
-    K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
-    L, info = torch.linalg.cholesky_ex(K)
-    yb = y[None, :, None].expand(K.shape[0], N, 1)
-    alpha = torch.cholesky_solve(yb, L)  # (B, N, 1)
-    quad = (yb * alpha).sum(dim=(1, 2))
-    logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(dim=-1)
+    if differentiable:
+        K = build_kernel_torch(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
+        L, info = torch.linalg.cholesky_ex(K)
+        yb = y[None, :, None].expand(K.shape[0], N, 1)
+        alpha = torch.cholesky_solve(yb, L)
+        quad = (yb * alpha).sum(dim=(1, 2))
+        logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(dim=-1)
+    else:
+        from dmosopt_amd import ops
+
+        K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
+        L, half_logdet_or_full, info = ops.chol_factor_batched(K)
+        # native path returns sum(log diag L); torch path the same
+        logdet = 2.0 * half_logdet_or_full
+        yb = y[None, :, None].expand(L.shape[0], N, 1).contiguous()
+        z = ops.tri_solve_forward(L, yb)  # L z = y
+        quad = (z * z).sum(dim=(1, 2))
     nmll = 0.5 * quad + 0.5 * logdet + 0.5 * N * LOG2PI
     bad = info != 0
     if bad.any():
         nmll = torch.where(bad, torch.full_like(nmll, float("inf")), nmll)
+    nmll = torch.where(torch.isfinite(nmll), nmll, torch.full_like(nmll, float("inf")))
     return nmll
 
 
@@ -138,41 +170,52 @@ class FittedGP:
         self.anisotropic = anisotropic
         self.y_mean = y_mean  # (m,)
         self.y_std = y_std  # (m,)
+        from dmosopt_amd import ops
+
         m, N = theta.shape[0], X.shape[0]
         K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
-        self.L, info = torch.linalg.cholesky_ex(K)  # (m, N, N)
+        self.L, _, info = ops.chol_factor_batched(K)  # (m, N, N)
         if int(info.sum()) != 0:
             # escalate jitter for failed objectives
             for _ in range(5):
                 bad = info != 0
-                if not bad.any():
+                if not bool(bad.any()):
                     break
                 jitter *= 100.0
                 K2 = build_kernel(
                     X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter
                 )
-                L2, info = torch.linalg.cholesky_ex(K2)
+                L2, _, info = ops.chol_factor_batched(K2)
                 self.L = torch.where(bad[:, None, None], L2, self.L)
         Yn = (Y - y_mean[None, :]) / y_std[None, :]  # (N, m) standardized
-        yb = Yn.T[:, :, None]  # (m, N, 1)
-        self.alpha = torch.cholesky_solve(yb, self.L)  # (m, N, 1)
+        yb = Yn.T[:, :, None].contiguous()  # (m, N, 1)
+        self.alpha = ops.chol_solve_batched(self.L, yb)  # (m, N, 1)
+        # On GPU, precompute K^-1 so per-generation posterior variance is a
+        # pair of batched GEMMs (rocBLAS/MFMA) instead of P triangular solves
+        self.Kinv: Optional[torch.Tensor] = None
+        if X.is_cuda:
+            eye = torch.eye(N, dtype=X.dtype, device=X.device)[None].expand(m, N, N).contiguous()
+            self.Kinv = ops.chol_solve_batched(self.L, eye)
 
     def predict(self, Xq: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
         """Posterior mean and variance at Xq (P, d) -> ((P, m), (P, m)).
 
         Variance matches sklearn's return_std**2: diag K(x*,x*) [incl. noise
-        from the White term] minus v^T v, scaled by y_std^2.
+        from the White term] minus k*^T K^-1 k*, scaled by y_std^2.
         """
-        m = self.theta.shape[0]
         Ks = build_kernel(Xq, self.X, self.theta, nu=self.nu, anisotropic=self.anisotropic)
         # (m, P, N)
-        mean_n = (Ks @ self.alpha)[:, :, 0]  # (m, P)
-        v = torch.linalg.solve_triangular(self.L, Ks.transpose(-1, -2), upper=False)
-        # (m, N, P)
+        mean_n = torch.bmm(Ks, self.alpha)[:, :, 0]  # (m, P)
         sf2 = torch.exp(self.theta[:, 0])
         noise = torch.exp(self.theta[:, -1])
         kss = (sf2 + noise)[:, None]  # (m, 1): k(x,x) = sf2*1 + noise
-        var_n = (kss - (v * v).sum(dim=1)).clamp_min(0.0)  # (m, P)
+        if self.Kinv is not None:
+            W = torch.bmm(Ks, self.Kinv)  # (m, P, N)
+            quad = (Ks * W).sum(dim=2)  # (m, P)
+        else:
+            v = torch.linalg.solve_triangular(self.L, Ks.transpose(-1, -2), upper=False)
+            quad = (v * v).sum(dim=1)
+        var_n = (kss - quad).clamp_min(0.0)  # (m, P)
         mean = self.y_mean[None, :] + self.y_std[None, :] * mean_n.T
         var = (self.y_std[None, :] ** 2) * var_n.T
         return mean, var

@@ -141,6 +141,60 @@ def remove_duplicates(x: torch.Tensor, y: torch.Tensor, eps: float = 1e-16):
     return x[~dup], y[~dup]
 
 
+# -------------------------------------------------------------------- GP ops
+def matern_train_kernel(X, theta, nu, anisotropic, jitter):
+    """Batched symmetric kernel matrices K (B,N,N) with noise+jitter diag."""
+    if _use_native(X) and X.dtype == torch.float32:
+        nu_arg = 0.0 if (nu is None or nu == float("inf")) else float(nu)
+        return _native.matern_train(
+            X.contiguous(), theta.contiguous().float(), nu_arg, bool(anisotropic), float(jitter)
+        )
+    from dmosopt_amd.models import gp_core
+
+    return gp_core.build_kernel_torch(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
+
+
+def matern_cross_kernel(Xq, X, theta, nu, anisotropic):
+    """Batched cross kernel K* (B,P,N), no diagonal noise."""
+    if _use_native(X) and X.dtype == torch.float32:
+        nu_arg = 0.0 if (nu is None or nu == float("inf")) else float(nu)
+        return _native.matern_cross(
+            Xq.contiguous(), X.contiguous(), theta.contiguous().float(), nu_arg, bool(anisotropic)
+        )
+    from dmosopt_amd.models import gp_core
+
+    return gp_core.build_kernel_torch(Xq, X, theta, nu=nu, anisotropic=anisotropic)
+
+
+def chol_factor_batched(K):
+    """Factor K (B,N,N) in place -> (L, logdet (B,), info (B,))."""
+    if _use_native(K) and K.dtype == torch.float32:
+        logdet, info = _native.cholesky_batched_(K)
+        return K, logdet, info
+    L, info = torch.linalg.cholesky_ex(K)
+    logdet = torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(dim=-1)
+    return L, logdet, info
+
+
+def chol_solve_batched(L, Y):
+    """Solve K X = Y given the factor L (B,N,N); Y (B,N,R) -> X (B,N,R)."""
+    if _use_native(L) and L.dtype == torch.float32:
+        Z = Y.contiguous().clone()
+        _native.forward_solve_(L.contiguous(), Z)
+        _native.backward_solve_(L.contiguous(), Z)
+        return Z
+    return torch.cholesky_solve(Y, L)
+
+
+def tri_solve_forward(L, Y):
+    """Solve L Z = Y (B,N,R)."""
+    if _use_native(L) and L.dtype == torch.float32:
+        Z = Y.contiguous().clone()
+        _native.forward_solve_(L.contiguous(), Z)
+        return Z
+    return torch.linalg.solve_triangular(L, Y, upper=False)
+
+
 __all__ = [
     "pareto_rank",
     "dominance_degree_matrix",

@@ -1,0 +1,179 @@
+// Pareto ranking (Dominance Degree Matrix) and crowding distance.
+//
+// Replaces reference dda.py:25-120 and indicators.py:12-51.
+//
+// pareto_rank design: D[i][j] = #objectives with Y[i][k] <= Y[j][k] is an
+// N x N x m comparison reduce computed tile-wise with both 32-row objective
+// slabs in LDS; the front peel (iterated column-max + mask) runs as a small
+// secondary kernel per front — front count is small (<= tens) and each peel
+// pass is a single coalesced N x N sweep.
+//
+// crowding design: one workgroup per objective dimension; (value, index)
+// pairs bitonic-sorted in LDS, boundary/interior gaps computed in place and
+// scattered into the output with atomics (m-way accumulation).
+
+#include "common.h"
+#include <math.h>
+
+#define PTILE 32
+#define PTPB 256
+
+__global__ void dominance_matrix_kernel(const float* __restrict__ Y,  // (N, m)
+                                        int* __restrict__ D,          // (N, N)
+                                        int N, int m) {
+  extern __shared__ float lds[];  // [2][PTILE][m]
+  float* yi = lds;                // rows i (tile_y)
+  float* yj = lds + PTILE * m;    // rows j (tile_x)
+  const int ti = blockIdx.y * PTILE;
+  const int tj = blockIdx.x * PTILE;
+  for (int idx = threadIdx.x; idx < PTILE * m; idx += PTPB) {
+    const int r = idx / m, c = idx % m;
+    yi[idx] = (ti + r < N) ? Y[(ti + r) * m + c] : 0.f;
+    yj[idx] = (tj + r < N) ? Y[(tj + r) * m + c] : 0.f;
+  }
+  __syncthreads();
+  const int ty = threadIdx.x / 16, tx = threadIdx.x % 16;
+#pragma unroll
+  for (int sy = 0; sy < 2; ++sy)
+#pragma unroll
+    for (int sx = 0; sx < 2; ++sx) {
+      const int li = ty * 2 + sy, lj = tx * 2 + sx;
+      const int gi = ti + li, gj = tj + lj;
+      if (gi >= N || gj >= N) continue;
+      int cnt = 0;
+      for (int k = 0; k < m; ++k)
+        cnt += (yi[li * m + k] <= yj[lj * m + k]) ? 1 : 0;
+      D[(long long)gi * N + gj] = cnt;
+    }
+}
+
+// Zero mutual-domination entries of identical rows: D[i][j] = 0 where
+// D[i][j] == m and D[j][i] == m (reference dda.py:47-49).
+__global__ void zero_identical_kernel(int* __restrict__ D, int N, int m) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long long)N * N) return;
+  const int i = (int)(idx / N), j = (int)(idx % N);
+  if (i < j && D[(long long)i * N + j] == m && D[(long long)j * N + i] == m) {
+    D[(long long)i * N + j] = 0;
+    D[(long long)j * N + i] = 0;
+  }
+}
+
+// One peel pass: for each alive column j compute max_i(D[i][j]) over alive
+// rows; columns with max < m belong to the current front.
+__global__ void peel_front_kernel(const int* __restrict__ D,
+                                  const unsigned char* __restrict__ alive,
+                                  unsigned char* __restrict__ front,  // out
+                                  int* __restrict__ n_front,          // out
+                                  int N, int m) {
+  const int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= N) return;
+  unsigned char f = 0;
+  if (alive[j]) {
+    int mx = -1;
+    const int* col = D + j;
+    for (int i = 0; i < N; ++i)
+      if (alive[i]) mx = max(mx, col[(long long)i * N]);
+    f = (mx < m) ? 1 : 0;
+  }
+  front[j] = f;
+  if (f) atomicAdd(n_front, 1);
+}
+
+__global__ void commit_front_kernel(const unsigned char* __restrict__ front,
+                                    unsigned char* __restrict__ alive,
+                                    int* __restrict__ rank, int k, int N) {
+  const int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= N) return;
+  if (front[j]) {
+    rank[j] = k;
+    alive[j] = 0;
+  }
+}
+
+extern "C" void launch_dominance_matrix(const float* Y, int* D, int N, int m,
+                                        hipStream_t stream) {
+  dim3 grid((N + PTILE - 1) / PTILE, (N + PTILE - 1) / PTILE);
+  size_t lds_bytes = 2 * PTILE * m * sizeof(float);
+  hipLaunchKernelGGL(dominance_matrix_kernel, grid, dim3(PTPB), lds_bytes,
+                     stream, Y, D, N, m);
+  long long total = (long long)N * N;
+  int blocks = (int)((total + 255) / 256);
+  hipLaunchKernelGGL(zero_identical_kernel, dim3(blocks), dim3(256), 0, stream,
+                     D, N, m);
+}
+
+extern "C" void launch_peel_front(const int* D, const unsigned char* alive,
+                                  unsigned char* front, int* n_front, int N,
+                                  int m, hipStream_t stream) {
+  int blocks = (N + 255) / 256;
+  hipLaunchKernelGGL(peel_front_kernel, dim3(blocks), dim3(256), 0, stream, D,
+                     alive, front, n_front, N, m);
+}
+
+extern "C" void launch_commit_front(const unsigned char* front,
+                                    unsigned char* alive, int* rank, int k,
+                                    int N, hipStream_t stream) {
+  int blocks = (N + 255) / 256;
+  hipLaunchKernelGGL(commit_front_kernel, dim3(blocks), dim3(256), 0, stream,
+                     front, alive, rank, k, N);
+}
+
+// ------------------------------------------------------------- crowding
+// One workgroup per objective dimension. Bitonic sort of (normalized value,
+// index) in LDS; N padded to the next power of two with +inf sentinels.
+#define CROWD_TPB 256
+
+__global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
+                                const float* __restrict__ lo,  // (m,)
+                                const float* __restrict__ span,  // (m,)
+                                float* __restrict__ out,  // (N,) pre-zeroed
+                                int N, int m, int npow2) {
+  extern __shared__ float sh[];
+  float* vals = sh;                       // npow2
+  int* idxs = (int*)(sh + npow2);         // npow2
+  const int j = blockIdx.x;               // objective dim
+  const float l = lo[j], s = span[j];
+
+  for (int i = threadIdx.x; i < npow2; i += CROWD_TPB) {
+    vals[i] = (i < N) ? (Y[i * m + j] - l) / s : INFINITY;
+    idxs[i] = i;
+  }
+  __syncthreads();
+
+  // bitonic sort ascending (stable order not required: equal normalized
+  // values produce zero gaps either way)
+  for (int ksz = 2; ksz <= npow2; ksz <<= 1) {
+    for (int jsz = ksz >> 1; jsz > 0; jsz >>= 1) {
+      for (int i = threadIdx.x; i < npow2; i += CROWD_TPB) {
+        const int ixj = i ^ jsz;
+        if (ixj > i) {
+          const bool up = ((i & ksz) == 0);
+          const bool swap = up ? (vals[i] > vals[ixj]) : (vals[i] < vals[ixj]);
+          if (swap) {
+            float tv = vals[i]; vals[i] = vals[ixj]; vals[ixj] = tv;
+            int tixx = idxs[i]; idxs[i] = idxs[ixj]; idxs[ixj] = tixx;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // gaps: boundary = 1, interior = vals[i+1] - vals[i-1]; scatter-add
+  for (int i = threadIdx.x; i < N; i += CROWD_TPB) {
+    float d = (i == 0 || i == N - 1) ? 1.f : vals[i + 1] - vals[i - 1];
+    if (isnan(d)) d = 0.f;
+    atomicAdd(&out[idxs[i]], d);
+  }
+}
+
+extern "C" void launch_crowding(const float* Y, const float* lo,
+                                const float* span, float* out, int N, int m,
+                                hipStream_t stream) {
+  int npow2 = 1;
+  while (npow2 < N) npow2 <<= 1;
+  size_t lds_bytes = npow2 * (sizeof(float) + sizeof(int));
+  hipLaunchKernelGGL(crowding_kernel, dim3(m), dim3(CROWD_TPB), lds_bytes,
+                     stream, Y, lo, span, out, N, m, npow2);
+}

@@ -1,0 +1,41 @@
+"""Build the in-tree gfx950 HIP extension: python setup.py build_ext --inplace.
+
+Compiles for gfx950 ONLY (MI355X / CDNA4); no CUDA path, no multi-arch
+fatbins. The built .so lands inside dmosopt_amd/ so it travels with the
+source tree to GPU boxes.
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+HIP_DIR = os.path.join(ROOT, "dmosopt_amd", "ops", "hip")
+
+ext = CUDAExtension(
+    name="dmosopt_amd._hipops",
+    sources=[
+        os.path.join("dmosopt_amd", "ops", "hip", "bindings.cpp"),
+        os.path.join("dmosopt_amd", "ops", "hip", "matern.hip"),
+        os.path.join("dmosopt_amd", "ops", "hip", "cholesky.hip"),
+        os.path.join("dmosopt_amd", "ops", "hip", "pareto.hip"),
+        os.path.join("dmosopt_amd", "ops", "hip", "variation.hip"),
+        os.path.join("dmosopt_amd", "ops", "hip", "hv_mc.hip"),
+    ],
+    include_dirs=[HIP_DIR],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+    },
+)
+
+setup(
+    name="dmosopt_amd",
+    version="0.1.0",
+    packages=["dmosopt_amd"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

@@ -1,0 +1,191 @@
+"""GPU numerics tests: gfx950 kernels vs the plain-PyTorch fp32/fp64
+reference implementations (the framework's kernel-vs-oracle contract,
+SURVEY.md section 4)."""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from dmosopt_amd import ops
+
+    assert ops.native_available(), "native extension must be built on a GPU box"
+    return torch.device("cuda", 0)
+
+
+def test_matern_train_matches_torch(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd.models.gp_core import build_kernel_torch
+
+    g = torch.Generator().manual_seed(0)
+    X = torch.rand(257, 30, generator=g).float().to(dev)
+    theta = torch.tensor(
+        [[0.3, math.log(0.5), math.log(1e-4)], [-0.2, math.log(2.0), math.log(1e-6)]]
+    ).float().to(dev)
+    K_native = _hipops.matern_train(X, theta, 2.5, False, 1e-6)
+    K_ref = build_kernel_torch(X.double(), None, theta.double(), nu=2.5, jitter=1e-6)
+    assert torch.allclose(K_native.double(), K_ref, atol=5e-5)
+
+
+def test_matern_cross_aniso_matches_torch(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd.models.gp_core import build_kernel_torch
+
+    g = torch.Generator().manual_seed(1)
+    Xq = torch.rand(100, 10, generator=g).float().to(dev)
+    X = torch.rand(211, 10, generator=g).float().to(dev)
+    theta = torch.cat(
+        [torch.zeros(3, 1), torch.randn(3, 10, generator=g) * 0.3, torch.full((3, 1), -9.0)],
+        dim=1,
+    ).float().to(dev)
+    K_native = _hipops.matern_cross(Xq, X, theta, 2.5, True)
+    K_ref = build_kernel_torch(Xq.double(), X.double(), theta.double(), nu=2.5, anisotropic=True)
+    assert torch.allclose(K_native.double(), K_ref, atol=5e-5)
+
+
+def test_cholesky_batched_matches_torch(dev):
+    from dmosopt_amd import _hipops
+
+    g = torch.Generator().manual_seed(2)
+    B, N = 4, 300
+    A = torch.randn(B, N, 8, generator=g)
+    K = (A @ A.transpose(1, 2) + 0.5 * torch.eye(N)).float().to(dev).contiguous()
+    L_ref = torch.linalg.cholesky(K.double().cpu())
+    logdet_ref = torch.log(torch.diagonal(L_ref, dim1=1, dim2=2)).sum(dim=1)
+    Kc = K.clone()
+    logdet, info = _hipops.cholesky_batched_(Kc)
+    assert (info == 0).all()
+    assert torch.allclose(Kc.double().cpu().tril(), L_ref, atol=2e-3, rtol=1e-3)
+    assert torch.allclose(logdet.double().cpu(), logdet_ref, rtol=1e-4)
+
+
+def test_solves_match_torch(dev):
+    from dmosopt_amd import _hipops
+
+    g = torch.Generator().manual_seed(3)
+    B, N, R = 3, 200, 5
+    A = torch.randn(B, N, 6, generator=g)
+    K = (A @ A.transpose(1, 2) + 0.3 * torch.eye(N)).double()
+    L = torch.linalg.cholesky(K).float().to(dev).contiguous()
+    Y = torch.randn(B, N, R, generator=g).float().to(dev).contiguous()
+    Z = Y.clone()
+    _hipops.forward_solve_(L, Z)
+    Z_ref = torch.linalg.solve_triangular(L.double().cpu(), Y.double().cpu(), upper=False)
+    assert torch.allclose(Z.double().cpu(), Z_ref, atol=1e-3, rtol=1e-3)
+    X2 = Z.clone()
+    _hipops.backward_solve_(L, X2)
+    X_ref = torch.cholesky_solve(Y.double().cpu(), L.double().cpu())
+    assert torch.allclose(X2.double().cpu(), X_ref, atol=5e-3, rtol=1e-2)
+
+
+def test_pareto_rank_matches_reference(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd.ops import torch_ref
+
+    g = torch.Generator().manual_seed(4)
+    for n, m in [(100, 2), (500, 3), (1000, 5)]:
+        Y = torch.rand(n, m, generator=g)
+        Y[: n // 10] = Y[n // 2 : n // 2 + n // 10]  # inject duplicates
+        r_gpu = _hipops.pareto_rank(Y.float().to(dev)).cpu()
+        r_ref = torch_ref.pareto_rank(Y.double())
+        assert torch.equal(r_gpu, r_ref)
+
+
+def test_crowding_matches_reference(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd.ops import torch_ref
+
+    g = torch.Generator().manual_seed(5)
+    for n, m in [(64, 2), (513, 3), (2000, 4)]:
+        Y = torch.rand(n, m, generator=g)
+        d_gpu = _hipops.crowding_distance(Y.float().to(dev)).cpu().double()
+        d_ref = torch_ref.crowding_distance(Y.double())
+        assert torch.allclose(d_gpu, d_ref, atol=1e-5)
+
+
+def test_variation_bounds_and_stats(dev):
+    from dmosopt_amd import _hipops
+
+    g = torch.Generator().manual_seed(6)
+    K, d, C, M = 50, 12, 40, 30
+    pool = torch.rand(K, d, generator=g).float().to(dev)
+    p1 = torch.randint(0, K, (C,), generator=g).int().to(dev)
+    p2 = torch.randint(0, K, (C,), generator=g).int().to(dev)
+    di = torch.full((d,), 1.0).float().to(dev)
+    lo = torch.zeros(d).float().to(dev)
+    hi = torch.ones(d).float().to(dev)
+    c1, c2 = _hipops.sbx_batch(pool, p1, p2, di, lo, hi, 12345)
+    assert (c1 >= 0).all() and (c1 <= 1).all()
+    # midpoint preservation (before clipping dominates): mean over genes
+    mid_p = (pool[p1.long()] + pool[p2.long()]).mean()
+    mid_c = (c1 + c2).mean()
+    assert abs(float(mid_p - mid_c)) < 0.05
+    mp = torch.randint(0, K, (M,), generator=g).int().to(dev)
+    dm = torch.full((d,), 20.0).float().to(dev)
+    child = _hipops.mutation_batch(pool, mp, dm, lo, hi, 0.5, 999)
+    assert (child >= 0).all() and (child <= 1).all()
+    assert float((child - pool[mp.long()]).abs().mean()) < 0.15
+    # determinism: same seed -> same output
+    c1b, _ = _hipops.sbx_batch(pool, p1, p2, di, lo, hi, 12345)
+    assert torch.equal(c1, c1b)
+
+
+def test_hv_mc_kernels_match_exact(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd.hv.exact import hv_3d
+
+    g = torch.Generator().manual_seed(7)
+    P = torch.rand(20, 3, generator=g).double()
+    ref = torch.full((3,), 1.1).double()
+    exact = hv_3d(P.numpy(), ref.numpy())
+    Pf = P.float().to(dev).contiguous()
+    reff = ref.float().to(dev)
+    ideal = Pf.min(dim=0).values.contiguous()
+    n = 2_000_000
+    hits = _hipops.hv_mc_uniform_hits(Pf, ideal, reff, n, 11)
+    box_vol = float(torch.prod(reff - ideal))
+    est = box_vol * hits / n
+    assert est == pytest.approx(exact, rel=0.02)
+    vols = torch.prod(reff[None, :] - Pf, dim=1)
+    cdf = (vols.cumsum(0) / vols.sum()).contiguous()
+    hits2 = _hipops.hv_fpras_hits(Pf, reff, cdf, n, 13)
+    est2 = float(vols.sum()) * hits2 / n
+    assert est2 == pytest.approx(exact, rel=0.02)
+
+
+def test_gp_end_to_end_gpu_vs_cpu(dev):
+    """Full GP fit+predict on GPU (native path) tracks the CPU fp64 fit."""
+    from dmosopt_amd.models.gp import GPRMatern
+    from dmosopt_amd.benchmarks.problems import zdt1
+
+    rng = np.random.default_rng(8)
+    X = rng.random((150, 30))
+    Y = zdt1(X).numpy()
+    gp_gpu = GPRMatern(X, Y, 30, 2, np.zeros(30), np.ones(30), optimizer="sceua",
+                       seed=1, device=dev)
+    gp_cpu = GPRMatern(X, Y, 30, 2, np.zeros(30), np.ones(30), optimizer="sceua",
+                       seed=1, device="cpu")
+    Xq = rng.random((64, 30))
+    mg, vg = gp_gpu.predict(Xq)
+    mc, vc = gp_cpu.predict(Xq)
+    Ytrue = zdt1(Xq).numpy()
+    rmse_gpu = np.sqrt(((mg - Ytrue) ** 2).mean())
+    rmse_cpu = np.sqrt(((mc - Ytrue) ** 2).mean())
+    assert rmse_gpu < max(2.5 * rmse_cpu, 0.05)
+
+
+def test_bench_one_epoch_runs(dev):
+    from bench import make_archive, one_epoch
+
+    X, Y = make_archive(seed=3)
+    x_res, y_res, hv = one_epoch(X[:100], Y[:100], pop=64, rank=0, world=1,
+                                 device=dev, seed=2, n_gen=5)
+    assert np.isfinite(hv)
