@@ -134,38 +134,42 @@ class NSGA2Optimizer(MOEA):
         if cross_pairs:
             i1 = torch.tensor([a for a, _ in cross_pairs], dtype=torch.long, device=pool.device)
             i2 = torch.tensor([b for _, b in cross_pairs], dtype=torch.long, device=pool.device)
-            children_c1, children_c2 = ops.sbx_crossover_batch(
-                pool[i1], pool[i2], di_c, xlb, xub, generator=self.torch_random
+            children_c1, children_c2 = ops.sbx_from_pool(
+                pool, i1, i2, di_c, xlb, xub,
+                seed=int(rng.integers(0, 2**62)), generator=self.torch_random,
             )
         if mut_parents:
             im = torch.tensor(mut_parents, dtype=torch.long, device=pool.device)
-            children_m = ops.polynomial_mutation_batch(
-                pool[im], di_m, xlb, xub, mutation_rate=p.mutation_rate,
-                generator=self.torch_random,
+            children_m = ops.mutation_from_pool(
+                pool, im, di_m, xlb, xub, p.mutation_rate,
+                seed=int(rng.integers(0, 2**62)), generator=self.torch_random,
             )
 
-        # Assemble children in event order so operator-success tracking has
-        # stable per-event slot indices.
-        rows = []
+        # Assemble children in event order with ONE device gather. Source
+        # tensor rows: [child1 (C) | child2 (C) | mutants (M)].
+        C = len(cross_pairs)
+        src_rows = []
         crossover_indices = []
         mutation_indices = []
         ci = mi = 0
         slot = 0
         for ev in order:
             if ev == "c":
-                rows.append(children_c1[ci : ci + 1])
-                rows.append(children_c2[ci : ci + 1])
+                src_rows.extend([ci, C + ci])
                 crossover_indices.extend([slot, slot + 1])
                 self.state.total_crossovers += 1
                 ci += 1
                 slot += 2
             else:
-                rows.append(children_m[mi : mi + 1])
+                src_rows.append(2 * C + mi)
                 mutation_indices.append(slot)
                 self.state.total_mutations += 1
                 mi += 1
                 slot += 1
-        x_gen = torch.cat(rows, dim=0)
+        parts = [t for t in (children_c1, children_c2, children_m) if t is not None]
+        src = torch.cat(parts, dim=0)
+        gather_idx = torch.tensor(src_rows, dtype=torch.long, device=src.device)
+        x_gen = src[gather_idx]
         return x_gen, {
             "crossover_indices": np.asarray(crossover_indices, dtype=int),
             "mutation_indices": np.asarray(mutation_indices, dtype=int),

@@ -141,6 +141,39 @@ def remove_duplicates(x: torch.Tensor, y: torch.Tensor, eps: float = 1e-16):
     return x[~dup], y[~dup]
 
 
+# ------------------------------------------------------------- variation ops
+def sbx_from_pool(pool, i1, i2, di, lo, hi, seed: int, generator=None):
+    """SBX on pool rows (i1, i2) -> (child1, child2), fused RNG on GPU."""
+    if _use_native(pool) and pool.dtype == torch.float32:
+        return _native.sbx_batch(
+            pool.contiguous(),
+            i1.to(torch.int32).contiguous(),
+            i2.to(torch.int32).contiguous(),
+            di.float().contiguous(),
+            lo.float().contiguous(),
+            hi.float().contiguous(),
+            int(seed),
+        )
+    return sbx_crossover_batch(pool[i1], pool[i2], di, lo, hi, generator=generator)
+
+
+def mutation_from_pool(pool, idx, di, lo, hi, mutation_rate: float, seed: int, generator=None):
+    """Polynomial mutation on pool rows idx, fused RNG on GPU."""
+    if _use_native(pool) and pool.dtype == torch.float32:
+        return _native.mutation_batch(
+            pool.contiguous(),
+            idx.to(torch.int32).contiguous(),
+            di.float().contiguous(),
+            lo.float().contiguous(),
+            hi.float().contiguous(),
+            float(mutation_rate),
+            int(seed),
+        )
+    return polynomial_mutation_batch(
+        pool[idx], di, lo, hi, mutation_rate=mutation_rate, generator=generator
+    )
+
+
 # -------------------------------------------------------------------- GP ops
 def matern_train_kernel(X, theta, nu, anisotropic, jitter):
     """Batched symmetric kernel matrices K (B,N,N) with noise+jitter diag."""

@@ -119,20 +119,27 @@ torch::Tensor pareto_rank(torch::Tensor Y) {
   auto front = torch::empty({N}, Y.options().dtype(torch::kUInt8));
   auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));
   auto n_front = torch::zeros({1}, Y.options().dtype(torch::kInt32));
+  // Peel in chases of CHASE fronts between host syncs: front k+1 depends
+  // only on device state from commit k, so the launches chain on-stream and
+  // one n_front readback per chase amortizes the sync. Extra launches after
+  // exhaustion are no-ops (alive all zero).
+  constexpr int CHASE = 8;
   int remaining = N;
   int k = 0;
-  while (remaining > 0) {
+  while (remaining > 0 && k < N + CHASE) {
     n_front.zero_();
-    launch_peel_front(D.data_ptr<int>(), alive.data_ptr<unsigned char>(),
-                      front.data_ptr<unsigned char>(), n_front.data_ptr<int>(),
-                      N, m, cur_stream());
-    launch_commit_front(front.data_ptr<unsigned char>(),
-                        alive.data_ptr<unsigned char>(), rank.data_ptr<int>(),
-                        k, N, cur_stream());
-    const int nf = n_front.item<int>();  // syncs
-    if (nf == 0) break;                  // safety
+    for (int c = 0; c < CHASE; ++c) {
+      launch_peel_front(D.data_ptr<int>(), alive.data_ptr<unsigned char>(),
+                        front.data_ptr<unsigned char>(),
+                        n_front.data_ptr<int>(), N, m, cur_stream());
+      launch_commit_front(front.data_ptr<unsigned char>(),
+                          alive.data_ptr<unsigned char>(),
+                          rank.data_ptr<int>(), k + c, N, cur_stream());
+    }
+    const int nf = n_front.item<int>();  // total over the chase; syncs
+    if (nf == 0) break;                  // safety against stalls
     remaining -= nf;
-    ++k;
+    k += CHASE;
   }
   return rank.to(torch::kLong);
 }

@@ -48,12 +48,14 @@ __global__ void dominance_matrix_kernel(const float* __restrict__ Y,  // (N, m)
 }
 
 // Zero mutual-domination entries of identical rows: D[i][j] = 0 where
-// D[i][j] == m and D[j][i] == m (reference dda.py:47-49).
+// D[i][j] == m and D[j][i] == m (reference dda.py:47-49). Includes the
+// diagonal (a row is always 'identical' to itself) — without this every
+// column max is >= m and no front would ever peel.
 __global__ void zero_identical_kernel(int* __restrict__ D, int N, int m) {
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= (long long)N * N) return;
   const int i = (int)(idx / N), j = (int)(idx % N);
-  if (i < j && D[(long long)i * N + j] == m && D[(long long)j * N + i] == m) {
+  if (i <= j && D[(long long)i * N + j] == m && D[(long long)j * N + i] == m) {
     D[(long long)i * N + j] = 0;
     D[(long long)j * N + i] = 0;
   }
