@@ -113,9 +113,10 @@ def one_epoch(X, Y, pop, rank, world, device, seed, n_gen=N_GEN):
     n_res = max(1, int(0.25 * pop))
     idx = torch.argsort(Dc, descending=True)[:n_res]
     x_res = best_x[idx]
-    shard = x_res[rank::world] if world > 1 else x_res
-    y_res = zdt1(shard.double())  # on-device synthetic objective
-    hv = hv_2d(res.best_y, np.array([11.0, 11.0]))
+    y_res = zdt1(x_res.double())  # on-device synthetic objective evaluation
+    # final hypervolume of the REAL-evaluated resample batch (the quality
+    # half of the headline metric)
+    hv = hv_2d(y_res.cpu().numpy(), np.array([11.0, 11.0]))
     return x_res, y_res, hv
 
 
