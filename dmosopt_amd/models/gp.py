@@ -118,16 +118,14 @@ class _GPRBase:
         Yn = (Y - y_mean[None, :]) / y_std[None, :]
 
         if optimizer in ("sceua", "dlib", None):
+            YnT = Yn.T.contiguous()  # (m, N)
+
             def nmll_func(theta_batch: torch.Tensor, stream: torch.Tensor):
-                out = torch.empty(theta_batch.shape[0], dtype=self.dtype, device=self.device)
-                for s in range(m):
-                    sel = stream == s
-                    if bool(sel.any()):
-                        out[sel] = batched_nmll(
-                            X, Yn[:, s], theta_batch[sel].to(self.dtype),
-                            nu=self.nu, anisotropic=anisotropic,
-                        )
-                return out
+                # one fused batched call for ALL streams: per-row y targets
+                return batched_nmll(
+                    X, YnT[stream], theta_batch.to(self.dtype),
+                    nu=self.nu, anisotropic=anisotropic,
+                )
 
             bestx, bestf, icall = sceua_batched(
                 nmll_func, bl, bu, nopt, n_streams=m, seed=seed,

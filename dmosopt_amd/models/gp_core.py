@@ -116,16 +116,18 @@ def batched_nmll(
 ) -> torch.Tensor:
     """Negative log marginal likelihood for a batch of theta.
 
-    X: (N, d), y: (N,) standardized. theta: (B, p). Returns (B,).
-    Failed factorizations get +inf (SCE-UA treats them as bad points).
-    ``differentiable=True`` forces the autograd-capable torch path (Adam
-    optimizer); the default dispatches to the fused gfx950 kernels.
+    X: (N, d), y standardized targets: (N,) shared by all candidates, or
+    (B, N) per-candidate rows (mixed-objective batches). theta: (B, p).
+    Returns (B,). Failed factorizations get +inf (SCE-UA treats them as bad
+    points). ``differentiable=True`` forces the autograd-capable torch path
+    (Adam optimizer); the default dispatches to the fused gfx950 kernels.
     """
     N = X.shape[0]
+    B = theta.shape[0]
+    yb = (y[None, :, None].expand(B, N, 1) if y.dim() == 1 else y[:, :, None]).contiguous()
     if differentiable:
         K = build_kernel_torch(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
         L, info = torch.linalg.cholesky_ex(K)
-        yb = y[None, :, None].expand(K.shape[0], N, 1)
         alpha = torch.cholesky_solve(yb, L)
         quad = (yb * alpha).sum(dim=(1, 2))
         logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(dim=-1)
@@ -133,16 +135,13 @@ def batched_nmll(
         from dmosopt_amd import ops
 
         K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
-        L, half_logdet_or_full, info = ops.chol_factor_batched(K)
-        # native path returns sum(log diag L); torch path the same
-        logdet = 2.0 * half_logdet_or_full
-        yb = y[None, :, None].expand(L.shape[0], N, 1).contiguous()
+        L, half_logdet, info = ops.chol_factor_batched(K)
+        # both paths return sum(log diag L)
+        logdet = 2.0 * half_logdet
         z = ops.tri_solve_forward(L, yb)  # L z = y
         quad = (z * z).sum(dim=(1, 2))
     nmll = 0.5 * quad + 0.5 * logdet + 0.5 * N * LOG2PI
-    bad = info != 0
-    if bad.any():
-        nmll = torch.where(bad, torch.full_like(nmll, float("inf")), nmll)
+    nmll = torch.where(info != 0, torch.full_like(nmll, float("inf")), nmll)
     nmll = torch.where(torch.isfinite(nmll), nmll, torch.full_like(nmll, float("inf")))
     return nmll
 

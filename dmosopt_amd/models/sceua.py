@@ -123,6 +123,7 @@ def sceua_batched(
         # partition: position p of complex g sits at row p*ngs+g
         cx = x.view(S, npg, G, nopt).permute(0, 2, 1, 3).contiguous()  # (S,G,npg,nopt)
         cf = xf.view(S, npg, G).permute(0, 2, 1).contiguous()  # (S,G,npg)
+        icall_dev = torch.zeros(S, dtype=torch.float64, device=device)
 
         for _step in range(nspl):
             lcs = torch.as_tensor(_select_simplex(nps, npg, rng), device=device)
@@ -133,31 +134,34 @@ def sceua_batched(
             fw = s_f[:, :, -1]  # (S,G)
             ce = s_pts[:, :, :-1, :].mean(dim=2)
 
-            snew = ce + (ce - sw)
-            oob = ((snew < bl_t) | (snew > bu_t)).any(dim=-1)  # (S,G)
-            if bool(oob.any()):
-                snew[oob] = rand_points((int(oob.sum()),))
-            fnew = func(snew.reshape(S * G, nopt), sid_sg).reshape(S, G)
-            icall[act_np] += G
+            # SPECULATIVE CCE: evaluate reflection, contraction and random
+            # fallback for every complex in ONE batched call, select on
+            # device with the sequential acceptance rule (reflect unless it
+            # fails the worst; then contraction; then random). No host syncs
+            # inside the stage; the actual-evaluation count is accumulated
+            # on device with the sequential semantics.
+            s_ref = ce + (ce - sw)
+            oob = ((s_ref < bl_t) | (s_ref > bu_t)).any(dim=-1, keepdim=True)
+            s_ref = torch.where(oob, rand_points((S, G)), s_ref)
+            s_con = sw + 0.5 * (ce - sw)
+            s_rnd = rand_points((S, G))
+            cand = torch.cat(
+                [s_ref.reshape(S * G, nopt), s_con.reshape(S * G, nopt),
+                 s_rnd.reshape(S * G, nopt)], dim=0)
+            sid3 = torch.cat([sid_sg, sid_sg, sid_sg], dim=0)
+            fall = func(cand, sid3).reshape(3, S, G)
+            f_ref, f_con, f_rnd = fall[0], fall[1], fall[2]
 
-            fail1 = fnew > fw
-            if bool(fail1.any()):
-                scon = sw[fail1] + 0.5 * (ce[fail1] - sw[fail1])  # (F, nopt)
-                sidf = sid_sg.reshape(S, G)[fail1]
-                fcon = func(scon, sidf)
-                snew = snew.clone()
-                fnew = fnew.clone()
-                snew[fail1] = scon
-                fnew[fail1] = fcon
-                icall += (fail1.sum(dim=1).cpu().numpy()) * act_np
-                fail2 = fnew > fw
-                if bool(fail2.any()):
-                    srnd = rand_points((int(fail2.sum()),))
-                    sidf2 = sid_sg.reshape(S, G)[fail2]
-                    frnd = func(srnd, sidf2)
-                    snew[fail2] = srnd
-                    fnew[fail2] = frnd
-                    icall += (fail2.sum(dim=1).cpu().numpy()) * act_np
+            use_con = f_ref > fw
+            use_rnd = use_con & (f_con > fw)
+            fnew = torch.where(use_rnd, f_rnd, torch.where(use_con, f_con, f_ref))
+            snew = torch.where(
+                use_rnd[..., None], s_rnd,
+                torch.where(use_con[..., None], s_con, s_ref))
+            icall_dev += (
+                G + use_con.sum(dim=1).to(torch.float64)
+                + use_rnd.sum(dim=1).to(torch.float64)
+            )
 
             # replace worst simplex point (only for active streams), reinsert
             upd = act[:, None]
@@ -172,6 +176,7 @@ def sceua_batched(
         x = cx.permute(0, 2, 1, 3).reshape(S, npt, nopt)
         xf = cf.permute(0, 2, 1).reshape(S, npt)
         x, xf = sort_pop(x, xf)
+        icall += icall_dev.cpu().numpy().astype(np.int64) * act_np
         gnrng = gnrng_of(x)
 
         bestf_now = xf[:, 0].cpu().numpy()
