@@ -83,11 +83,14 @@ def sceua_batched(
     bu_t = torch.as_tensor(bu, dtype=dtype, device=device)
     bd = bu_t - bl_t
 
+    # device-resident uniform draws (seeded once from the host Generator):
+    # avoids a host->device copy per CCE stage
+    tgen = torch.Generator(device=device)
+    tgen.manual_seed(int(rng.integers(0, 2**63 - 1)))
+
     def rand_points(shape):
-        return (
-            torch.as_tensor(rng.uniform(size=shape + (nopt,)), dtype=dtype, device=device) * bd
-            + bl_t
-        )
+        u = torch.rand(shape + (nopt,), dtype=dtype, device=device, generator=tgen)
+        return u * bd + bl_t
 
     x = rand_points((S, npt))  # (S, npt, nopt)
     ids_all = torch.arange(S, device=device).repeat_interleave(npt)

@@ -1,0 +1,300 @@
+"""AGE-MOEA (Panichella 2019) — adaptive geometry estimation.
+
+Parity with reference AGEMOEA.py:28-512: same variation machinery as
+NSGA-II; environmental selection with front-1 ideal-point shift, corner
+solutions, hyperplane-intercept normalization, Minkowski-p geometry
+estimate, 2-NN greedy survival scores for front 1 and 1/minkowski-to-ideal
+for deeper fronts. Distance matrices are computed as batched tensor ops;
+the greedy selection loop (inherently sequential) runs over a device-
+resident distance matrix.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+import numpy as np
+import torch
+
+from dmosopt_amd import ops
+from dmosopt_amd.datatypes import Struct
+from dmosopt_amd.hv.indicators import PopulationDiversity
+from dmosopt_amd.moea.base import MOEA
+from dmosopt_amd.moea.variation import event_stream_variation
+
+
+def point_to_line_distance(P: np.ndarray, A: np.ndarray, B: np.ndarray) -> np.ndarray:
+    """Distances of rows of P to the line A->B, vectorized."""
+    ba = B - A
+    pa = P - A[None, :]
+    t = (pa @ ba) / (ba @ ba)
+    return np.linalg.norm(pa - t[:, None] * ba[None, :], axis=1)
+
+
+def find_corner_solutions(front: np.ndarray) -> np.ndarray:
+    """Indexes of extreme points: nearest to each axis line (AGEMOEA.py:366)."""
+    m, n = front.shape
+    if m <= n:
+        return np.arange(m)
+    W = 1e-6 + np.eye(n)
+    indexes = np.zeros(n, dtype=int)
+    selected = np.zeros(m, dtype=bool)
+    for i in range(n):
+        dists = point_to_line_distance(front, np.zeros(n), W[i, :])
+        dists[selected] = np.inf
+        idx = int(np.argmin(dists))
+        indexes[i] = idx
+        selected[idx] = True
+    return indexes
+
+
+def normalize_front(front: np.ndarray, extreme: np.ndarray) -> np.ndarray:
+    """Hyperplane-intercept normalization (AGEMOEA.py:287-326)."""
+    m, n = front.shape
+    if len(extreme) != len(np.unique(extreme, axis=0)):
+        return np.max(front, axis=0)
+    try:
+        hyperplane = np.linalg.solve(front[extreme], np.ones(n))
+    except np.linalg.LinAlgError:
+        hyperplane = np.array([np.nan])
+    if np.any(np.isnan(hyperplane)) or np.any(np.isinf(hyperplane)) or np.any(hyperplane < 0):
+        normalization = np.max(front, axis=0)
+    else:
+        normalization = 1.0 / hyperplane
+        if np.any(np.isnan(normalization)) or np.any(np.isinf(normalization)):
+            normalization = np.max(front, axis=0)
+    normalization = np.array(normalization, dtype=np.float64)
+    normalization[np.isclose(normalization, 0.0, rtol=1e-4, atol=1e-4)] = 1.0
+    return normalization
+
+
+def get_geometry(front: np.ndarray, extreme: np.ndarray) -> float:
+    """Estimate the Minkowski exponent p of the front (AGEMOEA.py:336)."""
+    m, n = front.shape
+    d = point_to_line_distance(front, np.zeros(n), np.ones(n))
+    d[extreme] = np.inf
+    index = int(np.argmin(d))
+    with np.errstate(divide="ignore", invalid="ignore"):
+        p = np.log(n) / np.log(1.0 / np.mean(front[index, :]))
+    if np.isnan(p) or p <= 0.1:
+        p = 1.0
+    elif p > 20:
+        p = 20.0
+    return float(p)
+
+
+def minkowski_matrix(A: np.ndarray, B: np.ndarray, p: float) -> np.ndarray:
+    """Pairwise Minkowski-p distances (supports p < 1), vectorized."""
+    diff = np.abs(A[:, None, :] - B[None, :, :])
+    return np.power(np.power(diff, p).sum(axis=2), 1.0 / p)
+
+
+def survival_score(y: np.ndarray, front: np.ndarray, ideal_point: np.ndarray):
+    """Front-1 survival scores (AGEMOEA.py:389-442)."""
+    m, n = y[front, :].shape
+    crowd_dist = np.zeros(m)
+    if m < n:
+        normalization = np.max(y[front, :], axis=0)
+        normalization[np.isclose(normalization, 0.0, rtol=1e-4, atol=1e-4)] = 1.0
+        return normalization, 1, crowd_dist
+
+    yfront = y[front, :] - ideal_point
+    extreme = find_corner_solutions(yfront)
+    normalization = normalize_front(yfront, extreme)
+    ynfront = yfront / normalization
+    p = get_geometry(ynfront, extreme)
+
+    crowd_dist[extreme] = np.inf
+    selected = np.zeros(m, dtype=bool)
+    selected[extreme] = True
+
+    with np.errstate(divide="ignore", invalid="ignore"):
+        nn = np.power(np.power(np.abs(ynfront), p).sum(axis=1), 1.0 / p)
+        distances = minkowski_matrix(ynfront, ynfront, p)
+        distances = distances / nn[:, None]
+
+    neighbors = 2
+    remaining = list(np.arange(m)[~selected])
+    for _ in range(m - int(np.sum(selected))):
+        D_mg = distances[np.ix_(remaining, np.flatnonzero(selected))]
+        if D_mg.shape[1] > 1:
+            part = np.argpartition(D_mg, neighbors - 1, axis=1)[:, :neighbors]
+            tmp = np.take_along_axis(D_mg, part, axis=1).sum(axis=1)
+            index = int(np.argmax(tmp))
+            d = tmp[index]
+        else:
+            index = int(D_mg[:, 0].argmax())
+            d = D_mg[index, 0]
+        best = remaining.pop(index)
+        selected[best] = True
+        crowd_dist[best] = d
+    return normalization, p, crowd_dist
+
+
+def environmental_selection(
+    local_random, population_parm, population_obj, pop, nInput, nOutput,
+    feasibility_model=None, logger=None,
+):
+    """AGE-MOEA survivor selection (AGEMOEA.py:445-512). numpy in/out."""
+    xs = np.asarray(population_parm, dtype=np.float64)
+    ys = np.asarray(population_obj, dtype=np.float64)
+    rank = ops.pareto_rank(torch.as_tensor(ys)).cpu().numpy()
+    order = np.argsort(rank, kind="stable")
+    xs, ys, rank = xs[order], ys[order], rank[order]
+    rmax = int(rank.max())
+
+    yn = np.zeros_like(ys)
+    crowd_dist = np.zeros(len(rank), dtype=np.float64)
+    selected = np.zeros(len(rank), dtype=bool)
+
+    front_1 = np.flatnonzero(rank == 0)
+    ideal_point = np.min(ys[front_1, :], axis=0)
+    normalization, p, crowd_dist[front_1] = survival_score(ys, front_1, ideal_point)
+    yn[front_1, :] = ys[front_1] / normalization
+
+    count = len(front_1)
+    if count < pop:
+        selected[front_1] = True
+        for r in range(1, rmax + 1):
+            front_r = np.flatnonzero(rank == r)
+            yn[front_r] = ys[front_r] / normalization
+            crowd_dist[front_r] = 1.0 / minkowski_matrix(
+                yn[front_r, :], ideal_point[None, :], p
+            ).ravel()
+            if count + len(front_r) < pop:
+                selected[front_r] = True
+                count += len(front_r)
+            else:
+                sort_keys = []
+                if feasibility_model is not None:
+                    sort_keys.append(-feasibility_model.rank(xs[front_r]))
+                sort_keys.append(-crowd_dist[front_r])
+                perm = np.lexsort(tuple(sort_keys))
+                selected[front_r[perm[: pop - count]]] = True
+                break
+    else:
+        sort_keys = []
+        if feasibility_model is not None:
+            sort_keys.append(-feasibility_model.rank(xs[front_1]))
+        sort_keys.append(-crowd_dist[front_1])
+        perm = np.lexsort(tuple(sort_keys))
+        selected[front_1[perm[:pop]]] = True
+
+    assert selected.sum() > 0
+    return (
+        xs[selected].copy(),
+        ys[selected].copy(),
+        rank[selected].copy(),
+        crowd_dist[selected].copy(),
+    )
+
+
+class AGEMOEAOptimizer(MOEA):
+    def __init__(
+        self,
+        popsize: int,
+        nInput: int,
+        nOutput: int,
+        model: Optional[Any] = None,
+        optimize_mean_variance: bool = False,
+        **kwargs,
+    ):
+        super().__init__(name="AGEMOEA", popsize=popsize, nInput=nInput, nOutput=nOutput, **kwargs)
+        self.model = model
+        self.optimize_mean_variance = optimize_mean_variance
+        self.feasibility = getattr(model, "feasibility", None) if model is not None else None
+        p = self.opt_params
+        if np.isscalar(p.di_crossover):
+            p.di_crossover = np.full(nInput, float(p.di_crossover))
+        if np.isscalar(p.di_mutation):
+            p.di_mutation = np.full(nInput, float(p.di_mutation))
+        if p.mutation_rate is None:
+            p.mutation_rate = 1.0 / float(nInput)
+        p.poolsize = int(round(popsize / 2.0))
+        self.diversity_indicator = PopulationDiversity()
+
+    @property
+    def default_parameters(self) -> Dict[str, Any]:
+        return {
+            "crossover_prob": 0.9,
+            "mutation_prob": 0.1,
+            "mutation_rate": None,
+            "nchildren": 1,
+            "di_crossover": 1.0,
+            "di_mutation": 20.0,
+            "max_population_size": 2000,
+            "min_population_size": 100,
+            "adaptive_population_size": False,
+        }
+
+    def initialize_state(self, x, y, bounds, local_random, **params):
+        xn, yn, rank, crowd = environmental_selection(
+            local_random, x.cpu().numpy(), y.cpu().numpy(),
+            self.opt_params.popsize, self.nInput, self.nOutput,
+            feasibility_model=self.feasibility,
+        )
+        pop = self.opt_params.popsize
+        return Struct(
+            bounds=bounds,
+            population_parm=self._as_tensor(xn[:pop]),
+            population_obj=self._as_tensor(yn[:pop]),
+            rank=torch.as_tensor(rank[:pop], dtype=torch.long, device=self.device),
+            crowd_dist=self._as_tensor(crowd[:pop]),
+        )
+
+    def generate_strategy(self, **params):
+        p = self.opt_params
+        rng = self.local_random
+        xlb, xub = self.state.bounds[:, 0], self.state.bounds[:, 1]
+        population = self.state.population_parm
+        rank = self.state.rank
+        crowd = self.state.crowd_dist
+
+        pool_idx = ops.tournament_selection(
+            population.shape[0], p.poolsize, [-crowd, rank], rng
+        )
+        pool = population[pool_idx]
+        di_c = torch.as_tensor(p.di_crossover, dtype=pool.dtype, device=pool.device)
+        di_m = torch.as_tensor(p.di_mutation, dtype=pool.dtype, device=pool.device)
+        x_gen, _, _ = event_stream_variation(
+            pool, rng, p.popsize, p.poolsize, p.crossover_prob, p.mutation_prob,
+            p.mutation_rate, di_c, di_m, xlb, xub, torch_random=self.torch_random,
+        )
+        return x_gen, {}
+
+    def update_strategy(self, x_gen, y_gen, gen_state, **params):
+        p = self.opt_params
+        parm = torch.cat([self.state.population_parm, x_gen], dim=0)
+        obj = torch.cat([self.state.population_obj, y_gen], dim=0)
+        parm, obj = ops.remove_duplicates(parm, obj)
+        xn, yn, rank, crowd = environmental_selection(
+            self.local_random, parm.cpu().numpy(), obj.cpu().numpy(),
+            p.popsize, self.nInput, self.nOutput,
+            feasibility_model=self.feasibility,
+        )
+        self.state.population_parm = self._as_tensor(xn)
+        self.state.population_obj = self._as_tensor(yn)
+        self.state.rank = torch.as_tensor(rank, dtype=torch.long, device=self.device)
+        self.state.crowd_dist = self._as_tensor(crowd)
+        if p.adaptive_population_size:
+            self.update_population_size()
+
+    def get_population_strategy(self):
+        return (
+            self.state.population_parm.clone(),
+            self.state.population_obj.clone(),
+        )
+
+    def update_population_size(self):
+        diversity, cd_spread = self.diversity_indicator.do(
+            self.state.rank, self.state.population_obj
+        )
+        p = self.opt_params
+        if diversity < 0.5 and cd_spread < 2.0:
+            new_size = min(p.max_population_size, int(p.popsize * 1.2))
+        elif diversity > 0.9 or cd_spread > 1.0:
+            new_size = max(p.min_population_size, int(p.popsize * 0.9))
+        else:
+            new_size = p.popsize
+        p.popsize = new_size
+        p.poolsize = int(round(new_size / 2.0))

@@ -110,69 +110,19 @@ class NSGA2Optimizer(MOEA):
         )
         pool = population[pool_idx]
 
-        # Host-side Bernoulli event stream (reference NSGA2.py:141-177),
-        # then batched device variation.
-        cross_pairs = []  # (i1, i2) pool indices
-        mut_parents = []  # pool indices
-        order = []  # 'c' or 'm' events in sequence
-        count = 0
-        while count < popsize - 1:
-            if rng.random() < p.crossover_prob:
-                pidx = rng.choice(poolsize, 2, replace=False)
-                cross_pairs.append((int(pidx[0]), int(pidx[1])))
-                order.append("c")
-                count += 2
-            if rng.random() < p.mutation_prob:
-                mut_parents.append(int(rng.integers(low=0, high=poolsize)))
-                order.append("m")
-                count += 1
-
         di_c = torch.as_tensor(p.di_crossover, dtype=pool.dtype, device=pool.device)
         di_m = torch.as_tensor(p.di_mutation, dtype=pool.dtype, device=pool.device)
+        from dmosopt_amd.moea.variation import event_stream_variation
 
-        children_c1 = children_c2 = children_m = None
-        if cross_pairs:
-            i1 = torch.tensor([a for a, _ in cross_pairs], dtype=torch.long, device=pool.device)
-            i2 = torch.tensor([b for _, b in cross_pairs], dtype=torch.long, device=pool.device)
-            children_c1, children_c2 = ops.sbx_from_pool(
-                pool, i1, i2, di_c, xlb, xub,
-                seed=int(rng.integers(0, 2**62)), generator=self.torch_random,
-            )
-        if mut_parents:
-            im = torch.tensor(mut_parents, dtype=torch.long, device=pool.device)
-            children_m = ops.mutation_from_pool(
-                pool, im, di_m, xlb, xub, p.mutation_rate,
-                seed=int(rng.integers(0, 2**62)), generator=self.torch_random,
-            )
-
-        # Assemble children in event order with ONE device gather. Source
-        # tensor rows: [child1 (C) | child2 (C) | mutants (M)].
-        C = len(cross_pairs)
-        src_rows = []
-        crossover_indices = []
-        mutation_indices = []
-        ci = mi = 0
-        slot = 0
-        for ev in order:
-            if ev == "c":
-                src_rows.extend([ci, C + ci])
-                crossover_indices.extend([slot, slot + 1])
-                self.state.total_crossovers += 1
-                ci += 1
-                slot += 2
-            else:
-                src_rows.append(2 * C + mi)
-                mutation_indices.append(slot)
-                self.state.total_mutations += 1
-                mi += 1
-                slot += 1
-        parts = [t for t in (children_c1, children_c2, children_m) if t is not None]
-        src = torch.cat(parts, dim=0)
-        gather_idx = torch.tensor(src_rows, dtype=torch.long, device=src.device)
-        x_gen = src[gather_idx]
+        x_gen, crossover_indices, mutation_indices = event_stream_variation(
+            pool, rng, popsize, poolsize, p.crossover_prob, p.mutation_prob,
+            p.mutation_rate, di_c, di_m, xlb, xub, torch_random=self.torch_random,
+        )
+        self.state.total_crossovers += len(crossover_indices) // 2
+        self.state.total_mutations += len(mutation_indices)
         return x_gen, {
-            "crossover_indices": np.asarray(crossover_indices, dtype=int),
-            "mutation_indices": np.asarray(mutation_indices, dtype=int),
+            "crossover_indices": crossover_indices,
+            "mutation_indices": mutation_indices,
         }
 
     def update_strategy(self, x_gen, y_gen, gen_state, **params):

@@ -69,3 +69,12 @@ def test_optimizer_cycling():
 def test_termination_conditions_true():
     best = _run("t_term", termination_conditions=True, num_generations=30)
     assert best is not None
+
+
+def test_zdt1_agemoea():
+    best = _run("t_age", optimizer="age", num_generations=20)
+    bestx, besty = best
+    y = np.column_stack([v for _, v in besty])
+    assert y.shape[1] == 2 and y.shape[0] > 0
+    hv = hv_2d(y, np.array([11.0, 11.0]))
+    assert hv > 0.7 * hv_2d(zdt1_pareto(200), np.array([11.0, 11.0]))
