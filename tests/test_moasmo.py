@@ -78,3 +78,23 @@ def test_zdt1_agemoea():
     assert y.shape[1] == 2 and y.shape[0] > 0
     hv = hv_2d(y, np.array([11.0, 11.0]))
     assert hv > 0.7 * hv_2d(zdt1_pareto(200), np.array([11.0, 11.0]))
+
+
+def test_all_optimizers_run():
+    for name in ["smpso", "trs", "cmaes"]:
+        best = _run(f"t_{name}", optimizer=name, num_generations=8,
+                    population_size=30)
+        bestx, besty = best
+        y = np.column_stack([v for _, v in besty])
+        assert y.shape[1] == 2 and y.shape[0] > 0
+
+
+def test_optimizer_cycling_mixed():
+    best = _run(
+        "t_cycle2",
+        optimizer_name=["nsga2", "trs"],
+        optimizer_kwargs=[{"mutation_prob": 0.1, "crossover_prob": 0.9}, {}],
+        num_generations=8,
+        n_epochs=2,
+    )
+    assert best is not None
