@@ -236,7 +236,10 @@ class DistOptimizer:
         assert objective_names is not None
         self.objective_names = objective_names
 
-        has_problem_ids = problem_ids is not None
+        # A restored file always carries problem_ids=[0] for single-problem
+        # runs; treat {0} as the single-problem convention so the objective
+        # keeps its flat-dict signature across resume.
+        has_problem_ids = problem_ids is not None and set(problem_ids) != {0}
         if not has_problem_ids:
             problem_ids = set([0])
 

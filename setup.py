@@ -10,10 +10,13 @@ import os
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from setuptools import setup
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(ROOT, "dmosopt_amd", "ops", "hip")
+
+# libhdf5 location (this image ships it under /opt/conda)
+HDF5_PREFIX = os.environ.get("DMOSOPT_AMD_HDF5_PREFIX", "/opt/conda")
 
 ext = CUDAExtension(
     name="dmosopt_amd._hipops",
@@ -32,10 +35,20 @@ ext = CUDAExtension(
     },
 )
 
+h5ext = CppExtension(
+    name="dmosopt_amd._h5core",
+    sources=[os.path.join("dmosopt_amd", "storage", "h5cpp", "h5core.cpp")],
+    include_dirs=[os.path.join(HDF5_PREFIX, "include")],
+    library_dirs=[os.path.join(HDF5_PREFIX, "lib")],
+    libraries=["hdf5"],
+    extra_compile_args=["-O2", "-std=c++17"],
+    extra_link_args=[f"-Wl,-rpath,{os.path.join(HDF5_PREFIX, 'lib')}"],
+)
+
 setup(
     name="dmosopt_amd",
     version="0.1.0",
     packages=["dmosopt_amd"],
-    ext_modules=[ext],
+    ext_modules=[ext, h5ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )

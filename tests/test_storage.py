@@ -1,0 +1,152 @@
+"""HDF5 storage schema tests (role of reference tests/test_save_to_h5.py):
+enum/compound round-trips, append semantics, full save+restore."""
+
+import os
+
+import numpy as np
+import pytest
+
+import dmosopt_amd
+from dmosopt_amd.datatypes import ParameterSpace
+from dmosopt_amd.storage import h5 as h5store
+
+
+@pytest.fixture
+def space_fixture():
+    space = ParameterSpace.from_dict({"a": [0.0, 1.0], "b": [2.0, 5.0, True]})
+    pp = ParameterSpace.from_dict({"beta": 0.44}, is_value_only=True)
+    return space, pp
+
+
+def test_init_and_reload_types(tmp_path, space_fixture):
+    space, pp = space_fixture
+    fp = str(tmp_path / "t.h5")
+    h5store.init_h5(
+        "opt1", {0}, False, space, ["y1", "y2"], None, None, pp,
+        "meta-string", 42, fp,
+    )
+    seed, max_epoch, old_evals, ps, obj_names, feat, cons, pp2, pids = (
+        h5store.init_from_h5(fp, space.parameter_names, "opt1")
+    )
+    assert seed == 42
+    assert obj_names == ["y1", "y2"]
+    assert ps.parameter_names == ["a", "b"]
+    assert np.allclose(ps.bound1, [0.0, 2.0])
+    assert list(ps.is_integer) == [False, True]
+    assert pp2.parameter_names == ["beta"]
+    assert pp2.parameter_values[0] == pytest.approx(0.44, abs=1e-6)
+    assert pids == {0}
+
+
+def test_save_and_restore_evals(tmp_path, space_fixture):
+    space, pp = space_fixture
+    fp = str(tmp_path / "t.h5")
+    h5store.init_h5("opt1", {0}, False, space, ["y1", "y2"], None, None, pp, None, 7, fp)
+    rng = np.random.default_rng(0)
+    n = 12
+    xs = [rng.random(2) for _ in range(n)]
+    ys = [rng.random(2) for _ in range(n)]
+    preds = [[np.nan, np.nan]] * n
+    epochs = [0] * 6 + [1] * 6
+    h5store.save_to_h5(
+        "opt1", {0}, False, space.parameter_names, ["y1", "y2"], None, None,
+        False, {0: (epochs, xs, ys, None, None, preds)}, fp,
+    )
+    # second append
+    h5store.save_to_h5(
+        "opt1", {0}, False, space.parameter_names, ["y1", "y2"], None, None,
+        False, {0: ([2] * 3, xs[:3], ys[:3], None, None, preds[:3])}, fp,
+    )
+    seed, max_epoch, old_evals, *_ = h5store.init_from_h5(fp, space.parameter_names, "opt1")
+    evs = old_evals[0]
+    assert len(evs) == 15
+    assert max_epoch == 2
+    assert np.allclose(np.asarray(evs[0].parameters), xs[0], atol=1e-6)
+    assert np.allclose(np.asarray(evs[0].objectives), ys[0], atol=1e-6)
+
+
+def test_constraints_and_surrogate_evals(tmp_path):
+    space = ParameterSpace.from_dict({"x1": [0.0, 1.0]})
+    pp = ParameterSpace.from_dict({}, is_value_only=True)
+    fp = str(tmp_path / "c.h5")
+    h5store.init_h5("opt1", {0}, False, space, ["f"], None, ["c1"], pp, None, None, fp)
+    h5store.save_to_h5(
+        "opt1", {0}, False, ["x1"], ["f"], None, ["c1"], False,
+        {0: ([0], [np.array([0.5])], [np.array([1.0])], None,
+             [np.array([0.3])], [[np.nan]])}, fp,
+    )
+    h5store.save_surrogate_evals_to_h5(
+        "opt1", 0, 1, ["x1"], ["f"], np.arange(4, dtype=np.uint32),
+        np.random.random((4, 1)), np.random.random((4, 1)), fp,
+    )
+    _, old_evals, info = h5store.h5_load_all(fp, "opt1")
+    assert info["constraints"] == ["c1"]
+    assert old_evals[0][0].constraints == [pytest.approx(0.3, abs=1e-6)]
+
+
+def test_optimizer_params_and_stats(tmp_path, space_fixture):
+    space, pp = space_fixture
+    fp = str(tmp_path / "p.h5")
+    h5store.init_h5("opt1", {0}, False, space, ["y1"], None, None, pp, None, None, fp)
+    h5store.save_optimizer_params_to_h5(
+        "opt1", 0, 1, "nsga2",
+        {"crossover_prob": 0.9, "popsize": 100, "di_mutation": np.ones(3),
+         "name": "x", "skipme": None},
+        fp,
+    )
+    h5store.save_stats_to_h5("opt1", 0, 1, {"eval_mean": 0.5, "n": 3}, fp)
+
+
+def test_nested_space_roundtrip_h5(tmp_path):
+    space = ParameterSpace.from_dict(
+        {"soma": {"gk": [0.001, 0.1]}, "axon": {"gx": [0.5, 1.5]}}
+    )
+    pp = ParameterSpace.from_dict({}, is_value_only=True)
+    fp = str(tmp_path / "n.h5")
+    h5store.init_h5("opt1", {0}, False, space, ["y"], None, None, pp, None, None, fp)
+    raw_spec, _, info = h5store.h5_load_raw(fp, "opt1")
+    assert raw_spec["soma"]["gk"][0] == pytest.approx(0.001)
+    assert raw_spec["axon"]["gx"][1] == pytest.approx(1.5)
+    ps = ParameterSpace.from_dict(raw_spec)
+    assert ps.parameter_names == ["axon.gx", "soma.gk"]
+
+
+def test_run_with_save_and_resume(tmp_path):
+    fp = str(tmp_path / "run.h5")
+
+    def obj_fun(pp):
+        x = np.array([pp[f"x{i}"] for i in range(3)])
+        return np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+
+    params = {
+        "opt_id": "t_save",
+        "obj_fun": obj_fun,
+        "problem_parameters": {},
+        "space": {f"x{i}": [0.0, 1.0] for i in range(3)},
+        "objective_names": ["f1", "f2"],
+        "population_size": 16,
+        "num_generations": 4,
+        "surrogate_method_name": None,
+        "optimizer": "nsga2",
+        "n_initial": 2,
+        "n_epochs": 1,
+        "random_seed": 5,
+        "file_path": fp,
+        "save": True,
+        "save_eval": 5,
+    }
+    best = dmosopt_amd.run(params, verbose=False)
+    assert best is not None
+    assert os.path.isfile(fp)
+    seed, max_epoch, old_evals, ps, obj_names, *_ = h5store.init_from_h5(
+        fp, [f"x{i}" for i in range(3)], "t_save"
+    )
+    assert seed == 5
+    assert len(old_evals[0]) > 0
+    assert obj_names == ["f1", "f2"]
+
+    # resume: run again with the file present
+    params2 = dict(params)
+    params2["n_epochs"] = 1
+    best2 = dmosopt_amd.run(params2, verbose=False)
+    assert best2 is not None
