@@ -47,14 +47,19 @@ struct H5File {
     void* old_data;
     H5Eget_auto2(H5E_DEFAULT, &old_func, &old_data);
     H5Eset_auto2(H5E_DEFAULT, nullptr, nullptr);  // silence expected probes
+    // STRONG close degree: H5Fclose force-releases any stray object handle
+    // so the library's atexit teardown never spins on open ids.
+    hid_t fapl = H5Pcreate(H5P_FILE_ACCESS);
+    H5Pset_fclose_degree(fapl, H5F_CLOSE_STRONG);
     if (mode == "r") {
-      fid = H5Fopen(p.c_str(), H5F_ACC_RDONLY, H5P_DEFAULT);
+      fid = H5Fopen(p.c_str(), H5F_ACC_RDONLY, fapl);
     } else if (mode == "a" || mode == "r+") {
-      fid = H5Fopen(p.c_str(), H5F_ACC_RDWR, H5P_DEFAULT);
-      if (fid < 0) fid = H5Fcreate(p.c_str(), H5F_ACC_EXCL, H5P_DEFAULT, H5P_DEFAULT);
+      fid = H5Fopen(p.c_str(), H5F_ACC_RDWR, fapl);
+      if (fid < 0) fid = H5Fcreate(p.c_str(), H5F_ACC_EXCL, H5P_DEFAULT, fapl);
     } else if (mode == "w") {
-      fid = H5Fcreate(p.c_str(), H5F_ACC_TRUNC, H5P_DEFAULT, H5P_DEFAULT);
+      fid = H5Fcreate(p.c_str(), H5F_ACC_TRUNC, H5P_DEFAULT, fapl);
     }
+    H5Pclose(fapl);
     H5Eset_auto2(H5E_DEFAULT, old_func, old_data);
     check_id(fid, ("open " + p).c_str());
   }
@@ -413,6 +418,9 @@ struct H5File {
 
 PYBIND11_MODULE(_h5core, m) {
   m.doc() = "Native HDF5 storage core (libhdf5) for dmosopt_amd";
+  // Errors surface as Python exceptions via check()/check_id(); the default
+  // stderr error stack (incl. teardown diagnostics) is noise here.
+  H5Eset_auto2(H5E_DEFAULT, nullptr, nullptr);
   py::class_<H5File>(m, "H5File")
       .def(py::init<const std::string&, const std::string&>())
       .def("close", &H5File::close)
