@@ -36,17 +36,22 @@ def pairwise_sq_dists(X1: torch.Tensor, X2: torch.Tensor) -> torch.Tensor:
 
 
 def matern_from_d2(d2: torch.Tensor, nu: float) -> torch.Tensor:
-    """Matern correlation from squared scaled distance r^2 = d2."""
+    """Matern correlation from squared scaled distance r^2 = d2.
+
+    sqrt is taken of d2 + tiny so autograd through the kernel diagonal
+    (d2 == 0) yields finite gradients instead of the inf * 0 = NaN that
+    d(sqrt)/d(d2)|_0 produces; the value perturbation is ~1e-12 in r.
+    """
     if nu == 2.5:
-        r = torch.sqrt(d2)
+        r = torch.sqrt(d2 + 1e-24)
         s = SQRT5 * r
         return (1.0 + s + (5.0 / 3.0) * d2) * torch.exp(-s)
     if nu == 1.5:
-        r = torch.sqrt(d2)
+        r = torch.sqrt(d2 + 1e-24)
         s = SQRT3 * r
         return (1.0 + s) * torch.exp(-s)
     if nu == 0.5:
-        return torch.exp(-torch.sqrt(d2))
+        return torch.exp(-torch.sqrt(d2 + 1e-24))
     if nu == float("inf") or nu is None:  # RBF
         return torch.exp(-0.5 * d2)
     raise ValueError(f"Unsupported Matern nu={nu}")
