@@ -51,4 +51,11 @@ setup(
     packages=["dmosopt_amd"],
     ext_modules=[ext, h5ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+    entry_points={
+        "console_scripts": [
+            "dmosopt-analyze=dmosopt_amd.cli.analyze:main",
+            "dmosopt-train=dmosopt_amd.cli.train:main",
+            "dmosopt-onestep=dmosopt_amd.cli.onestep:main",
+        ]
+    },
 )

@@ -1,0 +1,112 @@
+"""CLI, constrained sampling, and discrepancy tests."""
+
+import numpy as np
+import pytest
+
+import dmosopt_amd
+from dmosopt_amd.sampling import discrepancy
+from dmosopt_amd.sampling.constrained import ParamSpacePoints, safe_eval_arith
+
+
+def _make_run_file(tmp_path, opt_id="t_cli"):
+    fp = str(tmp_path / "cli.h5")
+
+    def obj_fun(pp):
+        x = np.array([pp[f"x{i}"] for i in range(3)])
+        return np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+
+    params = {
+        "opt_id": opt_id,
+        "obj_fun": obj_fun,
+        "problem_parameters": {},
+        "space": {f"x{i}": [0.0, 1.0] for i in range(3)},
+        "objective_names": ["f1", "f2"],
+        "population_size": 16,
+        "num_generations": 4,
+        "surrogate_method_name": None,
+        "optimizer": "nsga2",
+        "n_initial": 2,
+        "n_epochs": 1,
+        "random_seed": 5,
+        "file_path": fp,
+        "save": True,
+        "save_eval": 5,
+    }
+    dmosopt_amd.run(params, verbose=False)
+    return fp
+
+
+def test_analyze_cli(tmp_path, capsys):
+    from dmosopt_amd.cli.analyze import main
+
+    fp = _make_run_file(tmp_path)
+    rc = main(["-p", fp, "--opt-id", "t_cli", "--sort-key", "f1", "--knn", "5"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "non-dominated solutions" in out
+
+
+def test_train_cli(tmp_path, capsys):
+    from dmosopt_amd.cli.train import main
+
+    fp = _make_run_file(tmp_path, "t_cli2")
+    rc = main(["-p", fp, "--opt-id", "t_cli2", "--surrogate-method", "gpr"])
+    assert rc == 0
+    assert "MAE" in capsys.readouterr().out
+
+
+def test_onestep_cli(tmp_path, capsys):
+    from dmosopt_amd.cli.onestep import main
+
+    fp = _make_run_file(tmp_path, "t_cli3")
+    rc = main(
+        ["-p", fp, "--opt-id", "t_cli3", "--population-size", "16",
+         "--num-generations", "4", "--seed", "3"]
+    )
+    assert rc == 0
+    assert "resample candidates" in capsys.readouterr().out
+
+
+# ----------------------------------------------------- constrained sampling
+def test_safe_eval_arith():
+    assert safe_eval_arith("0.5 * 2 + 1") == pytest.approx(2.0)
+    with pytest.raises(Exception):
+        safe_eval_arith("__import__('os')")
+
+
+def test_constrained_sampling_dag():
+    space = {
+        "a": [0.0, 1.0],
+        "b": {"abs": (0.0, 5.0), "lb": [("a", "* 2")], "ub": [("a", "* 2 + 1")],
+              "method": ("uniform",)},
+        "c": {"abs": (0.0, 10.0), "lb": [("b", "+ 0.5")], "method": ("uniform",)},
+    }
+    psp = ParamSpacePoints(50, space, seed=1)
+    vals = psp.as_dict()
+    a, b, c = vals["a"], vals["b"], vals["c"]
+    assert ((b >= 2 * a - 1e-9) & (b <= 2 * a + 1 + 1e-9)).all()
+    assert (c >= b + 0.5 - 1e-9).all() and (c <= 10.0 + 1e-9).all()
+
+
+def test_constrained_circular_raises():
+    space = {
+        "a": {"abs": (0, 1), "lb": [("b", "+ 0")], "method": ("uniform",)},
+        "b": {"abs": (0, 1), "lb": [("a", "+ 0")], "method": ("uniform",)},
+    }
+    with pytest.raises(ValueError):
+        ParamSpacePoints(10, space, seed=1)
+
+
+# ------------------------------------------------------------- discrepancy
+def test_discrepancy_metrics_prefer_lh(rng):
+    from dmosopt_amd import sampling as S
+
+    mc = S.mc(64, 4, np.random.default_rng(1))
+    lh = S.lh(64, 4, np.random.default_rng(1))
+    assert discrepancy.CD2(lh) < discrepancy.CD2(mc)
+    assert discrepancy.MD2(lh) < discrepancy.MD2(mc)
+    assert discrepancy.MinDist(lh) > 0
+    assert 0 <= discrepancy.corrscore(lh) <= 1
+    for fn in (discrepancy.SD2, discrepancy.WD2):
+        v = fn(lh)
+        assert np.isfinite(v) and v >= 0
