@@ -1,6 +1,7 @@
 """Summarize a rocprofv3 rocpd .db: per-kernel call count + time.
 
 Usage: python scripts_kernel_stats.py <results.db> [out.txt]
+Handles the GUID-suffixed table names of rocprofv3 7.x rocpd databases.
 """
 
 import sqlite3
@@ -12,38 +13,31 @@ def main():
     out_path = sys.argv[2] if len(sys.argv) > 2 else None
     con = sqlite3.connect(db)
     cur = con.cursor()
-    tables = [r[0] for r in cur.execute(
-        "SELECT name FROM sqlite_master WHERE type='table'"
-    )]
-    # rocpd schema: kernel dispatch rows reference kernel name via string table
-    q = None
-    if "rocpd_kernel_dispatch" in tables:
-        cols = [r[1] for r in cur.execute("PRAGMA table_info(rocpd_kernel_dispatch)")]
-        # find plausible column names
-        name_join = None
-        if "kernel_id" in cols and "rocpd_info_kernel_symbol" in tables:
-            kcols = [r[1] for r in cur.execute("PRAGMA table_info(rocpd_info_kernel_symbol)")]
-            sym = "display_name" if "display_name" in kcols else (
-                "kernel_name" if "kernel_name" in kcols else kcols[1]
-            )
-            q = f"""
-            SELECT k.{sym} AS name, COUNT(*) AS calls,
-                   SUM(d.end - d.start) AS total_ns,
-                   AVG(d.end - d.start) AS mean_ns
-            FROM rocpd_kernel_dispatch d
-            JOIN rocpd_info_kernel_symbol k ON d.kernel_id = k.id
-            GROUP BY name ORDER BY total_ns DESC LIMIT 40
-            """
-    if q is None:
-        print("tables:", tables)
+    tables = [r[0] for r in cur.execute("SELECT name FROM sqlite_master WHERE type='table'")]
+
+    def find(prefix):
         for t in tables:
-            cols = [r[1] for r in cur.execute(f"PRAGMA table_info({t})")]
-            print(t, cols)
-        return
+            if t.startswith(prefix):
+                return t
+        raise RuntimeError(f"table {prefix}* not found in {tables}")
+
+    disp = find("rocpd_kernel_dispatch")
+    sym = find("rocpd_info_kernel_symbol")
+    q = f"""
+    SELECT k.display_name AS name, COUNT(*) AS calls,
+           SUM(d.end - d.start) AS total_ns,
+           AVG(d.end - d.start) AS mean_ns
+    FROM {disp} d
+    JOIN {sym} k ON d.kernel_id = k.id
+    GROUP BY name ORDER BY total_ns DESC LIMIT 45
+    """
     lines = [f"{'calls':>8} {'total_ms':>12} {'mean_us':>10}  kernel"]
+    total_gpu_ns = 0
     for name, calls, total_ns, mean_ns in cur.execute(q):
-        short = str(name).split("(")[0][:100]
+        total_gpu_ns += total_ns
+        short = str(name).split("(")[0][:110]
         lines.append(f"{calls:>8} {total_ns / 1e6:>12.3f} {mean_ns / 1e3:>10.2f}  {short}")
+    lines.insert(1, f"# total GPU kernel time: {total_gpu_ns / 1e6:.1f} ms")
     text = "\n".join(lines)
     print(text)
     if out_path:
