@@ -4,118 +4,170 @@
 // reference model.py:1246-1265).
 //
 // Design: ONE workgroup (256 threads) per matrix, right-looking blocked
-// factorization with the BSxBS diagonal block staged in LDS. Sized for the
-// MO-ASMO regime: B = (objectives x SCE-UA complexes) matrices of
-// N <= ~4k factorized concurrently, one block each — batch parallelism
-// fills the 256 CUs, so per-matrix ILP matters less than launch count.
-// All math exact fp32 (f32 VALU); failure (non-PD pivot) is recorded per
-// batch in info[] and the pivot clamped so the factorization completes.
+// factorization (BS = 32). v2 data path (gfx950):
+//   * diagonal block factored in LDS (stride-33 rows: conflict-free),
+//   * panel solve with each ROW held in 32 VGPRs (float4 global loads),
+//     back-substituted against the LDS diagonal block,
+//   * trailing SYRK update C -= P P^T fed ENTIRELY from LDS: the panel is
+//     staged in chunk pairs (CHUNK x 32 floats each) so every multiply
+//     reads LDS, with 2x2 register tiles per thread.
+// Sized for the MO-ASMO regime: B = objectives x SCE-UA complexes matrices
+// of N <= ~4k factorized concurrently, one CU each; batch parallelism fills
+// the chip. Exact fp32; non-PD pivots clamp and set info[b].
 
 #include "common.h"
 #include <math.h>
 
 #define CHOL_BS 32
 #define CHOL_TPB 256
+#define CHOL_CHUNK 384  // panel rows staged per LDS buffer (384*32*4 = 48 KiB)
 
-__global__ void cholesky_batched_kernel(float* __restrict__ A,  // (B, N, N)
-                                        float* __restrict__ logdet,  // (B,)
-                                        int* __restrict__ info,      // (B,)
-                                        int N) {
-  __shared__ float S[CHOL_BS][CHOL_BS + 1];  // +1 pad: stride-33 banks
-  __shared__ float ld_accum;
+struct CholLds {
+  float S[CHOL_BS][CHOL_BS + 1];  // diagonal block, padded row stride
+  float Pi[CHOL_CHUNK][CHOL_BS];  // i-side panel chunk
+  float Pj[CHOL_CHUNK][CHOL_BS];  // j-side panel chunk
+  float ld_accum;
+};
+
+// ~100 KiB of LDS: above the 64 KiB static default, so allocated
+// dynamically (gfx950 has 160 KiB per CU).
+__global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
+    float* __restrict__ A,       // (B, N, N)
+    float* __restrict__ logdet,  // (B,)
+    int* __restrict__ info,      // (B,)
+    int N) {
+  extern __shared__ char smem[];
+  CholLds& L = *reinterpret_cast<CholLds*>(smem);
   const int b = blockIdx.x;
   float* M = A + (long long)b * N * N;
   const int tid = threadIdx.x;
-  if (tid == 0) ld_accum = 0.f;
+  if (tid == 0) L.ld_accum = 0.f;
   __syncthreads();
 
   for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
     const int bs = min(CHOL_BS, N - k0);
 
-    // --- load diagonal block
+    // --- load + factor diagonal block in LDS
     for (int idx = tid; idx < bs * bs; idx += CHOL_TPB)
-      S[idx / bs][idx % bs] = M[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
+      L.S[idx / bs][idx % bs] = M[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
     __syncthreads();
-
-    // --- factor diagonal block (small sequential-over-column loop)
     for (int j = 0; j < bs; ++j) {
       if (tid == 0) {
-        float djj = S[j][j];
+        float djj = L.S[j][j];
         if (djj <= 0.f) {
-          atomicExch(&info[b], 1);
+          info[b] = 1;
           djj = 1e-30f;
         }
-        S[j][j] = sqrtf(djj);
-        ld_accum += __logf(S[j][j]);
+        L.S[j][j] = sqrtf(djj);
+        L.ld_accum += __logf(L.S[j][j]);
       }
       __syncthreads();
-      // column scale + rank-1 update of the remaining block, parallel
-      for (int i = j + 1 + tid; i < bs; i += CHOL_TPB) S[i][j] /= S[j][j];
+      for (int i = j + 1 + tid; i < bs; i += CHOL_TPB) L.S[i][j] /= L.S[j][j];
       __syncthreads();
-      for (int idx = tid; idx < (bs - j - 1) * (bs - j - 1); idx += CHOL_TPB) {
-        const int r = j + 1 + idx / (bs - j - 1);
-        const int c = j + 1 + idx % (bs - j - 1);
-        if (c <= r) S[r][c] -= S[r][j] * S[c][j];
+      const int rem_d = bs - j - 1;
+      for (int idx = tid; idx < rem_d * rem_d; idx += CHOL_TPB) {
+        const int r = j + 1 + idx / rem_d;
+        const int c = j + 1 + idx % rem_d;
+        if (c <= r) L.S[r][c] -= L.S[r][j] * L.S[c][j];
       }
       __syncthreads();
     }
-
-    // --- write back diagonal block (lower); zero strict upper
     for (int idx = tid; idx < bs * bs; idx += CHOL_TPB) {
       const int r = idx / bs, c = idx % bs;
-      M[(long long)(k0 + r) * N + k0 + c] = (c <= r) ? S[r][c] : 0.f;
+      M[(long long)(k0 + r) * N + k0 + c] = (c <= r) ? L.S[r][c] : 0.f;
     }
     __syncthreads();
 
     const int rem = N - k0 - bs;
     if (rem <= 0) continue;
 
-    // --- panel solve: rows i in [k0+bs, N): row_i <- row_i * L11^-T
+    // --- panel solve: row_i <- row_i * L11^-T with the row in registers.
+    // float4 loads need 16-B alignment: row offset i*N + k0 is 16-B aligned
+    // for all i iff N % 4 == 0 (k0 is a multiple of 32).
+    const bool aligned4 = (N & 3) == 0;
     for (int i = k0 + bs + tid; i < N; i += CHOL_TPB) {
-      float* row = M + (long long)i * N + k0;
+      float* grow = M + (long long)i * N + k0;
+      float row[CHOL_BS];
+      if (aligned4) {
+#pragma unroll
+        for (int t = 0; t < CHOL_BS; t += 4) {
+          const float4 v = *(const float4*)(grow + t);
+          row[t] = v.x; row[t + 1] = v.y; row[t + 2] = v.z; row[t + 3] = v.w;
+        }
+      } else {
+#pragma unroll
+        for (int t = 0; t < CHOL_BS; ++t) row[t] = grow[t];
+      }
       for (int j = 0; j < bs; ++j) {
         float v = row[j];
-        for (int t = 0; t < j; ++t) v -= row[t] * S[j][t];
-        row[j] = v / S[j][j];
+        for (int t = 0; t < j; ++t) v = fmaf(-row[t], L.S[j][t], v);
+        row[j] = v / L.S[j][j];
+      }
+      if (aligned4) {
+#pragma unroll
+        for (int t = 0; t < CHOL_BS; t += 4) {
+          float4 v;
+          v.x = row[t]; v.y = row[t + 1]; v.z = row[t + 2]; v.w = row[t + 3];
+          *(float4*)(grow + t) = v;
+        }
+      } else {
+#pragma unroll
+        for (int t = 0; t < CHOL_BS; ++t) grow[t] = row[t];
       }
     }
     __syncthreads();
 
-    // --- trailing update: A22 -= P P^T (P = panel rows), lower triangle.
-    // Each thread owns a 2x2 patch of the trailing block; panel rows come
-    // through L2 (the panel is re-read ~rem/64 times; N<=4k keeps it hot).
-    const int tiles = (rem + 1) / 2;
-    for (long long t = tid; t < (long long)tiles * tiles; t += CHOL_TPB) {
-      const int ti = (int)(t / tiles);
-      const int tj = (int)(t % tiles);
-      if (tj > ti) continue;  // lower-triangular tiles only
-      const int i0 = k0 + bs + ti * 2;
-      const int j0 = k0 + bs + tj * 2;
-      float acc[2][2] = {{0.f, 0.f}, {0.f, 0.f}};
-      const float* Pi0 = M + (long long)i0 * N + k0;
-      const float* Pi1 = M + (long long)min(i0 + 1, N - 1) * N + k0;
-      const float* Pj0 = M + (long long)j0 * N + k0;
-      const float* Pj1 = M + (long long)min(j0 + 1, N - 1) * N + k0;
-      for (int s = 0; s < bs; ++s) {
-        const float a0 = Pi0[s], a1 = Pi1[s];
-        const float b0 = Pj0[s], b1 = Pj1[s];
-        acc[0][0] = fmaf(a0, b0, acc[0][0]);
-        acc[0][1] = fmaf(a0, b1, acc[0][1]);
-        acc[1][0] = fmaf(a1, b0, acc[1][0]);
-        acc[1][1] = fmaf(a1, b1, acc[1][1]);
+    // --- trailing SYRK: A22 -= P P^T, LDS-fed chunk pairs, lower triangle
+    for (int jc = 0; jc < rem; jc += CHOL_CHUNK) {
+      const int jrows = min(CHOL_CHUNK, rem - jc);
+      for (int idx = tid; idx < jrows * CHOL_BS; idx += CHOL_TPB) {
+        const int r = idx / CHOL_BS, c = idx % CHOL_BS;
+        L.Pj[r][c] = M[(long long)(k0 + bs + jc + r) * N + k0 + c];
       }
-#pragma unroll
-      for (int di = 0; di < 2; ++di)
-#pragma unroll
-        for (int dj = 0; dj < 2; ++dj) {
-          const int gi = i0 + di, gj = j0 + dj;
-          if (gi < N && gj < N && gj <= gi)
-            M[(long long)gi * N + gj] -= acc[di][dj];
+      __syncthreads();
+      for (int ic = jc; ic < rem; ic += CHOL_CHUNK) {
+        const int irows = min(CHOL_CHUNK, rem - ic);
+        const bool same = (ic == jc);
+        if (!same) {
+          for (int idx = tid; idx < irows * CHOL_BS; idx += CHOL_TPB) {
+            const int r = idx / CHOL_BS, c = idx % CHOL_BS;
+            L.Pi[r][c] = M[(long long)(k0 + bs + ic + r) * N + k0 + c];
+          }
         }
+        __syncthreads();
+        const float(*PI)[CHOL_BS] = same ? L.Pj : L.Pi;
+        // 2x2 register tiles over (irows x jrows)
+        const int ti = (irows + 1) / 2, tj = (jrows + 1) / 2;
+        for (int t = tid; t < ti * tj; t += CHOL_TPB) {
+          const int tir = t / tj, tjr = t % tj;
+          const int i0 = tir * 2, j0 = tjr * 2;
+          const int gi0 = k0 + bs + ic + i0;
+          const int gj0 = k0 + bs + jc + j0;
+          if (same && gj0 > gi0 + 1) continue;  // fully-upper tile
+          float acc00 = 0.f, acc01 = 0.f, acc10 = 0.f, acc11 = 0.f;
+          const int i1 = min(i0 + 1, irows - 1);
+          const int j1 = min(j0 + 1, jrows - 1);
+#pragma unroll 8
+          for (int s = 0; s < CHOL_BS; ++s) {
+            const float a0 = PI[i0][s], a1 = PI[i1][s];
+            const float b0 = L.Pj[j0][s], b1 = L.Pj[j1][s];
+            acc00 = fmaf(a0, b0, acc00);
+            acc01 = fmaf(a0, b1, acc01);
+            acc10 = fmaf(a1, b0, acc10);
+            acc11 = fmaf(a1, b1, acc11);
+          }
+          const int gi1 = k0 + bs + ic + i1;
+          const int gj1 = k0 + bs + jc + j1;
+          if (gj0 <= gi0) M[(long long)gi0 * N + gj0] -= acc00;
+          if (j1 != j0 && gj1 <= gi0) M[(long long)gi0 * N + gj1] -= acc01;
+          if (i1 != i0 && gj0 <= gi1) M[(long long)gi1 * N + gj0] -= acc10;
+          if (i1 != i0 && j1 != j0 && gj1 <= gi1) M[(long long)gi1 * N + gj1] -= acc11;
+        }
+        __syncthreads();
+      }
     }
-    __syncthreads();
   }
-  if (tid == 0) logdet[b] = ld_accum;
+  if (tid == 0) logdet[b] = L.ld_accum;
 }
 
 // Forward substitution: solve L z = y for R right-hand sides.
@@ -128,9 +180,6 @@ __global__ void forward_solve_batched_kernel(const float* __restrict__ L,
   if (r >= R) return;
   const float* Lb = L + (long long)b * N * N;
   float* y = Y + (long long)b * N * R;
-  // each (b, r) column handled by one thread-row walking sequentially;
-  // threads in x cooperate on the AXPY update
-  // simple layout: threadIdx.x == 0 does division; all do updates
   for (int j = 0; j < N; ++j) {
     const float zj = y[j * R + r] / Lb[(long long)j * N + j];
     __syncthreads();
@@ -162,8 +211,15 @@ __global__ void backward_solve_batched_kernel(const float* __restrict__ L,
 
 extern "C" void launch_cholesky_batched(float* A, float* logdet, int* info,
                                         int B, int N, hipStream_t stream) {
-  hipLaunchKernelGGL(cholesky_batched_kernel, dim3(B), dim3(CHOL_TPB), 0,
-                     stream, A, logdet, info, N);
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)cholesky_batched_kernel,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        (int)sizeof(CholLds));
+    attr_set = true;
+  }
+  hipLaunchKernelGGL(cholesky_batched_kernel, dim3(B), dim3(CHOL_TPB),
+                     sizeof(CholLds), stream, A, logdet, info, N);
 }
 
 extern "C" void launch_forward_solve_batched(const float* L, float* Y, int B,
