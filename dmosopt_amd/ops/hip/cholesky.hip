@@ -23,9 +23,11 @@
 #define CHOL_CHUNK 384  // panel rows staged per LDS buffer (384*32*4 = 48 KiB)
 
 struct CholLds {
-  float S[CHOL_BS][CHOL_BS + 1];  // diagonal block, padded row stride
-  float Pi[CHOL_CHUNK][CHOL_BS];  // i-side panel chunk
-  float Pj[CHOL_CHUNK][CHOL_BS];  // j-side panel chunk
+  // all row strides padded +1: an unpadded 32-float (128 B) stride puts
+  // every row on the same b32 bank group -> 32-way conflicts in the SYRK
+  float S[CHOL_BS][CHOL_BS + 1];
+  float Pi[CHOL_CHUNK][CHOL_BS + 1];  // i-side panel chunk
+  float Pj[CHOL_CHUNK][CHOL_BS + 1];  // j-side panel chunk
   float ld_accum;
 };
 
