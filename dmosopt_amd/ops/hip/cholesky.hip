@@ -137,7 +137,7 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
           }
         }
         __syncthreads();
-        const float(*PI)[CHOL_BS] = same ? L.Pj : L.Pi;
+        const float(*PI)[CHOL_BS + 1] = same ? L.Pj : L.Pi;
         // 2x2 register tiles over (irows x jrows)
         const int ti = (irows + 1) / 2, tj = (jrows + 1) / 2;
         for (int t = tid; t < ti * tj; t += CHOL_TPB) {
