@@ -22,14 +22,12 @@
 #define CHOL_TPB 256
 #define CHOL_CHUNK 384  // panel rows staged per LDS buffer (384*32*4 = 48 KiB)
 
-#define CHOL_PSTRIDE (CHOL_BS + 4)  // 36 floats: 16-B aligned rows (float4
-// LDS reads) AND bank-spread — an unpadded 32-float (128 B) stride puts
-// every row on the same bank group (32-way conflicts in the SYRK)
-
 struct CholLds {
+  // all row strides padded +1: an unpadded 32-float (128 B) stride puts
+  // every row on the same b32 bank group -> 32-way conflicts in the SYRK
   float S[CHOL_BS][CHOL_BS + 1];
-  float Pi[CHOL_CHUNK][CHOL_PSTRIDE];  // i-side panel chunk
-  float Pj[CHOL_CHUNK][CHOL_PSTRIDE];  // j-side panel chunk
+  float Pi[CHOL_CHUNK][CHOL_BS + 1];  // i-side panel chunk
+  float Pj[CHOL_CHUNK][CHOL_BS + 1];  // j-side panel chunk
   float ld_accum;
 };
 
@@ -121,67 +119,51 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
     }
     __syncthreads();
 
-    // --- trailing SYRK: A22 -= P P^T, LDS-fed chunk pairs, lower triangle.
-    // 4x4 register tiles, float4 LDS reads; chunks zero-padded to a
-    // multiple of 4 rows so the inner loop has no ragged-edge branches
-    // (guarded writes handle validity).
+    // --- trailing SYRK: A22 -= P P^T, LDS-fed chunk pairs, lower triangle
     for (int jc = 0; jc < rem; jc += CHOL_CHUNK) {
       const int jrows = min(CHOL_CHUNK, rem - jc);
-      const int jrows_pad = (jrows + 3) & ~3;
-      for (int idx = tid; idx < jrows_pad * CHOL_BS; idx += CHOL_TPB) {
+      for (int idx = tid; idx < jrows * CHOL_BS; idx += CHOL_TPB) {
         const int r = idx / CHOL_BS, c = idx % CHOL_BS;
-        L.Pj[r][c] =
-            (r < jrows) ? M[(long long)(k0 + bs + jc + r) * N + k0 + c] : 0.f;
+        L.Pj[r][c] = M[(long long)(k0 + bs + jc + r) * N + k0 + c];
       }
       __syncthreads();
       for (int ic = jc; ic < rem; ic += CHOL_CHUNK) {
         const int irows = min(CHOL_CHUNK, rem - ic);
-        const int irows_pad = (irows + 3) & ~3;
         const bool same = (ic == jc);
         if (!same) {
-          for (int idx = tid; idx < irows_pad * CHOL_BS; idx += CHOL_TPB) {
+          for (int idx = tid; idx < irows * CHOL_BS; idx += CHOL_TPB) {
             const int r = idx / CHOL_BS, c = idx % CHOL_BS;
-            L.Pi[r][c] =
-                (r < irows) ? M[(long long)(k0 + bs + ic + r) * N + k0 + c] : 0.f;
+            L.Pi[r][c] = M[(long long)(k0 + bs + ic + r) * N + k0 + c];
           }
         }
         __syncthreads();
-        const float(*PI)[CHOL_PSTRIDE] = same ? L.Pj : L.Pi;
-        const int ti = irows_pad / 4, tj = jrows_pad / 4;
+        const float(*PI)[CHOL_BS + 1] = same ? L.Pj : L.Pi;
+        // 2x2 register tiles over (irows x jrows)
+        const int ti = (irows + 1) / 2, tj = (jrows + 1) / 2;
         for (int t = tid; t < ti * tj; t += CHOL_TPB) {
-          const int i0 = (t / tj) * 4, j0 = (t % tj) * 4;
+          const int tir = t / tj, tjr = t % tj;
+          const int i0 = tir * 2, j0 = tjr * 2;
           const int gi0 = k0 + bs + ic + i0;
           const int gj0 = k0 + bs + jc + j0;
-          if (same && gj0 > gi0 + 3) continue;  // fully-upper tile
-          float acc[4][4] = {};
-#pragma unroll
-          for (int s = 0; s < CHOL_BS; s += 4) {
-            float4 av[4], bv[4];
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              av[r] = *(const float4*)&PI[i0 + r][s];
-              bv[r] = *(const float4*)&L.Pj[j0 + r][s];
-            }
-#pragma unroll
-            for (int r = 0; r < 4; ++r)
-#pragma unroll
-              for (int c = 0; c < 4; ++c) {
-                acc[r][c] = fmaf(av[r].x, bv[c].x, acc[r][c]);
-                acc[r][c] = fmaf(av[r].y, bv[c].y, acc[r][c]);
-                acc[r][c] = fmaf(av[r].z, bv[c].z, acc[r][c]);
-                acc[r][c] = fmaf(av[r].w, bv[c].w, acc[r][c]);
-              }
+          if (same && gj0 > gi0 + 1) continue;  // fully-upper tile
+          float acc00 = 0.f, acc01 = 0.f, acc10 = 0.f, acc11 = 0.f;
+          const int i1 = min(i0 + 1, irows - 1);
+          const int j1 = min(j0 + 1, jrows - 1);
+#pragma unroll 8
+          for (int s = 0; s < CHOL_BS; ++s) {
+            const float a0 = PI[i0][s], a1 = PI[i1][s];
+            const float b0 = L.Pj[j0][s], b1 = L.Pj[j1][s];
+            acc00 = fmaf(a0, b0, acc00);
+            acc01 = fmaf(a0, b1, acc01);
+            acc10 = fmaf(a1, b0, acc10);
+            acc11 = fmaf(a1, b1, acc11);
           }
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            if (i0 + r >= irows) break;
-            const long long gi = gi0 + r;
-#pragma unroll
-            for (int c = 0; c < 4; ++c) {
-              const long long gj = gj0 + c;
-              if (j0 + c < jrows && gj <= gi) M[gi * N + gj] -= acc[r][c];
-            }
-          }
+          const int gi1 = k0 + bs + ic + i1;
+          const int gj1 = k0 + bs + jc + j1;
+          if (gj0 <= gi0) M[(long long)gi0 * N + gj0] -= acc00;
+          if (j1 != j0 && gj1 <= gi0) M[(long long)gi0 * N + gj1] -= acc01;
+          if (i1 != i0 && gj0 <= gi1) M[(long long)gi1 * N + gj0] -= acc10;
+          if (i1 != i0 && j1 != j0 && gj1 <= gi1) M[(long long)gi1 * N + gj1] -= acc11;
         }
         __syncthreads();
       }
