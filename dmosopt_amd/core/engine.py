@@ -450,6 +450,65 @@ def get_best(
     return best_x, best_y, best_f, best_c, best_epoch, perm
 
 
+def get_feasible(x, y, f, c, nInput, nOutput, epochs=None):
+    """Feasible filter + full rank/epoch cross-indexing (MOASMO.py:642-700).
+
+    Returns (perm_arrs, rnk_arrs, epc_arrs, rnk_epc_idx): the permuted
+    arrays, unique-rank and unique-epoch index groupings, and their
+    intersection matrix.
+    """
+    xtmp, ytmp = np.copy(x), np.copy(y)
+    feas = None
+    if c is not None:
+        feas = np.argwhere(np.all(c > 0.0, axis=1))
+        if len(feas) > 0:
+            feas = feas.ravel()
+            xtmp, ytmp = xtmp[feas, :], ytmp[feas, :]
+            if f is not None:
+                f = f[feas]
+            c = c[feas, :]
+            if epochs is not None:
+                epochs = epochs[feas]
+
+    xt = torch.as_tensor(xtmp, dtype=torch.float64)
+    yt = torch.as_tensor(ytmp, dtype=torch.float64)
+    perm_t, rank_t, _ = ops.order_mo(xt, yt)
+    perm = perm_t.cpu().numpy()
+    rank = rank_t.cpu().numpy()
+    perm_x, perm_y = xtmp[perm], ytmp[perm]
+    perm_f = f[perm] if f is not None else None
+    perm_epoch = epochs[perm] if epochs is not None else None
+    perm_c = c[perm] if c is not None else None
+
+    uniq_rank, rnk_inv, rnk_cnt = np.unique(rank, return_inverse=True, return_counts=True)
+    rank_idx = np.array(
+        [np.flatnonzero(rnk_inv == i) for i in range(len(uniq_rank))], dtype=object
+    )
+    if perm_epoch is not None:
+        uniq_epc, epc_inv, epc_cnt = np.unique(
+            perm_epoch, return_inverse=True, return_counts=True
+        )
+        epc_idx = np.array(
+            [np.flatnonzero(epc_inv == i) for i in range(len(uniq_epc))], dtype=object
+        )
+    else:
+        uniq_epc = np.array([0])
+        epc_cnt = np.array([len(perm)])
+        epc_idx = np.array([np.arange(len(perm))], dtype=object)
+
+    rnk_epc_idx = np.empty((len(uniq_rank), len(uniq_epc)), dtype=object)
+    for i in range(len(uniq_rank)):
+        for j in range(len(uniq_epc)):
+            rnk_epc_idx[i, j] = np.intersect1d(
+                rank_idx[i], epc_idx[j], assume_unique=True
+            )
+
+    perm_arrs = (perm_x, perm_y, perm_f, perm_epoch, perm, feas)
+    rnk_arrs = (uniq_rank, rank_idx, rnk_cnt)
+    epc_arrs = (uniq_epc, epc_idx, epc_cnt)
+    return perm_arrs, rnk_arrs, epc_arrs, rnk_epc_idx
+
+
 def epsilon_get_best(x, y, f, c, feasible=True, delete_duplicates=True, epsilons=None):
     """Epsilon-box Pareto archive selection (MOASMO.py:703-758)."""
     from dmosopt_amd.moea.epsilon import EpsilonSort

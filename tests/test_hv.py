@@ -123,3 +123,43 @@ def test_ehvi_matches_reference_formula(rng):
     order = np.argsort(-expected, kind="stable")
     assert np.array_equal(sel, order)
     assert np.allclose(vals, expected[order], rtol=1e-10)
+
+
+def test_adaptive_router_and_hybrid(rng):
+    from dmosopt_amd.hv.adaptive import AdaptiveHyperVolume, DominanceAnalysis, estimate_overlap
+
+    pts = rng.random((15, 3))
+    ref = np.ones(3) * 1.1
+    exact = hv_3d(pts, ref)
+    # exact routing below threshold
+    ahv = AdaptiveHyperVolume(ref, mc_dim_threshold=10)
+    assert ahv.compute(pts) == pytest.approx(exact, rel=1e-12)
+    # MC routing above threshold agrees within tolerance
+    ahv_mc = AdaptiveHyperVolume(ref, mc_dim_threshold=0, mc_method="hybrid", seed=3)
+    val, conf = ahv_mc.compute_with_confidence(pts)
+    assert val == pytest.approx(exact, rel=0.06)
+    assert 0 < conf <= 1
+
+    # dominance analysis agrees with brute force
+    da = DominanceAnalysis(pts)
+    samples = rng.random((500, 3)) * 1.1
+    brute = (pts[None, :, :] <= samples[:, None, :]).all(axis=2).any(axis=1)
+    assert np.array_equal(da.dominates_any(samples), brute)
+
+    ov = estimate_overlap(pts, ref)
+    assert 0 < ov <= 1
+
+
+def test_engine_get_feasible(rng):
+    from dmosopt_amd.core.engine import get_feasible
+
+    x = rng.random((40, 4))
+    y = rng.random((40, 2))
+    c = rng.random((40, 1)) - 0.3  # some infeasible
+    epochs = rng.integers(0, 3, 40)
+    perm_arrs, rnk_arrs, epc_arrs, rnk_epc = get_feasible(x, y, None, c, 4, 2, epochs)
+    perm_x, perm_y, _, perm_epoch, perm, feas = perm_arrs
+    assert perm_x.shape[0] == (c > 0).all(axis=1).sum()
+    uniq_rank, rank_idx, rnk_cnt = rnk_arrs
+    assert sum(rnk_cnt) == perm_x.shape[0]
+    assert rnk_epc.shape == (len(uniq_rank), len(epc_arrs[0]))
