@@ -49,8 +49,24 @@ def batched_cholesky_update(A, Ainv, pc, z, psucc, cc, ccov, pthresh):
     """Rank-1 Cholesky update for K individuals at once (CMAES.py:489-537).
 
     A, Ainv: (K, d, d); pc, z: (K, d); psucc: (K,). Returns updated copies.
+    On GPU this is ONE fused gfx950 kernel launch (ops/hip/cmaes_update.hip);
+    the torch path below is the CPU/oracle implementation.
     """
     K, d, _ = A.shape
+    if A.is_cuda and A.dtype == torch.float32 and d <= 256:
+        from dmosopt_amd import ops as _ops
+
+        if _ops.native_available():
+            from dmosopt_amd import _hipops
+
+            A2 = A.contiguous().clone()
+            Ainv2 = Ainv.contiguous().clone()
+            pc2 = pc.contiguous().clone()
+            _hipops.cmaes_update_(
+                A2, Ainv2, pc2, z.contiguous().float(), psucc.contiguous().float(),
+                float(cc), float(ccov), float(pthresh),
+            )
+            return A2, Ainv2, pc2
     below = (psucc < pthresh)[:, None]
     pc_new = torch.where(
         below,

@@ -38,6 +38,8 @@ void launch_hv_mc_uniform(const float*, const float*, const float*,
 void launch_hv_fpras(const float*, const float*, const float*,
                      unsigned long long*, long long, int, int,
                      unsigned long long, hipStream_t);
+void launch_cmaes_update(float*, float*, float*, const float*, const float*,
+                         int, int, float, float, float, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -232,6 +234,20 @@ torch::Tensor get_duplicates(torch::Tensor X, double eps) {
   return std::get<0>((D <= eps).max(1)).to(torch::kBool);
 }
 
+void cmaes_update_(torch::Tensor A, torch::Tensor Ainv, torch::Tensor pc,
+                   torch::Tensor z, torch::Tensor psucc, double cc,
+                   double ccov, double pthresh) {
+  CHECK_GPU(A);
+  CHECK_GPU(Ainv);
+  CHECK_GPU(pc);
+  const int K = A.size(0), d = A.size(1);
+  TORCH_CHECK(d <= 256, "cmaes_update supports d <= 256");
+  launch_cmaes_update(A.data_ptr<float>(), Ainv.data_ptr<float>(),
+                      pc.data_ptr<float>(), z.data_ptr<float>(),
+                      psucc.data_ptr<float>(), K, d, (float)cc, (float)ccov,
+                      (float)pthresh, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
@@ -247,4 +263,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hv_mc_uniform_hits", &hv_mc_uniform_hits);
   m.def("hv_fpras_hits", &hv_fpras_hits);
   m.def("get_duplicates", &get_duplicates);
+  m.def("cmaes_update_", &cmaes_update_,
+        "In-place batched MO-CMA-ES rank-1 Cholesky update");
 }

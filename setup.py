@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join("dmosopt_amd", "ops", "hip", "pareto.hip"),
         os.path.join("dmosopt_amd", "ops", "hip", "variation.hip"),
         os.path.join("dmosopt_amd", "ops", "hip", "hv_mc.hip"),
+        os.path.join("dmosopt_amd", "ops", "hip", "cmaes_update.hip"),
     ],
     include_dirs=[HIP_DIR],
     extra_compile_args={

@@ -189,3 +189,31 @@ def test_bench_one_epoch_runs(dev):
     x_res, y_res, hv = one_epoch(X[:100], Y[:100], pop=64, rank=0, world=1,
                                  device=dev, seed=2, n_gen=5)
     assert np.isfinite(hv)
+
+
+def test_cmaes_update_matches_torch(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd.moea import cmaes as cm
+
+    g = torch.Generator().manual_seed(9)
+    K, d = 64, 30
+    A0 = torch.eye(d)[None].repeat(K, 1, 1) + 0.01 * torch.randn(K, d, d, generator=g).tril()
+    Ainv0 = torch.linalg.inv(A0)
+    pc0 = 0.1 * torch.randn(K, d, generator=g)
+    z = torch.randn(K, d, generator=g)
+    ps = torch.rand(K, generator=g)
+    cc, ccov, pthresh = 2.0 / (d + 2.0), 2.0 / (d**2 + 6.0), 0.44
+
+    # fp64 torch oracle
+    A_ref, Ainv_ref, pc_ref = cm.batched_cholesky_update(
+        A0.double(), Ainv0.double(), pc0.double(), z.double(), ps.double(),
+        cc, ccov, pthresh,
+    )
+    A_g = A0.float().to(dev).contiguous()
+    Ai_g = Ainv0.float().to(dev).contiguous()
+    pc_g = pc0.float().to(dev).contiguous()
+    _hipops.cmaes_update_(A_g, Ai_g, pc_g, z.float().to(dev), ps.float().to(dev),
+                          cc, ccov, pthresh)
+    assert torch.allclose(A_g.cpu().double(), A_ref, atol=1e-5)
+    assert torch.allclose(Ai_g.cpu().double(), Ainv_ref, atol=1e-4)
+    assert torch.allclose(pc_g.cpu().double(), pc_ref, atol=1e-6)
