@@ -125,7 +125,7 @@ torch::Tensor pareto_rank(torch::Tensor Y) {
   // only on device state from commit k, so the launches chain on-stream and
   // one n_front readback per chase amortizes the sync. Extra launches after
   // exhaustion are no-ops (alive all zero).
-  constexpr int CHASE = 8;
+  constexpr int CHASE = 16;
   int remaining = N;
   int k = 0;
   while (remaining > 0 && k < N + CHASE) {

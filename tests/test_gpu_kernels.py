@@ -217,3 +217,25 @@ def test_cmaes_update_matches_torch(dev):
     assert torch.allclose(A_g.cpu().double(), A_ref, atol=1e-5)
     assert torch.allclose(Ai_g.cpu().double(), Ainv_ref, atol=1e-4)
     assert torch.allclose(pc_g.cpu().double(), pc_ref, atol=1e-6)
+
+
+def test_cmaes_optimizer_gpu_e2e(dev):
+    """MO-CMA-ES on GPU with the fused update kernel on the hot path."""
+    import numpy as np
+    from dmosopt_amd.moea.cmaes import CMAESOptimizer
+    from dmosopt_amd.benchmarks.problems import zdt1
+
+    rng = np.random.default_rng(3)
+    d, pop = 10, 32
+    opt = CMAESOptimizer(popsize=pop, nInput=d, nOutput=2, model=None)
+    opt.set_device(dev)
+    bounds = np.stack([np.zeros(d), np.ones(d)], axis=1)
+    x0 = opt.generate_initial(bounds, rng)
+    y0 = zdt1(x0).numpy()
+    opt.initialize_strategy(x0, y0, bounds, rng)
+    for _ in range(5):
+        x_gen, st = opt.generate()
+        y_gen = zdt1(x_gen.double().cpu()).to(dev).float()
+        opt.update(x_gen, y_gen, st)
+    px, py = opt.population_objectives
+    assert torch.isfinite(py).all()
