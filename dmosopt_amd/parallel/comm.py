@@ -55,16 +55,61 @@ def is_controller() -> bool:
     return (not dist_is_initialized()) or dist.get_rank() == 0
 
 
+class _FarmStats:
+    """Per-rank evaluation accounting (role of the reference's distwq
+    controller stats, dmosopt.py:855-882)."""
+
+    def __init__(self, world: int):
+        self.world = world
+        self.n_processed = np.zeros(world, dtype=np.int64)
+        self.total_time = np.zeros(world, dtype=np.float64)
+        self.call_times: List[float] = []
+
+    def record(self, rank: int, n: int, t: float):
+        self.n_processed[rank] += n
+        self.total_time[rank] += t
+
+    def summary(self) -> Dict:
+        out = {
+            "results_collected": int(self.n_processed.sum()),
+            "total_evaluation_time": float(self.total_time.sum()),
+        }
+        if self.call_times:
+            ct = np.asarray(self.call_times)
+            out.update(
+                mean_time_per_call=float(ct.mean()),
+                stdev_time_per_call=float(ct.std()),
+            )
+        if self.world > 1:
+            out.update(
+                mean_calls_per_worker=float(self.n_processed.mean()),
+                stdev_calls_per_worker=float(self.n_processed.std()),
+                min_calls_per_worker=int(self.n_processed.min()),
+                max_calls_per_worker=int(self.n_processed.max()),
+                mean_time_per_worker=float(self.total_time.mean()),
+                stdev_time_per_worker=float(self.total_time.std()),
+            )
+        return out
+
+
 class LocalFarm:
     """Single-process evaluation (controller evaluates everything)."""
 
     def __init__(self, eval_funs: Dict[str, Callable]):
         self.eval_funs = eval_funs
+        self._stats = _FarmStats(1)
 
     def evaluate(self, opt_id: str, points: List) -> List:
         """points: list of per-request eval_fun arguments (space-vals dicts)."""
         fn = self.eval_funs[opt_id]
-        return [fn(p) for p in points]
+        t0 = time.time()
+        out = [fn(p) for p in points]
+        self._stats.record(0, len(points), time.time() - t0)
+        self._stats.call_times.append(time.time() - t0)
+        return out
+
+    def stats(self) -> Dict:
+        return self._stats.summary()
 
     def shutdown(self):
         pass
@@ -86,10 +131,12 @@ class TorchDistFarm:
         self.eval_funs = eval_funs
         self.rank = dist.get_rank()
         self.world = dist.get_world_size()
+        self._stats = _FarmStats(self.world)
 
     # ------------------------------------------------------------ controller
     def evaluate(self, opt_id: str, points: List) -> List:
         assert self.rank == 0
+        t_call = time.time()
         cmd = [("eval", opt_id, points)]
         dist.broadcast_object_list(cmd, src=0)
         my_results = self._eval_shard(opt_id, points)
@@ -98,9 +145,17 @@ class TorchDistFarm:
         # interleave shards back into original order
         out: List = [None] * len(points)
         for r, shard in enumerate(gathered):
+            shard_t = 0.0
             for j, res in enumerate(shard):
                 out[r + j * self.world] = res
+                if isinstance(res, dict) and "time" in res:
+                    shard_t += res["time"]
+            self._stats.record(r, len(shard), shard_t)
+        self._stats.call_times.append(time.time() - t_call)
         return out
+
+    def stats(self) -> Dict:
+        return self._stats.summary()
 
     def shutdown(self):
         if self.rank == 0:

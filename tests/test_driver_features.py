@@ -1,0 +1,148 @@
+"""Driver feature coverage: mean-variance mode, problem_ids, features,
+nested spaces, dynamic initial sampling, time limit, farm stats."""
+
+import numpy as np
+import pytest
+
+import dmosopt_amd
+
+
+def _base(opt_id, d=4, **over):
+    def obj_fun(pp):
+        x = np.array([pp[f"x{i}"] for i in range(d)])
+        return np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+
+    params = {
+        "opt_id": opt_id,
+        "obj_fun": obj_fun,
+        "problem_parameters": {},
+        "space": {f"x{i}": [0.0, 1.0] for i in range(d)},
+        "objective_names": ["f1", "f2"],
+        "population_size": 20,
+        "num_generations": 6,
+        "optimizer": "nsga2",
+        "n_initial": 2,
+        "n_epochs": 2,
+        "random_seed": 7,
+    }
+    params.update(over)
+    return params
+
+
+def test_optimize_mean_variance_mode():
+    best = dmosopt_amd.run(
+        _base("t_mv", optimize_mean_variance=True,
+              surrogate_method_kwargs={"anisotropic": False, "optimizer": "sceua"}),
+        verbose=False,
+    )
+    bestx, besty = best
+    y = np.column_stack([v for _, v in besty])
+    assert y.shape[1] == 2 and np.isfinite(y).all()
+
+
+def test_problem_ids_multi():
+    def obj_fun_mp(mpp):
+        out = {}
+        for pid, pp in mpp.items():
+            x = np.array([pp[f"x{i}"] for i in range(4)])
+            shift = 0.1 * pid
+            out[pid] = np.array([np.sum((x - shift) ** 2), np.sum((x - 1) ** 2)])
+        return out
+
+    params = _base("t_mp", obj_fun=obj_fun_mp, problem_ids={1, 2},
+                   surrogate_method_name=None, num_generations=3)
+    best = dmosopt_amd.run(params, verbose=False)
+    assert set(best.keys()) == {1, 2}
+    for pid in (1, 2):
+        prms, objs = best[pid]
+        assert len(objs) == 2
+
+
+def test_features_roundtrip(tmp_path):
+    feature_dtypes = [("aux", np.float32)]
+
+    def obj_fun(pp):
+        x = np.array([pp[f"x{i}"] for i in range(4)])
+        y = np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+        f = np.array([(np.float32(x.mean()),)], dtype=feature_dtypes)
+        return y, f
+
+    fp = str(tmp_path / "feat.h5")
+    params = _base(
+        "t_feat", obj_fun=obj_fun, feature_dtypes=feature_dtypes,
+        surrogate_method_name=None, num_generations=3,
+        file_path=fp, save=True, save_eval=4,
+    )
+    best = dmosopt_amd.run(params, return_features=True, verbose=False)
+    bestx, besty, bestf = best
+    assert bestf is not None and len(bestf) > 0
+
+    from dmosopt_amd.storage import h5 as h5store
+
+    _, old_evals, info = h5store.h5_load_all(fp, "t_feat")
+    assert info["features"] == ["aux"]
+    assert old_evals[0][0].features is not None
+
+
+def test_nested_parameter_space_e2e():
+    def obj_fun(pp):
+        x = np.array([pp["grp"]["a"], pp["grp"]["b"], pp["other"]])
+        return np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+
+    params = {
+        "opt_id": "t_nested",
+        "obj_fun": obj_fun,
+        "problem_parameters": {},
+        "space": {"grp": {"a": [0.0, 1.0], "b": [0.0, 1.0]}, "other": [0.0, 1.0]},
+        "objective_names": ["f1", "f2"],
+        "nested_parameter_space": True,
+        "population_size": 16,
+        "num_generations": 4,
+        "surrogate_method_name": None,
+        "optimizer": "nsga2",
+        "n_initial": 2,
+        "n_epochs": 1,
+        "random_seed": 3,
+    }
+    best = dmosopt_amd.run(params, verbose=False)
+    assert best is not None
+
+
+def test_dynamic_initial_sampling():
+    calls = []
+
+    def dyn(file_path, iteration, evaluated_samples, next_samples, sampler, **kw):
+        calls.append(iteration)
+        if iteration >= 2:
+            return None
+        return next_samples[:4]
+
+    params = _base("t_dyn", dynamic_initial_sampling=dyn,
+                   surrogate_method_name=None, num_generations=3)
+    best = dmosopt_amd.run(params, verbose=False)
+    assert best is not None
+    assert calls == [0, 1, 2]
+
+
+def test_time_limit_stops_early():
+    import time as _t
+
+    def slow_obj(pp):
+        _t.sleep(0.02)
+        x = np.array([pp[f"x{i}"] for i in range(4)])
+        return np.array([x.sum(), (1 - x).sum()])
+
+    params = _base("t_tl", obj_fun=slow_obj, surrogate_method_name=None,
+                   n_epochs=50, num_generations=50)
+    t0 = _t.time()
+    dmosopt_amd.run(params, time_limit=3, verbose=False)
+    assert _t.time() - t0 < 30
+
+
+def test_farm_stats_present():
+    params = _base("t_stats", surrogate_method_name=None, num_generations=3)
+    dmosopt_amd.run(params, verbose=False)
+    dopt = dmosopt_amd.sopt_dict["t_stats"]
+    stats = dopt.get_stats()
+    assert stats["results_collected"] > 0
+    assert "total_evaluation_time" in stats

@@ -179,3 +179,49 @@ def test_tournament_selection_distribution(rng):
         idx = ops.tournament_selection(pop, 5, metrics, rng).numpy()
         counts[idx] += 1
     assert counts[:5].sum() > counts[15:].sum()
+
+
+def test_agemoea_survival_matches_reference_greedy(rng):
+    """Incremental 2-NN greedy selection == reference O(m^3) formulation
+    (AGEMOEA.py:389-442)."""
+    from dmosopt_amd.moea.agemoea import (
+        find_corner_solutions, get_geometry, minkowski_matrix, normalize_front,
+        survival_score,
+    )
+
+    y = rng.random((40, 3)) + 0.1
+    front = np.arange(40)
+    ideal = y.min(axis=0)
+
+    # reference-style oracle
+    yf = y - ideal
+    extreme = find_corner_solutions(yf)
+    normalization = normalize_front(yf, extreme)
+    yn = yf / normalization
+    p = get_geometry(yn, extreme)
+    m = 40
+    crowd = np.zeros(m)
+    crowd[extreme] = np.inf
+    selected = np.zeros(m, dtype=bool)
+    selected[extreme] = True
+    nn_norm = np.power(np.power(np.abs(yn), p).sum(axis=1), 1.0 / p)
+    distances = minkowski_matrix(yn, yn, p) / nn_norm[:, None]
+    remaining = list(np.arange(m)[~selected])
+    for _ in range(m - selected.sum()):
+        D_mg = distances[np.ix_(remaining, np.flatnonzero(selected))]
+        if D_mg.shape[1] > 1:
+            part = np.argpartition(D_mg, 1, axis=1)[:, :2]
+            tmp = np.take_along_axis(D_mg, part, axis=1).sum(axis=1)
+            index = int(np.argmax(tmp))
+            d = tmp[index]
+        else:
+            index = int(D_mg[:, 0].argmax())
+            d = D_mg[index, 0]
+        best = remaining.pop(index)
+        selected[best] = True
+        crowd[best] = d
+
+    _, _, crowd_fast = survival_score(y, front, ideal)
+    finite = np.isfinite(crowd)
+    assert np.allclose(crowd_fast[finite], crowd[finite], atol=1e-10)
+    assert np.array_equal(np.isinf(crowd_fast), np.isinf(crowd))
