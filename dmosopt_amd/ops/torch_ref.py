@@ -33,10 +33,11 @@ def dominance_degree_matrix(Y: Tensor) -> Tensor:
 def pareto_rank(Y: Tensor) -> Tensor:
     """0-based Pareto front index per row of Y (minimization).
 
-    DDA ranking with iterative max-reduction front peel — the GPU-friendly
-    variant of the reference's dda_non_dominated_sort (dda.py:34-76); it
-    produces the same front partition as dda_ens up to identical semantics
-    of duplicate rows.
+    DDA ranking (dda.py:34-76 semantics: D[i][j] == m, after zeroing
+    mutual-domination entries of identical rows, means i dominates j) with
+    a dominator-COUNT front peel: n_dom[j] = #alive dominators; a front is
+    n_dom == 0; peeling subtracts only the peeled rows' contributions —
+    O(N^2) total instead of a full-matrix max per front.
     """
     n, d = Y.shape
     if n == 0:
@@ -44,20 +45,22 @@ def pareto_rank(Y: Tensor) -> Tensor:
     D = dominance_degree_matrix(Y)
     # zero out mutual-domination entries for identical objective rows
     identical = (D == d) & (D.T == d)
-    D = D.masked_fill(identical, 0)
+    Dom = (D == d) & ~identical  # Dom[i][j]: i dominates j
+    n_dom = Dom.sum(dim=0)  # (n,) alive dominator counts
     rank = torch.zeros(n, dtype=torch.long, device=Y.device)
     alive = torch.ones(n, dtype=torch.bool, device=Y.device)
     k = 0
-    while bool(alive.any()):
-        maxD = D.max(dim=0).values  # column max: strongest dominator count
-        front = alive & (maxD < d)
+    remaining = n
+    while remaining > 0:
+        front = alive & (n_dom == 0)
         if not bool(front.any()):
-            # numerical safety: should not happen, but avoid infinite loop
-            front = alive
+            front = alive  # numerical safety: avoid infinite loop
         rank[front] = k
-        alive = alive & ~front
-        # peel: remove front rows and columns from D
-        D = D.masked_fill(front[:, None], -1).masked_fill(front[None, :], -1)
+        alive &= ~front
+        idx = front.nonzero(as_tuple=True)[0]
+        remaining -= int(idx.numel())
+        if remaining > 0:
+            n_dom = n_dom - Dom[idx].sum(dim=0)
         k += 1
     return rank
 
