@@ -50,34 +50,60 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
   for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
     const int bs = min(CHOL_BS, N - k0);
 
-    // --- load + factor diagonal block in LDS
-    for (int idx = tid; idx < bs * bs; idx += CHOL_TPB)
-      L.S[idx / bs][idx % bs] = M[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
-    __syncthreads();
-    for (int j = 0; j < bs; ++j) {
-      if (tid == 0) {
-        float djj = L.S[j][j];
-        if (djj <= 0.f) {
-          info[b] = 1;
-          djj = 1e-30f;
+    // --- factor the diagonal block WAVE-SYNCHRONOUSLY in wave 0: lane l
+    // holds row l in registers; column values broadcast with shuffles —
+    // zero barriers inside the 32-step j loop (the previous LDS version
+    // paid ~100 workgroup barriers per k-step).
+    if (tid < WAVE_SIZE) {
+      const int lane = tid;
+      // the j/c loops are FULLY unrolled (compile-time indices) so r[]
+      // stays in registers — runtime-indexed register arrays spill
+      float r[CHOL_BS];
+#pragma unroll
+      for (int c = 0; c < CHOL_BS; ++c)
+        r[c] = (lane < bs && c < bs) ? M[(long long)(k0 + lane) * N + k0 + c]
+                                     : 0.f;
+      float ld_part = 0.f;
+      int bad = 0;
+#pragma unroll
+      for (int j = 0; j < CHOL_BS; ++j) {
+        if (j >= bs) break;
+        // lane j holds the fully-updated pivot
+        float piv = __shfl(r[j], j);
+        if (piv <= 0.f) {
+          bad = 1;
+          piv = 1e-30f;
         }
-        L.S[j][j] = sqrtf(djj);
-        L.ld_accum += __logf(L.S[j][j]);
+        const float sjj = sqrtf(piv);
+        if (lane == j) {
+          r[j] = sjj;
+          ld_part += __logf(sjj);
+        }
+        const float lij = (lane > j && lane < bs) ? r[j] / sjj : 0.f;
+        if (lane > j && lane < bs) r[j] = lij;
+        // rank-1 update: a[i][c] -= L[i][j] * L[c][j] (lower triangle)
+#pragma unroll
+        for (int c = j + 1; c < CHOL_BS; ++c) {
+          if (c >= bs) break;
+          const float lcj = __shfl(lij, c);
+          if (lane >= c && lane < bs) r[c] = fmaf(-lij, lcj, r[c]);
+        }
       }
-      __syncthreads();
-      for (int i = j + 1 + tid; i < bs; i += CHOL_TPB) L.S[i][j] /= L.S[j][j];
-      __syncthreads();
-      const int rem_d = bs - j - 1;
-      for (int idx = tid; idx < rem_d * rem_d; idx += CHOL_TPB) {
-        const int r = j + 1 + idx / rem_d;
-        const int c = j + 1 + idx % rem_d;
-        if (c <= r) L.S[r][c] -= L.S[r][j] * L.S[c][j];
+      if (lane < bs) {
+#pragma unroll
+        for (int c = 0; c < CHOL_BS; ++c) {
+          const float v = (c <= lane) ? r[c] : 0.f;
+          if (c < bs) {
+            L.S[lane][c] = v;
+            M[(long long)(k0 + lane) * N + k0 + c] = v;
+          }
+        }
       }
-      __syncthreads();
-    }
-    for (int idx = tid; idx < bs * bs; idx += CHOL_TPB) {
-      const int r = idx / bs, c = idx % bs;
-      M[(long long)(k0 + r) * N + k0 + c] = (c <= r) ? L.S[r][c] : 0.f;
+      const float ld_sum = warp_reduce_sum(ld_part);
+      if (lane == 0) {
+        L.ld_accum += ld_sum;
+        if (bad) info[b] = 1;  // bad is wave-uniform (piv is broadcast)
+      }
     }
     __syncthreads();
 
