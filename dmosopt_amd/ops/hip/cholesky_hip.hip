@@ -34,6 +34,7 @@ struct CholLds {
 
 // ~100 KiB of LDS: above the 64 KiB static default, so allocated
 // dynamically (gfx950 has 160 KiB per CU).
+template <bool WAVE_DIAG>
 __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
     float* __restrict__ A,       // (B, N, N)
     float* __restrict__ logdet,  // (B,)
@@ -54,7 +55,7 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
     // holds row l in registers; column values broadcast with shuffles —
     // zero barriers inside the 32-step j loop (the previous LDS version
     // paid ~100 workgroup barriers per k-step).
-    if (tid < WAVE_SIZE) {
+    if (WAVE_DIAG && tid < WAVE_SIZE) {
       const int lane = tid;
       // the j/c loops are FULLY unrolled (compile-time indices) so r[]
       // stays in registers — runtime-indexed register arrays spill
@@ -103,6 +104,38 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
       if (lane == 0) {
         L.ld_accum += ld_sum;
         if (bad) info[b] = 1;  // bad is wave-uniform (piv is broadcast)
+      }
+    }
+    if (!WAVE_DIAG) {
+      // LDS/barrier variant of the diagonal-block factorization
+      for (int idx = tid; idx < bs * bs; idx += CHOL_TPB)
+        L.S[idx / bs][idx % bs] =
+            M[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
+      __syncthreads();
+      for (int j = 0; j < bs; ++j) {
+        if (tid == 0) {
+          float djj = L.S[j][j];
+          if (djj <= 0.f) {
+            info[b] = 1;
+            djj = 1e-30f;
+          }
+          L.S[j][j] = sqrtf(djj);
+          L.ld_accum += __logf(L.S[j][j]);
+        }
+        __syncthreads();
+        for (int i = j + 1 + tid; i < bs; i += CHOL_TPB) L.S[i][j] /= L.S[j][j];
+        __syncthreads();
+        const int rem_d = bs - j - 1;
+        for (int idx = tid; idx < rem_d * rem_d; idx += CHOL_TPB) {
+          const int rr = j + 1 + idx / rem_d;
+          const int cc = j + 1 + idx % rem_d;
+          if (cc <= rr) L.S[rr][cc] -= L.S[rr][j] * L.S[cc][j];
+        }
+        __syncthreads();
+      }
+      for (int idx = tid; idx < bs * bs; idx += CHOL_TPB) {
+        const int rr = idx / bs, cc = idx % bs;
+        M[(long long)(k0 + rr) * N + k0 + cc] = (cc <= rr) ? L.S[rr][cc] : 0.f;
       }
     }
     __syncthreads();
@@ -238,17 +271,27 @@ __global__ void backward_solve_batched_kernel(const float* __restrict__ L,
   }
 }
 
+#include <stdlib.h>
+
 extern "C" void launch_cholesky_batched(float* A, float* logdet, int* info,
                                         int B, int N, hipStream_t stream) {
-  static bool attr_set = false;
-  if (!attr_set) {
-    hipFuncSetAttribute((const void*)cholesky_batched_kernel,
+  static int mode = -1;
+  if (mode < 0) {
+    const char* env = getenv("DMOSOPT_CHOL_DIAG");  // "lds" selects variant B
+    mode = (env && env[0] == 'l') ? 0 : 1;
+    hipFuncSetAttribute((const void*)cholesky_batched_kernel<true>,
                         hipFuncAttributeMaxDynamicSharedMemorySize,
                         (int)sizeof(CholLds));
-    attr_set = true;
+    hipFuncSetAttribute((const void*)cholesky_batched_kernel<false>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        (int)sizeof(CholLds));
   }
-  hipLaunchKernelGGL(cholesky_batched_kernel, dim3(B), dim3(CHOL_TPB),
-                     sizeof(CholLds), stream, A, logdet, info, N);
+  if (mode)
+    hipLaunchKernelGGL(cholesky_batched_kernel<true>, dim3(B), dim3(CHOL_TPB),
+                       sizeof(CholLds), stream, A, logdet, info, N);
+  else
+    hipLaunchKernelGGL(cholesky_batched_kernel<false>, dim3(B), dim3(CHOL_TPB),
+                       sizeof(CholLds), stream, A, logdet, info, N);
 }
 
 extern "C" void launch_forward_solve_batched(const float* L, float* Y, int B,
