@@ -29,13 +29,12 @@ from dmosopt_amd.models.sceua import sceua_batched
 
 def _top_k_mo(x: np.ndarray, y: np.ndarray, top_k):
     """Keep the top_k rows by non-dominated sort (reference MOEA.py:350-372)."""
-    if not isinstance(top_k, int) or x.shape[0] <= top_k:
-        return x, y
-    xt = torch.as_tensor(x, dtype=torch.float64)
-    yt = torch.as_tensor(y, dtype=torch.float64)
-    perm, _, _ = ops.order_mo(xt, yt)
-    perm = perm[:top_k].cpu().numpy()
-    return x[perm], y[perm]
+    xt, yt = ops.top_k_mo(
+        torch.as_tensor(x, dtype=torch.float64),
+        torch.as_tensor(y, dtype=torch.float64),
+        top_k,
+    )
+    return xt.numpy() if xt is not x else x, yt.numpy() if yt is not y else y
 
 
 class _GPRBase:

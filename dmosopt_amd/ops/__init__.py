@@ -130,6 +130,17 @@ def remove_worst(
     return population_parm[perm], population_obj[perm], rank[:pop], perm
 
 
+def top_k_mo(x: torch.Tensor, y: torch.Tensor, top_k=None):
+    """Top-k rows by non-dominated sort (reference MOEA.top_k_MO,
+    MOEA.py:350-372); returns (x, y) unchanged when top_k is not an int or
+    the population is already small enough."""
+    if not isinstance(top_k, int) or x.shape[0] <= top_k:
+        return x, y
+    perm, _, _ = order_mo(x, y)
+    perm = perm[:top_k]
+    return x[perm], y[perm]
+
+
 def get_duplicates(X: torch.Tensor, eps: float = 1e-16) -> torch.Tensor:
     if _use_native(X):
         return _native.get_duplicates(X.contiguous().float(), eps)
@@ -235,6 +246,7 @@ __all__ = [
     "euclidean_distance_metric",
     "order_mo",
     "remove_worst",
+    "top_k_mo",
     "get_duplicates",
     "remove_duplicates",
     "lexsort",

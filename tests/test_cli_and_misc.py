@@ -110,3 +110,24 @@ def test_discrepancy_metrics_prefer_lh(rng):
     for fn in (discrepancy.SD2, discrepancy.WD2):
         v = fn(lh)
         assert np.isfinite(v) and v >= 0
+
+
+def test_early_stopping_utils():
+    from dmosopt_amd.models.early_stopping import (
+        AdaptiveEarlyStopping, EarlyStoppingConfig, ModelType,
+        analyze_loss_trajectory, suggest_hyperparameters,
+    )
+
+    cfg = EarlyStoppingConfig.for_model_type(ModelType.DEEP_GP)
+    assert cfg.min_iterations > EarlyStoppingConfig().min_iterations
+    es = AdaptiveEarlyStopping(EarlyStoppingConfig(min_iterations=5, patience=3))
+    losses = [1.0 / (i + 1) for i in range(4)] + [0.2] * 30
+    stopped_at = None
+    for i, l in enumerate(losses):
+        if es.should_stop(l):
+            stopped_at = i
+            break
+    assert stopped_at is not None and stopped_at < len(losses) - 1
+    stats = analyze_loss_trajectory(losses[: stopped_at + 1])
+    assert stats["n"] == stopped_at + 1
+    assert "lr" in suggest_hyperparameters(losses, 0.1)
