@@ -155,11 +155,11 @@ torch::Tensor crowding_distance(torch::Tensor Y) {
   auto hi = std::get<0>(Y.max(0)).contiguous();
   auto span = (hi - lo).contiguous();
   span = torch::where(span == 0, torch::ones_like(span), span).contiguous();
-  auto out = torch::zeros({N}, Y.options());
+  auto per_dim = torch::zeros({m, N}, Y.options());
   launch_crowding(Y.data_ptr<float>(), lo.data_ptr<float>(),
-                  span.data_ptr<float>(), out.data_ptr<float>(), N, m,
+                  span.data_ptr<float>(), per_dim.data_ptr<float>(), N, m,
                   cur_stream());
-  return out;
+  return per_dim.sum(0);  // fixed-order reduction: deterministic
 }
 
 std::vector<torch::Tensor> sbx_batch(torch::Tensor pool, torch::Tensor p1,

@@ -163,14 +163,18 @@ __global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
     }
   }
 
-  // gaps: boundary = 1, interior = vals[i+1] - vals[i-1]; scatter-add
+  // gaps: boundary = 1, interior = vals[i+1] - vals[i-1].
+  // Each dim writes its own row of out (m, N) — NO atomics, so the sum
+  // order is fixed and results are bit-identical across replicated ranks
+  // (float atomicAdd order would not be).
   for (int i = threadIdx.x; i < N; i += CROWD_TPB) {
     float d = (i == 0 || i == N - 1) ? 1.f : vals[i + 1] - vals[i - 1];
     if (isnan(d)) d = 0.f;
-    atomicAdd(&out[idxs[i]], d);
+    out[(long long)j * N + idxs[i]] = d;
   }
 }
 
+// out must be an (m, N) buffer; the host sums over dim 0 deterministically.
 extern "C" void launch_crowding(const float* Y, const float* lo,
                                 const float* span, float* out, int N, int m,
                                 hipStream_t stream) {
