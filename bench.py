@@ -189,6 +189,7 @@ def main():
                 "num_generations": args.gens,
                 "archive_size": ARCHIVE_N,
                 "final_hypervolume_ref11": hv,
+                "reference_measured_ms_per_epoch_cpu": 19200.0,
             },
         }
         print(json.dumps(out))
