@@ -9,6 +9,8 @@ import time
 import numpy as np
 import torch
 
+torch.set_num_threads(min(8, os.cpu_count() or 8))
+
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 from dmosopt_amd.benchmarks.problems import tnk, zdt3
