@@ -71,8 +71,39 @@ def _use_native(*tensors: torch.Tensor) -> bool:
 # ------------------------------------------------------------------ ranking
 def pareto_rank(Y: torch.Tensor) -> torch.Tensor:
     if _use_native(Y):
-        return _native.pareto_rank(Y.contiguous().float())
+        return _pareto_rank_gpu(Y)
     return torch_ref.pareto_rank(Y)
+
+
+def _pareto_rank_gpu(Y: torch.Tensor) -> torch.Tensor:
+    """GPU ranking: native fused dominance-degree matrix, then a
+    dominator-count peel driven by one float matvec per front (O(N^2) GEMV
+    work per front is microseconds on-device; the former per-front full
+    column-max scan was O(N^2) int traffic x #fronts — 208 ms at N=8192 on
+    many-front problems). Syncs amortized over 16-front chases."""
+    n, m = Y.shape
+    if n == 0:
+        return torch.zeros(0, dtype=torch.long, device=Y.device)
+    D = _native.dominance_degree_matrix(Y.contiguous().float())
+    # identical rows already zeroed by the native kernel (incl. diagonal)
+    Dom_f = (D == m).to(torch.float32)  # Dom[i][j]: i dominates j
+    n_dom = Dom_f.sum(dim=0)  # float counts (exact for n < 2^24)
+    rank = torch.zeros(n, dtype=torch.long, device=Y.device)
+    alive = torch.ones(n, dtype=torch.bool, device=Y.device)
+    k = 0
+    remaining = n
+    CHASE = 16
+    while remaining > 0 and k < n + CHASE:
+        for c in range(CHASE):
+            front = alive & (n_dom == 0)
+            rank = torch.where(front, torch.full_like(rank, k + c), rank)
+            alive &= ~front
+            # subtract peeled dominators' contributions with one matvec
+            n_dom = n_dom - front.to(torch.float32) @ Dom_f
+            n_dom = torch.where(alive, n_dom, torch.ones_like(n_dom))
+        k += CHASE
+        remaining = int(alive.sum().item())
+    return rank
 
 
 def dominance_degree_matrix(Y: torch.Tensor) -> torch.Tensor:

@@ -87,16 +87,25 @@ def test_solves_match_torch(dev):
 
 
 def test_pareto_rank_matches_reference(dev):
-    from dmosopt_amd import _hipops
+    from dmosopt_amd import _hipops, ops
     from dmosopt_amd.ops import torch_ref
 
     g = torch.Generator().manual_seed(4)
     for n, m in [(100, 2), (500, 3), (1000, 5)]:
         Y = torch.rand(n, m, generator=g)
         Y[: n // 10] = Y[n // 2 : n // 2 + n // 10]  # inject duplicates
-        r_gpu = _hipops.pareto_rank(Y.float().to(dev)).cpu()
         r_ref = torch_ref.pareto_rank(Y.double())
+        # native column-max peel
+        r_native = _hipops.pareto_rank(Y.float().to(dev)).cpu()
+        assert torch.equal(r_native, r_ref)
+        # dispatched matvec dominator-count peel
+        r_gpu = ops.pareto_rank(Y.float().to(dev)).cpu()
         assert torch.equal(r_gpu, r_ref)
+    # many-front stress (d=2 spread points, the TNK-like regime)
+    Y2 = torch.rand(4096, 2, generator=g)
+    assert torch.equal(
+        ops.pareto_rank(Y2.float().to(dev)).cpu(), torch_ref.pareto_rank(Y2.double())
+    )
 
 
 def test_crowding_matches_reference(dev):
