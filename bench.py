@@ -24,6 +24,10 @@ import time
 import numpy as np
 import torch
 
+# host-side control code thrashes with one thread per core on many-core
+# GPU boxes (small CPU tensor ops); the compute path is the GPU anyway
+torch.set_num_threads(min(8, os.cpu_count() or 8))
+
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 from dmosopt_amd.benchmarks.problems import zdt1
