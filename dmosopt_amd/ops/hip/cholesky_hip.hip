@@ -20,7 +20,9 @@
 #include <math.h>
 
 #define CHOL_BS 32
-#define CHOL_TPB 256
+#ifndef CHOL_TPB
+#define CHOL_TPB 1024
+#endif
 #define CHOL_CHUNK 384  // panel rows staged per LDS buffer (384*32*4 = 48 KiB)
 
 struct CholLds {
