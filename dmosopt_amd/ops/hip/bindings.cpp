@@ -40,6 +40,7 @@ void launch_hv_fpras(const float*, const float*, const float*,
                      unsigned long long, hipStream_t);
 void launch_cmaes_update(float*, float*, float*, const float*, const float*,
                          int, int, float, float, float, hipStream_t);
+void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -117,6 +118,13 @@ torch::Tensor pareto_rank(torch::Tensor Y) {
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
   auto D = dominance_degree_matrix(Y);
+  if (N <= 2048) {
+    // single-launch persistent peel
+    auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));
+    launch_peel_single_block(D.data_ptr<int>(), rank.data_ptr<int>(), N, m,
+                             cur_stream());
+    return rank.to(torch::kLong);
+  }
   auto alive = torch::ones({N}, Y.options().dtype(torch::kUInt8));
   auto front = torch::empty({N}, Y.options().dtype(torch::kUInt8));
   auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));
