@@ -146,3 +146,19 @@ def test_farm_stats_present():
     stats = dopt.get_stats()
     assert stats["results_collected"] > 0
     assert "total_evaluation_time" in stats
+
+
+def test_seed_reproducibility():
+    """Same random_seed => identical archives (self-reproducibility,
+    SURVEY.md section 7 RNG discipline)."""
+    import dmosopt_amd
+
+    outs = []
+    for tag in ("a", "b"):
+        params = _base(f"t_repro_{tag}", surrogate_method_name=None,
+                       num_generations=4, random_seed=123)
+        dmosopt_amd.run(params, verbose=False)
+        x, y = dmosopt_amd.sopt_dict[f"t_repro_{tag}"].optimizer_dict[0].get_evals()
+        outs.append((x.copy(), y.copy()))
+    assert np.array_equal(outs[0][0], outs[1][0])
+    assert np.array_equal(outs[0][1], outs[1][1])
