@@ -59,6 +59,13 @@ class ShardedGPObjective:
         self.world = world
         self.device = device
 
+    def evaluate_tensor(self, x):
+        if self.world == 1:
+            return self.gp.evaluate_tensor(x)
+        return torch.as_tensor(
+            self.evaluate(x.detach().cpu().numpy()), dtype=x.dtype, device=x.device
+        )
+
     def evaluate(self, x):
         import torch.distributed as dist
 

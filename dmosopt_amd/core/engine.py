@@ -72,8 +72,11 @@ def xinit(
 
 def _surrogate_eval(mdl: Model, x, optimize_mean_variance: bool):
     if optimize_mean_variance:
-        y_mean, y_var = mdl.objective.evaluate(x)
+        y_mean, y_var = mdl.objective.evaluate(_to_np(x))
         return np.column_stack((y_mean, np.round(y_var, 6)))
+    # device-resident fast path: no numpy round trip per generation
+    if isinstance(x, torch.Tensor) and hasattr(mdl.objective, "evaluate_tensor"):
+        return mdl.objective.evaluate_tensor(x)
     return mdl.objective.evaluate(_to_np(x))
 
 
@@ -144,7 +147,7 @@ def optimize_loop(
         optimizer.update(x_gen, y_gen, gen_state)
         n_eval += x_gen.shape[0]
         x_new.append(_to_np(x_gen))
-        y_new.append(_to_np(y_gen))
+        y_new.append(_to_np(y_gen))  # host archive copies (saved to H5)
         gen_indexes.append(np.full((x_gen.shape[0],), i, dtype=np.uint32))
 
     gen_index = np.concatenate(gen_indexes)

@@ -210,6 +210,12 @@ class _GPRBase:
             return mean, var
         return mean
 
+    def evaluate_tensor(self, x: torch.Tensor) -> torch.Tensor:
+        """Device-resident evaluate: tensor in, tensor out, no host trip."""
+        xq = self.normalize_query(x.to(self.device, self.dtype))
+        mean, _ = self._fitted.predict(xq)
+        return mean
+
 
 class GPRMatern(_GPRBase):
     """Registry name 'gpr' — Matern-5/2 exact GP (reference model.py:1182)."""

@@ -198,11 +198,20 @@ def polynomial_mutation_batch(
     return child.clamp(xlb, xub)
 
 
+_TOURNAMENT_PROB_CACHE = {}
+
+
 def tournament_prob_vector(n: int, p: float = 0.5) -> torch.Tensor:
-    """Geometric selection probabilities p(1-p)^i over sorted candidates."""
-    i = torch.arange(n, dtype=torch.float64)
-    prob = p * (1.0 - p) ** i
-    return prob / prob.sum()
+    """Geometric selection probabilities p(1-p)^i over sorted candidates
+    (cached per (n, p): rebuilt thousands of times per epoch otherwise)."""
+    key = (n, p)
+    out = _TOURNAMENT_PROB_CACHE.get(key)
+    if out is None:
+        i = torch.arange(n, dtype=torch.float64)
+        prob = p * (1.0 - p) ** i
+        out = prob / prob.sum()
+        _TOURNAMENT_PROB_CACHE[key] = out
+    return out
 
 
 def tournament_selection(
@@ -220,7 +229,7 @@ def tournament_selection(
     """
     dev_metrics = [m if isinstance(m, torch.Tensor) else torch.as_tensor(m) for m in metrics]
     sorted_candidates = lexsort(dev_metrics)
-    prob = tournament_prob_vector(pop).numpy()
+    prob = tournament_prob_vector(pop).numpy()  # cached
     pool_pos = np_random.choice(pop, size=poolsize, p=prob, replace=False)
     pool_pos = torch.as_tensor(pool_pos, dtype=torch.long, device=sorted_candidates.device)
     return sorted_candidates[pool_pos]
