@@ -49,7 +49,7 @@ optimizer_registry = {
 surrogate_registry = {
     "gpr": "dmosopt_amd.models.gp.GPRMatern",
     "egp": "dmosopt_amd.models.gp.EGPMatern",
-    "megp": "dmosopt_amd.models.gp.MEGPMatern",
+    "megp": "dmosopt_amd.models.multitask_gp.MEGPMaternICM",
     "mdgp": "dmosopt_amd.models.deep_gp.MDGPMatern",
     "mdspp": "dmosopt_amd.models.deep_gp.MDSPPMatern",
     "vgp": "dmosopt_amd.models.variational_gp.VGPMatern",
