@@ -247,6 +247,7 @@ __global__ void forward_solve_batched_kernel(const float* __restrict__ L,
                                              int N, int R) {
   __shared__ float S[TRSV_BS][TRSV_BS + 1];
   __shared__ float z[TRSV_BS];
+  __shared__ float yp[TRSV_BS];
   const int b = blockIdx.x;
   const int r = blockIdx.y;
   if (r >= R) return;
@@ -255,16 +256,27 @@ __global__ void forward_solve_batched_kernel(const float* __restrict__ L,
   const int tid = threadIdx.x;
   for (int k0 = 0; k0 < N; k0 += TRSV_BS) {
     const int bs = min(TRSV_BS, N - k0);
+    // stage panel of L and the panel's rhs entries into LDS in parallel
     for (int idx = tid; idx < bs * bs; idx += blockDim.x)
       S[idx / bs][idx % bs] = Lb[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
+    if (tid < bs) yp[tid] = y[(k0 + tid) * R + r];
     __syncthreads();
-    if (tid == 0) {
-      for (int j = 0; j < bs; ++j) {
-        float v = y[(k0 + j) * R + r];
-        for (int t = 0; t < j; ++t) v = fmaf(-S[j][t], z[t], v);
-        z[j] = v / S[j][j];
-        y[(k0 + j) * R + r] = z[j];
+    // wave-parallel diagonal solve: lane j owns panel row j; per column t
+    // the solved z_t is broadcast with __shfl — no block barrier, no
+    // global traffic inside the serial dependency chain.
+    if (tid < 64) {
+      const int j = tid;  // lanes >= bs just follow along
+      float v = (j < bs) ? yp[j] : 0.0f;
+      for (int t = 0; t < bs; ++t) {
+        const float zt = __shfl(v, t) / S[t][t];
+        if (j == t) {
+          v = zt;
+          z[t] = zt;
+        } else if (j > t && j < bs) {
+          v = fmaf(-S[j][t], zt, v);
+        }
       }
+      if (j < bs) y[(k0 + j) * R + r] = v;
     }
     __syncthreads();
     // rank-bs update of the remaining rows
@@ -285,6 +297,7 @@ __global__ void backward_solve_batched_kernel(const float* __restrict__ L,
                                               int N, int R) {
   __shared__ float S[TRSV_BS][TRSV_BS + 1];
   __shared__ float z[TRSV_BS];
+  __shared__ float yp[TRSV_BS];
   const int b = blockIdx.x;
   const int r = blockIdx.y;
   if (r >= R) return;
@@ -296,14 +309,22 @@ __global__ void backward_solve_batched_kernel(const float* __restrict__ L,
     const int bs = min(TRSV_BS, N - k0);
     for (int idx = tid; idx < bs * bs; idx += blockDim.x)
       S[idx / bs][idx % bs] = Lb[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
+    if (tid < bs) yp[tid] = y[(k0 + tid) * R + r];
     __syncthreads();
-    if (tid == 0) {
-      for (int j = bs - 1; j >= 0; --j) {
-        float v = y[(k0 + j) * R + r];
-        for (int t = j + 1; t < bs; ++t) v = fmaf(-S[t][j], z[t], v);
-        z[j] = v / S[j][j];
-        y[(k0 + j) * R + r] = z[j];
+    // wave-parallel diagonal solve against S^T, columns walked bottom-up
+    if (tid < 64) {
+      const int j = tid;
+      float v = (j < bs) ? yp[j] : 0.0f;
+      for (int t = bs - 1; t >= 0; --t) {
+        const float zt = __shfl(v, t) / S[t][t];
+        if (j == t) {
+          v = zt;
+          z[t] = zt;
+        } else if (j < t) {
+          v = fmaf(-S[t][j], zt, v);
+        }
       }
+      if (j < bs) y[(k0 + j) * R + r] = v;
     }
     __syncthreads();
     // update rows above the panel: x_i -= sum_t L[k0+t][i] * z[t]
