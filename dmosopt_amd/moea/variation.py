@@ -2,10 +2,14 @@
 
 The reference NSGA2/AGEMOEA generate loop (NSGA2.py:141-177,
 AGEMOEA.py:146-180): a Bernoulli event stream of crossovers (2 children)
-and mutations (1 child) until popsize-1 children exist. Here the stream is
-drawn on the host and ALL variation executes as two fused device launches
-(Philox SBX + polynomial mutation kernels on gfx950), assembled into event
-order with a single gather.
+and mutations (1 child) until popsize-1 children exist. The stream here is
+drawn VECTORIZED on the host (one `rng.random((n, 2))` draw instead of a
+per-event Python loop), all index bookkeeping is assembled with numpy, a
+single H2D transfer carries every index array, and all variation executes
+as two fused device launches (Philox SBX + polynomial mutation kernels on
+gfx950) assembled into event order with one gather. Pair selection is
+uniform over ordered distinct pairs (statistically equivalent to the
+reference's `choice(poolsize, 2, replace=False)`).
 """
 
 from __future__ import annotations
@@ -16,6 +20,35 @@ import numpy as np
 import torch
 
 from dmosopt_amd import ops
+
+
+def _draw_event_stream(rng, popsize: int, crossover_prob: float, mutation_prob: float):
+    """Vectorized replica of the reference's while-loop event stream.
+
+    Each iteration draws (u_c, u_m); a crossover event adds 2 children, a
+    mutation event adds 1; the loop body runs while count < popsize - 1 at
+    iteration start. Returns boolean arrays (c_ev, m_ev) over the executed
+    iterations.
+    """
+    target = popsize - 1
+    c_parts, m_parts = [], []
+    cum = 0
+    while True:
+        chunk = max(64, popsize)
+        u = rng.random((chunk, 2))
+        c = u[:, 0] < crossover_prob
+        m = u[:, 1] < mutation_prob
+        inc = 2 * c.astype(np.int64) + m.astype(np.int64)
+        cs = cum + np.cumsum(inc)
+        j0 = int(np.searchsorted(cs, target, side="left"))
+        if j0 < chunk:  # iteration j0 is the first to START with count >= target
+            c_parts.append(c[: j0 + 1])
+            m_parts.append(m[: j0 + 1])
+            break
+        c_parts.append(c)
+        m_parts.append(m)
+        cum = int(cs[-1])
+    return np.concatenate(c_parts), np.concatenate(m_parts)
 
 
 def event_stream_variation(
@@ -31,60 +64,66 @@ def event_stream_variation(
     xlb: torch.Tensor,
     xub: torch.Tensor,
     torch_random=None,
-) -> Tuple[torch.Tensor, np.ndarray, np.ndarray]:
-    """Returns (x_gen, crossover_slot_indices, mutation_slot_indices)."""
-    cross_pairs = []
-    mut_parents = []
-    order = []
-    count = 0
-    while count < popsize - 1:
-        if rng.random() < crossover_prob:
-            pidx = rng.choice(poolsize, 2, replace=False)
-            cross_pairs.append((int(pidx[0]), int(pidx[1])))
-            order.append("c")
-            count += 2
-        if rng.random() < mutation_prob:
-            mut_parents.append(int(rng.integers(low=0, high=poolsize)))
-            order.append("m")
-            count += 1
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Returns (x_gen, crossover_slot_indices, mutation_slot_indices).
 
-    children_c1 = children_c2 = children_m = None
-    if cross_pairs:
-        i1 = torch.tensor([a for a, _ in cross_pairs], dtype=torch.long, device=pool.device)
-        i2 = torch.tensor([b for _, b in cross_pairs], dtype=torch.long, device=pool.device)
-        children_c1, children_c2 = ops.sbx_from_pool(
-            pool, i1, i2, di_crossover, xlb, xub,
+    The slot-index outputs are int64 tensors on ``pool.device`` so callers
+    can track operator success without a device->host sync.
+    """
+    c_ev, m_ev = _draw_event_stream(rng, popsize, crossover_prob, mutation_prob)
+
+    C = int(c_ev.sum())
+    M = int(m_ev.sum())
+    if C:
+        i1 = rng.integers(0, poolsize, C)
+        i2 = rng.integers(0, poolsize - 1, C)
+        i2 = i2 + (i2 >= i1)
+    else:
+        i1 = i2 = np.empty(0, dtype=np.int64)
+    im = rng.integers(0, poolsize, M) if M else np.empty(0, dtype=np.int64)
+
+    # events in iteration order: within an iteration the crossover precedes
+    # the mutation (matches the loop body order)
+    ev_is_c = np.stack([c_ev, np.zeros_like(c_ev)], 1).ravel()[
+        np.stack([c_ev, m_ev], 1).ravel()
+    ]
+    sizes = np.where(ev_is_c, 2, 1)
+    starts = np.cumsum(sizes) - sizes  # slot offset of each event
+    total = int(sizes.sum())
+
+    crossover_indices = np.repeat(starts[ev_is_c], 2)
+    crossover_indices[1::2] += 1
+    mutation_indices = starts[~ev_is_c]
+
+    src_rows = np.empty(total, dtype=np.int64)
+    src_rows[crossover_indices[0::2]] = np.arange(C)
+    src_rows[crossover_indices[1::2]] = C + np.arange(C)
+    src_rows[mutation_indices] = 2 * C + np.arange(M)
+
+    # single H2D transfer for every index array of the generation
+    combined = np.concatenate([i1, i2, im, src_rows, crossover_indices, mutation_indices])
+    dev = torch.as_tensor(combined, dtype=torch.long, device=pool.device)
+    o = 0
+    i1_t = dev[o : o + C]; o += C
+    i2_t = dev[o : o + C]; o += C
+    im_t = dev[o : o + M]; o += M
+    gather_idx = dev[o : o + total]; o += total
+    c_idx_t = dev[o : o + 2 * C]; o += 2 * C
+    m_idx_t = dev[o : o + M]
+
+    parts = []
+    if C:
+        c1, c2 = ops.sbx_from_pool(
+            pool, i1_t, i2_t, di_crossover, xlb, xub,
             seed=int(rng.integers(0, 2**62)), generator=torch_random,
         )
-    if mut_parents:
-        im = torch.tensor(mut_parents, dtype=torch.long, device=pool.device)
-        children_m = ops.mutation_from_pool(
-            pool, im, di_mutation, xlb, xub, mutation_rate,
-            seed=int(rng.integers(0, 2**62)), generator=torch_random,
+        parts += [c1, c2]
+    if M:
+        parts.append(
+            ops.mutation_from_pool(
+                pool, im_t, di_mutation, xlb, xub, mutation_rate,
+                seed=int(rng.integers(0, 2**62)), generator=torch_random,
+            )
         )
-
-    C = len(cross_pairs)
-    src_rows = []
-    crossover_indices = []
-    mutation_indices = []
-    ci = mi = 0
-    slot = 0
-    for ev in order:
-        if ev == "c":
-            src_rows.extend([ci, C + ci])
-            crossover_indices.extend([slot, slot + 1])
-            ci += 1
-            slot += 2
-        else:
-            src_rows.append(2 * C + mi)
-            mutation_indices.append(slot)
-            mi += 1
-            slot += 1
-    parts = [t for t in (children_c1, children_c2, children_m) if t is not None]
-    src = torch.cat(parts, dim=0)
-    gather_idx = torch.tensor(src_rows, dtype=torch.long, device=src.device)
-    return (
-        src[gather_idx],
-        np.asarray(crossover_indices, dtype=int),
-        np.asarray(mutation_indices, dtype=int),
-    )
+    src = torch.cat(parts, dim=0) if len(parts) > 1 else parts[0]
+    return src[gather_idx], c_idx_t, m_idx_t
