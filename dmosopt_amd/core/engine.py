@@ -146,13 +146,21 @@ def optimize_loop(
             y_gen = _surrogate_eval(mdl, x_gen, optimize_mean_variance)
         optimizer.update(x_gen, y_gen, gen_state)
         n_eval += x_gen.shape[0]
-        x_new.append(_to_np(x_gen))
-        y_new.append(_to_np(y_gen))  # host archive copies (saved to H5)
+        # archive ON DEVICE; the host copy (for H5) happens once after the
+        # loop — a per-generation .cpu() here serializes every generation
+        # against the GPU queue
+        x_new.append(x_gen)
+        y_new.append(y_gen)
         gen_indexes.append(np.full((x_gen.shape[0],), i, dtype=np.uint32))
 
+    def _stack_np(first, parts):
+        if parts and all(isinstance(p, torch.Tensor) for p in parts):
+            return np.vstack([first, _to_np(torch.cat(parts, dim=0)).astype(first.dtype)])
+        return np.vstack([first] + [np.asarray(_to_np(p), dtype=first.dtype) for p in parts])
+
     gen_index = np.concatenate(gen_indexes)
-    x_all = np.vstack([x] + x_new)
-    y_all = np.vstack([y] + y_new)
+    x_all = _stack_np(x, x_new)
+    y_all = _stack_np(y, y_new)
     bestx, besty = optimizer.population_objectives
     return EpochResults(_to_np(bestx), _to_np(besty), gen_index, x_all, y_all, optimizer)
 

@@ -124,7 +124,7 @@ class NSGA2Optimizer(MOEA):
         rank = self.state.rank
 
         pool_idx = ops.tournament_selection(
-            population.shape[0], poolsize, [rank], rng
+            population.shape[0], poolsize, [rank], rng, generator=self.torch_random
         )
         pool = population[pool_idx]
 
@@ -154,16 +154,25 @@ class NSGA2Optimizer(MOEA):
             x_dists=self._x_dists(population_parm),
             y_distance_metrics=self.y_distance_metrics,
         )
-        # device-side survivor accounting (torch.isin against the surviving
-        # permutation; no host round-trip)
+        # device-side survivor accounting without a host round-trip: the
+        # generation's children occupy rows [0, n_children) of the
+        # concatenated population, so survival is just `perm < n_children`
+        # plus a boolean gather over the slot-type mask (fixed-shape ops
+        # only — masked_select/isin would force a sync or a sort)
         c_idx = gen_state["crossover_indices"]
         m_idx = gen_state["mutation_indices"]
         if not isinstance(c_idx, torch.Tensor):
             c_idx = torch.as_tensor(np.asarray(c_idx), dtype=torch.long, device=perm.device)
         if not isinstance(m_idx, torch.Tensor):
             m_idx = torch.as_tensor(np.asarray(m_idx), dtype=torch.long, device=perm.device)
-        self.state.successful_crossovers += torch.isin(c_idx, perm).sum() // 2
-        self.state.successful_mutations += torch.isin(m_idx, perm).sum()
+        n_children = x_gen.shape[0]
+        is_cross = torch.zeros(n_children, dtype=torch.bool, device=perm.device)
+        is_cross[c_idx] = True
+        child = perm < n_children
+        slot = torch.where(child, perm, torch.zeros_like(perm))
+        surv_cross = is_cross[slot] & child
+        self.state.successful_crossovers += surv_cross.sum() // 2
+        self.state.successful_mutations += ((~is_cross[slot]) & child).sum()
 
         self.state.population_parm = parm
         self.state.population_obj = obj

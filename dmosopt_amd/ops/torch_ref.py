@@ -219,20 +219,42 @@ def tournament_selection(
     poolsize: int,
     metrics: Sequence[Tensor],
     np_random,
+    generator=None,
 ) -> Tensor:
     """Tournament selection into the mating pool (reference MOEA.py:385-395).
 
     Sorts candidates by lexsort(metrics) and samples `poolsize` without
-    replacement with geometric probability over sorted position. The draw
-    itself uses the host numpy Generator for reproducibility of the control
-    stream; the heavy sort keys may live on device.
+    replacement with geometric probability over sorted position. With a
+    device ``generator`` and device metrics the weighted draw runs on the
+    GPU via the Gumbel top-k trick (exact weighted sampling without
+    replacement: argmax_k of log(p_i) + Gumbel noise); otherwise the host
+    numpy Generator draws, matching the reference's control stream.
     """
     dev_metrics = [m if isinstance(m, torch.Tensor) else torch.as_tensor(m) for m in metrics]
     sorted_candidates = lexsort(dev_metrics)
+    dev = sorted_candidates.device
+    if generator is not None and dev.type == "cuda":
+        logp = _tournament_logp(pop, dev)
+        u = torch.rand(pop, dtype=torch.float32, device=dev, generator=generator)
+        keys = logp - torch.log(-torch.log(u.clamp_min(1e-12)).clamp_min(1e-12))
+        pool_pos = torch.topk(keys, poolsize).indices
+        return sorted_candidates[pool_pos]
     prob = tournament_prob_vector(pop).numpy()  # cached
     pool_pos = np_random.choice(pop, size=poolsize, p=prob, replace=False)
-    pool_pos = torch.as_tensor(pool_pos, dtype=torch.long, device=sorted_candidates.device)
+    pool_pos = torch.as_tensor(pool_pos, dtype=torch.long, device=dev)
     return sorted_candidates[pool_pos]
+
+
+_TOURNAMENT_LOGP_CACHE = {}
+
+
+def _tournament_logp(pop: int, device) -> Tensor:
+    key = (pop, str(device))
+    out = _TOURNAMENT_LOGP_CACHE.get(key)
+    if out is None:
+        out = tournament_prob_vector(pop).log().to(device=device, dtype=torch.float32)
+        _TOURNAMENT_LOGP_CACHE[key] = out
+    return out
 
 
 # ----------------------------------------------------------------- distance

@@ -19,4 +19,6 @@ print(f"epoch(50 gens) wall: {time.time()-t0:.2f}s")
 s = io.StringIO()
 ps = pstats.Stats(pr, stream=s).sort_stats("cumulative")
 ps.print_stats(35)
+ps.print_callers("as_tensor")
+ps.print_callers("'cpu'")
 print(s.getvalue())
