@@ -108,9 +108,9 @@ def sceua_batched(
 
     def gnrng_of(x_):
         rngs = (x_.max(dim=1).values - x_.min(dim=1).values) / bd
-        return torch.exp(torch.log(rngs.clamp_min(1e-300)).mean(dim=1)).cpu().numpy()
+        return torch.exp(torch.log(rngs.clamp_min(1e-300)).mean(dim=1))
 
-    gnrng = gnrng_of(x)
+    gnrng = gnrng_of(x).cpu().numpy()
     criter = [[] for _ in range(S)]
     criter_change = np.full(S, 1e5)
     sid_sg = torch.arange(S, device=device).repeat_interleave(G)
@@ -123,13 +123,21 @@ def sceua_batched(
         nloop += 1
         act = torch.as_tensor(act_np, device=device)
 
+        # all simplex selections of this shuffle drawn up front: ONE small
+        # H2D instead of one per CCE stage (a pageable-memory copy inside
+        # the stage loop near-synchronizes the stream and kills pipelining)
+        lcs_all = torch.as_tensor(
+            np.stack([_select_simplex(nps, npg, rng) for _ in range(nspl)]),
+            device=device,
+        )
+
         # partition: position p of complex g sits at row p*ngs+g
         cx = x.view(S, npg, G, nopt).permute(0, 2, 1, 3).contiguous()  # (S,G,npg,nopt)
         cf = xf.view(S, npg, G).permute(0, 2, 1).contiguous()  # (S,G,npg)
         icall_dev = torch.zeros(S, dtype=torch.float64, device=device)
 
         for _step in range(nspl):
-            lcs = torch.as_tensor(_select_simplex(nps, npg, rng), device=device)
+            lcs = lcs_all[_step]
             s_pts = cx[:, :, lcs, :]  # (S,G,nps,nopt)
             s_f = cf[:, :, lcs]  # (S,G,nps)
 
@@ -179,10 +187,13 @@ def sceua_batched(
         x = cx.permute(0, 2, 1, 3).reshape(S, npt, nopt)
         xf = cf.permute(0, 2, 1).reshape(S, npt)
         x, xf = sort_pop(x, xf)
-        icall += icall_dev.cpu().numpy().astype(np.int64) * act_np
-        gnrng = gnrng_of(x)
-
-        bestf_now = xf[:, 0].cpu().numpy()
+        # ONE device->host transfer per shuffle for every termination input
+        pack = torch.cat(
+            [icall_dev, gnrng_of(x), xf[:, 0].to(torch.float64)]
+        ).cpu().numpy()
+        icall += pack[:S].astype(np.int64) * act_np
+        gnrng = pack[S : 2 * S]
+        bestf_now = pack[2 * S :]
         for s in range(S):
             if act_np[s]:
                 criter[s].append(float(bestf_now[s]))
