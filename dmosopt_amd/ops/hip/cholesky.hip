@@ -233,6 +233,125 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
   if (tid == 0) logdet[b] = L.ld_accum;
 }
 
+// ---------------------------------------------------------------------------
+// Multi-launch right-looking factorization for SMALL batch counts.
+//
+// The single-block kernel above runs one block per matrix: at the headline
+// shape (B ~ 12 systems of N=300 from the speculative SCE-UA stages) that
+// leaves 244 of 256 CUs idle. Here each 32-column step is two stream-ordered
+// launches — a panel kernel (grid B) and a trailing-update SYRK kernel
+// (grid B x tile-pairs, 64x64 tiles) — so the SYRK, which is ~90% of the
+// FLOPs, spreads across B * O((N/64)^2) workgroups.
+
+#define CHOLP_TPB 256
+#define SYRK_TS 64  // trailing-update tile edge
+
+// Factor the 32x32 diagonal block at (k0,k0) and panel-solve the rows below
+// it. One block per matrix; also accumulates the step's logdet contribution
+// deterministically (no atomics: k-steps are stream-ordered).
+__global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
+    float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
+    int N, int k0) {
+  __shared__ float S[CHOL_BS][CHOL_BS + 1];
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  float* Ab = A + (long long)b * N * N;
+  const int bs = min(CHOL_BS, N - k0);
+
+  for (int idx = tid; idx < bs * bs; idx += blockDim.x)
+    S[idx / bs][idx % bs] = Ab[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
+  __syncthreads();
+
+  for (int j = 0; j < bs; ++j) {
+    if (tid == 0) {
+      float d = S[j][j];
+      if (d <= 0.0f || !isfinite(d)) {
+        if (info[b] == 0) info[b] = k0 + j + 1;
+        d = 1e-30f;
+      }
+      S[j][j] = sqrtf(d);
+    }
+    __syncthreads();
+    if (tid > j && tid < bs) S[tid][j] /= S[j][j];
+    __syncthreads();
+    const int rem = bs - j - 1;
+    for (int idx = tid; idx < rem * rem; idx += blockDim.x) {
+      const int i = j + 1 + idx / rem;
+      const int t = j + 1 + idx % rem;
+      if (t <= i) S[i][t] -= S[i][j] * S[t][j];
+    }
+    __syncthreads();
+  }
+  if (tid == 0) {
+    float s = 0.0f;
+    for (int t = 0; t < bs; ++t) s += logf(S[t][t]);
+    logdet[b] += s;
+  }
+  // write the factored diagonal block back (lower triangle)
+  for (int idx = tid; idx < bs * bs; idx += blockDim.x) {
+    const int i = idx / bs, t = idx % bs;
+    if (t <= i) Ab[(long long)(k0 + i) * N + k0 + t] = S[i][t];
+  }
+  __syncthreads();
+  // panel solve: each thread owns whole rows below the block — its running
+  // row values stay in registers, the diag factor is read from LDS, so the
+  // 32 sequential columns need no further barriers
+  for (int i = k0 + bs + tid; i < N; i += blockDim.x) {
+    float rv[CHOL_BS];
+    float* arow = Ab + (long long)i * N + k0;
+    for (int t = 0; t < bs; ++t) rv[t] = arow[t];
+    for (int j = 0; j < bs; ++j) {
+      float v = rv[j];
+      for (int t = 0; t < j; ++t) v = fmaf(-rv[t], S[j][t], v);
+      rv[j] = v / S[j][j];
+    }
+    for (int t = 0; t < bs; ++t) arow[t] = rv[t];
+  }
+}
+
+// Trailing update A[ti,tj] -= P_i P_j^T over 64x64 tiles of the submatrix
+// below/right of the panel; blockIdx.y enumerates lower-triangular tile
+// pairs, each 256-thread block computes a 4x4 register tile per thread.
+__global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
+    float* __restrict__ A, int N, int k0, int nt) {
+  __shared__ float Pi[SYRK_TS][CHOL_BS + 1];
+  __shared__ float Pj[SYRK_TS][CHOL_BS + 1];
+  const int b = blockIdx.x;
+  float* Ab = A + (long long)b * N * N;
+  const int r0 = k0 + CHOL_BS;  // first trailing row
+  // decode lower-triangular pair p -> (ti, tj), tj <= ti
+  int p = blockIdx.y, ti = 0;
+  while (p > ti) { p -= ti + 1; ++ti; }
+  const int tj = p;
+  const int i0 = r0 + ti * SYRK_TS, j0 = r0 + tj * SYRK_TS;
+  const int tid = threadIdx.x;
+
+  for (int idx = tid; idx < SYRK_TS * CHOL_BS; idx += blockDim.x) {
+    const int r = idx / CHOL_BS, c = idx % CHOL_BS;
+    Pi[r][c] = (i0 + r < N) ? Ab[(long long)(i0 + r) * N + k0 + c] : 0.0f;
+    Pj[r][c] = (j0 + r < N) ? Ab[(long long)(j0 + r) * N + k0 + c] : 0.0f;
+  }
+  __syncthreads();
+
+  const int ty = tid / 16, tx = tid % 16;  // 16x16 threads, 4x4 each
+  float acc[4][4] = {};
+  for (int kk = 0; kk < CHOL_BS; ++kk) {
+    float a[4], c[4];
+    for (int r = 0; r < 4; ++r) a[r] = Pi[ty * 4 + r][kk];
+    for (int r = 0; r < 4; ++r) c[r] = Pj[tx * 4 + r][kk];
+    for (int r = 0; r < 4; ++r)
+      for (int q = 0; q < 4; ++q) acc[r][q] = fmaf(a[r], c[q], acc[r][q]);
+  }
+  for (int r = 0; r < 4; ++r) {
+    const int i = i0 + ty * 4 + r;
+    if (i >= N) continue;
+    for (int q = 0; q < 4; ++q) {
+      const int j = j0 + tx * 4 + q;
+      if (j < N && j <= i) Ab[(long long)i * N + j] -= acc[r][q];
+    }
+  }
+}
+
 // Blocked forward substitution: solve L z = y for R right-hand sides.
 // One block per (batch, rhs). Per 32-column panel: thread 0 solves the
 // 32x32 diagonal block serially from LDS while the other threads wait,
@@ -339,8 +458,37 @@ __global__ void backward_solve_batched_kernel(const float* __restrict__ L,
 
 #include <stdlib.h>
 
+extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
+                                       int B, int N, hipStream_t stream) {
+  hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
+  for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
+    hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0, stream,
+                       A, logdet, info, N, k0);
+    const int trailing = N - k0 - CHOL_BS;
+    if (trailing > 0) {
+      const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
+      hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, nt * (nt + 1) / 2),
+                         dim3(CHOLP_TPB), 0, stream, A, N, k0, nt);
+    }
+  }
+}
+
 extern "C" void launch_cholesky_batched(float* A, float* logdet, int* info,
                                         int B, int N, hipStream_t stream) {
+  static int multik_max_b = -2;
+  if (multik_max_b == -2) {
+    // below this batch count the one-block-per-matrix kernel cannot fill
+    // the 256 CUs, so the multi-launch right-looking path wins; override
+    // with DMOSOPT_CHOL_MODE=single|multik for A/B measurement
+    const char* env = getenv("DMOSOPT_CHOL_MODE");
+    if (env && env[0] == 's') multik_max_b = 0;
+    else if (env && env[0] == 'm') multik_max_b = 1 << 30;
+    else multik_max_b = 48;
+  }
+  if (B <= multik_max_b && N > CHOL_BS) {
+    launch_cholesky_multik(A, logdet, info, B, N, stream);
+    return;
+  }
   static int mode = -1;
   if (mode < 0) {
     // same-box A/B (profiles/README.md): LDS/barrier diag factor measured
