@@ -259,39 +259,62 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
   float* Ab = A + (long long)b * N * N;
   const int bs = min(CHOL_BS, N - k0);
 
-  for (int idx = tid; idx < bs * bs; idx += blockDim.x)
-    S[idx / bs][idx % bs] = Ab[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
-  __syncthreads();
-
-  for (int j = 0; j < bs; ++j) {
-    if (tid == 0) {
-      float d = S[j][j];
+  // Wave-parallel diagonal factor: lane i of the first wave owns row i of
+  // the 32x32 block in registers; column j's pivot and multipliers move by
+  // __shfl broadcast. Zero block barriers in the serial dependency chain
+  // (the barrier version spent ~96 barriers here and dominated the whole
+  // multi-launch path at 54 us per panel).
+  if (tid < 64) {
+    const int lane = tid;
+    float r[CHOL_BS];
+#pragma unroll
+    for (int t = 0; t < CHOL_BS; ++t)
+      r[t] = (lane < bs && t < bs)
+                 ? Ab[(long long)(k0 + lane) * N + k0 + t]
+                 : 0.0f;
+    float mylog = 0.0f;
+    int bad = 0;
+#pragma unroll
+    for (int j = 0; j < CHOL_BS; ++j) {
+      if (j >= bs) continue;
+      float d = __shfl(r[j], j);
       if (d <= 0.0f || !isfinite(d)) {
-        if (info[b] == 0) info[b] = k0 + j + 1;
+        bad = bad ? bad : (k0 + j + 1);
         d = 1e-30f;
       }
-      S[j][j] = sqrtf(d);
+      d = sqrtf(d);
+      if (lane == j) {
+        r[j] = d;
+        mylog = logf(d);
+      } else if (lane > j) {
+        r[j] /= d;
+      }
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t) {
+        if (t <= j || t >= bs) continue;
+        const float stj = __shfl(r[j], t);
+        if (lane >= t) r[t] = fmaf(-r[j], stj, r[t]);
+      }
     }
-    __syncthreads();
-    if (tid > j && tid < bs) S[tid][j] /= S[j][j];
-    __syncthreads();
-    const int rem = bs - j - 1;
-    for (int idx = tid; idx < rem * rem; idx += blockDim.x) {
-      const int i = j + 1 + idx / rem;
-      const int t = j + 1 + idx % rem;
-      if (t <= i) S[i][t] -= S[i][j] * S[t][j];
+    // stage the factored block for the panel solve + write back lower part
+    if (lane < bs) {
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t) {
+        if (t >= bs) continue;
+        S[lane][t] = r[t];
+        if (t <= lane) Ab[(long long)(k0 + lane) * N + k0 + t] = r[t];
+      }
     }
-    __syncthreads();
-  }
-  if (tid == 0) {
-    float s = 0.0f;
-    for (int t = 0; t < bs; ++t) s += logf(S[t][t]);
-    logdet[b] += s;
-  }
-  // write the factored diagonal block back (lower triangle)
-  for (int idx = tid; idx < bs * bs; idx += blockDim.x) {
-    const int i = idx / bs, t = idx % bs;
-    if (t <= i) Ab[(long long)(k0 + i) * N + k0 + t] = S[i][t];
+    // wave-reduce the logdet contribution and error flag
+    for (int off = 32; off > 0; off >>= 1) {
+      mylog += __shfl_down(mylog, off);
+      const int ob = __shfl_down(bad, off);
+      bad = bad ? bad : ob;
+    }
+    if (lane == 0) {
+      logdet[b] += mylog;
+      if (bad && info[b] == 0) info[b] = bad;
+    }
   }
   __syncthreads();
   // panel solve: each thread owns whole rows below the block — its running
