@@ -264,14 +264,18 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
   // __shfl broadcast. Zero block barriers in the serial dependency chain
   // (the barrier version spent ~96 barriers here and dominated the whole
   // multi-launch path at 54 us per panel).
+  // coalesced stage of the block into LDS by the whole workgroup first —
+  // a single wave doing the 32x32 global load directly issues ~1000
+  // uncoalesced (1200 B stride) loads and stalls on memory latency
+  for (int idx = tid; idx < bs * bs; idx += blockDim.x)
+    S[idx / bs][idx % bs] = Ab[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
+  __syncthreads();
   if (tid < 64) {
     const int lane = tid;
     float r[CHOL_BS];
 #pragma unroll
     for (int t = 0; t < CHOL_BS; ++t)
-      r[t] = (lane < bs && t < bs)
-                 ? Ab[(long long)(k0 + lane) * N + k0 + t]
-                 : 0.0f;
+      r[t] = (lane < bs && t < bs) ? S[lane][t] : 0.0f;
     float mylog = 0.0f;
     int bad = 0;
 #pragma unroll
@@ -296,13 +300,13 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
         if (lane >= t) r[t] = fmaf(-r[j], stj, r[t]);
       }
     }
-    // stage the factored block for the panel solve + write back lower part
+    // stage the factored block back to LDS (global writeback below is
+    // done coalesced by the whole workgroup)
     if (lane < bs) {
 #pragma unroll
       for (int t = 0; t < CHOL_BS; ++t) {
         if (t >= bs) continue;
         S[lane][t] = r[t];
-        if (t <= lane) Ab[(long long)(k0 + lane) * N + k0 + t] = r[t];
       }
     }
     // wave-reduce the logdet contribution and error flag
@@ -317,6 +321,10 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     }
   }
   __syncthreads();
+  for (int idx = tid; idx < bs * bs; idx += blockDim.x) {
+    const int i = idx / bs, t = idx % bs;
+    if (t <= i) Ab[(long long)(k0 + i) * N + k0 + t] = S[i][t];
+  }
   // panel solve: each thread owns whole rows below the block — its running
   // row values stay in registers, the diag factor is read from LDS, so the
   // 32 sequential columns need no further barriers
