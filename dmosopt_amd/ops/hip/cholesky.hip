@@ -326,17 +326,25 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
   }
   // panel solve: each thread owns whole rows below the block — its running
   // row values stay in registers, the diag factor is read from LDS, so the
-  // 32 sequential columns need no further barriers
-  for (int i = k0 + bs + tid; i < N; i += blockDim.x) {
+  // 32 sequential columns need no further barriers. Trailing rows exist
+  // only under a FULL panel (k0 + bs < N implies bs == CHOL_BS), so the
+  // trip counts are compile-time: fully unrolled, the 32 row loads issue
+  // independently instead of one load-wait per loop iteration (measured
+  // 60 us -> this is the fix for the constant-latency panel cost).
+  for (int i = k0 + CHOL_BS + tid; i < N; i += blockDim.x) {
     float rv[CHOL_BS];
     float* arow = Ab + (long long)i * N + k0;
-    for (int t = 0; t < bs; ++t) rv[t] = arow[t];
-    for (int j = 0; j < bs; ++j) {
+#pragma unroll
+    for (int t = 0; t < CHOL_BS; ++t) rv[t] = arow[t];
+#pragma unroll
+    for (int j = 0; j < CHOL_BS; ++j) {
       float v = rv[j];
+#pragma unroll
       for (int t = 0; t < j; ++t) v = fmaf(-rv[t], S[j][t], v);
       rv[j] = v / S[j][j];
     }
-    for (int t = 0; t < bs; ++t) arow[t] = rv[t];
+#pragma unroll
+    for (int t = 0; t < CHOL_BS; ++t) arow[t] = rv[t];
   }
 }
 
@@ -428,11 +436,13 @@ __global__ void forward_solve_batched_kernel(const float* __restrict__ L,
       if (j < bs) y[(k0 + j) * R + r] = v;
     }
     __syncthreads();
-    // rank-bs update of the remaining rows
-    for (int i = k0 + bs + tid; i < N; i += blockDim.x) {
+    // rank-32 update of the remaining rows: rows below exist only under a
+    // FULL panel, so the trip count is compile-time (unrolled loads)
+    for (int i = k0 + TRSV_BS + tid; i < N; i += blockDim.x) {
       float acc = y[i * R + r];
       const float* row = Lb + (long long)i * N + k0;
-      for (int t = 0; t < bs; ++t) acc = fmaf(-row[t], z[t], acc);
+#pragma unroll
+      for (int t = 0; t < TRSV_BS; ++t) acc = fmaf(-row[t], z[t], acc);
       y[i * R + r] = acc;
     }
     __syncthreads();
@@ -477,11 +487,22 @@ __global__ void backward_solve_batched_kernel(const float* __restrict__ L,
     }
     __syncthreads();
     // update rows above the panel: x_i -= sum_t L[k0+t][i] * z[t]
-    for (int i = tid; i < k0; i += blockDim.x) {
-      float acc = y[i * R + r];
-      for (int t = 0; t < bs; ++t)
-        acc = fmaf(-Lb[(long long)(k0 + t) * N + i], z[t], acc);
-      y[i * R + r] = acc;
+    // (specialized full-panel path so the column loads unroll)
+    if (bs == TRSV_BS) {
+      for (int i = tid; i < k0; i += blockDim.x) {
+        float acc = y[i * R + r];
+#pragma unroll
+        for (int t = 0; t < TRSV_BS; ++t)
+          acc = fmaf(-Lb[(long long)(k0 + t) * N + i], z[t], acc);
+        y[i * R + r] = acc;
+      }
+    } else {
+      for (int i = tid; i < k0; i += blockDim.x) {
+        float acc = y[i * R + r];
+        for (int t = 0; t < bs; ++t)
+          acc = fmaf(-Lb[(long long)(k0 + t) * N + i], z[t], acc);
+        y[i * R + r] = acc;
+      }
     }
     __syncthreads();
   }
