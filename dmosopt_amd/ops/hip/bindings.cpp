@@ -40,6 +40,7 @@ void launch_hv_fpras(const float*, const float*, const float*,
                      unsigned long long, hipStream_t);
 void launch_cmaes_update(float*, float*, float*, const float*, const float*,
                          int, int, float, float, float, hipStream_t);
+int launch_peel_from_y(const float*, int*, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
 }
 
@@ -117,14 +118,19 @@ torch::Tensor dominance_degree_matrix(torch::Tensor Y) {
 torch::Tensor pareto_rank(torch::Tensor Y) {
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
-  auto D = dominance_degree_matrix(Y);
   if (N <= 2048) {
-    // single-launch persistent peel
+    // single-launch peel straight from Y (no dominance matrix at all)
+    auto Yc = Y.contiguous().to(torch::kFloat32);
     auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));
+    if (launch_peel_from_y(Yc.data_ptr<float>(), rank.data_ptr<int>(), N, m,
+                           cur_stream()) == 0)
+      return rank.to(torch::kLong);
+    auto D = dominance_degree_matrix(Y);
     launch_peel_single_block(D.data_ptr<int>(), rank.data_ptr<int>(), N, m,
                              cur_stream());
     return rank.to(torch::kLong);
   }
+  auto D = dominance_degree_matrix(Y);
   auto alive = torch::ones({N}, Y.options().dtype(torch::kUInt8));
   auto front = torch::empty({N}, Y.options().dtype(torch::kUInt8));
   auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));

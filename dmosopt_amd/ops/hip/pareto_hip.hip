@@ -150,6 +150,87 @@ extern "C" void launch_peel_single_block(const int* D, int* rank, int N, int m,
                      stream, D, rank, N, m);
 }
 
+// Whole ranking from Y in ONE launch: Y (N x m) is staged in LDS and
+// dominance is recomputed on the fly — the (N x N) int32 dominance matrix
+// (640 KB at N=400) never exists, so the single CU running this block does
+// LDS compares instead of pulling the matrix from HBM twice. Semantics
+// identical to dominance_degree_matrix + zero_identical + peel: i dominates
+// j iff all objectives <= and not all equal.
+__global__ __launch_bounds__(PEEL1_TPB) void peel_from_y_kernel(
+    const float* __restrict__ Y, int* __restrict__ rank, int N, int m) {
+  extern __shared__ char sh_raw[];
+  float* Ys = (float*)sh_raw;                 // N * m
+  int* n_dom = (int*)(Ys + (size_t)N * m);    // N
+  int* front = n_dom + N;                     // N
+  int* ctrl = front + N;                      // [front_sz, remaining]
+  const int tid = threadIdx.x;
+
+  for (int i = tid; i < N * m; i += PEEL1_TPB) Ys[i] = Y[i];
+  for (int j = tid; j < N; j += PEEL1_TPB) n_dom[j] = 0;
+  if (tid == 0) ctrl[1] = N;
+  __syncthreads();
+
+  // dominator counts over all pairs (row-major order, LDS reads only)
+  for (long long p = tid; p < (long long)N * N; p += PEEL1_TPB) {
+    const int i = (int)(p / N), j = (int)(p % N);
+    if (i == j) continue;
+    bool le = true, lt = false;
+    for (int t = 0; t < m; ++t) {
+      const float a = Ys[i * m + t], b = Ys[j * m + t];
+      le &= (a <= b);
+      lt |= (a < b);
+    }
+    if (le && lt) atomicAdd(&n_dom[j], 1);
+  }
+  __syncthreads();
+
+  for (int k = 0; ctrl[1] > 0 && k <= N; ++k) {
+    if (tid == 0) ctrl[0] = 0;
+    __syncthreads();
+    for (int j = tid; j < N; j += PEEL1_TPB) {
+      if (n_dom[j] == 0) {
+        rank[j] = k;
+        n_dom[j] = -1;
+        front[atomicAdd(&ctrl[0], 1)] = j;
+      }
+    }
+    __syncthreads();
+    const int fs = ctrl[0];
+    if (fs == 0) break;
+    for (long long t = tid; t < (long long)fs * N; t += PEEL1_TPB) {
+      const int f = front[t / N];
+      const int j = (int)(t % N);
+      if (n_dom[j] > 0) {
+        bool le = true, lt = false;
+        for (int c = 0; c < m; ++c) {
+          const float a = Ys[f * m + c], b = Ys[j * m + c];
+          le &= (a <= b);
+          lt |= (a < b);
+        }
+        if (le && lt) atomicSub(&n_dom[j], 1);
+      }
+    }
+    __syncthreads();
+    if (tid == 0) ctrl[1] -= fs;
+    __syncthreads();
+  }
+}
+
+extern "C" int launch_peel_from_y(const float* Y, int* rank, int N, int m,
+                                  hipStream_t stream) {
+  const size_t lds = (size_t)N * m * sizeof(float) + (2 * N + 2) * sizeof(int);
+  if (lds > 144 * 1024) return -1;  // caller falls back to the D-matrix path
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)peel_from_y_kernel,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    attr_set = true;
+  }
+  hipLaunchKernelGGL(peel_from_y_kernel, dim3(1), dim3(PEEL1_TPB), lds, stream,
+                     Y, rank, N, m);
+  return 0;
+}
+
 extern "C" void launch_dominance_matrix(const float* Y, int* D, int N, int m,
                                         hipStream_t stream) {
   dim3 grid((N + PTILE - 1) / PTILE, (N + PTILE - 1) / PTILE);
