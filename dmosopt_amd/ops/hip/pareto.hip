@@ -155,7 +155,8 @@ extern "C" void launch_peel_single_block(const int* D, int* rank, int N, int m,
 // LDS compares instead of pulling the matrix from HBM twice. Semantics
 // identical to dominance_degree_matrix + zero_identical + peel: i dominates
 // j iff all objectives <= and not all equal.
-__global__ __launch_bounds__(PEEL1_TPB) void peel_from_y_kernel(
+template <int TPB>
+__global__ __launch_bounds__(TPB) void peel_from_y_kernel(
     const float* __restrict__ Y, int* __restrict__ rank, int N, int m) {
   extern __shared__ char sh_raw[];
   float* Ys = (float*)sh_raw;                 // N * m
@@ -164,13 +165,13 @@ __global__ __launch_bounds__(PEEL1_TPB) void peel_from_y_kernel(
   int* ctrl = front + N;                      // [front_sz, remaining]
   const int tid = threadIdx.x;
 
-  for (int i = tid; i < N * m; i += PEEL1_TPB) Ys[i] = Y[i];
-  for (int j = tid; j < N; j += PEEL1_TPB) n_dom[j] = 0;
+  for (int i = tid; i < N * m; i += TPB) Ys[i] = Y[i];
+  for (int j = tid; j < N; j += TPB) n_dom[j] = 0;
   if (tid == 0) ctrl[1] = N;
   __syncthreads();
 
   // dominator counts over all pairs (row-major order, LDS reads only)
-  for (long long p = tid; p < (long long)N * N; p += PEEL1_TPB) {
+  for (long long p = tid; p < (long long)N * N; p += TPB) {
     const int i = (int)(p / N), j = (int)(p % N);
     if (i == j) continue;
     bool le = true, lt = false;
@@ -186,7 +187,7 @@ __global__ __launch_bounds__(PEEL1_TPB) void peel_from_y_kernel(
   for (int k = 0; ctrl[1] > 0 && k <= N; ++k) {
     if (tid == 0) ctrl[0] = 0;
     __syncthreads();
-    for (int j = tid; j < N; j += PEEL1_TPB) {
+    for (int j = tid; j < N; j += TPB) {
       if (n_dom[j] == 0) {
         rank[j] = k;
         n_dom[j] = -1;
@@ -196,7 +197,7 @@ __global__ __launch_bounds__(PEEL1_TPB) void peel_from_y_kernel(
     __syncthreads();
     const int fs = ctrl[0];
     if (fs == 0) break;
-    for (long long t = tid; t < (long long)fs * N; t += PEEL1_TPB) {
+    for (long long t = tid; t < (long long)fs * N; t += TPB) {
       const int f = front[t / N];
       const int j = (int)(t % N);
       if (n_dom[j] > 0) {
@@ -221,12 +222,22 @@ extern "C" int launch_peel_from_y(const float* Y, int* rank, int N, int m,
   if (lds > 144 * 1024) return -1;  // caller falls back to the D-matrix path
   static bool attr_set = false;
   if (!attr_set) {
-    hipFuncSetAttribute((const void*)peel_from_y_kernel,
+    hipFuncSetAttribute((const void*)peel_from_y_kernel<64>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    hipFuncSetAttribute((const void*)peel_from_y_kernel<PEEL1_TPB>,
                         hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
     attr_set = true;
   }
-  hipLaunchKernelGGL(peel_from_y_kernel, dim3(1), dim3(PEEL1_TPB), lds, stream,
-                     Y, rank, N, m);
+  // a converged population peels into MANY single-digit fronts: the round
+  // loop is then bound by block-barrier latency, which for a 16-wave block
+  // is ~1 us x ~3 barriers x #fronts. A single-wave block makes barriers
+  // ~free; its lower count-pass parallelism only matters for large N.
+  if (N <= 640)
+    hipLaunchKernelGGL(peel_from_y_kernel<64>, dim3(1), dim3(64), lds, stream,
+                       Y, rank, N, m);
+  else
+    hipLaunchKernelGGL(peel_from_y_kernel<PEEL1_TPB>, dim3(1), dim3(PEEL1_TPB),
+                       lds, stream, Y, rank, N, m);
   return 0;
 }
 
