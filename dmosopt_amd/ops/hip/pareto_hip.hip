@@ -171,17 +171,20 @@ __global__ __launch_bounds__(TPB) void peel_from_y_kernel(
   if (tid == 0) ctrl[1] = N;
   __syncthreads();
 
-  // dominator counts over all pairs (row-major order, LDS reads only)
-  for (long long p = tid; p < (long long)N * N; p += TPB) {
-    const int i = (int)(p / N), j = (int)(p % N);
-    if (i == j) continue;
-    bool le = true, lt = false;
-    for (int t = 0; t < m; ++t) {
-      const float a = Ys[i * m + t], b = Ys[j * m + t];
-      le &= (a <= b);
-      lt |= (a < b);
+  // dominator counts: thread OWNS columns j (register accumulate — no
+  // atomics, no 64-bit div/mod in the pair loop)
+  for (int j = tid; j < N; j += TPB) {
+    int cnt = 0;
+    for (int i = 0; i < N; ++i) {
+      bool le = true, lt = false;
+      for (int t = 0; t < m; ++t) {
+        const float a = Ys[i * m + t], b = Ys[j * m + t];
+        le &= (a <= b);
+        lt |= (a < b);
+      }
+      cnt += (le && lt) ? 1 : 0;
     }
-    if (le && lt) atomicAdd(&n_dom[j], 1);
+    n_dom[j] = cnt;
   }
   __syncthreads();
 
@@ -198,18 +201,21 @@ __global__ __launch_bounds__(TPB) void peel_from_y_kernel(
     __syncthreads();
     const int fs = ctrl[0];
     if (fs == 0) break;
-    for (long long t = tid; t < (long long)fs * N; t += TPB) {
-      const int f = front[t / N];
-      const int j = (int)(t % N);
-      if (n_dom[j] > 0) {
+    // column-owned update: subtract each peeled row's domination of j
+    for (int j = tid; j < N; j += TPB) {
+      if (n_dom[j] <= 0) continue;
+      int dec = 0;
+      for (int q = 0; q < fs; ++q) {
+        const int f = front[q];
         bool le = true, lt = false;
         for (int c = 0; c < m; ++c) {
           const float a = Ys[f * m + c], b = Ys[j * m + c];
           le &= (a <= b);
           lt |= (a < b);
         }
-        if (le && lt) atomicSub(&n_dom[j], 1);
+        dec += (le && lt) ? 1 : 0;
       }
+      n_dom[j] -= dec;
     }
     __syncthreads();
     if (tid == 0) ctrl[1] -= fs;
@@ -225,6 +231,8 @@ extern "C" int launch_peel_from_y(const float* Y, int* rank, int N, int m,
   if (!attr_set) {
     hipFuncSetAttribute((const void*)peel_from_y_kernel<64>,
                         hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    hipFuncSetAttribute((const void*)peel_from_y_kernel<256>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
     hipFuncSetAttribute((const void*)peel_from_y_kernel<PEEL1_TPB>,
                         hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
     attr_set = true;
@@ -233,9 +241,14 @@ extern "C" int launch_peel_from_y(const float* Y, int* rank, int N, int m,
   // loop is then bound by block-barrier latency, which for a 16-wave block
   // is ~1 us x ~3 barriers x #fronts. A single-wave block makes barriers
   // ~free; its lower count-pass parallelism only matters for large N.
-  if (N <= 640)
+  const char* tpb_env = getenv("DMOSOPT_PEEL_TPB");
+  const int tpb = tpb_env ? atoi(tpb_env) : (N <= 1024 ? 256 : PEEL1_TPB);
+  if (tpb <= 64)
     hipLaunchKernelGGL(peel_from_y_kernel<64>, dim3(1), dim3(64), lds, stream,
                        Y, rank, N, m);
+  else if (tpb <= 256)
+    hipLaunchKernelGGL(peel_from_y_kernel<256>, dim3(1), dim3(256), lds,
+                       stream, Y, rank, N, m);
   else
     hipLaunchKernelGGL(peel_from_y_kernel<PEEL1_TPB>, dim3(1), dim3(PEEL1_TPB),
                        lds, stream, Y, rank, N, m);
