@@ -41,6 +41,7 @@ void launch_hv_fpras(const float*, const float*, const float*,
 void launch_cmaes_update(float*, float*, float*, const float*, const float*,
                          int, int, float, float, float, hipStream_t);
 int launch_peel_from_y(const float*, int*, int, int, hipStream_t);
+int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
 }
 
@@ -119,9 +120,20 @@ torch::Tensor pareto_rank(torch::Tensor Y) {
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
   if (N <= 2048) {
-    // single-launch peel straight from Y (no dominance matrix at all)
     auto Yc = Y.contiguous().to(torch::kFloat32);
     auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));
+    // bit-matrix path: grid-wide packed dominator build + popcount peel
+    const int W = (N + 31) / 32;
+    const size_t bits_lds = ((size_t)N * W + W) * sizeof(unsigned int) +
+                            (N + 2) * sizeof(int);
+    if (bits_lds <= 144 * 1024) {
+      auto Dbits = torch::empty({N, W}, Y.options().dtype(torch::kInt32));
+      if (launch_peel_bits(Yc.data_ptr<float>(),
+                           (unsigned int*)Dbits.data_ptr<int>(),
+                           rank.data_ptr<int>(), N, m, cur_stream()) == 0)
+        return rank.to(torch::kLong);
+    }
+    // fallback: one-launch peel recomputing dominance from Y in LDS
     if (launch_peel_from_y(Yc.data_ptr<float>(), rank.data_ptr<int>(), N, m,
                            cur_stream()) == 0)
       return rank.to(torch::kLong);

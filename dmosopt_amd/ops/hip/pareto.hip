@@ -222,6 +222,103 @@ __global__ __launch_bounds__(TPB) void peel_from_y_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Bit-matrix ranking: dominators of column j packed 32/word, built GRID-WIDE
+// (the count pass was the single-CU bottleneck of the one-launch peel), then
+// a single small block peels fronts with popcount(dom_mask & front_mask) —
+// each round costs ~N/TPB * N/32 word-ops instead of N^2/TPB pair compares.
+__global__ void dom_bits_kernel(const float* __restrict__ Y,
+                                unsigned int* __restrict__ Dbits,  // (N, W)
+                                int N, int m, int W) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long long)N * W) return;
+  const int j = (int)(idx / W), w = (int)(idx % W);
+  unsigned int bits = 0;
+  const int i0 = w * 32;
+  const int iend = min(32, N - i0);
+  for (int b = 0; b < iend; ++b) {
+    const int i = i0 + b;
+    if (i == j) continue;
+    bool le = true, lt = false;
+    for (int t = 0; t < m; ++t) {
+      const float a = Y[i * m + t], c = Y[j * m + t];
+      le &= (a <= c);
+      lt |= (a < c);
+    }
+    if (le && lt) bits |= (1u << b);
+  }
+  Dbits[idx] = bits;
+}
+
+#define PEELB_TPB 256
+
+__global__ __launch_bounds__(PEELB_TPB) void peel_bits_kernel(
+    const unsigned int* __restrict__ Dbits, int* __restrict__ rank, int N,
+    int W) {
+  extern __shared__ char sh_raw[];
+  unsigned int* Db = (unsigned int*)sh_raw;      // N * W
+  unsigned int* fmask = Db + (size_t)N * W;      // W
+  int* n_dom = (int*)(fmask + W);                // N
+  int* ctrl = n_dom + N;                         // [front_sz, remaining]
+  const int tid = threadIdx.x;
+
+  for (int i = tid; i < N * W; i += PEELB_TPB) Db[i] = Dbits[i];
+  if (tid == 0) ctrl[1] = N;
+  __syncthreads();
+  for (int j = tid; j < N; j += PEELB_TPB) {
+    int c = 0;
+    for (int w = 0; w < W; ++w) c += __popc(Db[j * W + w]);
+    n_dom[j] = c;
+  }
+  __syncthreads();
+
+  for (int k = 0; ctrl[1] > 0 && k <= N; ++k) {
+    if (tid == 0) ctrl[0] = 0;
+    for (int w = tid; w < W; w += PEELB_TPB) fmask[w] = 0u;
+    __syncthreads();
+    for (int j = tid; j < N; j += PEELB_TPB) {
+      if (n_dom[j] == 0) {
+        rank[j] = k;
+        n_dom[j] = -1;
+        atomicOr(&fmask[j >> 5], 1u << (j & 31));
+        atomicAdd(&ctrl[0], 1);
+      }
+    }
+    __syncthreads();
+    const int fs = ctrl[0];
+    if (fs == 0) break;
+    for (int j = tid; j < N; j += PEELB_TPB) {
+      if (n_dom[j] <= 0) continue;
+      int dec = 0;
+      for (int w = 0; w < W; ++w) dec += __popc(Db[j * W + w] & fmask[w]);
+      n_dom[j] -= dec;
+    }
+    __syncthreads();
+    if (tid == 0) ctrl[1] -= fs;
+    __syncthreads();
+  }
+}
+
+extern "C" int launch_peel_bits(const float* Y, unsigned int* Dbits_scratch,
+                                int* rank, int N, int m, hipStream_t stream) {
+  const int W = (N + 31) / 32;
+  const size_t lds = ((size_t)N * W + W) * sizeof(unsigned int) +
+                     (N + 2) * sizeof(int);
+  if (lds > 144 * 1024) return -1;
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)peel_bits_kernel,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    attr_set = true;
+  }
+  const long long total = (long long)N * W;
+  hipLaunchKernelGGL(dom_bits_kernel, dim3((int)((total + 255) / 256)),
+                     dim3(256), 0, stream, Y, Dbits_scratch, N, m, W);
+  hipLaunchKernelGGL(peel_bits_kernel, dim3(1), dim3(PEELB_TPB), lds, stream,
+                     Dbits_scratch, rank, N, W);
+  return 0;
+}
+
 extern "C" int launch_peel_from_y(const float* Y, int* rank, int N, int m,
                                   hipStream_t stream) {
   const size_t lds = (size_t)N * m * sizeof(float) + (2 * N + 2) * sizeof(int);
