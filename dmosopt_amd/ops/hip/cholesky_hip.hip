@@ -325,27 +325,36 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     const int i = idx / bs, t = idx % bs;
     if (t <= i) Ab[(long long)(k0 + i) * N + k0 + t] = S[i][t];
   }
-  // panel solve: each thread owns whole rows below the block — its running
-  // row values stay in registers, the diag factor is read from LDS, so the
-  // 32 sequential columns need no further barriers. Trailing rows exist
-  // only under a FULL panel (k0 + bs < N implies bs == CHOL_BS), so the
-  // trip counts are compile-time: fully unrolled, the 32 row loads issue
-  // independently instead of one load-wait per loop iteration (measured
-  // 60 us -> this is the fix for the constant-latency panel cost).
-  for (int i = k0 + CHOL_BS + tid; i < N; i += blockDim.x) {
-    float rv[CHOL_BS];
-    float* arow = Ab + (long long)i * N + k0;
-#pragma unroll
-    for (int t = 0; t < CHOL_BS; ++t) rv[t] = arow[t];
-#pragma unroll
-    for (int j = 0; j < CHOL_BS; ++j) {
-      float v = rv[j];
-#pragma unroll
-      for (int t = 0; t < j; ++t) v = fmaf(-rv[t], S[j][t], v);
-      rv[j] = v / S[j][j];
+  // panel solve, chunked through LDS: per-thread direct row loads are
+  // uncoalesced (64 lanes touch 64 cache lines per load instruction, a
+  // constant ~27 us regardless of row count), so each 256-row chunk is
+  // staged with a COALESCED copy, solved entirely in LDS (one row per
+  // thread; the +1-padded stride keeps lanes on distinct banks), and
+  // written back coalesced. Trailing rows exist only under a FULL panel
+  // (k0 + bs < N implies bs == CHOL_BS): constant trip counts throughout.
+  __shared__ float P[CHOLP_TPB][CHOL_BS + 1];
+  for (int c0 = k0 + CHOL_BS; c0 < N; c0 += CHOLP_TPB) {
+    const int rows = min(CHOLP_TPB, N - c0);
+    for (int idx = tid; idx < rows * CHOL_BS; idx += CHOLP_TPB) {
+      const int r = idx / CHOL_BS, c = idx % CHOL_BS;
+      P[r][c] = Ab[(long long)(c0 + r) * N + k0 + c];
     }
+    __syncthreads();
+    if (tid < rows) {
 #pragma unroll
-    for (int t = 0; t < CHOL_BS; ++t) arow[t] = rv[t];
+      for (int j = 0; j < CHOL_BS; ++j) {
+        float v = P[tid][j];
+#pragma unroll
+        for (int t = 0; t < j; ++t) v = fmaf(-P[tid][t], S[j][t], v);
+        P[tid][j] = v / S[j][j];
+      }
+    }
+    __syncthreads();
+    for (int idx = tid; idx < rows * CHOL_BS; idx += CHOLP_TPB) {
+      const int r = idx / CHOL_BS, c = idx % CHOL_BS;
+      Ab[(long long)(c0 + r) * N + k0 + c] = P[r][c];
+    }
+    __syncthreads();
   }
 }
 
