@@ -415,7 +415,6 @@ __global__ void forward_solve_batched_kernel(const float* __restrict__ L,
   __shared__ float S[TRSV_BS][TRSV_BS + 1];
   __shared__ float z[TRSV_BS];
   __shared__ float yp[TRSV_BS];
-  __shared__ float FP[256][TRSV_BS + 1];
   const int b = blockIdx.x;
   const int r = blockIdx.y;
   if (r >= R) return;
@@ -447,25 +446,16 @@ __global__ void forward_solve_batched_kernel(const float* __restrict__ L,
       if (j < bs) y[(k0 + j) * R + r] = v;
     }
     __syncthreads();
-    // rank-32 update of the remaining rows (rows below exist only under a
-    // FULL panel): the L row segments are staged per 256-row chunk with a
-    // COALESCED copy — per-thread direct row reads put 64 lanes on 64
-    // cache lines per load instruction and latency-bound the kernel.
-    for (int c0 = k0 + TRSV_BS; c0 < N; c0 += 256) {
-      const int rows = min(256, N - c0);
-      for (int idx = tid; idx < rows * TRSV_BS; idx += blockDim.x) {
-        const int rr = idx / TRSV_BS, cc = idx % TRSV_BS;
-        FP[rr][cc] = Lb[(long long)(c0 + rr) * N + k0 + cc];
-      }
-      __syncthreads();
-      if (tid < rows) {
-        float acc = y[(c0 + tid) * R + r];
+    // rank-32 update of the remaining rows: rows below exist only under a
+    // FULL panel, so the trip count is compile-time (unrolled loads)
+    for (int i = k0 + TRSV_BS + tid; i < N; i += blockDim.x) {
+      float acc = y[i * R + r];
+      const float* row = Lb + (long long)i * N + k0;
 #pragma unroll
-        for (int t = 0; t < TRSV_BS; ++t) acc = fmaf(-FP[tid][t], z[t], acc);
-        y[(c0 + tid) * R + r] = acc;
-      }
-      __syncthreads();
+      for (int t = 0; t < TRSV_BS; ++t) acc = fmaf(-row[t], z[t], acc);
+      y[i * R + r] = acc;
     }
+    __syncthreads();
   }
 }
 
