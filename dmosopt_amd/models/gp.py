@@ -213,7 +213,7 @@ class _GPRBase:
     def evaluate_tensor(self, x: torch.Tensor) -> torch.Tensor:
         """Device-resident evaluate: tensor in, tensor out, no host trip."""
         xq = self.normalize_query(x.to(self.device, self.dtype))
-        mean, _ = self._fitted.predict(xq)
+        mean, _ = self._fitted.predict(xq, return_var=self.return_mean_variance)
         return mean
 
 

@@ -201,15 +201,19 @@ class FittedGP:
             eye = torch.eye(N, dtype=X.dtype, device=X.device)[None].expand(m, N, N).contiguous()
             self.Kinv = ops.chol_solve_batched(self.L, eye)
 
-    def predict(self, Xq: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    def predict(self, Xq: torch.Tensor, return_var: bool = True):
         """Posterior mean and variance at Xq (P, d) -> ((P, m), (P, m)).
 
         Variance matches sklearn's return_std**2: diag K(x*,x*) [incl. noise
-        from the White term] minus k*^T K^-1 k*, scaled by y_std^2.
+        from the White term] minus k*^T K^-1 k*, scaled by y_std^2. With
+        ``return_var=False`` (the per-generation surrogate-evaluate path)
+        the quadratic term is skipped and (mean, None) returned.
         """
         Ks = build_kernel(Xq, self.X, self.theta, nu=self.nu, anisotropic=self.anisotropic)
         # (m, P, N)
         mean_n = torch.bmm(Ks, self.alpha)[:, :, 0]  # (m, P)
+        if not return_var:
+            return self.y_mean[None, :] + self.y_std[None, :] * mean_n.T, None
         sf2 = torch.exp(self.theta[:, 0])
         noise = torch.exp(self.theta[:, -1])
         kss = (sf2 + noise)[:, None]  # (m, 1): k(x,x) = sf2*1 + noise
