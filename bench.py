@@ -62,9 +62,22 @@ class ShardedGPObjective:
     def evaluate_tensor(self, x):
         if self.world == 1:
             return self.gp.evaluate_tensor(x)
-        return torch.as_tensor(
-            self.evaluate(x.detach().cpu().numpy()), dtype=x.dtype, device=x.device
-        )
+        import torch.distributed as dist
+
+        # device-resident shard -> predict -> ONE all_gather over RCCL/xGMI
+        # (no host round trip inside the generation loop)
+        P = x.shape[0]
+        pad = (self.world - P % self.world) % self.world
+        if pad:
+            x = torch.cat([x, x[-1:].expand(pad, -1)], dim=0)
+        shard = x[self.rank :: self.world]
+        xq = self.gp.normalize_query(shard.to(self.gp.device, self.gp.dtype))
+        mean, _ = self.gp._fitted.predict(xq, return_var=False)
+        mean = mean.to(torch.float32).contiguous()
+        out = [torch.empty_like(mean) for _ in range(self.world)]
+        dist.all_gather(out, mean)
+        full = torch.stack(out, dim=1).reshape(-1, mean.shape[1])
+        return full[:P].to(dtype=x.dtype, device=x.device)
 
     def evaluate(self, x):
         import torch.distributed as dist

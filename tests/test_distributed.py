@@ -45,6 +45,52 @@ def _worker(rank, world_size, port, out_q):
     out_q.put((rank, best is not None))
 
 
+def _shard_worker(rank, world_size, port, out_q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    from bench import ShardedGPObjective, make_archive
+    from dmosopt_amd.models.gp import GPRMatern
+
+    X, Y = make_archive(seed=5)
+    dev = torch.device("cpu")
+    gp = GPRMatern(X, Y, 30, 2, np.zeros(30), np.ones(30),
+                   optimizer="sceua", seed=7, device=dev)
+    obj = ShardedGPObjective(gp, rank, world_size, dev)
+    rng = np.random.default_rng(3)
+    xq = torch.as_tensor(rng.random((13, 30)))  # odd count exercises padding
+    got = obj.evaluate_tensor(xq)
+    want = gp.evaluate_tensor(xq)  # single-model reference on every rank
+    err = float((got.double() - want.double()).abs().max())
+    dist.destroy_process_group()
+    out_q.put((rank, err))
+
+
+def test_sharded_gp_objective_matches_single_rank():
+    """The bench's rank-sharded surrogate prediction (the path the driver
+    scales to 8 GPUs) must reassemble to the single-rank result exactly
+    (identical GP on every rank; all_gather interleave + padding)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_shard_worker, args=(r, 2, 29737, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    errs = []
+    for _ in range(2):
+        _, err = q.get(timeout=300)
+        errs.append(err)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert max(errs) < 1e-5, errs
+
+
 def test_two_rank_farm():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
