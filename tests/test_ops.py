@@ -225,3 +225,49 @@ def test_agemoea_survival_matches_reference_greedy(rng):
     finite = np.isfinite(crowd)
     assert np.allclose(crowd_fast[finite], crowd[finite], atol=1e-10)
     assert np.array_equal(np.isinf(crowd_fast), np.isinf(crowd))
+
+
+def test_pareto_rank_beats_naive_reference_walltime():
+    """Performance-regression gate in the reference's style
+    (test_dda_performance.py / test_hv_performance.py: fast path within a
+    wall-clock ratio of a reference implementation). Our vectorized DDA
+    ranking must beat a pure-Python per-pair peel outright at N=512 — a
+    generous inversion bound that survives noisy CI boxes."""
+    import time
+
+    rng = np.random.default_rng(11)
+    Y = torch.as_tensor(rng.random((512, 3)))
+
+    t0 = time.perf_counter()
+    fast = ops.pareto_rank(Y)
+    t_fast = time.perf_counter() - t0
+
+    def naive_rank(Yn):
+        n = Yn.shape[0]
+        rank = np.full(n, -1)
+        alive = np.ones(n, bool)
+        k = 0
+        while alive.any():
+            front = []
+            for j in np.flatnonzero(alive):
+                dominated = False
+                for i in np.flatnonzero(alive):
+                    if i == j:
+                        continue
+                    if np.all(Yn[i] <= Yn[j]) and np.any(Yn[i] < Yn[j]):
+                        dominated = True
+                        break
+                if not dominated:
+                    front.append(j)
+            for j in front:
+                rank[j] = k
+                alive[j] = False
+            k += 1
+        return rank
+
+    t0 = time.perf_counter()
+    want = naive_rank(Y.numpy())
+    t_naive = time.perf_counter() - t0
+
+    assert np.array_equal(fast.numpy(), want)
+    assert t_fast < t_naive, (t_fast, t_naive)
