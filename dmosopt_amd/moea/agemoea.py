@@ -83,8 +83,19 @@ def get_geometry(front: np.ndarray, extreme: np.ndarray) -> float:
     return float(p)
 
 
+def _minkowski_device(A: np.ndarray, B: np.ndarray, p: float) -> np.ndarray:
+    """Large Minkowski distance matrices routed through torch.cdist on the
+    GPU (np.power with a float exponent costs ~100 ns/element on the host;
+    a front of 1024 means 2M elements per generation)."""
+    At = torch.as_tensor(A, dtype=torch.float64, device="cuda")
+    Bt = torch.as_tensor(B, dtype=torch.float64, device="cuda")
+    return torch.cdist(At, Bt, p=float(p)).cpu().numpy()
+
+
 def minkowski_matrix(A: np.ndarray, B: np.ndarray, p: float) -> np.ndarray:
     """Pairwise Minkowski-p distances (supports p < 1), vectorized."""
+    if A.shape[0] * B.shape[0] >= 1 << 16 and torch.cuda.is_available() and np.isfinite(p):
+        return _minkowski_device(A, B, p)
     diff = np.abs(A[:, None, :] - B[None, :, :])
     return np.power(np.power(diff, p).sum(axis=2), 1.0 / p)
 
@@ -148,10 +159,20 @@ def environmental_selection(
     local_random, population_parm, population_obj, pop, nInput, nOutput,
     feasibility_model=None, logger=None,
 ):
-    """AGE-MOEA survivor selection (AGEMOEA.py:445-512). numpy in/out."""
-    xs = np.asarray(population_parm, dtype=np.float64)
-    ys = np.asarray(population_obj, dtype=np.float64)
-    rank = ops.pareto_rank(torch.as_tensor(ys)).cpu().numpy()
+    """AGE-MOEA survivor selection (AGEMOEA.py:445-512).
+
+    Accepts torch tensors (ranked on their device — the gfx950 bit-matrix
+    peel when on GPU) or numpy arrays; the greedy survival bookkeeping runs
+    on the host either way (serial by construction)."""
+    if isinstance(population_obj, torch.Tensor):
+        rank_t = ops.pareto_rank(population_obj.double())
+        xs = population_parm.detach().double().cpu().numpy()
+        ys = population_obj.detach().double().cpu().numpy()
+        rank = rank_t.cpu().numpy()
+    else:
+        xs = np.asarray(population_parm, dtype=np.float64)
+        ys = np.asarray(population_obj, dtype=np.float64)
+        rank = ops.pareto_rank(torch.as_tensor(ys)).cpu().numpy()
     order = np.argsort(rank, kind="stable")
     xs, ys, rank = xs[order], ys[order], rank[order]
     rmax = int(rank.max())
@@ -242,7 +263,7 @@ class AGEMOEAOptimizer(MOEA):
 
     def initialize_state(self, x, y, bounds, local_random, **params):
         xn, yn, rank, crowd = environmental_selection(
-            local_random, x.cpu().numpy(), y.cpu().numpy(),
+            local_random, x, y,
             self.opt_params.popsize, self.nInput, self.nOutput,
             feasibility_model=self.feasibility,
         )
@@ -281,7 +302,7 @@ class AGEMOEAOptimizer(MOEA):
         obj = torch.cat([self.state.population_obj, y_gen], dim=0)
         parm, obj = ops.remove_duplicates(parm, obj)
         xn, yn, rank, crowd = environmental_selection(
-            self.local_random, parm.cpu().numpy(), obj.cpu().numpy(),
+            self.local_random, parm, obj,
             p.popsize, self.nInput, self.nOutput,
             feasibility_model=self.feasibility,
         )
