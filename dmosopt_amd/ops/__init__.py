@@ -147,6 +147,23 @@ def order_mo(
             else:
                 raise RuntimeError(f"order_mo: unknown distance metric {metric}")
     x_dist_vals = list(x_dists) if x_dists else []
+    if (
+        not x_dist_vals
+        and len(y_dist_vals) == 1
+        and y.device.type == "cuda"
+    ):
+        # fused sort key for the common (rank, -crowding) case: one stable
+        # argsort on (rank << 32 | ~monotone_float32_bits(crowding))
+        # instead of two radix sorts + gathers per generation. Non-negative
+        # IEEE floats compare like their bit patterns, so descending
+        # crowding == ascending complemented bits; ties fall back to index
+        # order exactly like np.lexsort.
+        d = y_dist_vals[0].float().clamp_min(0.0)
+        d = torch.nan_to_num(d, nan=0.0, posinf=float(torch.finfo(torch.float32).max))
+        bits = d.view(torch.int32).to(torch.int64)
+        key = (rank.to(torch.int64) << 32) | ((0x7FFFFFFF - bits) & 0xFFFFFFFF)
+        perm = torch.argsort(key, stable=True)
+        return perm, rank[perm], (y_dist_vals[0][perm],)
     keys = [-d for d in x_dist_vals] + [-d for d in y_dist_vals] + [rank.to(y.dtype)]
     perm = lexsort(keys)
     return perm, rank[perm], tuple(d[perm] for d in y_dist_vals)
