@@ -253,7 +253,6 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
     int N, int k0) {
   __shared__ float S[CHOL_BS][CHOL_BS + 1];
-  __shared__ float colbuf[CHOL_BS];
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   float* Ab = A + (long long)b * N * N;
@@ -293,16 +292,10 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
       } else if (lane > j) {
         r[j] /= d;
       }
-      // broadcast column j through LDS: ONE ds_write, then the rank-1
-      // update reads colbuf[t] with plain pipelined LDS loads — the
-      // per-t __shfl chain (~530 bpermutes per 32x32 factor) was ~12 us
-      // of the panel kernel. Same-wave LDS ops are program-ordered, so
-      // no barrier is needed; volatile stops the compiler caching.
-      if (lane < bs) ((volatile float*)colbuf)[lane] = r[j];
 #pragma unroll
       for (int t = 0; t < CHOL_BS; ++t) {
         if (t <= j || t >= bs) continue;
-        const float stj = ((volatile float*)colbuf)[t];
+        const float stj = __shfl(r[j], t);
         if (lane >= t) r[t] = fmaf(-r[j], stj, r[t]);
       }
     }
