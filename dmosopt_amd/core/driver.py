@@ -141,6 +141,13 @@ class DistOptimizer:
         device=None,
         **kwargs,
     ) -> None:
+        if device is None:
+            import torch
+
+            if torch.cuda.is_available():
+                # a GPU box runs the whole stack on the GPU unless told
+                # otherwise (surrogate fit AND the MOEA populations)
+                device = torch.device("cuda")
         if (random_seed is not None) and (local_random is not None):
             raise RuntimeError(
                 "Both random_seed and local_random are specified! Only one may be."

@@ -248,3 +248,43 @@ def test_cmaes_optimizer_gpu_e2e(dev):
         opt.update(x_gen, y_gen, st)
     px, py = opt.population_objectives
     assert torch.isfinite(py).all()
+
+
+@pytest.mark.gpu
+def test_full_run_on_gpu(tmp_path):
+    """Complete dmosopt_amd.run() stack on the GPU: GP surrogate fit with
+    the native Cholesky/SCE-UA path, device-resident NSGA2, H5 output."""
+    import dmosopt_amd
+    from dmosopt_amd.benchmarks.problems import zdt1
+
+    def objfun(pp):
+        names = sorted(pp.keys())
+        x = np.array([pp[k] for k in names])
+        return zdt1(x[None, :]).numpy()[0]
+
+    fp = str(tmp_path / "gpu_run.h5")
+    params = {
+        "opt_id": "gpu_e2e",
+        "obj_fun": objfun,
+        "problem_parameters": {},
+        "space": {f"x{i:02d}": [0.0, 1.0] for i in range(6)},
+        "objective_names": ["f1", "f2"],
+        "population_size": 48,
+        "num_generations": 8,
+        "initial_maxiter": 2,
+        "n_initial": 3,
+        "n_epochs": 2,
+        "surrogate_method_name": "gpr",
+        "surrogate_method_kwargs": {"anisotropic": False, "optimizer": "sceua"},
+        "optimizer": "nsga2",
+        "random_seed": 21,
+        "file_path": fp,
+        "save": True,
+    }
+    best = dmosopt_amd.run(params, verbose=False)
+    assert best is not None
+    bx, by = best
+    assert by.shape[1] == 2 and bx.shape[0] > 0
+    import os
+
+    assert os.path.exists(fp)
