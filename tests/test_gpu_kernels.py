@@ -284,7 +284,8 @@ def test_full_run_on_gpu(tmp_path):
     best = dmosopt_amd.run(params, verbose=False)
     assert best is not None
     bx, by = best
-    assert by.shape[1] == 2 and bx.shape[0] > 0
+    y = np.column_stack([v for _, v in by])
+    assert y.shape[1] == 2 and y.shape[0] > 0
     import os
 
     assert os.path.exists(fp)
