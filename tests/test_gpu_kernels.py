@@ -289,3 +289,11 @@ def test_full_run_on_gpu(tmp_path):
     import os
 
     assert os.path.exists(fp)
+    # drop the module-global driver reference NOW: device tensors held in a
+    # module global are otherwise destroyed during interpreter shutdown,
+    # after the HIP runtime has torn down (exit-time SIGSEGV)
+    import gc
+
+    dmosopt_amd.sopt_dict.clear()
+    gc.collect()
+    torch.cuda.synchronize()
