@@ -421,6 +421,11 @@ PYBIND11_MODULE(_h5core, m) {
   // Errors surface as Python exceptions via check()/check_id(); the default
   // stderr error stack (incl. teardown diagnostics) is noise here.
   H5Eset_auto2(H5E_DEFAULT, nullptr, nullptr);
+  // Every file is closed explicitly with H5F_CLOSE_STRONG, so HDF5's own
+  // atexit teardown has nothing to do — and on a GPU box it runs in
+  // undefined order against the HIP runtime's teardown (observed exit-time
+  // SIGSEGV after a run that both used the GPU and saved results).
+  H5dont_atexit();
   py::class_<H5File>(m, "H5File")
       .def(py::init<const std::string&, const std::string&>())
       .def("close", &H5File::close)
