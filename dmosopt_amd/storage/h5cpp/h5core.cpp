@@ -420,12 +420,14 @@ PYBIND11_MODULE(_h5core, m) {
   m.doc() = "Native HDF5 storage core (libhdf5) for dmosopt_amd";
   // Errors surface as Python exceptions via check()/check_id(); the default
   // stderr error stack (incl. teardown diagnostics) is noise here.
-  H5Eset_auto2(H5E_DEFAULT, nullptr, nullptr);
   // Every file is closed explicitly with H5F_CLOSE_STRONG, so HDF5's own
   // atexit teardown has nothing to do — and on a GPU box it runs in
   // undefined order against the HIP runtime's teardown (observed exit-time
-  // SIGSEGV after a run that both used the GPU and saved results).
+  // SIGSEGV after a run that both used the GPU and saved results). MUST be
+  // the first HDF5 call: any other call initializes the library and
+  // registers the atexit handler.
   H5dont_atexit();
+  H5Eset_auto2(H5E_DEFAULT, nullptr, nullptr);
   py::class_<H5File>(m, "H5File")
       .def(py::init<const std::string&, const std::string&>())
       .def("close", &H5File::close)
