@@ -129,8 +129,14 @@ def batched_nmll(
     """
     N = X.shape[0]
     B = theta.shape[0]
-    yb = (y[None, :, None].expand(B, N, 1) if y.dim() == 1 else y[:, :, None]).contiguous()
+
+    def _yb():
+        return (
+            y[None, :, None].expand(B, N, 1) if y.dim() == 1 else y[:, :, None]
+        ).contiguous()
+
     if differentiable:
+        yb = _yb()
         K = build_kernel_torch(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
         L, info = torch.linalg.cholesky_ex(K)
         alpha = torch.cholesky_solve(yb, L)
@@ -139,11 +145,16 @@ def batched_nmll(
     else:
         from dmosopt_amd import ops
 
+        fused = ops.gp_nmll_fused(
+            X, theta, y if y.dim() == 2 else y, nu, anisotropic, jitter
+        )
+        if fused is not None:
+            return fused
         K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
         L, half_logdet, info = ops.chol_factor_batched(K)
         # both paths return sum(log diag L)
         logdet = 2.0 * half_logdet
-        z = ops.tri_solve_forward(L, yb)  # L z = y
+        z = ops.tri_solve_forward(L, _yb())  # L z = y
         quad = (z * z).sum(dim=(1, 2))
     nmll = 0.5 * quad + 0.5 * logdet + 0.5 * N * LOG2PI
     nmll = torch.where(info != 0, torch.full_like(nmll, float("inf")), nmll)

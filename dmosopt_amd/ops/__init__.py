@@ -239,6 +239,18 @@ def mutation_from_pool(pool, idx, di, lo, hi, mutation_rate: float, seed: int, g
 
 
 # -------------------------------------------------------------------- GP ops
+def gp_nmll_fused(X, theta, y, nu, anisotropic, jitter):
+    """Fused batched GP NMLL (assemble + Cholesky + solve + reduce) in one
+    extension call; None if the native path does not apply."""
+    if _use_native(X) and X.dtype == torch.float32:
+        nu_arg = 0.0 if (nu is None or nu == float("inf")) else float(nu)
+        return _native.gp_nmll(
+            X.contiguous(), theta.contiguous().float(), y.contiguous().float(),
+            nu_arg, bool(anisotropic), float(jitter),
+        )
+    return None
+
+
 def matern_train_kernel(X, theta, nu, anisotropic, jitter):
     """Batched symmetric kernel matrices K (B,N,N) with noise+jitter diag."""
     if _use_native(X) and X.dtype == torch.float32:

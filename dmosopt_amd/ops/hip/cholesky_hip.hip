@@ -520,6 +520,39 @@ __global__ void backward_solve_batched_kernel(const float* __restrict__ L,
 
 #include <stdlib.h>
 
+// Final NMLL assembly: out[b] = 0.5 * sum(Z_b^2) + half_logdet[b] + c,
+// inf where the factorization failed or the value is non-finite — fuses the
+// (z*z).sum + scalar-combine + masking chain (a handful of torch dispatches
+// per SCE-UA stage) into one launch.
+__global__ void nmll_reduce_kernel(const float* __restrict__ Z,
+                                   const float* __restrict__ half_logdet,
+                                   const int* __restrict__ info,
+                                   float* __restrict__ out, int N, float c) {
+  __shared__ float part[256];
+  const int b = blockIdx.x;
+  const float* z = Z + (long long)b * N;
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < N; i += blockDim.x) acc += z[i] * z[i];
+  part[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) part[threadIdx.x] += part[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    float v = 0.5f * part[0] + half_logdet[b] + c;
+    if (info[b] != 0 || !isfinite(v)) v = INFINITY;
+    out[b] = v;
+  }
+}
+
+extern "C" void launch_nmll_reduce(const float* Z, const float* half_logdet,
+                                   const int* info, float* out, int B, int N,
+                                   float c, hipStream_t stream) {
+  hipLaunchKernelGGL(nmll_reduce_kernel, dim3(B), dim3(256), 0, stream, Z,
+                     half_logdet, info, out, N, c);
+}
+
 extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
                                        int B, int N, hipStream_t stream) {
   hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
