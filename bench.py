@@ -25,8 +25,11 @@ import numpy as np
 import torch
 
 # host-side control code thrashes with one thread per core on many-core
-# GPU boxes (small CPU tensor ops); the compute path is the GPU anyway
-torch.set_num_threads(min(8, os.cpu_count() or 8))
+# GPU boxes (small CPU tensor ops); the compute path is the GPU anyway.
+# Divide by the local world size: N co-located ranks each spinning 8 OMP
+# threads oversubscribe small hosts 16x (measured on an 8-core box).
+_local_world = int(os.environ.get("LOCAL_WORLD_SIZE", os.environ.get("WORLD_SIZE", "1")))
+torch.set_num_threads(max(1, min(8, (os.cpu_count() or 8) // max(1, _local_world))))
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
