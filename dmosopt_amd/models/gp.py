@@ -200,9 +200,15 @@ class _GPRBase:
         return self._fitted.predict(Xq_normalized)
 
     def normalize_query(self, xin: torch.Tensor) -> torch.Tensor:
-        lb = torch.as_tensor(self.xlb, dtype=xin.dtype, device=xin.device)
-        rg = torch.as_tensor(self.xrg, dtype=xin.dtype, device=xin.device)
-        return (xin - lb) / rg
+        # cache the bounds tensors: re-uploading two tiny arrays is two H2D
+        # transfers per surrogate evaluation (every generation)
+        cache = getattr(self, "_nq_cache", None)
+        key = (xin.dtype, xin.device)
+        if cache is None or cache[0] != key:
+            lb = torch.as_tensor(self.xlb, dtype=xin.dtype, device=xin.device)
+            rg = torch.as_tensor(self.xrg, dtype=xin.dtype, device=xin.device)
+            self._nq_cache = cache = (key, lb, rg)
+        return (xin - cache[1]) / cache[2]
 
     def evaluate(self, x):
         mean, var = self.predict(x)
