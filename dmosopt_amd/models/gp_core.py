@@ -220,6 +220,15 @@ class FittedGP:
         ``return_var=False`` (the per-generation surrogate-evaluate path)
         the quadratic term is skipped and (mean, None) returned.
         """
+        if not return_var:
+            from dmosopt_amd import ops
+
+            fused = ops.gp_predict_mean_fused(
+                Xq, self.X, self.theta, self.alpha, self.y_mean, self.y_std,
+                self.nu, self.anisotropic,
+            )
+            if fused is not None:
+                return fused, None
         Ks = build_kernel(Xq, self.X, self.theta, nu=self.nu, anisotropic=self.anisotropic)
         # (m, P, N)
         mean_n = torch.bmm(Ks, self.alpha)[:, :, 0]  # (m, P)

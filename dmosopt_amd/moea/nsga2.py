@@ -145,15 +145,30 @@ class NSGA2Optimizer(MOEA):
     def update_strategy(self, x_gen, y_gen, gen_state, **params):
         p = self.opt_params
         popsize = p.popsize
-        population_parm = torch.cat([x_gen, self.state.population_parm], dim=0)
-        population_obj = torch.cat([y_gen, self.state.population_obj], dim=0)
-        parm, obj, rank, perm = ops.remove_worst(
-            population_parm,
-            population_obj,
-            popsize,
-            x_dists=self._x_dists(population_parm),
-            y_distance_metrics=self.y_distance_metrics,
-        )
+        if (
+            x_gen.device.type == "cuda"
+            and self.x_distance_fns is None
+            and self.y_distance_metrics == ["crowding"]
+            and ops.native_available()
+        ):
+            # fused native path: one extension call per generation
+            from dmosopt_amd import _hipops
+
+            parm, obj, rank, perm = _hipops.nsga2_select(
+                x_gen.float().contiguous(), y_gen.float().contiguous(),
+                self.state.population_parm.float().contiguous(),
+                self.state.population_obj.float().contiguous(), popsize,
+            )
+        else:
+            population_parm = torch.cat([x_gen, self.state.population_parm], dim=0)
+            population_obj = torch.cat([y_gen, self.state.population_obj], dim=0)
+            parm, obj, rank, perm = ops.remove_worst(
+                population_parm,
+                population_obj,
+                popsize,
+                x_dists=self._x_dists(population_parm),
+                y_distance_metrics=self.y_distance_metrics,
+            )
         # device-side survivor accounting without a host round-trip: the
         # generation's children occupy rows [0, n_children) of the
         # concatenated population, so survival is just `perm < n_children`

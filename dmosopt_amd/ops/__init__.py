@@ -239,6 +239,18 @@ def mutation_from_pool(pool, idx, di, lo, hi, mutation_rate: float, seed: int, g
 
 
 # -------------------------------------------------------------------- GP ops
+def gp_predict_mean_fused(Xq, X, theta, alpha, y_mean, y_std, nu, anisotropic):
+    """Fused normalized-query posterior mean; None when native doesn't apply."""
+    if _use_native(Xq) and X.dtype == torch.float32:
+        nu_arg = 0.0 if (nu is None or nu == float("inf")) else float(nu)
+        return _native.gp_predict_mean(
+            Xq.to(torch.float32).contiguous(), X.contiguous(), theta.contiguous().float(),
+            alpha.contiguous().float(), y_mean.float(), y_std.float(),
+            nu_arg, bool(anisotropic),
+        )
+    return None
+
+
 def gp_nmll_fused(X, theta, y, nu, anisotropic, jitter):
     """Fused batched GP NMLL (assemble + Cholesky + solve + reduce) in one
     extension call; None if the native path does not apply."""

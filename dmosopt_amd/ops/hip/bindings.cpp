@@ -117,6 +117,26 @@ torch::Tensor gp_nmll(torch::Tensor X, torch::Tensor theta, torch::Tensor y,
   return out;
 }
 
+// Fused posterior-mean predict for the inner-MOEA surrogate evaluate:
+// cross-kernel assembly + bmm(alpha) + de-standardization queued from one
+// python call (the torch op chain costs ~0.1 ms of dispatch per
+// generation). Xq must already be normalized to the unit box.
+torch::Tensor gp_predict_mean(torch::Tensor Xq, torch::Tensor X,
+                              torch::Tensor theta, torch::Tensor alpha,
+                              torch::Tensor y_mean, torch::Tensor y_std,
+                              double nu, bool aniso) {
+  CHECK_GPU(Xq);
+  CHECK_GPU(X);
+  const int P = Xq.size(0), N = X.size(0), D = X.size(1), B = theta.size(0);
+  auto Ks = torch::empty({B, P, N}, X.options());
+  launch_matern_assemble(Xq.data_ptr<float>(), X.data_ptr<float>(),
+                         theta.data_ptr<float>(), Ks.data_ptr<float>(), B, P,
+                         N, D, theta.size(1), 0.0f, nu_code(nu), aniso ? 1 : 0,
+                         0, cur_stream());
+  auto mean_n = torch::bmm(Ks, alpha).squeeze(-1);  // (m, P)
+  return y_mean.unsqueeze(1).addcmul(y_std.unsqueeze(1), mean_n).transpose(0, 1);
+}
+
 std::vector<torch::Tensor> cholesky_batched_(torch::Tensor A) {
   CHECK_GPU(A);
   const int B = A.size(0), N = A.size(1);
@@ -222,6 +242,30 @@ torch::Tensor crowding_distance(torch::Tensor Y) {
   return per_dim.sum(0);  // fixed-order reduction: deterministic
 }
 
+// Fused NSGA2 survivor selection: concatenate children+parents, pareto-rank,
+// crowding, packed-key stable sort ((rank << 32) | ~float_bits(crowding)),
+// truncate to pop and gather — one python call instead of ~15 dispatched
+// ops per generation. Mirrors ops.remove_worst with the crowding metric.
+std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
+                                        torch::Tensor y_gen,
+                                        torch::Tensor pop_parm,
+                                        torch::Tensor pop_obj, int64_t pop) {
+  auto parm = torch::cat({x_gen, pop_parm}, 0);
+  auto obj = torch::cat({y_gen, pop_obj}, 0);
+  auto rank = pareto_rank(obj);                    // (N,) long
+  auto crowd = crowding_distance(obj).to(torch::kFloat32);  // (N,)
+  auto d = torch::nan_to_num(crowd.clamp_min(0.0), 0.0,
+                             std::numeric_limits<float>::max());
+  auto bits = d.view(torch::kInt32).to(torch::kLong);
+  auto key = rank.to(torch::kLong).__lshift__(32)
+                 .__or__((0x7FFFFFFFLL - bits).__and__(0xFFFFFFFFLL));
+  auto perm = torch::argsort(key, /*stable=*/true, /*dim=*/-1,
+                             /*descending=*/false)
+                  .slice(0, 0, pop);
+  return {parm.index_select(0, perm), obj.index_select(0, perm),
+          rank.index_select(0, perm), perm};
+}
+
 std::vector<torch::Tensor> sbx_batch(torch::Tensor pool, torch::Tensor p1,
                                      torch::Tensor p2, torch::Tensor di,
                                      torch::Tensor lo, torch::Tensor hi,
@@ -311,6 +355,8 @@ void cmaes_update_(torch::Tensor A, torch::Tensor Ainv, torch::Tensor pc,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
+  m.def("nsga2_select", &nsga2_select, "Fused survivor selection (cat+rank+crowding+sort+gather)");
+  m.def("gp_predict_mean", &gp_predict_mean, "Fused cross-kernel + posterior mean");
   m.def("gp_nmll", &gp_nmll, "Fused batched GP NMLL (assemble+chol+solve+reduce)");
   m.def("cholesky_batched_", &cholesky_batched_,
         "In-place batched Cholesky; returns (logdet, info)");
