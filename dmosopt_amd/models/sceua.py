@@ -106,6 +106,20 @@ def sceua_batched(
 
     x, xf = sort_pop(x, xf)
 
+    # fused CCE stage path: propose + accept/resort run as two extension
+    # kernels (Philox RNG in-kernel) instead of ~30 torch launches/stage
+    use_native_stage = False
+    if device.type == "cuda" and dtype == torch.float32:
+        from dmosopt_amd import ops as _ops
+
+        if _ops.native_available():
+            lds = (npg * 2 + npg * nopt) * 4
+            use_native_stage = lds <= 60 * 1024
+    if use_native_stage:
+        from dmosopt_amd import _hipops as _hip
+        bl_f = bl_t.to(torch.float32).contiguous()
+        bu_f = bu_t.to(torch.float32).contiguous()
+
     def gnrng_of(x_):
         rngs = (x_.max(dim=1).values - x_.min(dim=1).values) / bd
         return torch.exp(torch.log(rngs.clamp_min(1e-300)).mean(dim=1))
@@ -114,6 +128,7 @@ def sceua_batched(
     criter = [[] for _ in range(S)]
     criter_change = np.full(S, 1e5)
     sid_sg = torch.arange(S, device=device).repeat_interleave(G)
+    sid3 = torch.cat([sid_sg, sid_sg, sid_sg], dim=0)
 
     nloop = 0
     while True:
@@ -127,7 +142,9 @@ def sceua_batched(
         # H2D instead of one per CCE stage (a pageable-memory copy inside
         # the stage loop near-synchronizes the stream and kills pipelining)
         lcs_all = torch.as_tensor(
-            np.stack([_select_simplex(nps, npg, rng) for _ in range(nspl)]),
+            np.stack([_select_simplex(nps, npg, rng) for _ in range(nspl)]).astype(
+                np.int32
+            ),
             device=device,
         )
 
@@ -136,8 +153,24 @@ def sceua_batched(
         cf = xf.view(S, npg, G).permute(0, 2, 1).contiguous()  # (S,G,npg)
         icall_dev = torch.zeros(S, dtype=torch.float64, device=device)
 
-        for _step in range(nspl):
-            lcs = lcs_all[_step]
+        if use_native_stage:
+            act_i32 = act.to(torch.int32).contiguous()
+            icall_i32 = torch.zeros(S, dtype=torch.int32, device=device)
+            for _step in range(nspl):
+                lcs = lcs_all[_step].contiguous()
+                cand = _hip.sceua_propose(
+                    cx, lcs, bl_f, bu_f, nps, int(rng.integers(0, 2**62))
+                )
+                fall = func(cand.reshape(3 * S * G, nopt), sid3)
+                ok = _hip.sceua_accept(
+                    cx, cf, cand, fall.to(torch.float32).contiguous(), lcs,
+                    act_i32, icall_i32, nps,
+                )
+                assert ok
+            icall_dev = icall_i32.to(torch.float64)
+        else:
+          for _step in range(nspl):
+            lcs = lcs_all[_step].long()
             s_pts = cx[:, :, lcs, :]  # (S,G,nps,nopt)
             s_f = cf[:, :, lcs]  # (S,G,nps)
 
@@ -159,7 +192,6 @@ def sceua_batched(
             cand = torch.cat(
                 [s_ref.reshape(S * G, nopt), s_con.reshape(S * G, nopt),
                  s_rnd.reshape(S * G, nopt)], dim=0)
-            sid3 = torch.cat([sid_sg, sid_sg, sid_sg], dim=0)
             fall = func(cand, sid3).reshape(3, S, G)
             f_ref, f_con, f_rnd = fall[0], fall[1], fall[2]
 

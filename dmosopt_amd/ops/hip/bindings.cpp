@@ -42,6 +42,8 @@ void launch_cmaes_update(float*, float*, float*, const float*, const float*,
                          int, int, float, float, float, hipStream_t);
 int launch_peel_from_y(const float*, int*, int, int, hipStream_t);
 void launch_nmll_reduce(const float*, const float*, const int*, float*, int, int, float, hipStream_t);
+void launch_sceua_propose(const float*, const int*, const float*, const float*, float*, int, int, int, int, int, unsigned long long, hipStream_t);
+int launch_sceua_accept(float*, float*, const float*, const float*, const int*, const int*, int*, int, int, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
 }
@@ -135,6 +137,35 @@ torch::Tensor gp_predict_mean(torch::Tensor Xq, torch::Tensor X,
                          0, cur_stream());
   auto mean_n = torch::bmm(Ks, alpha).squeeze(-1);  // (m, P)
   return y_mean.unsqueeze(1).addcmul(y_std.unsqueeze(1), mean_n).transpose(0, 1);
+}
+
+// Fused SCE-UA CCE stage: candidate proposal and acceptance+resort, one
+// extension call each instead of ~30 torch kernels per stage.
+torch::Tensor sceua_propose(torch::Tensor cx, torch::Tensor lcs,
+                            torch::Tensor bl, torch::Tensor bu, int64_t nps,
+                            int64_t seed) {
+  CHECK_GPU(cx);
+  const int S = cx.size(0), G = cx.size(1), npg = cx.size(2),
+            nopt = cx.size(3);
+  auto cand = torch::empty({3, (long)S * G, nopt}, cx.options());
+  launch_sceua_propose(cx.data_ptr<float>(), lcs.data_ptr<int>(),
+                       bl.data_ptr<float>(), bu.data_ptr<float>(),
+                       cand.data_ptr<float>(), S, G, npg, nopt, (int)nps,
+                       (unsigned long long)seed, cur_stream());
+  return cand;
+}
+
+bool sceua_accept(torch::Tensor cx, torch::Tensor cf, torch::Tensor cand,
+                  torch::Tensor fall, torch::Tensor lcs, torch::Tensor act,
+                  torch::Tensor icall, int64_t nps) {
+  CHECK_GPU(cx);
+  const int S = cx.size(0), G = cx.size(1), npg = cx.size(2),
+            nopt = cx.size(3);
+  return launch_sceua_accept(cx.data_ptr<float>(), cf.data_ptr<float>(),
+                             cand.data_ptr<float>(), fall.data_ptr<float>(),
+                             lcs.data_ptr<int>(), act.data_ptr<int>(),
+                             icall.data_ptr<int>(), S, G, npg, nopt, (int)nps,
+                             cur_stream()) == 0;
 }
 
 std::vector<torch::Tensor> cholesky_batched_(torch::Tensor A) {
@@ -355,6 +386,8 @@ void cmaes_update_(torch::Tensor A, torch::Tensor Ainv, torch::Tensor pc,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
+  m.def("sceua_propose", &sceua_propose);
+  m.def("sceua_accept", &sceua_accept);
   m.def("nsga2_select", &nsga2_select, "Fused survivor selection (cat+rank+crowding+sort+gather)");
   m.def("gp_predict_mean", &gp_predict_mean, "Fused cross-kernel + posterior mean");
   m.def("gp_nmll", &gp_nmll, "Fused batched GP NMLL (assemble+chol+solve+reduce)");

@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join("dmosopt_amd", "ops", "hip", "variation.hip"),
         os.path.join("dmosopt_amd", "ops", "hip", "hv_mc.hip"),
         os.path.join("dmosopt_amd", "ops", "hip", "cmaes_update.hip"),
+        os.path.join("dmosopt_amd", "ops", "hip", "sceua_stage.hip"),
     ],
     include_dirs=[HIP_DIR],
     extra_compile_args={
