@@ -21,6 +21,23 @@ import torch
 
 from dmosopt_amd import ops
 
+_PIN_CACHE = {}
+
+
+def _to_device_pinned(arr: np.ndarray, device: torch.device) -> torch.Tensor:
+    if device.type != "cuda":
+        return torch.as_tensor(arr, dtype=torch.long, device=device)
+    n = arr.shape[0]
+    buf = _PIN_CACHE.get(device)
+    if buf is None or buf.shape[0] < n:
+        cap = max(1024, 2 * n)
+        buf = torch.empty(cap, dtype=torch.long, pin_memory=True)
+        _PIN_CACHE[device] = buf
+    buf[:n].copy_(torch.from_numpy(arr))
+    out = torch.empty(n, dtype=torch.long, device=device)
+    out.copy_(buf[:n], non_blocking=True)
+    return out
+
 
 def _draw_event_stream(rng, popsize: int, crossover_prob: float, mutation_prob: float):
     """Vectorized replica of the reference's while-loop event stream.
@@ -100,9 +117,12 @@ def event_stream_variation(
     src_rows[crossover_indices[1::2]] = C + np.arange(C)
     src_rows[mutation_indices] = 2 * C + np.arange(M)
 
-    # single H2D transfer for every index array of the generation
+    # single H2D transfer for every index array of the generation, staged
+    # through a cached PINNED buffer with a non-blocking copy — a pageable
+    # torch.as_tensor(...) H2D blocks the host until the stream drains,
+    # serializing every generation against the previous one's GPU work
     combined = np.concatenate([i1, i2, im, src_rows, crossover_indices, mutation_indices])
-    dev = torch.as_tensor(combined, dtype=torch.long, device=pool.device)
+    dev = _to_device_pinned(combined, pool.device)
     o = 0
     i1_t = dev[o : o + C]; o += C
     i2_t = dev[o : o + C]; o += C
