@@ -21,21 +21,17 @@ import torch
 
 from dmosopt_amd import ops
 
-_PIN_CACHE = {}
-
-
 def _to_device_pinned(arr: np.ndarray, device: torch.device) -> torch.Tensor:
     if device.type != "cuda":
         return torch.as_tensor(arr, dtype=torch.long, device=device)
-    n = arr.shape[0]
-    buf = _PIN_CACHE.get(device)
-    if buf is None or buf.shape[0] < n:
-        cap = max(1024, 2 * n)
-        buf = torch.empty(cap, dtype=torch.long, pin_memory=True)
-        _PIN_CACHE[device] = buf
-    buf[:n].copy_(torch.from_numpy(arr))
-    out = torch.empty(n, dtype=torch.long, device=device)
-    out.copy_(buf[:n], non_blocking=True)
+    # torch's caching host allocator recycles the pinned block only after
+    # the async copy's stream event completes, so a fresh pin_memory tensor
+    # per call is both cheap and safe even when the host runs generations
+    # ahead of the GPU (a manual reused buffer would race)
+    buf = torch.empty(arr.shape[0], dtype=torch.long, pin_memory=True)
+    buf.copy_(torch.from_numpy(arr))
+    out = torch.empty(arr.shape[0], dtype=torch.long, device=device)
+    out.copy_(buf, non_blocking=True)
     return out
 
 
