@@ -123,10 +123,22 @@ class NSGA2Optimizer(MOEA):
         population = self.state.population_parm
         rank = self.state.rank
 
-        pool_idx = ops.tournament_selection(
-            population.shape[0], poolsize, [rank], rng, generator=self.torch_random
-        )
-        pool = population[pool_idx]
+        pool = None
+        if population.device.type == "cuda" and ops.native_available():
+            from dmosopt_amd import _hipops
+
+            res = _hipops.tournament_pool(
+                population.float().contiguous(), rank.long().contiguous(),
+                poolsize, 0.5, int(rng.integers(0, 2**62)),
+            )
+            if res[0] is not None:
+                pool = res[0].to(population.dtype)
+        if pool is None:
+            pool_idx = ops.tournament_selection(
+                population.shape[0], poolsize, [rank], rng,
+                generator=self.torch_random,
+            )
+            pool = population[pool_idx]
 
         di_c, di_m = self._di_tensors(pool)
         from dmosopt_amd.moea.variation import event_stream_variation
@@ -181,13 +193,23 @@ class NSGA2Optimizer(MOEA):
         if not isinstance(m_idx, torch.Tensor):
             m_idx = torch.as_tensor(np.asarray(m_idx), dtype=torch.long, device=perm.device)
         n_children = x_gen.shape[0]
-        is_cross = torch.zeros(n_children, dtype=torch.bool, device=perm.device)
-        is_cross[c_idx] = True
-        child = perm < n_children
-        slot = torch.where(child, perm, torch.zeros_like(perm))
-        surv_cross = is_cross[slot] & child
-        self.state.successful_crossovers += surv_cross.sum() // 2
-        self.state.successful_mutations += ((~is_cross[slot]) & child).sum()
+        counted = False
+        if perm.device.type == "cuda" and ops.native_available() and n_children <= 2048:
+            from dmosopt_amd import _hipops
+
+            sc, sm = self.state.successful_crossovers, self.state.successful_mutations
+            if sc.device.type == "cuda":
+                counted = _hipops.survivor_count(
+                    perm.contiguous(), c_idx.contiguous(), n_children, sc, sm
+                )
+        if not counted:
+            is_cross = torch.zeros(n_children, dtype=torch.bool, device=perm.device)
+            is_cross[c_idx] = True
+            child = perm < n_children
+            slot = torch.where(child, perm, torch.zeros_like(perm))
+            surv_cross = is_cross[slot] & child
+            self.state.successful_crossovers += surv_cross.sum() // 2
+            self.state.successful_mutations += ((~is_cross[slot]) & child).sum()
 
         self.state.population_parm = parm
         self.state.population_obj = obj

@@ -44,6 +44,10 @@ int launch_peel_from_y(const float*, int*, int, int, hipStream_t);
 void launch_nmll_reduce(const float*, const float*, const int*, float*, int, int, float, hipStream_t);
 void launch_sceua_propose(const float*, const int*, const float*, const float*, float*, int, int, int, int, int, unsigned long long, hipStream_t);
 int launch_sceua_accept(float*, float*, const float*, const float*, const int*, const int*, int*, int, int, int, int, int, hipStream_t);
+int launch_tournament(const float*, const long long*, float*, long long*, int, int, int, float, unsigned long long, hipStream_t);
+int launch_survivor_count(const long long*, const long long*, int, int, int, long long*, long long*, hipStream_t);
+void launch_pack_rank_crowd(const long long*, const float*, long long*, int, hipStream_t);
+void launch_gather3(const float*, const float*, const long long*, const long long*, float*, float*, long long*, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
 }
@@ -141,6 +145,39 @@ torch::Tensor gp_predict_mean(torch::Tensor Xq, torch::Tensor X,
 
 // Fused SCE-UA CCE stage: candidate proposal and acceptance+resort, one
 // extension call each instead of ~30 torch kernels per stage.
+// Fused tournament selection: stable rank sort + Gumbel top-k weighted
+// sampling without replacement + pool gather in ONE launch. Returns
+// (pool, pool_idx); empty tensors when N exceeds the LDS sort bound.
+std::vector<torch::Tensor> tournament_pool(torch::Tensor population,
+                                           torch::Tensor rank,
+                                           int64_t poolsize, double p_sel,
+                                           int64_t seed) {
+  CHECK_GPU(population);
+  const int N = population.size(0), d = population.size(1);
+  auto pool = torch::empty({poolsize, d}, population.options());
+  auto pool_idx =
+      torch::empty({poolsize}, population.options().dtype(torch::kLong));
+  const float log1mp = logf(1.0f - (float)p_sel);
+  if (launch_tournament(population.data_ptr<float>(),
+                        rank.data_ptr<long long>(), pool.data_ptr<float>(),
+                        pool_idx.data_ptr<long long>(), N, d, (int)poolsize,
+                        log1mp, (unsigned long long)seed, cur_stream()) != 0)
+    return {torch::Tensor(), torch::Tensor()};
+  return {pool, pool_idx};
+}
+
+// Device-side operator-success accounting (no host sync, one launch).
+bool survivor_count(torch::Tensor perm, torch::Tensor c_idx,
+                    int64_t n_children, torch::Tensor succ_cross,
+                    torch::Tensor succ_mut) {
+  CHECK_GPU(perm);
+  return launch_survivor_count(
+             perm.data_ptr<long long>(), c_idx.data_ptr<long long>(),
+             perm.size(0), c_idx.size(0), (int)n_children,
+             succ_cross.data_ptr<long long>(), succ_mut.data_ptr<long long>(),
+             cur_stream()) == 0;
+}
+
 torch::Tensor sceua_propose(torch::Tensor cx, torch::Tensor lcs,
                             torch::Tensor bl, torch::Tensor bu, int64_t nps,
                             int64_t seed) {
@@ -285,16 +322,23 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   auto obj = torch::cat({y_gen, pop_obj}, 0);
   auto rank = pareto_rank(obj);                    // (N,) long
   auto crowd = crowding_distance(obj).to(torch::kFloat32);  // (N,)
-  auto d = torch::nan_to_num(crowd.clamp_min(0.0), 0.0,
-                             std::numeric_limits<float>::max());
-  auto bits = d.view(torch::kInt32).to(torch::kLong);
-  auto key = rank.to(torch::kLong).__lshift__(32)
-                 .__or__((0x7FFFFFFFLL - bits).__and__(0xFFFFFFFFLL));
+  const int N = obj.size(0), d = parm.size(1), m = obj.size(1);
+  auto key = torch::empty({N}, rank.options());
+  launch_pack_rank_crowd(rank.data_ptr<long long>(), crowd.data_ptr<float>(),
+                         key.data_ptr<long long>(), N, cur_stream());
   auto perm = torch::argsort(key, /*stable=*/true, /*dim=*/-1,
                              /*descending=*/false)
-                  .slice(0, 0, pop);
-  return {parm.index_select(0, perm), obj.index_select(0, perm),
-          rank.index_select(0, perm), perm};
+                  .slice(0, 0, pop)
+                  .contiguous();
+  const int P = perm.size(0);
+  auto parm_o = torch::empty({P, d}, parm.options());
+  auto obj_o = torch::empty({P, m}, obj.options());
+  auto rank_o = torch::empty({P}, rank.options());
+  launch_gather3(parm.data_ptr<float>(), obj.data_ptr<float>(),
+                 rank.data_ptr<long long>(), perm.data_ptr<long long>(),
+                 parm_o.data_ptr<float>(), obj_o.data_ptr<float>(),
+                 rank_o.data_ptr<long long>(), P, d, m, cur_stream());
+  return {parm_o, obj_o, rank_o, perm};
 }
 
 std::vector<torch::Tensor> sbx_batch(torch::Tensor pool, torch::Tensor p1,
@@ -386,6 +430,8 @@ void cmaes_update_(torch::Tensor A, torch::Tensor Ainv, torch::Tensor pc,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
+  m.def("tournament_pool", &tournament_pool);
+  m.def("survivor_count", &survivor_count);
   m.def("sceua_propose", &sceua_propose);
   m.def("sceua_accept", &sceua_accept);
   m.def("nsga2_select", &nsga2_select, "Fused survivor selection (cat+rank+crowding+sort+gather)");

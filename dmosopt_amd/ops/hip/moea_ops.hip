@@ -1,0 +1,205 @@
+// Small fused MOEA kernels for the per-generation loop.
+//
+// The generation loop is launch-count bound (~9-12 us of in-stream gap per
+// kernel on this stack, measured from rocpd timelines), so these kernels
+// exist to collapse chains of tiny torch ops into single launches:
+//   tournament_kernel: stable rank sort + Gumbel top-k weighted sampling
+//                      without replacement + pool gather  (~8 launches -> 1)
+//   survivor_count_kernel: operator-success accounting against the
+//                      survivor permutation                (~8 launches -> 1)
+
+#include "common.h"
+#include <math.h>
+
+#define TOUR_TPB 256
+#define TOUR_NMAX 2048  // npow2 cap; LDS fits the 64 KB default
+
+// population (N, d) f32; rank (N,) int64. Writes pool (poolsize, d) and
+// pool_idx (poolsize,) int64.
+// Sampling: candidates sorted by (rank, index) ascending; candidate at
+// sorted position i gets weight p*(1-p)^i (reference MOEA.py:385-395);
+// weighted sampling WITHOUT replacement via Gumbel top-k on
+// key_i = i*log(1-p) + Gumbel_i (constants cancel under top-k).
+__global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
+    const float* __restrict__ population, const long long* __restrict__ rank,
+    float* __restrict__ pool, long long* __restrict__ pool_idx, int N, int d,
+    int poolsize, int npow2, float log1mp, unsigned long long seed) {
+  extern __shared__ char sh_raw[];
+  long long* key = (long long*)sh_raw;        // npow2 (composite sort key)
+  int* idx = (int*)(key + npow2);             // npow2
+  float* gkey = (float*)(idx + npow2);        // npow2 (gumbel keys)
+
+  const int tid = threadIdx.x;
+  // stage 1: stable sort by rank — composite integer key rank*N + index
+  for (int i = tid; i < npow2; i += TOUR_TPB) {
+    key[i] = (i < N) ? (rank[i] * (long long)N + i) : 0x7FFFFFFFFFFFFFFFLL;
+    idx[i] = i;
+  }
+  __syncthreads();
+  for (int ks = 2; ks <= npow2; ks <<= 1) {
+    for (int js = ks >> 1; js > 0; js >>= 1) {
+      for (int i = tid; i < npow2; i += TOUR_TPB) {
+        const int ixj = i ^ js;
+        if (ixj > i) {
+          const bool up = ((i & ks) == 0);
+          if (up ? (key[i] > key[ixj]) : (key[i] < key[ixj])) {
+            long long tk = key[i]; key[i] = key[ixj]; key[ixj] = tk;
+            int ti = idx[i]; idx[i] = idx[ixj]; idx[ixj] = ti;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+  // idx[i] now = candidate at sorted position i. Stage 2: Gumbel keys per
+  // position, sort DESCENDING, take first poolsize.
+  for (int i = tid; i < npow2; i += TOUR_TPB) {
+    if (i < N) {
+      Philox4 r = philox4x32(seed, (unsigned long long)i);
+      const float u = u01(r.c0);
+      gkey[i] = (float)i * log1mp - logf(-logf(u));
+    } else {
+      gkey[i] = -INFINITY;
+    }
+  }
+  __syncthreads();
+  for (int ks = 2; ks <= npow2; ks <<= 1) {
+    for (int js = ks >> 1; js > 0; js >>= 1) {
+      for (int i = tid; i < npow2; i += TOUR_TPB) {
+        const int ixj = i ^ js;
+        if (ixj > i) {
+          const bool up = ((i & ks) == 0);
+          // descending overall
+          if (up ? (gkey[i] < gkey[ixj]) : (gkey[i] > gkey[ixj])) {
+            float tg = gkey[i]; gkey[i] = gkey[ixj]; gkey[ixj] = tg;
+            int ti = idx[i]; idx[i] = idx[ixj]; idx[ixj] = ti;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+  // stage 3: gather the pool rows
+  for (int j = tid; j < poolsize; j += TOUR_TPB) pool_idx[j] = idx[j];
+  __syncthreads();
+  for (long long t = tid; t < (long long)poolsize * d; t += TOUR_TPB) {
+    const int r = (int)(t / d), c = (int)(t % d);
+    pool[t] = population[(long long)idx[r] * d + c];
+  }
+}
+
+extern "C" int launch_tournament(const float* population, const long long* rank,
+                                 float* pool, long long* pool_idx, int N,
+                                 int d, int poolsize, float log1mp,
+                                 unsigned long long seed, hipStream_t stream) {
+  int npow2 = 1;
+  while (npow2 < N) npow2 <<= 1;
+  if (npow2 > TOUR_NMAX) return -1;
+  const size_t lds = (size_t)npow2 * (8 + 4 + 4);
+  hipLaunchKernelGGL(tournament_kernel, dim3(1), dim3(TOUR_TPB), lds, stream,
+                     population, rank, pool, pool_idx, N, d, poolsize, npow2,
+                     log1mp, seed);
+  return 0;
+}
+
+// Survivor accounting: children occupied rows [0, n_children) of the
+// concatenated population; c_idx lists the crossover child slots. Adds
+// (#surviving crossover children)/2 and #surviving mutation children to the
+// int64 scalar counters.
+__global__ void survivor_count_kernel(const long long* __restrict__ perm,
+                                      const long long* __restrict__ c_idx,
+                                      int n_perm, int n_cross, int n_children,
+                                      long long* __restrict__ succ_cross,
+                                      long long* __restrict__ succ_mut) {
+  __shared__ unsigned int is_cross[(2048 + 31) / 32];
+  __shared__ int cnt_c, cnt_m;
+  const int tid = threadIdx.x;
+  const int words = (n_children + 31) / 32;
+  for (int w = tid; w < words; w += blockDim.x) is_cross[w] = 0u;
+  if (tid == 0) { cnt_c = 0; cnt_m = 0; }
+  __syncthreads();
+  for (int i = tid; i < n_cross; i += blockDim.x) {
+    const int s = (int)c_idx[i];
+    atomicOr(&is_cross[s >> 5], 1u << (s & 31));
+  }
+  __syncthreads();
+  int lc = 0, lm = 0;
+  for (int i = tid; i < n_perm; i += blockDim.x) {
+    const long long e = perm[i];
+    if (e < n_children) {
+      if (is_cross[(int)(e >> 5)] & (1u << ((int)e & 31))) ++lc;
+      else ++lm;
+    }
+  }
+  atomicAdd(&cnt_c, lc);
+  atomicAdd(&cnt_m, lm);
+  __syncthreads();
+  if (tid == 0) {
+    atomicAdd((unsigned long long*)succ_cross, (unsigned long long)(cnt_c / 2));
+    atomicAdd((unsigned long long*)succ_mut, (unsigned long long)cnt_m);
+  }
+}
+
+extern "C" int launch_survivor_count(const long long* perm,
+                                     const long long* c_idx, int n_perm,
+                                     int n_cross, int n_children,
+                                     long long* succ_cross,
+                                     long long* succ_mut,
+                                     hipStream_t stream) {
+  if (n_children > 2048) return -1;
+  hipLaunchKernelGGL(survivor_count_kernel, dim3(1), dim3(256), 0, stream,
+                     perm, c_idx, n_perm, n_cross, n_children, succ_cross,
+                     succ_mut);
+  return 0;
+}
+
+// key[i] = (rank[i] << 32) | (0x7FFFFFFF - float_bits(max(crowd,0)))
+// (one launch instead of the 5-op torch packing chain in survivor select)
+__global__ void pack_rank_crowd_kernel(const long long* __restrict__ rank,
+                                       const float* __restrict__ crowd,
+                                       long long* __restrict__ key, int N) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= N) return;
+  float d = crowd[i];
+  if (!(d >= 0.f)) d = 0.f;               // clamp + NaN
+  if (isinf(d)) d = 3.4028235e38f;        // +inf -> fmax
+  unsigned int bits = __float_as_uint(d);
+  key[i] = (rank[i] << 32) | (long long)(0x7FFFFFFFu - bits);
+}
+
+extern "C" void launch_pack_rank_crowd(const long long* rank,
+                                       const float* crowd, long long* key,
+                                       int N, hipStream_t stream) {
+  hipLaunchKernelGGL(pack_rank_crowd_kernel, dim3((N + 255) / 256), dim3(256),
+                     0, stream, rank, crowd, key, N);
+}
+
+// One launch gathering the three survivor outputs.
+__global__ void gather3_kernel(const float* __restrict__ parm,
+                               const float* __restrict__ obj,
+                               const long long* __restrict__ rank,
+                               const long long* __restrict__ perm,
+                               float* __restrict__ parm_o,
+                               float* __restrict__ obj_o,
+                               long long* __restrict__ rank_o, int pop, int d,
+                               int m) {
+  const long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long total = (long long)pop * (d + m + 1);
+  if (t >= total) return;
+  const int r = (int)(t / (d + m + 1));
+  const int c = (int)(t % (d + m + 1));
+  const long long src = perm[r];
+  if (c < d) parm_o[(long long)r * d + c] = parm[src * d + c];
+  else if (c < d + m) obj_o[(long long)r * m + (c - d)] = obj[src * m + (c - d)];
+  else rank_o[r] = rank[src];
+}
+
+extern "C" void launch_gather3(const float* parm, const float* obj,
+                               const long long* rank, const long long* perm,
+                               float* parm_o, float* obj_o, long long* rank_o,
+                               int pop, int d, int m, hipStream_t stream) {
+  const long long total = (long long)pop * (d + m + 1);
+  hipLaunchKernelGGL(gather3_kernel, dim3((int)((total + 255) / 256)),
+                     dim3(256), 0, stream, parm, obj, rank, perm, parm_o,
+                     obj_o, rank_o, pop, d, m);
+}

@@ -297,3 +297,43 @@ def test_full_run_on_gpu(tmp_path):
     dmosopt_amd.sopt_dict.clear()
     gc.collect()
     torch.cuda.synchronize()
+
+
+@pytest.mark.gpu
+def test_fused_moea_kernels(dev):
+    from dmosopt_amd import _hipops
+
+    torch.manual_seed(3)
+    # survivor_count against the reference boolean accounting
+    n_children, n_pop = 201, 200
+    perm = torch.randperm(401, device=dev)[:n_pop]
+    c_idx = torch.arange(0, 120, device=dev)
+    sc = torch.zeros((), dtype=torch.long, device=dev)
+    sm = torch.zeros((), dtype=torch.long, device=dev)
+    assert _hipops.survivor_count(perm.contiguous(), c_idx.contiguous(), n_children, sc, sm)
+    is_cross = torch.zeros(n_children, dtype=torch.bool, device=dev)
+    is_cross[c_idx] = True
+    child = perm < n_children
+    slot = torch.where(child, perm, torch.zeros_like(perm))
+    want_c = (is_cross[slot] & child).sum() // 2
+    want_m = ((~is_cross[slot]) & child).sum()
+    assert int(sc) == int(want_c) and int(sm) == int(want_m)
+
+    # tournament_pool: valid indices, geometric bias toward best ranks
+    N, d, poolsize = 400, 12, 200
+    population = torch.randn(N, d, device=dev)
+    rank = torch.randint(0, 6, (N,), device=dev, dtype=torch.long)
+    picks_best = 0
+    for seed in range(20):
+        pool, pool_idx = _hipops.tournament_pool(
+            population.contiguous(), rank.contiguous(), poolsize, 0.5, seed
+        )
+        assert pool.shape == (poolsize, d)
+        assert int(pool_idx.min()) >= 0 and int(pool_idx.max()) < N
+        assert len(set(pool_idx.tolist())) == poolsize  # without replacement
+        assert torch.allclose(pool, population[pool_idx])
+        picks_best += int((rank[pool_idx] == 0).sum())
+    # best-rank candidates must be strongly overrepresented vs uniform
+    frac_best_pop = float((rank == 0).sum()) / N
+    frac_best_pool = picks_best / (20 * poolsize)
+    assert frac_best_pool > 1.2 * frac_best_pop
