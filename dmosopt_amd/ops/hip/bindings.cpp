@@ -159,8 +159,8 @@ std::vector<torch::Tensor> tournament_pool(torch::Tensor population,
       torch::empty({poolsize}, population.options().dtype(torch::kLong));
   const float log1mp = logf(1.0f - (float)p_sel);
   if (launch_tournament(population.data_ptr<float>(),
-                        rank.data_ptr<long long>(), pool.data_ptr<float>(),
-                        pool_idx.data_ptr<long long>(), N, d, (int)poolsize,
+                        (long long*)rank.data_ptr<int64_t>(), pool.data_ptr<float>(),
+                        (long long*)pool_idx.data_ptr<int64_t>(), N, d, (int)poolsize,
                         log1mp, (unsigned long long)seed, cur_stream()) != 0)
     return {torch::Tensor(), torch::Tensor()};
   return {pool, pool_idx};
@@ -172,9 +172,9 @@ bool survivor_count(torch::Tensor perm, torch::Tensor c_idx,
                     torch::Tensor succ_mut) {
   CHECK_GPU(perm);
   return launch_survivor_count(
-             perm.data_ptr<long long>(), c_idx.data_ptr<long long>(),
+             (long long*)perm.data_ptr<int64_t>(), (long long*)c_idx.data_ptr<int64_t>(),
              perm.size(0), c_idx.size(0), (int)n_children,
-             succ_cross.data_ptr<long long>(), succ_mut.data_ptr<long long>(),
+             (long long*)succ_cross.data_ptr<int64_t>(), (long long*)succ_mut.data_ptr<int64_t>(),
              cur_stream()) == 0;
 }
 
@@ -324,8 +324,8 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   auto crowd = crowding_distance(obj).to(torch::kFloat32);  // (N,)
   const int N = obj.size(0), d = parm.size(1), m = obj.size(1);
   auto key = torch::empty({N}, rank.options());
-  launch_pack_rank_crowd(rank.data_ptr<long long>(), crowd.data_ptr<float>(),
-                         key.data_ptr<long long>(), N, cur_stream());
+  launch_pack_rank_crowd((long long*)rank.data_ptr<int64_t>(), crowd.data_ptr<float>(),
+                         (long long*)key.data_ptr<int64_t>(), N, cur_stream());
   auto perm = torch::argsort(key, /*stable=*/true, /*dim=*/-1,
                              /*descending=*/false)
                   .slice(0, 0, pop)
@@ -335,9 +335,9 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   auto obj_o = torch::empty({P, m}, obj.options());
   auto rank_o = torch::empty({P}, rank.options());
   launch_gather3(parm.data_ptr<float>(), obj.data_ptr<float>(),
-                 rank.data_ptr<long long>(), perm.data_ptr<long long>(),
+                 (long long*)rank.data_ptr<int64_t>(), (long long*)perm.data_ptr<int64_t>(),
                  parm_o.data_ptr<float>(), obj_o.data_ptr<float>(),
-                 rank_o.data_ptr<long long>(), P, d, m, cur_stream());
+                 (long long*)rank_o.data_ptr<int64_t>(), P, d, m, cur_stream());
   return {parm_o, obj_o, rank_o, perm};
 }
 
