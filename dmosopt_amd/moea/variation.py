@@ -127,6 +127,20 @@ def event_stream_variation(
     c_idx_t = dev[o : o + 2 * C]; o += 2 * C
     m_idx_t = dev[o : o + M]
 
+    if pool.device.type == "cuda" and ops.native_available():
+        # whole-generation variation in ONE launch (bitwise identical to the
+        # split sbx/mutation kernels + gather)
+        from dmosopt_amd import _hipops
+
+        x_gen = _hipops.variation_slots(
+            pool.float().contiguous(), gather_idx, i1_t, i2_t, im_t,
+            di_crossover.float().contiguous(), di_mutation.float().contiguous(),
+            xlb.float().contiguous(), xub.float().contiguous(),
+            float(mutation_rate), C,
+            int(rng.integers(0, 2**62)), int(rng.integers(0, 2**62)),
+        )
+        return x_gen.to(pool.dtype), c_idx_t, m_idx_t
+
     parts = []
     if C:
         c1, c2 = ops.sbx_from_pool(

@@ -47,6 +47,7 @@ int launch_sceua_accept(float*, float*, const float*, const float*, const int*, 
 int launch_tournament(const float*, const long long*, float*, long long*, int, int, int, float, unsigned long long, hipStream_t);
 int launch_survivor_count(const long long*, const long long*, int, int, int, long long*, long long*, hipStream_t);
 void launch_pack_rank_crowd(const long long*, const float*, long long*, int, hipStream_t);
+void launch_variation_slots(const float*, const long long*, const long long*, const long long*, const long long*, const float*, const float*, const float*, const float*, float*, int, int, int, float, unsigned long long, unsigned long long, hipStream_t);
 void launch_gather3(const float*, const float*, const long long*, const long long*, float*, float*, long long*, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
@@ -176,6 +177,27 @@ bool survivor_count(torch::Tensor perm, torch::Tensor c_idx,
              perm.size(0), c_idx.size(0), (int)n_children,
              (long long*)succ_cross.data_ptr<int64_t>(), (long long*)succ_mut.data_ptr<int64_t>(),
              cur_stream()) == 0;
+}
+
+// Whole-generation variation in one launch (see variation.hip).
+torch::Tensor variation_slots(torch::Tensor pool, torch::Tensor src_rows,
+                              torch::Tensor p1, torch::Tensor p2,
+                              torch::Tensor im, torch::Tensor di_c,
+                              torch::Tensor di_m, torch::Tensor lo,
+                              torch::Tensor hi, double mutation_rate,
+                              int64_t C, int64_t seed_sbx, int64_t seed_mut) {
+  CHECK_GPU(pool);
+  const int total = src_rows.size(0), d = pool.size(1);
+  auto out = torch::empty({total, d}, pool.options());
+  launch_variation_slots(
+      pool.data_ptr<float>(), (long long*)src_rows.data_ptr<int64_t>(),
+      (long long*)p1.data_ptr<int64_t>(), (long long*)p2.data_ptr<int64_t>(),
+      (long long*)im.data_ptr<int64_t>(), di_c.data_ptr<float>(),
+      di_m.data_ptr<float>(), lo.data_ptr<float>(), hi.data_ptr<float>(),
+      out.data_ptr<float>(), total, (int)C, d, (float)mutation_rate,
+      (unsigned long long)seed_sbx, (unsigned long long)seed_mut,
+      cur_stream());
+  return out;
 }
 
 torch::Tensor sceua_propose(torch::Tensor cx, torch::Tensor lcs,
@@ -430,6 +452,7 @@ void cmaes_update_(torch::Tensor A, torch::Tensor Ainv, torch::Tensor pc,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
+  m.def("variation_slots", &variation_slots);
   m.def("tournament_pool", &tournament_pool);
   m.def("survivor_count", &survivor_count);
   m.def("sceua_propose", &sceua_propose);

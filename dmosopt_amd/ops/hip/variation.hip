@@ -81,3 +81,61 @@ extern "C" void launch_mutation_batch(const float* pool, const int* parents,
   hipLaunchKernelGGL(mutation_batch_kernel, dim3(blocks), dim3(256), 0, stream,
                      pool, parents, di, lo, hi, out, M, d, mutation_rate, seed);
 }
+
+// Whole-generation variation in ONE launch: slot s of the output is child
+// src_rows[s] of the virtual [sbx_c1 | sbx_c2 | mutation] stack — decoded
+// on the fly, so the separate sbx + mutation + cat + gather launches (and
+// their ~10 us in-stream gaps each) collapse into one kernel. Philox
+// counters match the split kernels exactly: identical output bits.
+__global__ void variation_slots_kernel(
+    const float* __restrict__ pool, const long long* __restrict__ src_rows,
+    const long long* __restrict__ p1, const long long* __restrict__ p2,
+    const long long* __restrict__ im, const float* __restrict__ di_c,
+    const float* __restrict__ di_m, const float* __restrict__ lo,
+    const float* __restrict__ hi, float* __restrict__ out, int total, int C,
+    int d, float mutation_rate, unsigned long long seed_sbx,
+    unsigned long long seed_mut) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long long)total * d) return;
+  const int s = (int)(idx / d);
+  const int g = (int)(idx % d);
+  const long long row = src_rows[s];
+  float v;
+  if (row < 2 * C) {
+    const int c = (int)(row < C ? row : row - C);
+    const float a = pool[p1[c] * d + g];
+    const float b = pool[p2[c] * d + g];
+    const Philox4 r = philox4x32(seed_sbx, (unsigned long long)c * d + g);
+    const float u = u01(r.c0);
+    const float e = 1.f / (di_c[g] + 1.f);
+    const float beta = (u <= 0.5f) ? __powf(2.f * u, e)
+                                   : __powf(1.f / (2.f * (1.f - u)), e);
+    v = (row < C) ? 0.5f * ((1.f - beta) * a + (1.f + beta) * b)
+                  : 0.5f * ((1.f + beta) * a + (1.f - beta) * b);
+  } else {
+    const int m = (int)(row - 2 * C);
+    const float p = pool[im[m] * d + g];
+    const Philox4 ph =
+        philox4x32(seed_mut ^ 0x5deece66dULL, (unsigned long long)m * d + g);
+    const float u = u01(ph.c0);
+    const float e = 1.f / (di_m[g] + 1.f);
+    const float delta = (u < mutation_rate)
+                            ? __powf(2.f * u, e) - 1.f
+                            : 1.f - __powf(2.f * (1.f - u), e);
+    v = p + (hi[g] - lo[g]) * delta;
+  }
+  out[idx] = fminf(fmaxf(v, lo[g]), hi[g]);
+}
+
+extern "C" void launch_variation_slots(
+    const float* pool, const long long* src_rows, const long long* p1,
+    const long long* p2, const long long* im, const float* di_c,
+    const float* di_m, const float* lo, const float* hi, float* out,
+    int total, int C, int d, float mutation_rate, unsigned long long seed_sbx,
+    unsigned long long seed_mut, hipStream_t stream) {
+  long long n = (long long)total * d;
+  hipLaunchKernelGGL(variation_slots_kernel, dim3((int)((n + 255) / 256)),
+                     dim3(256), 0, stream, pool, src_rows, p1, p2, im, di_c,
+                     di_m, lo, hi, out, total, C, d, mutation_rate, seed_sbx,
+                     seed_mut);
+}
