@@ -337,3 +337,33 @@ def test_fused_moea_kernels(dev):
     frac_best_pop = float((rank == 0).sum()) / N
     frac_best_pool = picks_best / (20 * poolsize)
     assert frac_best_pool > 1.2 * frac_best_pop
+
+
+@pytest.mark.gpu
+def test_variation_slots_matches_split_path(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd import ops as dops
+
+    torch.manual_seed(5)
+    K, d, C, M = 100, 30, 45, 22
+    pool = torch.rand(K, d, device=dev)
+    i1 = torch.randint(0, K, (C,), device=dev)
+    i2 = (i1 + 1 + torch.randint(0, K - 1, (C,), device=dev)) % K
+    im = torch.randint(0, K, (M,), device=dev)
+    di_c = torch.full((d,), 1.0, device=dev)
+    di_m = torch.full((d,), 20.0, device=dev)
+    lo = torch.zeros(d, device=dev)
+    hi = torch.ones(d, device=dev)
+    total = 2 * C + M
+    src_rows = torch.randperm(total, device=dev)
+
+    s1, s2 = 1234, 9876
+    c1, c2 = dops.sbx_from_pool(pool, i1, i2, di_c, lo, hi, seed=s1)
+    mm = dops.mutation_from_pool(pool, im, di_m, lo, hi, 0.1, seed=s2)
+    want = torch.cat([c1, c2, mm], dim=0)[src_rows]
+
+    got = _hipops.variation_slots(
+        pool.contiguous(), src_rows.contiguous(), i1.contiguous(),
+        i2.contiguous(), im.contiguous(), di_c, di_m, lo, hi, 0.1, C, s1, s2,
+    )
+    assert torch.equal(got, want)
