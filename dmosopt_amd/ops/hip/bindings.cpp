@@ -11,6 +11,7 @@
   TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 
 extern "C" {
+void launch_matern_assemble_affine(const float*, const float*, const float*, float*, int, int, int, int, int, float, int, int, int, const float*, const float*, hipStream_t);
 void launch_matern_assemble(const float*, const float*, const float*, float*,
                             int, int, int, int, int, float, int, int, int,
                             hipStream_t);
@@ -131,15 +132,19 @@ torch::Tensor gp_nmll(torch::Tensor X, torch::Tensor theta, torch::Tensor y,
 torch::Tensor gp_predict_mean(torch::Tensor Xq, torch::Tensor X,
                               torch::Tensor theta, torch::Tensor alpha,
                               torch::Tensor y_mean, torch::Tensor y_std,
-                              double nu, bool aniso) {
+                              double nu, bool aniso,
+                              c10::optional<torch::Tensor> q_lb = c10::nullopt,
+                              c10::optional<torch::Tensor> q_invrg = c10::nullopt) {
   CHECK_GPU(Xq);
   CHECK_GPU(X);
   const int P = Xq.size(0), N = X.size(0), D = X.size(1), B = theta.size(0);
   auto Ks = torch::empty({B, P, N}, X.options());
-  launch_matern_assemble(Xq.data_ptr<float>(), X.data_ptr<float>(),
-                         theta.data_ptr<float>(), Ks.data_ptr<float>(), B, P,
-                         N, D, theta.size(1), 0.0f, nu_code(nu), aniso ? 1 : 0,
-                         0, cur_stream());
+  // optional per-dim affine normalizes RAW queries inside the kernel load
+  launch_matern_assemble_affine(
+      Xq.data_ptr<float>(), X.data_ptr<float>(), theta.data_ptr<float>(),
+      Ks.data_ptr<float>(), B, P, N, D, theta.size(1), 0.0f, nu_code(nu),
+      aniso ? 1 : 0, 0, q_lb ? q_lb->data_ptr<float>() : nullptr,
+      q_invrg ? q_invrg->data_ptr<float>() : nullptr, cur_stream());
   auto mean_n = torch::bmm(Ks, alpha).squeeze(-1);  // (m, P)
   return y_mean.unsqueeze(1).addcmul(y_std.unsqueeze(1), mean_n).transpose(0, 1);
 }
@@ -458,7 +463,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sceua_propose", &sceua_propose);
   m.def("sceua_accept", &sceua_accept);
   m.def("nsga2_select", &nsga2_select, "Fused survivor selection (cat+rank+crowding+sort+gather)");
-  m.def("gp_predict_mean", &gp_predict_mean, "Fused cross-kernel + posterior mean");
+  m.def("gp_predict_mean", &gp_predict_mean, "Fused cross-kernel + posterior mean",
+        py::arg("Xq"), py::arg("X"), py::arg("theta"), py::arg("alpha"),
+        py::arg("y_mean"), py::arg("y_std"), py::arg("nu"), py::arg("aniso"),
+        py::arg("q_lb") = py::none(), py::arg("q_invrg") = py::none());
   m.def("gp_nmll", &gp_nmll, "Fused batched GP NMLL (assemble+chol+solve+reduce)");
   m.def("cholesky_batched_", &cholesky_batched_,
         "In-place batched Cholesky; returns (logdet, info)");

@@ -39,7 +39,10 @@ __global__ void matern_assemble_kernel(
     const float* __restrict__ X,    // (N, D)
     const float* __restrict__ theta,  // (B, p)
     float* __restrict__ K,          // (B, P, N)
-    int P, int N, int D, int theta_stride, float jitter) {
+    int P, int N, int D, int theta_stride, float jitter,
+    const float* __restrict__ q_lb = nullptr,   // optional per-dim affine
+    const float* __restrict__ q_invrg = nullptr  // applied to Xq at load
+    ) {
   extern __shared__ float lds[];  // [2][TILE][D] scaled row slabs
   float* q_tile = lds;
   float* x_tile = lds + TILE * D;
@@ -58,7 +61,9 @@ __global__ void matern_assemble_kernel(
         ANISO ? __expf(-theta[b * theta_stride + 1 + col])
               : __expf(-theta[b * theta_stride + 1]);
     const int gq = tile_p + row;
-    q_tile[idx] = (gq < P) ? Xq[gq * D + col] * inv_ell : 0.f;
+    float qv = (gq < P) ? Xq[gq * D + col] : 0.f;
+    if (q_lb != nullptr) qv = (qv - q_lb[col]) * q_invrg[col];
+    q_tile[idx] = qv * inv_ell;
     const int gx = tile_n + row;
     x_tile[idx] = (gx < N) ? X[gx * D + col] * inv_ell : 0.f;
   }
@@ -101,17 +106,18 @@ __global__ void matern_assemble_kernel(
   }
 }
 
-extern "C" void launch_matern_assemble(
+extern "C" void launch_matern_assemble_affine(
     const float* Xq, const float* X, const float* theta, float* K, int B,
     int P, int N, int D, int theta_stride, float jitter, int nu_code,
-    int aniso, int symmetric, hipStream_t stream) {
+    int aniso, int symmetric, const float* q_lb, const float* q_invrg,
+    hipStream_t stream) {
   dim3 grid((N + TILE - 1) / TILE, (P + TILE - 1) / TILE, B);
   dim3 block(TPB);
   size_t lds_bytes = 2 * TILE * D * sizeof(float);
   #define DISPATCH(NU, AN, SY)                                              \
     hipLaunchKernelGGL((matern_assemble_kernel<NU, AN, SY>), grid, block,   \
                        lds_bytes, stream, Xq, X, theta, K, P, N, D,         \
-                       theta_stride, jitter)
+                       theta_stride, jitter, q_lb, q_invrg)
   #define DISPATCH_AN(NU)                                                   \
     if (aniso) {                                                            \
       if (symmetric) DISPATCH(NU, true, true); else DISPATCH(NU, true, false); \
@@ -126,4 +132,13 @@ extern "C" void launch_matern_assemble(
   }
   #undef DISPATCH_AN
   #undef DISPATCH
+}
+
+extern "C" void launch_matern_assemble(
+    const float* Xq, const float* X, const float* theta, float* K, int B,
+    int P, int N, int D, int theta_stride, float jitter, int nu_code,
+    int aniso, int symmetric, hipStream_t stream) {
+  launch_matern_assemble_affine(Xq, X, theta, K, B, P, N, D, theta_stride,
+                                jitter, nu_code, aniso, symmetric, nullptr,
+                                nullptr, stream);
 }
