@@ -25,8 +25,7 @@ void launch_peel_front(const int*, const unsigned char*, unsigned char*, int*,
                        int, int, hipStream_t);
 void launch_commit_front(const unsigned char*, unsigned char*, int*, int, int,
                          hipStream_t);
-void launch_crowding(const float*, const float*, const float*, float*, int,
-                     int, hipStream_t);
+void launch_crowding(const float*, float*, int, int, hipStream_t);
 void launch_sbx_batch(const float*, const int*, const int*, const float*,
                       const float*, const float*, float*, int, int,
                       unsigned long long, hipStream_t);
@@ -326,13 +325,8 @@ torch::Tensor crowding_distance(torch::Tensor Y) {
   const int N = Y.size(0), m = Y.size(1);
   if (N == 1) return torch::ones({1}, Y.options());
   TORCH_CHECK(N <= 16384, "crowding_distance HIP kernel supports N <= 16384");
-  auto lo = std::get<0>(Y.min(0)).contiguous();
-  auto hi = std::get<0>(Y.max(0)).contiguous();
-  auto span = (hi - lo).contiguous();
-  span = torch::where(span == 0, torch::ones_like(span), span).contiguous();
-  auto per_dim = torch::zeros({m, N}, Y.options());
-  launch_crowding(Y.data_ptr<float>(), lo.data_ptr<float>(),
-                  span.data_ptr<float>(), per_dim.data_ptr<float>(), N, m,
+  auto per_dim = torch::empty({m, N}, Y.options());
+  launch_crowding(Y.data_ptr<float>(), per_dim.data_ptr<float>(), N, m,
                   cur_stream());
   return per_dim.sum(0);  // fixed-order reduction: deterministic
 }

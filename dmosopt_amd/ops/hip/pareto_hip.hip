@@ -386,18 +386,18 @@ extern "C" void launch_commit_front(const unsigned char* front,
 #define CROWD_TPB 256
 
 __global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
-                                const float* __restrict__ lo,  // (m,)
-                                const float* __restrict__ span,  // (m,)
-                                float* __restrict__ out,  // (N,) pre-zeroed
+                                float* __restrict__ out,  // (m, N)
                                 int N, int m, int npow2) {
   extern __shared__ float sh[];
   float* vals = sh;                       // npow2
   int* idxs = (int*)(sh + npow2);         // npow2
   const int j = blockIdx.x;               // objective dim
-  const float l = lo[j], s = span[j];
 
+  // RAW column values: sorting is affine-invariant, and after the sort the
+  // column min/max are vals[0]/vals[N-1] — the former host-side min/max/
+  // span prep (~6 torch launches per call) is free here.
   for (int i = threadIdx.x; i < npow2; i += CROWD_TPB) {
-    vals[i] = (i < N) ? (Y[i * m + j] - l) / s : INFINITY;
+    vals[i] = (i < N) ? Y[i * m + j] : INFINITY;
     idxs[i] = i;
   }
   __syncthreads();
@@ -421,24 +421,27 @@ __global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
     }
   }
 
-  // gaps: boundary = 1, interior = vals[i+1] - vals[i-1].
+  // gaps: boundary = 1, interior = (vals[i+1] - vals[i-1]) / span.
   // Each dim writes its own row of out (m, N) — NO atomics, so the sum
   // order is fixed and results are bit-identical across replicated ranks
   // (float atomicAdd order would not be).
+  const float l = vals[0];
+  const float h = vals[N - 1];
+  float s = h - l;
+  if (!(s > 0.f) || !isfinite(s)) s = 1.f;
   for (int i = threadIdx.x; i < N; i += CROWD_TPB) {
-    float d = (i == 0 || i == N - 1) ? 1.f : vals[i + 1] - vals[i - 1];
+    float d = (i == 0 || i == N - 1) ? 1.f : (vals[i + 1] - vals[i - 1]) / s;
     if (isnan(d)) d = 0.f;
     out[(long long)j * N + idxs[i]] = d;
   }
 }
 
 // out must be an (m, N) buffer; the host sums over dim 0 deterministically.
-extern "C" void launch_crowding(const float* Y, const float* lo,
-                                const float* span, float* out, int N, int m,
+extern "C" void launch_crowding(const float* Y, float* out, int N, int m,
                                 hipStream_t stream) {
   int npow2 = 1;
   while (npow2 < N) npow2 <<= 1;
   size_t lds_bytes = npow2 * (sizeof(float) + sizeof(int));
   hipLaunchKernelGGL(crowding_kernel, dim3(m), dim3(CROWD_TPB), lds_bytes,
-                     stream, Y, lo, span, out, N, m, npow2);
+                     stream, Y, out, N, m, npow2);
 }
