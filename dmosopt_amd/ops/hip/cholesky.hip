@@ -253,6 +253,7 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
     int N, int k0) {
   __shared__ float S[CHOL_BS][CHOL_BS + 1];
+  __shared__ float colbuf[CHOL_BS];
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   float* Ab = A + (long long)b * N * N;
@@ -292,11 +293,15 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
       } else if (lane > j) {
         r[j] /= d;
       }
+      // broadcast column j through LDS with ONE workgroup fence: the
+      // (31-j) reads then pipeline freely, unlike a per-t __shfl chain
+      // (ds_bpermute each) or a volatile pointer (per-access ordering)
+      if (lane < bs) colbuf[lane] = r[j];
+      __threadfence_block();
 #pragma unroll
       for (int t = 0; t < CHOL_BS; ++t) {
         if (t <= j || t >= bs) continue;
-        const float stj = __shfl(r[j], t);
-        if (lane >= t) r[t] = fmaf(-r[j], stj, r[t]);
+        if (lane >= t) r[t] = fmaf(-r[j], colbuf[t], r[t]);
       }
     }
     // stage the factored block back to LDS (global writeback below is
