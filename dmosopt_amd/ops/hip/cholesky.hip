@@ -366,16 +366,27 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
 // below/right of the panel; blockIdx.y enumerates lower-triangular tile
 // pairs, each 256-thread block computes a 4x4 register tile per thread.
 __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
-    float* __restrict__ A, int N, int k0, int nt) {
+    float* __restrict__ A, int N, int k0, int nt, int tj_fixed, int off) {
   __shared__ float Pi[SYRK_TS][CHOL_BS + 1];
   __shared__ float Pj[SYRK_TS][CHOL_BS + 1];
   const int b = blockIdx.x;
   float* Ab = A + (long long)b * N * N;
   const int r0 = k0 + CHOL_BS;  // first trailing row
-  // decode lower-triangular pair p -> (ti, tj), tj <= ti
-  int p = blockIdx.y, ti = 0;
-  while (p > ti) { p -= ti + 1; ++ti; }
-  const int tj = p;
+  // tile-pair selection: tj_fixed >= 0 enumerates one tile COLUMN
+  // (ti = blockIdx.y + tj_fixed); otherwise blockIdx.y walks the lower
+  // triangle, shifted by `off` columns/rows (used to split the first tile
+  // column from the rest for cross-stream overlap)
+  int ti, tj;
+  if (tj_fixed >= 0) {
+    tj = tj_fixed;
+    ti = blockIdx.y + tj_fixed;
+  } else {
+    int p = blockIdx.y;
+    ti = 0;
+    while (p > ti) { p -= ti + 1; ++ti; }
+    tj = p + off;
+    ti += off;
+  }
   const int i0 = r0 + ti * SYRK_TS, j0 = r0 + tj * SYRK_TS;
   const int tid = threadIdx.x;
 
@@ -559,17 +570,67 @@ extern "C" void launch_nmll_reduce(const float* Z, const float* half_logdet,
 
 extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
                                        int B, int N, hipStream_t stream) {
+  // Cross-stream software pipeline: panel(k+1) only depends on the FIRST
+  // tile column of step k's trailing update (its 64 columns cover the next
+  // panel), so the remaining tile pairs run on a side stream concurrently
+  // with the next panel. Regions are disjoint: tj0(k) writes cols
+  // [k0+32, k0+95]; rest(k) writes cols >= k0+96; panel(k+1) writes cols
+  // [k0+32, k0+63] AFTER tj0(k) on the main stream.
+  static int overlap = -1;
+  static hipStream_t s2 = nullptr;
+  static hipEvent_t evP[2], evR[2];
+  if (overlap < 0) {
+    const char* env = getenv("DMOSOPT_CHOL_OVERLAP");
+    overlap = (env && env[0] == '0') ? 0 : 1;
+    if (overlap) {
+      hipStreamCreateWithFlags(&s2, hipStreamNonBlocking);
+      for (int i = 0; i < 2; ++i) {
+        hipEventCreateWithFlags(&evP[i], hipEventDisableTiming);
+        hipEventCreateWithFlags(&evR[i], hipEventDisableTiming);
+      }
+    }
+  }
   hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
-  for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
+  if (!overlap) {
+    for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
+      hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0,
+                         stream, A, logdet, info, N, k0);
+      const int trailing = N - k0 - CHOL_BS;
+      if (trailing > 0) {
+        const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
+        hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, nt * (nt + 1) / 2),
+                           dim3(CHOLP_TPB), 0, stream, A, N, k0, nt, -1, 0);
+      }
+    }
+    return;
+  }
+  int kstep = 0;
+  bool have_rest_prev = false;
+  for (int k0 = 0; k0 < N; k0 += CHOL_BS, ++kstep) {
     hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0, stream,
                        A, logdet, info, N, k0);
     const int trailing = N - k0 - CHOL_BS;
-    if (trailing > 0) {
-      const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
-      hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, nt * (nt + 1) / 2),
-                         dim3(CHOLP_TPB), 0, stream, A, N, k0, nt);
+    if (trailing <= 0) continue;
+    const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
+    const int slot = kstep & 1;
+    if (nt > 1) hipEventRecord(evP[slot], stream);
+    // tj0(k) waits for rest(k-1): their column regions overlap
+    if (have_rest_prev) hipStreamWaitEvent(stream, evR[slot ^ 1], 0);
+    hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, nt), dim3(CHOLP_TPB), 0,
+                       stream, A, N, k0, nt, 0, 0);
+    if (nt > 1) {
+      const int rest_pairs = (nt - 1) * nt / 2;
+      hipStreamWaitEvent(s2, evP[slot], 0);
+      hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, rest_pairs),
+                         dim3(CHOLP_TPB), 0, s2, A, N, k0, nt, -1, 1);
+      hipEventRecord(evR[slot], s2);
+      have_rest_prev = true;
+    } else {
+      have_rest_prev = false;
     }
   }
+  // rejoin the side stream before anything downstream reads the factor
+  if (have_rest_prev) hipStreamWaitEvent(stream, evR[(kstep - 1) & 1], 0);
 }
 
 extern "C" void launch_cholesky_batched(float* A, float* logdet, int* info,
