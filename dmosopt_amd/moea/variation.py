@@ -5,9 +5,10 @@ AGEMOEA.py:146-180): a Bernoulli event stream of crossovers (2 children)
 and mutations (1 child) until popsize-1 children exist. The stream here is
 drawn VECTORIZED on the host (one `rng.random((n, 2))` draw instead of a
 per-event Python loop), all index bookkeeping is assembled with numpy, a
-single H2D transfer carries every index array, and all variation executes
-as two fused device launches (Philox SBX + polynomial mutation kernels on
-gfx950) assembled into event order with one gather. Pair selection is
+single pinned H2D transfer carries every index array, and on GPU ALL
+variation (SBX + mutation + event-order assembly) executes as ONE
+slot-decoded gfx950 launch (CPU fallback: split Philox SBX/mutation
+kernels + gather, bitwise-identical output). Pair selection is
 uniform over ordered distinct pairs (statistically equivalent to the
 reference's `choice(poolsize, 2, replace=False)`).
 """

@@ -71,9 +71,10 @@ def _use_native(*tensors: torch.Tensor) -> bool:
 # ------------------------------------------------------------------ ranking
 def pareto_rank(Y: torch.Tensor) -> torch.Tensor:
     if _use_native(Y):
-        # small N: the native column-max peel (2 launches/front, cheap
-        # scans); large N: matvec dominator-count peel (per-front O(N^2)
-        # int traffic would dominate on many-front problems)
+        # small N: bit-matrix ranking (grid-wide packed dominator build +
+        # one-block popcount peel); large N: matvec dominator-count peel
+        # (per-front O(N^2) int traffic would dominate on many-front
+        # problems)
         if Y.shape[0] <= 2048:
             return _native.pareto_rank(Y.contiguous().float())
         return _pareto_rank_gpu(Y)

@@ -3,15 +3,19 @@
 //
 // Replaces reference dda.py:25-120 and indicators.py:12-51.
 //
-// pareto_rank design: D[i][j] = #objectives with Y[i][k] <= Y[j][k] is an
-// N x N x m comparison reduce computed tile-wise with both 32-row objective
-// slabs in LDS; the front peel (iterated column-max + mask) runs as a small
-// secondary kernel per front — front count is small (<= tens) and each peel
-// pass is a single coalesced N x N sweep.
+// pareto_rank design (three size regimes, dispatched in bindings.cpp):
+//   N <= ~2048: bit-matrix path — dom_bits_kernel packs "i dominates j"
+//     grid-wide into 32 bits/word, then peel_bits_kernel peels all fronts
+//     in one small block with popcount(dom_mask & front_mask);
+//   fallbacks: single-launch from-Y peel (LDS-staged objectives), and for
+//     large N the DDA matrix (tile-wise, 32-row objective slabs in LDS)
+//     with chased peel_front/commit_front launches or a matvec
+//     dominator-count peel driven from the host.
 //
 // crowding design: one workgroup per objective dimension; (value, index)
-// pairs bitonic-sorted in LDS, boundary/interior gaps computed in place and
-// scattered into the output with atomics (m-way accumulation).
+// pairs bitonic-sorted in LDS (column min/max read off the sorted ends);
+// per-dim gap rows are written WITHOUT atomics and summed in fixed order
+// so results are bit-identical across replicated ranks.
 
 #include "common.h"
 #include <math.h>
