@@ -131,3 +131,28 @@ def test_early_stopping_utils():
     stats = analyze_loss_trajectory(losses[: stopped_at + 1])
     assert stats["n"] == stopped_at + 1
     assert "lr" in suggest_hyperparameters(losses, 0.1)
+
+
+def test_normalization_roundtrip():
+    """ZeroToOneNormalization forward/backward inverse; degenerate bounds
+    handled (reference normalization.py semantics)."""
+    from dmosopt_amd.normalization import (
+        NoNormalization,
+        PreNormalization,
+        ZeroToOneNormalization,
+    )
+
+    rng = np.random.default_rng(0)
+    xl = np.array([0.0, -2.0, 5.0])
+    xu = np.array([1.0, 2.0, 5.0])  # last dim degenerate
+    z = ZeroToOneNormalization(xl, xu)
+    X = rng.random((20, 3)) * (xu - xl) + xl
+    N = z.forward(X)
+    assert N.min() >= -1e-9 and N[:, :2].max() <= 1 + 1e-9
+    assert np.allclose(z.backward(N), X, atol=1e-9)
+
+    n = NoNormalization()
+    assert np.allclose(n.forward(X), X) and np.allclose(n.backward(X), X)
+
+    p = PreNormalization(zero_to_one=True, ideal=xl, nadir=xu)
+    assert p.ideal is not None
