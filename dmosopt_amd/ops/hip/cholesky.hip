@@ -397,21 +397,32 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
   }
   __syncthreads();
 
-  const int ty = tid / 16, tx = tid % 16;  // 16x16 threads, 4x4 each
-  float acc[4][4] = {};
-  for (int kk = 0; kk < CHOL_BS; ++kk) {
-    float a[4], c[4];
-    for (int r = 0; r < 4; ++r) a[r] = Pi[ty * 4 + r][kk];
-    for (int r = 0; r < 4; ++r) c[r] = Pj[tx * 4 + r][kk];
-    for (int r = 0; r < 4; ++r)
-      for (int q = 0; q < 4; ++q) acc[r][q] = fmaf(a[r], c[q], acc[r][q]);
-  }
-  for (int r = 0; r < 4; ++r) {
-    const int i = i0 + ty * 4 + r;
-    if (i >= N) continue;
-    for (int q = 0; q < 4; ++q) {
-      const int j = j0 + tx * 4 + q;
-      if (j < N && j <= i) Ab[(long long)i * N + j] -= acc[r][q];
+  // MFMA tile core: C -= Pi * Pj^T on the matrix units.
+  // v_mfma_f32_16x16x4_f32 is EXACT fp32 (identical to an fmaf chain) at
+  // the f32 vector rate, but one instruction carries a whole 16x16x4 tile
+  // — 16x fewer issue slots than the 4x4-register-tile FMA version, which
+  // matters for this latency-bound kernel. Operand map (cdna4_isa §10):
+  // A: lane l -> A[l&15][l>>4]; B: lane l -> B[l>>4][l&15];
+  // C/D (f32x4): col = lane&15, row = (lane>>4)*4 + reg.
+  typedef __attribute__((ext_vector_type(4))) float f32x4;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int lr = lane & 15, lk = lane >> 4;
+#pragma unroll
+  for (int sIdx = 0; sIdx < 4; ++sIdx) {
+    const int sub = wave * 4 + sIdx;        // 4x4 grid of 16x16 subtiles
+    const int r16 = (sub >> 2) * 16, c16 = (sub & 3) * 16;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int k = 0; k < CHOL_BS; k += 4) {
+      const float a = Pi[r16 + lr][k + lk];   // A[i][k]
+      const float b = Pj[c16 + lr][k + lk];   // B[k][j] = Pj[j][k]
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int i = i0 + r16 + lk * 4 + r;
+      const int j = j0 + c16 + lr;
+      if (i < N && j < N && j <= i) Ab[(long long)i * N + j] -= acc[r];
     }
   }
 }
