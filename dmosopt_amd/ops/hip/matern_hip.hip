@@ -5,11 +5,15 @@
 // One launch assembles B hyperparameter candidates' kernel matrices:
 //   K[b] = sf2[b] * matern_nu(||x_i/ell - x_j/ell||) (+ (noise+jitter) I)
 //
-// Design (gfx950): 32x32 output tile per 256-thread block, both 32-row
-// input slabs staged in LDS with the 1/ell scaling fused into the load, each
-// thread computing a 2x2 sub-tile. The d-loop is a register FMA chain; for
-// the d<=128 regime of this framework the kernel is bandwidth-trivial and
-// the fusion (no separate cdist + transform passes) is the point.
+// Design (gfx950): 32x32 output tile per 256-thread block (4 waves), both
+// 32-row input slabs staged in LDS with the 1/ell scaling fused into the
+// load. The pairwise cross terms run on the MATRIX UNITS:
+// d2(i,j) = |q_i|^2 + |x_j|^2 - 2 q_i.x_j, with the q.x dot products
+// computed by v_mfma_f32_16x16x4_f32 (exact fp32 at the f32 vector rate,
+// one instruction per 16x16x4 tile — ~16x fewer issue slots than the
+// per-pair FMA chain). d2 is clamped at 0 against cancellation before the
+// Matern transform. Fusion (no separate cdist + transform passes) is the
+// point; for the d <= 128 regime the kernel is bandwidth-trivial.
 
 #include "common.h"
 #include <math.h>
@@ -69,40 +73,44 @@ __global__ void matern_assemble_kernel(
   }
   __syncthreads();
 
-  const int ty = threadIdx.x / 16;  // 0..15
-  const int tx = threadIdx.x % 16;
+  // per-row squared norms of the scaled slabs
+  float* qn = x_tile + TILE * D;  // TILE floats
+  float* xn = qn + TILE;          // TILE floats
+  if (threadIdx.x < 2 * TILE) {
+    const bool is_q = threadIdx.x < TILE;
+    const int row = is_q ? threadIdx.x : threadIdx.x - TILE;
+    const float* src = (is_q ? q_tile : x_tile) + row * D;
+    float acc = 0.f;
+    for (int k = 0; k < D; ++k) acc = fmaf(src[k], src[k], acc);
+    (is_q ? qn : xn)[row] = acc;
+  }
+  __syncthreads();
 
+  // cross terms on the matrix units: wave w computes 16x16 subtile
+  // (w>>1, w&1). A: lane -> q[(r16 + (lane&15))][k + (lane>>4)];
+  // B[k][j] = x[j][k]; C/D: col = lane&15, row = (lane>>4)*4 + reg.
+  typedef __attribute__((ext_vector_type(4))) float f32x4;
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int lr = lane & 15, lk = lane >> 4;
+  const int r16 = (wave >> 1) * 16, c16 = (wave & 1) * 16;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int k = 0; k < D; k += 4) {
+    const bool ink = (k + lk) < D;
+    const float a = ink ? q_tile[(r16 + lr) * D + k + lk] : 0.f;
+    const float bv = ink ? x_tile[(c16 + lr) * D + k + lk] : 0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+  }
 #pragma unroll
-  for (int sy = 0; sy < 2; ++sy) {
-#pragma unroll
-    for (int sx = 0; sx < 2; ++sx) {
-      const int lp = ty * 2 + sy;   // local row in [0, TILE)
-      const int ln = tx * 2 + sx;
-      const int gp = tile_p + lp;
-      const int gn = tile_n + ln;
-      if (gp >= P || gn >= N) continue;
-      const float* qa = q_tile + lp * D;
-      const float* xb = x_tile + ln * D;
-      float d2 = 0.f;
-      int k = 0;
-      for (; k + 4 <= D; k += 4) {
-        float t0 = qa[k] - xb[k];
-        float t1 = qa[k + 1] - xb[k + 1];
-        float t2 = qa[k + 2] - xb[k + 2];
-        float t3 = qa[k + 3] - xb[k + 3];
-        d2 = fmaf(t0, t0, d2);
-        d2 = fmaf(t1, t1, d2);
-        d2 = fmaf(t2, t2, d2);
-        d2 = fmaf(t3, t3, d2);
-      }
-      for (; k < D; ++k) {
-        float t = qa[k] - xb[k];
-        d2 = fmaf(t, t, d2);
-      }
-      float v = sf2 * matern_transform<NU>(d2);
-      if (SYMMETRIC && gp == gn) v += noise + jitter;
-      K[((long long)b * P + gp) * N + gn] = v;
-    }
+  for (int r = 0; r < 4; ++r) {
+    const int lp = r16 + lk * 4 + r;
+    const int ln = c16 + lr;
+    const int gp = tile_p + lp;
+    const int gn = tile_n + ln;
+    if (gp >= P || gn >= N) continue;
+    const float d2 = fmaxf(qn[lp] + xn[ln] - 2.f * acc[r], 0.f);
+    float v = sf2 * matern_transform<NU>(d2);
+    if (SYMMETRIC && gp == gn) v += noise + jitter;
+    K[((long long)b * P + gp) * N + gn] = v;
   }
 }
 
@@ -113,7 +121,7 @@ extern "C" void launch_matern_assemble_affine(
     hipStream_t stream) {
   dim3 grid((N + TILE - 1) / TILE, (P + TILE - 1) / TILE, B);
   dim3 block(TPB);
-  size_t lds_bytes = 2 * TILE * D * sizeof(float);
+  size_t lds_bytes = (2 * TILE * D + 2 * TILE) * sizeof(float);
   #define DISPATCH(NU, AN, SY)                                              \
     hipLaunchKernelGGL((matern_assemble_kernel<NU, AN, SY>), grid, block,   \
                        lds_bytes, stream, Xq, X, theta, K, P, N, D,         \
