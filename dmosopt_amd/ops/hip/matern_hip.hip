@@ -73,6 +73,38 @@ __global__ void matern_assemble_kernel(
   }
   __syncthreads();
 
+  if (SYMMETRIC) {
+    // TRAIN kernel: exact (q-x)^2 accumulation. The norms+MFMA form
+    // perturbs d2 by ~1e-6 (cancellation), which flips SCE-UA accept
+    // decisions bit-wise and changes the fit's iteration count — the
+    // train path stays bitwise-stable; the cross kernel (predictions,
+    // tolerance-level effect) uses the matrix units below.
+    const int ty = threadIdx.x / 16;
+    const int tx = threadIdx.x % 16;
+#pragma unroll
+    for (int sy = 0; sy < 2; ++sy) {
+#pragma unroll
+      for (int sx = 0; sx < 2; ++sx) {
+        const int lp = ty * 2 + sy;
+        const int ln = tx * 2 + sx;
+        const int gp = tile_p + lp;
+        const int gn = tile_n + ln;
+        if (gp >= P || gn >= N) continue;
+        const float* qa = q_tile + lp * D;
+        const float* xb = x_tile + ln * D;
+        float d2 = 0.f;
+        for (int k = 0; k < D; ++k) {
+          const float t = qa[k] - xb[k];
+          d2 = fmaf(t, t, d2);
+        }
+        float v = sf2 * matern_transform<NU>(d2);
+        if (gp == gn) v += noise + jitter;
+        K[((long long)b * P + gp) * N + gn] = v;
+      }
+    }
+    return;
+  }
+
   // per-row squared norms of the scaled slabs
   float* qn = x_tile + TILE * D;  // TILE floats
   float* xn = qn + TILE;          // TILE floats
