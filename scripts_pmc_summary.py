@@ -15,36 +15,31 @@ def find(prefix):
     raise RuntimeError(f"{prefix}* not in {tables}")
 
 
-try:
-    cr = find("rocpd_counter")  # counter records
-except RuntimeError:
-    print("tables:", tables)
-    raise
-# discover schema
-cols = [r[1] for r in cur.execute(f"PRAGMA table_info({cr})")]
-print("counter table:", cr, cols, file=sys.stderr)
+cr = find("rocpd_pmc_event")
 disp = find("rocpd_kernel_dispatch")
 sym = find("rocpd_info_kernel_symbol")
-q = f"""
-SELECT k.display_name, c.counter_id, SUM(c.value)
-FROM {cr} c
-JOIN {disp} d ON c.dispatch_id = d.dispatch_id
+ci = find("rocpd_info_pmc")
+for t in (cr, disp, ci):
+    print(t.split("_0000")[0], [r[1] for r in cur.execute(f"PRAGMA table_info({t})")],
+          file=sys.stderr)
+cr_cols = [r[1] for r in cur.execute(f"PRAGMA table_info({cr})")]
+ci_cols = [r[1] for r in cur.execute(f"PRAGMA table_info({ci})")]
+# column-name guesses across rocpd versions
+val_col = "value" if "value" in cr_cols else cr_cols[-1]
+pmc_key = [c for c in cr_cols if "pmc" in c and c != val_col][0]
+evt_key = [c for c in cr_cols if "event" in c or "dispatch" in c][0]
+name_col = "name" if "name" in ci_cols else ci_cols[1]
+ci_id = "id" if "id" in ci_cols else ci_cols[0]
+disp_cols = [r[1] for r in cur.execute(f"PRAGMA table_info({disp})")]
+d_evt = [c for c in disp_cols if "event" in c or c == "dispatch_id" or c == "id"][0]
+print("join:", val_col, pmc_key, evt_key, name_col, d_evt, file=sys.stderr)
+rows = cur.execute(f"""
+SELECT k.display_name, c.{pmc_key}, SUM(c.{val_col})
+FROM {cr} c JOIN {disp} d ON c.{evt_key} = d.{d_evt}
 JOIN {sym} k ON d.kernel_id = k.id
-GROUP BY k.display_name, c.counter_id
-"""
-try:
-    rows = cur.execute(q).fetchall()
-except Exception as e:
-    print("query failed:", e)
-    for t in tables:
-        print(t, [r[1] for r in cur.execute(f"PRAGMA table_info({t})")])
-    sys.exit(1)
-# counter names
-try:
-    ci = find("rocpd_info_counter")
-    names = dict(cur.execute(f"SELECT id, name FROM {ci}"))
-except Exception:
-    names = {}
+GROUP BY k.display_name, c.{pmc_key}
+""").fetchall()
+names = dict(cur.execute(f"SELECT {ci_id}, {name_col} FROM {ci}"))
 agg = {}
 for kname, cid, val in rows:
     agg.setdefault(kname.split("(")[0][:60], {})[names.get(cid, cid)] = val
