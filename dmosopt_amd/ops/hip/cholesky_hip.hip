@@ -368,8 +368,12 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
 // pairs, each 256-thread block computes a 4x4 register tile per thread.
 __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
     float* __restrict__ A, int N, int k0, int nt, int tj_fixed, int off) {
-  __shared__ float Pi[SYRK_TS][CHOL_BS + 1];
-  __shared__ float Pj[SYRK_TS][CHOL_BS + 1];
+  // stride 32 with an XOR swizzle on the k-column: the MFMA operand read
+  // (lane -> [row + lr][k + lk]) hits banks (row+lk) mod 32 under a +1 pad,
+  // an up-to-8-way conflict (PMC: 2.2 extra cycles per LDS instruction);
+  // c = k ^ ((row & 7) << 2) spreads (lr, lk) over all 32 banks.
+  __shared__ float Pi[SYRK_TS][CHOL_BS];
+  __shared__ float Pj[SYRK_TS][CHOL_BS];
   const int b = blockIdx.x;
   float* Ab = A + (long long)b * N * N;
   const int r0 = k0 + CHOL_BS;  // first trailing row
@@ -393,8 +397,9 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
 
   for (int idx = tid; idx < SYRK_TS * CHOL_BS; idx += blockDim.x) {
     const int r = idx / CHOL_BS, c = idx % CHOL_BS;
-    Pi[r][c] = (i0 + r < N) ? Ab[(long long)(i0 + r) * N + k0 + c] : 0.0f;
-    Pj[r][c] = (j0 + r < N) ? Ab[(long long)(j0 + r) * N + k0 + c] : 0.0f;
+    const int cs = c ^ ((r & 7) << 2);  // swizzled physical column
+    Pi[r][cs] = (i0 + r < N) ? Ab[(long long)(i0 + r) * N + k0 + c] : 0.0f;
+    Pj[r][cs] = (j0 + r < N) ? Ab[(long long)(j0 + r) * N + k0 + c] : 0.0f;
   }
   __syncthreads();
 
@@ -415,8 +420,9 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int k = 0; k < CHOL_BS; k += 4) {
-      const float a = Pi[r16 + lr][k + lk];   // A[i][k]
-      const float b = Pj[c16 + lr][k + lk];   // B[k][j] = Pj[j][k]
+      const int ra = r16 + lr, rb = c16 + lr;
+      const float a = Pi[ra][(k + lk) ^ ((ra & 7) << 2)];  // A[i][k]
+      const float b = Pj[rb][(k + lk) ^ ((rb & 7) << 2)];  // B[k][j]=Pj[j][k]
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
 #pragma unroll

@@ -46,9 +46,10 @@ __global__ void matern_assemble_kernel(
     const float* __restrict__ q_lb = nullptr,   // optional per-dim affine
     const float* __restrict__ q_invrg = nullptr  // applied to Xq at load
     ) {
-  extern __shared__ float lds[];  // [2][TILE][D] scaled row slabs
+  extern __shared__ float lds[];  // [2][TILE][DS] scaled row slabs
+  const int DS = D | 1;  // odd stride: row base spreads over all 32 banks
   float* q_tile = lds;
-  float* x_tile = lds + TILE * D;
+  float* x_tile = lds + TILE * DS;
 
   const int b = blockIdx.z;
   const int tile_p = blockIdx.y * TILE;
@@ -66,9 +67,9 @@ __global__ void matern_assemble_kernel(
     const int gq = tile_p + row;
     float qv = (gq < P) ? Xq[gq * D + col] : 0.f;
     if (q_lb != nullptr) qv = (qv - q_lb[col]) * q_invrg[col];
-    q_tile[idx] = qv * inv_ell;
+    q_tile[row * DS + col] = qv * inv_ell;
     const int gx = tile_n + row;
-    x_tile[idx] = (gx < N) ? X[gx * D + col] * inv_ell : 0.f;
+    x_tile[row * DS + col] = (gx < N) ? X[gx * D + col] * inv_ell : 0.f;
   }
   __syncthreads();
 
@@ -89,8 +90,8 @@ __global__ void matern_assemble_kernel(
         const int gp = tile_p + lp;
         const int gn = tile_n + ln;
         if (gp >= P || gn >= N) continue;
-        const float* qa = q_tile + lp * D;
-        const float* xb = x_tile + ln * D;
+        const float* qa = q_tile + lp * DS;
+        const float* xb = x_tile + ln * DS;
         float d2 = 0.f;
         for (int k = 0; k < D; ++k) {
           const float t = qa[k] - xb[k];
@@ -105,12 +106,12 @@ __global__ void matern_assemble_kernel(
   }
 
   // per-row squared norms of the scaled slabs
-  float* qn = x_tile + TILE * D;  // TILE floats
+  float* qn = x_tile + TILE * DS;  // TILE floats
   float* xn = qn + TILE;          // TILE floats
   if (threadIdx.x < 2 * TILE) {
     const bool is_q = threadIdx.x < TILE;
     const int row = is_q ? threadIdx.x : threadIdx.x - TILE;
-    const float* src = (is_q ? q_tile : x_tile) + row * D;
+    const float* src = (is_q ? q_tile : x_tile) + row * DS;
     float acc = 0.f;
     for (int k = 0; k < D; ++k) acc = fmaf(src[k], src[k], acc);
     (is_q ? qn : xn)[row] = acc;
@@ -127,8 +128,8 @@ __global__ void matern_assemble_kernel(
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   for (int k = 0; k < D; k += 4) {
     const bool ink = (k + lk) < D;
-    const float a = ink ? q_tile[(r16 + lr) * D + k + lk] : 0.f;
-    const float bv = ink ? x_tile[(c16 + lr) * D + k + lk] : 0.f;
+    const float a = ink ? q_tile[(r16 + lr) * DS + k + lk] : 0.f;
+    const float bv = ink ? x_tile[(c16 + lr) * DS + k + lk] : 0.f;
     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
   }
 #pragma unroll
@@ -152,7 +153,7 @@ extern "C" void launch_matern_assemble_affine(
     hipStream_t stream) {
   dim3 grid((N + TILE - 1) / TILE, (P + TILE - 1) / TILE, B);
   dim3 block(TPB);
-  size_t lds_bytes = (2 * TILE * D + 2 * TILE) * sizeof(float);
+  size_t lds_bytes = (2 * TILE * (D | 1) + 2 * TILE) * sizeof(float);
   #define DISPATCH(NU, AN, SY)                                              \
     hipLaunchKernelGGL((matern_assemble_kernel<NU, AN, SY>), grid, block,   \
                        lds_bytes, stream, Xq, X, theta, K, P, N, D,         \
