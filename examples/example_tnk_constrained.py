@@ -1,6 +1,11 @@
 """TNK constrained problem with MO-CMA-ES + logistic feasibility model."""
 
+import os
+import sys
+
 import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import dmosopt_amd
 

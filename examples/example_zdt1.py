@@ -1,6 +1,11 @@
 """ZDT1 d=30 with NSGA-II + GP surrogate (the canonical README config)."""
 
+import os
+import sys
+
 import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import dmosopt_amd
 
