@@ -271,3 +271,30 @@ def test_pareto_rank_beats_naive_reference_walltime():
 
     assert np.array_equal(fast.numpy(), want)
     assert t_fast < t_naive, (t_fast, t_naive)
+
+
+def test_packed_rank_crowding_key_orders_like_lexsort():
+    """The GPU survivor-selection path sorts ONE packed int64 key
+    (rank << 32 | ~float32_bits(crowding)) instead of two stable sorts;
+    this guards the bit-trick's ordering equivalence (non-negative IEEE
+    floats compare like their bit patterns; +inf boundary points first
+    within a rank; ties resolved by index exactly like np.lexsort)."""
+    torch.manual_seed(0)
+    N = 600
+    rank = torch.randint(0, 12, (N,))
+    d = torch.rand(N).double() * 10
+    d[torch.randint(0, N, (25,))] = float("inf")
+    d[torch.randint(0, N, (10,))] = 0.0
+
+    df = torch.nan_to_num(
+        d.float().clamp_min(0.0), nan=0.0, posinf=float(torch.finfo(torch.float32).max)
+    )
+    bits = df.view(torch.int32).to(torch.int64)
+    key = (rank.to(torch.int64) << 32) | ((0x7FFFFFFF - bits) & 0xFFFFFFFF)
+    perm_fast = torch.argsort(key, stable=True)
+
+    perm = torch.arange(N)
+    for k in [(-df.double()), rank.double()]:  # np.lexsort order: last primary
+        order = torch.argsort(k[perm], stable=True)
+        perm = perm[order]
+    assert torch.equal(perm_fast, perm)
