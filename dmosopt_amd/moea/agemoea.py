@@ -285,7 +285,8 @@ class AGEMOEAOptimizer(MOEA):
         crowd = self.state.crowd_dist
 
         pool_idx = ops.tournament_selection(
-            population.shape[0], p.poolsize, [-crowd, rank], rng
+            population.shape[0], p.poolsize, [-crowd, rank], rng,
+            generator=self.torch_random,
         )
         pool = population[pool_idx]
         di_c = torch.as_tensor(p.di_crossover, dtype=pool.dtype, device=pool.device)
