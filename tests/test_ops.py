@@ -298,3 +298,42 @@ def test_packed_rank_crowding_key_orders_like_lexsort():
         order = torch.argsort(k[perm], stable=True)
         perm = perm[order]
     assert torch.equal(perm_fast, perm)
+
+
+def test_epsilon_sort_box_semantics():
+    """Epsilon-box archive (reference MOEA.py:470-595 semantics): box
+    dominance, same-box corner-distance replacement, eviction."""
+    from dmosopt_amd.moea.epsilon import EpsilonSort
+
+    s = EpsilonSort([1.0, 1.0])
+    s.sortinto(np.array([2.3, 2.3]), tagalong="a")   # box (2,2)
+    s.sortinto(np.array([2.6, 2.6]), tagalong="b")   # same box, farther corner
+    assert s.tagalongs == ["a"]                      # a kept (closer to corner)
+    s.sortinto(np.array([2.1, 2.1]), tagalong="c")   # same box, closer
+    assert s.tagalongs == ["c"]
+    s.sortinto(np.array([0.5, 3.5]), tagalong="d")   # box (0,3): nondominated
+    assert set(s.tagalongs) == {"c", "d"}
+    s.sortinto(np.array([0.5, 0.5]), tagalong="e")   # box (0,0): dominates both
+    assert s.tagalongs == ["e"]
+    s.sortinto(np.array([5.0, 5.0]), tagalong="f")   # dominated by e's box
+    assert s.tagalongs == ["e"]
+
+
+def test_epsilon_get_best_selects_archive():
+    from dmosopt_amd.core.engine import epsilon_get_best
+
+    rng = np.random.default_rng(4)
+    x = rng.random((60, 3))
+    f1 = rng.random(60)
+    y = np.column_stack([f1, 1.0 - f1 + 0.02 * rng.standard_normal(60)])
+    bx, by, _, _, eps = epsilon_get_best(x, y, None, None, epsilons=[0.1, 0.1])
+    assert by.shape[0] > 1
+    # archive members pairwise non-dominated at epsilon-box resolution
+    boxes = np.floor(by / np.asarray(eps))
+    for i in range(len(boxes)):
+        for j in range(len(boxes)):
+            if i == j:
+                continue
+            assert not (
+                (boxes[i] <= boxes[j]).all() and (boxes[i] < boxes[j]).any()
+            )
