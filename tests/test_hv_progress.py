@@ -1,0 +1,88 @@
+"""HV-progress termination subsystem (reference hv_termination.py parity):
+precision scheduler, algorithm router, multi-fidelity tracker, convergence
+detector, and the SlidingWindow-based termination."""
+
+import numpy as np
+import pytest
+
+from dmosopt_amd.termination.hv_progress import (
+    ConvergenceDetector,
+    HVEstimate,
+    HVAlgorithmRouter,
+    HypervolumeProgressTermination,
+    MultiFidelityHVTracker,
+    ProgressivePrecisionScheduler,
+)
+
+
+def test_precision_scheduler_phases_and_adaptation():
+    s = ProgressivePrecisionScheduler(coarse_until=50, medium_until=150)
+    assert s.get_precision_config(0)["fidelity"] == "coarse"
+    assert s.get_precision_config(100)["fidelity"] == "medium"
+    assert s.get_precision_config(200)["fidelity"] == "fine"
+    # stagnation escalates precision sooner
+    s.adapt_to_progress(1e-6)
+    assert s.get_precision_config(35)["fidelity"] == "medium"
+    # strong progress de-escalates the boost
+    s.adapt_to_progress(0.5)
+    s.adapt_to_progress(0.5)
+    assert s.get_precision_config(35)["fidelity"] == "coarse"
+    # finer fidelities use tighter eps, sparser cadence
+    assert s.configs["fine"]["eps"] < s.configs["coarse"]["eps"]
+    assert s.configs["fine"]["cadence"] > s.configs["coarse"]["cadence"]
+
+
+def test_algorithm_router_dimensionality():
+    r = HVAlgorithmRouter()
+    assert r.select_algorithm(2, 100, 0.05) == "box"
+    assert r.select_algorithm(12, 100, 0.05) == "adaptive_mc"
+    assert r.select_algorithm(25, 100, 0.05) == "reduced_mc"
+    # exact 2D value through the router
+    front = np.array([[1.0, 3.0], [2.0, 2.0], [3.0, 1.0]])
+    hv = r.compute_hypervolume(front, np.array([4.0, 4.0]), eps=0.05)
+    assert hv == pytest.approx(6.0)
+
+
+def test_multifidelity_tracker_cadence():
+    sched = ProgressivePrecisionScheduler(coarse_until=50, medium_until=150)
+    tracker = MultiFidelityHVTracker(np.array([4.0, 4.0]), sched, HVAlgorithmRouter())
+    front = np.array([[1.0, 3.0], [2.0, 2.0], [3.0, 1.0]])
+    e0 = tracker.compute_and_update(front, 0)
+    assert e0 is not None and e0.fidelity == "coarse"
+    assert tracker.compute_and_update(front, 1) is None  # within cadence 2
+    assert tracker.compute_and_update(front, 2) is not None
+    assert tracker.get_best_estimate().generation == 2
+
+
+def test_convergence_detector():
+    det = ConvergenceDetector(min_generations=20, stagnation_window=5)
+    flat = [HVEstimate(7.0, "fine", g) for g in range(30)]
+    res = det.check_convergence(flat, generation=30)
+    assert res["converged"] and res["confidence"] > 0.5
+    rising = [HVEstimate(1.0 + 0.5 * g, "fine", g) for g in range(30)]
+    assert not det.check_convergence(rising, generation=30)["converged"]
+    assert not det.check_convergence(flat[:3], generation=30)["converged"]
+
+
+def test_hv_progress_termination_stops_on_stagnant_front():
+    class Problem:
+        n_obj = 2
+
+    term = HypervolumeProgressTermination(
+        Problem(), ref_point=np.array([11.0, 11.0]), n_last=4, nth_gen=1
+    )
+
+    class Hist:
+        def __init__(self, gen, y):
+            self.n_gen = gen
+            self.gen_index = gen
+            self.y = y
+            self.x = np.zeros((len(y), 3))
+
+    y = np.array([[1.0, 3.0], [2.0, 2.0], [3.0, 1.0]])
+    stopped = False
+    for g in range(1, 40):
+        if term.has_terminated(Hist(g, y)):
+            stopped = True
+            break
+    assert stopped, "stagnant front must terminate"
