@@ -402,3 +402,44 @@ def test_all_optimizers_one_epoch_gpu(dev):
         )
         assert res.best_x.shape[1] == d and res.best_y.shape[1] == m
         assert np.isfinite(res.best_y).all(), name
+
+
+@pytest.mark.gpu
+def test_gp_nmll_fused_matches_torch_oracle(dev):
+    from dmosopt_amd import _hipops
+    from dmosopt_amd.models.gp_core import batched_nmll
+
+    torch.manual_seed(2)
+    N, d, B = 120, 8, 10
+    X = torch.rand(N, d, device=dev)
+    y = torch.randn(N, device=dev)
+    theta = torch.stack(
+        [
+            torch.zeros(B, device=dev),                      # log sf2
+            torch.rand(B, device=dev) * 1.5 - 1.0,           # log ell
+            torch.full((B,), -3.0, device=dev),              # log noise
+        ],
+        dim=1,
+    )
+    got = _hipops.gp_nmll(X.contiguous(), theta.contiguous(), y.contiguous(),
+                          2.5, False, 1e-6)
+    want = batched_nmll(X.double(), y.double(), theta.double(), nu=2.5,
+                        anisotropic=False, jitter=1e-6, differentiable=True)
+    assert torch.allclose(got.double().cpu(), want.cpu(), rtol=2e-3, atol=2e-3)
+
+
+@pytest.mark.gpu
+def test_gp_predict_mean_fused_matches_oracle(dev):
+    import numpy as np
+    from dmosopt_amd.models.gp import GPRMatern
+    from dmosopt_amd.benchmarks.problems import zdt1
+
+    rng = np.random.default_rng(3)
+    X = rng.random((90, 6))
+    Y = zdt1(X).numpy()
+    gp = GPRMatern(X, Y, 6, 2, np.zeros(6), np.ones(6), optimizer="sceua",
+                   seed=11, device=dev)
+    xq = torch.as_tensor(rng.random((33, 6)), device=dev)
+    got = gp.evaluate_tensor(xq).cpu().numpy()          # fused affine path
+    want, _ = gp.predict(xq.cpu().numpy())               # host oracle path
+    assert np.allclose(got, want, rtol=1e-3, atol=1e-3)
