@@ -158,6 +158,12 @@ std::vector<torch::Tensor> tournament_pool(torch::Tensor population,
                                            int64_t poolsize, double p_sel,
                                            int64_t seed) {
   CHECK_GPU(population);
+  TORCH_CHECK(population.dtype() == torch::kFloat32 &&
+                  rank.dtype() == torch::kLong,
+              "tournament_pool: (float32 population, int64 rank) required");
+  TORCH_CHECK(rank.size(0) == population.size(0) &&
+                  poolsize <= population.size(0),
+              "tournament_pool: shape mismatch");
   const int N = population.size(0), d = population.size(1);
   auto pool = torch::empty({poolsize, d}, population.options());
   auto pool_idx =
@@ -176,6 +182,10 @@ bool survivor_count(torch::Tensor perm, torch::Tensor c_idx,
                     int64_t n_children, torch::Tensor succ_cross,
                     torch::Tensor succ_mut) {
   CHECK_GPU(perm);
+  TORCH_CHECK(perm.dtype() == torch::kLong && c_idx.dtype() == torch::kLong &&
+                  succ_cross.dtype() == torch::kLong &&
+                  succ_mut.dtype() == torch::kLong,
+              "survivor_count: int64 tensors required");
   return launch_survivor_count(
              (long long*)perm.data_ptr<int64_t>(), (long long*)c_idx.data_ptr<int64_t>(),
              perm.size(0), c_idx.size(0), (int)n_children,
@@ -191,6 +201,10 @@ torch::Tensor variation_slots(torch::Tensor pool, torch::Tensor src_rows,
                               torch::Tensor hi, double mutation_rate,
                               int64_t C, int64_t seed_sbx, int64_t seed_mut) {
   CHECK_GPU(pool);
+  TORCH_CHECK(pool.dtype() == torch::kFloat32 &&
+                  src_rows.dtype() == torch::kLong &&
+                  p1.size(0) == C && p2.size(0) == C,
+              "variation_slots: f32 pool, int64 indices, |p1|=|p2|=C");
   const int total = src_rows.size(0), d = pool.size(1);
   auto out = torch::empty({total, d}, pool.options());
   launch_variation_slots(
@@ -208,6 +222,9 @@ torch::Tensor sceua_propose(torch::Tensor cx, torch::Tensor lcs,
                             torch::Tensor bl, torch::Tensor bu, int64_t nps,
                             int64_t seed) {
   CHECK_GPU(cx);
+  TORCH_CHECK(cx.dtype() == torch::kFloat32 && cx.dim() == 4 &&
+                  lcs.dtype() == torch::kInt32,
+              "sceua_propose: cx (S,G,npg,nopt) f32 + int32 lcs required");
   const int S = cx.size(0), G = cx.size(1), npg = cx.size(2),
             nopt = cx.size(3);
   auto cand = torch::empty({3, (long)S * G, nopt}, cx.options());
@@ -339,6 +356,15 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
                                         torch::Tensor y_gen,
                                         torch::Tensor pop_parm,
                                         torch::Tensor pop_obj, int64_t pop) {
+  CHECK_GPU(x_gen);
+  TORCH_CHECK(x_gen.dtype() == torch::kFloat32 &&
+                  pop_parm.dtype() == torch::kFloat32,
+              "nsga2_select: float32 inputs required");
+  TORCH_CHECK(x_gen.size(1) == pop_parm.size(1) &&
+                  y_gen.size(1) == pop_obj.size(1) &&
+                  x_gen.size(0) == y_gen.size(0) &&
+                  pop_parm.size(0) == pop_obj.size(0),
+              "nsga2_select: shape mismatch");
   auto parm = torch::cat({x_gen, pop_parm}, 0);
   auto obj = torch::cat({y_gen, pop_obj}, 0);
   auto rank = pareto_rank(obj);                    // (N,) long
