@@ -1,0 +1,79 @@
+"""Configuration-matrix shakeout: run() across optimizer x surrogate x
+constraint combinations (miniature budgets). Guards the glue paths the
+focused tests don't cross (reference exercises these through its
+integration scripts, tests/test_zdt1_*.py etc.)."""
+
+import numpy as np
+import pytest
+
+import dmosopt_amd
+
+
+def _sphere2(pp):
+    names = sorted(pp.keys())
+    x = np.array([pp[k] for k in names])
+    return np.array([np.sum(x**2), np.sum((x - 1.0) ** 2)])
+
+
+def _constrained(pp):
+    names = sorted(pp.keys())
+    x = np.array([pp[k] for k in names])
+    y = np.array([np.sum(x**2), np.sum((x - 1.0) ** 2)])
+    c = np.array([x[0] + x[1] - 0.2])  # feasible when sum > 0.2
+    return y, c
+
+
+def _params(opt_id, **over):
+    p = {
+        "opt_id": opt_id,
+        "obj_fun": _sphere2,
+        "problem_parameters": {},
+        "space": {f"x{i}": [0.0, 1.0] for i in range(4)},
+        "objective_names": ["f1", "f2"],
+        "population_size": 16,
+        "num_generations": 4,
+        "initial_maxiter": 2,
+        "n_initial": 2,
+        "n_epochs": 2,
+        "surrogate_method_name": "gpr",
+        "surrogate_method_kwargs": {"anisotropic": False, "optimizer": "sceua"},
+        "random_seed": 5,
+    }
+    p.update(over)
+    return p
+
+
+@pytest.mark.parametrize("optimizer", ["nsga2", "age", "smpso", "cmaes", "trs"])
+def test_every_optimizer_with_gpr_surrogate(optimizer):
+    best = dmosopt_amd.run(_params(f"m_{optimizer}", optimizer=optimizer),
+                           verbose=False)
+    assert best is not None
+    _, by = best
+    y = np.column_stack([v for _, v in by])
+    assert np.isfinite(y).all()
+
+
+@pytest.mark.parametrize("surrogate", ["egp", "vgp", None])
+def test_surrogate_variants_end_to_end(surrogate):
+    over = {"surrogate_method_name": surrogate}
+    if surrogate == "egp":
+        over["surrogate_method_kwargs"] = {"n_iter": 30}
+    elif surrogate == "vgp":
+        over["surrogate_method_kwargs"] = {"n_iter": 30, "num_inducing": 16}
+    best = dmosopt_amd.run(_params(f"m_s_{surrogate}", **over), verbose=False)
+    assert best is not None
+
+
+def test_constrained_with_feasibility_model():
+    p = _params(
+        "m_constr",
+        obj_fun=_constrained,
+        constraint_names=["c1"],
+        feasibility_method_name="logreg",
+        num_generations=5,
+    )
+    best = dmosopt_amd.run(p, verbose=False)
+    assert best is not None
+    _, by = best
+    y = np.column_stack([v for _, v in by])
+    assert np.isfinite(y).all()
