@@ -116,6 +116,12 @@ class _GPRBase:
         )
         Yn = (Y - y_mean[None, :]) / y_std[None, :]
 
+        if optimizer == "dlib" and logger is not None:
+            logger.warning(
+                "GP optimizer 'dlib' (reference model.py:1367-1416 optional "
+                "dlib.find_min_global path) is served by the batched SCE-UA "
+                "search here; dlib is not a dependency of this framework"
+            )
         if optimizer in ("sceua", "dlib", None):
             YnT = Yn.T.contiguous()  # (m, N)
 
