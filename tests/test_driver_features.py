@@ -193,3 +193,31 @@ def test_reduce_fun_applied_to_results():
     x, y = dmosopt_amd.sopt_dict["t_reduce"].optimizer_dict[0].get_evals()
     want = np.stack([np.sum(x**2, axis=1), np.sum((x - 1) ** 2, axis=1)], axis=1)
     assert np.allclose(y, want, atol=1e-5)
+
+
+def test_get_best_and_print_best_feasibility_filters(capsys):
+    """get_best feasible filtering + features/constraints passthrough
+    (reference dmosopt.py get_best/print_best surface)."""
+
+    def objfun(pp):
+        names = sorted(pp.keys())
+        x = np.array([pp[k] for k in names])
+        y = np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+        c = np.array([x[0] - 0.5])  # feasible iff x0 > 0.5
+        return y, c
+
+    params = _base("t_best", obj_fun=objfun, constraint_names=["c1"])
+    dmosopt_amd.run(params, verbose=False)
+    dopt = dmosopt_amd.sopt_dict["t_best"]
+
+    # get_best returns ([(pname, col)...], [(oname, col)...][, f][, c])
+    prms, objs, bc = dopt.get_best(feasible=True, return_constraints=True)
+    assert (bc > 0).all(), "feasible filter must keep only c > 0"
+    n_feas = len(objs[0][1])
+    prms_all, objs_all = dopt.get_best(feasible=False)
+    assert len(objs_all[0][1]) >= n_feas
+    # parameter columns align with the space names
+    assert sorted(n for n, _ in prms) == sorted(f"x{i}" for i in range(4))
+
+    dopt.print_best(feasible=True)  # must not raise
+    assert len(objs) == 2
