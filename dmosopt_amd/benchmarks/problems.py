@@ -374,6 +374,7 @@ _PROBLEMS = {
     "maf1": (maf1, None, (0.0, 1.0)),
     "maf2": (maf2, None, (0.0, 1.0)),
     "maf4": (maf4, None, (0.0, 1.0)),
+    "sphere": (sphere, None, (-5.0, 5.0)),
 }
 
 

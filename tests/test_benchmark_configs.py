@@ -101,3 +101,48 @@ def test_problem_space_generation():
     assert len(space) == 12 and space["x1"] == [0.0, 1.0]
     meta = bp.get_problem_metadata("zdt3", 2)
     assert meta["pareto_front_type"] == "disconnected"
+
+
+@pytest.mark.parametrize(
+    "name,n_obj",
+    [("zdt1", 2), ("zdt2", 2), ("zdt3", 2), ("zdt4", 2), ("zdt6", 2),
+     ("dtlz1", 3), ("dtlz2", 3), ("dtlz3", 3), ("dtlz4", 3), ("dtlz5", 3),
+     ("dtlz7", 3), ("wfg1", 3), ("wfg4", 3), ("maf1", 5), ("maf2", 5),
+     ("maf4", 5), ("sphere", 2)],
+)
+def test_problem_zoo_shapes_and_finiteness(name, n_obj):
+    """Every benchmark problem evaluates batched, with the declared number
+    of objectives, finite values, and known anchor behavior where cheap."""
+    from dmosopt_amd.benchmarks import problems as P
+
+    fn = P.get_problem(name)
+    rng = np.random.default_rng(0)
+    d = 12
+    x = rng.random((17, d))
+    kwargs = {} if name.startswith("zdt") or name == "sphere" else {"n_obj": n_obj}
+    y = fn(x, **kwargs) if kwargs else fn(x)
+    y = y.numpy()
+    assert y.shape == (17, n_obj)
+    assert np.isfinite(y).all()
+
+
+def test_zdt1_known_values():
+    from dmosopt_amd.benchmarks.problems import zdt1
+
+    # on the Pareto front (tail dims zero): f2 = 1 - sqrt(f1)
+    x = np.zeros((3, 30))
+    x[:, 0] = [0.0, 0.25, 1.0]
+    y = zdt1(x).numpy()
+    assert np.allclose(y[:, 0], [0.0, 0.25, 1.0])
+    assert np.allclose(y[:, 1], 1.0 - np.sqrt([0.0, 0.25, 1.0]))
+
+
+def test_constrained_problems_shapes():
+    from dmosopt_amd.benchmarks.problems import constr, osy, srn, tnk
+
+    rng = np.random.default_rng(1)
+    for fn, d, n_c in [(tnk, 2, 2), (constr, 2, 2), (srn, 2, 2), (osy, 6, 6)]:
+        x = rng.random((9, d))
+        y, c = fn(x)
+        assert y.shape[0] == 9 and c.shape == (9, n_c)
+        assert np.isfinite(y.numpy()).all()
