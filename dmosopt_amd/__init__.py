@@ -17,8 +17,23 @@ def run(dopt_params, time_limit=None, feasible=True, return_features=False,
         spawn_startup_wait=None, spawn_executable=None, spawn_args=[],
         nprocs_per_worker=1, collective_mode="gather", verbose=True,
         worker_debug=False):
-    """Run a distributed optimization. See dmosopt_amd.api.run."""
+    """Run a distributed optimization. See dmosopt_amd.api.run.
+
+    The MPI-spawn arguments of the reference surface (spawn_workers,
+    sequential_spawn, spawn_startup_wait, spawn_executable, spawn_args,
+    nprocs_per_worker, collective_mode, worker_debug) are accepted for API
+    compatibility; distribution here is torch.distributed (one process per
+    GPU, launched externally via torchrun), so they have no effect.
+    """
     from dmosopt_amd.api import run as _run
+
+    if spawn_workers or nprocs_per_worker != 1:
+        import logging
+
+        logging.getLogger("dmosopt_amd").warning(
+            "spawn_workers/nprocs_per_worker are MPI-spawn options of the "
+            "reference API; launch ranks with torchrun instead (ignored)"
+        )
 
     return _run(
         dopt_params,
