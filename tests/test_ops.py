@@ -337,3 +337,18 @@ def test_epsilon_get_best_selects_archive():
             assert not (
                 (boxes[i] <= boxes[j]).all() and (boxes[i] < boxes[j]).any()
             )
+
+
+def test_ops_degenerate_shapes():
+    """Empty and single-row populations must not crash any ranking op."""
+    assert ops.pareto_rank(torch.zeros(0, 2)).shape == (0,)
+    assert int(ops.pareto_rank(torch.zeros(1, 2))[0]) == 0
+    assert ops.crowding_distance(torch.zeros(1, 2)).shape == (1,)
+    assert ops.get_duplicates(torch.zeros(0, 3)).shape == (0,)
+    assert ops.lexsort([torch.zeros(0)]).shape == (0,)
+    # identical rows: same rank, neither dominates the other
+    Y = torch.ones(4, 2)
+    assert torch.equal(ops.pareto_rank(Y), torch.zeros(4, dtype=torch.long))
+    # single objective column: total order by value
+    y1 = torch.tensor([[3.0], [1.0], [2.0]])
+    assert ops.pareto_rank(y1).tolist() == [2, 0, 1]
