@@ -319,6 +319,7 @@ class DistOptimizer:
             self.eval_fun, logger=self.logger,
         )
         dim = len(self.param_names)
+        any_restored = False
         for problem_id in self.problem_ids:
             initial = None
             if problem_id in self.old_evals and len(self.old_evals[problem_id]) > 0:
@@ -368,8 +369,9 @@ class DistOptimizer:
                 device=self.device,
             )
             self.storage_dict[problem_id] = []
-            if initial is not None:
-                self.print_best()
+            any_restored = any_restored or (initial is not None)
+        if any_restored:
+            self.print_best()
 
     # ------------------------------------------------------------- requests
     def _process_requests(self):

@@ -150,3 +150,50 @@ def test_run_with_save_and_resume(tmp_path):
     params2["n_epochs"] = 1
     best2 = dmosopt_amd.run(params2, verbose=False)
     assert best2 is not None
+
+
+def test_multi_problem_save_and_resume(tmp_path):
+    """Multi-problem runs persist per-problem datasets and resume with the
+    problem split intact (reference init_from_h5 multi-problem path)."""
+    import dmosopt_amd
+
+    def obj_fun_mp(mpp):
+        out = {}
+        for pid, pp in mpp.items():
+            x = np.array([pp[k] for k in sorted(pp.keys())])
+            out[pid] = np.array([np.sum((x - 0.1 * pid) ** 2), np.sum((x - 1) ** 2)])
+        return out
+
+    fp = str(tmp_path / "mp.h5")
+
+    def params(resume):
+        return {
+            "opt_id": "t_mp_h5",
+            "obj_fun": obj_fun_mp,
+            "problem_parameters": {},
+            "space": {f"x{i}": [0.0, 1.0] for i in range(3)},
+            "objective_names": ["f1", "f2"],
+            "problem_ids": {1, 2},
+            "population_size": 12,
+            "num_generations": 3,
+            "surrogate_method_name": None,
+            "n_initial": 2,
+            "n_epochs": 1,
+            "random_seed": 13,
+            "file_path": fp,
+            "save": True,
+            "resume": resume,
+        }
+
+    best = dmosopt_amd.run(params(False), verbose=False)
+    assert set(best.keys()) == {1, 2}
+    n1 = dmosopt_amd.sopt_dict["t_mp_h5"].optimizer_dict[1].get_evals()[0].shape[0]
+    dmosopt_amd.sopt_dict.clear()
+
+    best2 = dmosopt_amd.run(params(True), verbose=False)
+    assert set(best2.keys()) == {1, 2}
+    dopt = dmosopt_amd.sopt_dict["t_mp_h5"]
+    assert dopt.has_problem_ids
+    # resumed archive includes the first run's evaluations
+    n1b = dopt.optimizer_dict[1].get_evals()[0].shape[0]
+    assert n1b >= n1
