@@ -416,6 +416,13 @@ class DistOptimizer:
             ):
                 self.save_evals()
                 self.saved_eval_count = self.eval_count
+        # flush the tail below the cadence (reference dmosopt.py:1329-1335:
+        # nothing unsaved survives the end of a request pump)
+        if self.save and 0 < self.saved_eval_count < self.eval_count or (
+            self.save and self.saved_eval_count == 0 and self.eval_count > 0
+        ):
+            self.save_evals()
+            self.saved_eval_count = self.eval_count
         return self.eval_count, self.saved_eval_count
 
     # ---------------------------------------------------------------- epoch
