@@ -209,3 +209,53 @@ def test_multi_problem_save_and_resume(tmp_path):
     for pid in (1, 2):
         prms, objs = best2[pid]
         assert len(objs) == 2 and len(objs[0][1]) > 0
+
+
+def test_surrogate_evals_and_optimizer_params_saved(tmp_path):
+    """save_surrogate_evals + save_optimizer_params land in the schema's
+    surrogate_evals/{...} and optimizer_params/{epoch} groups and survive
+    resume (reference dmosopt.py:2186-2270)."""
+    import dmosopt_amd
+    from dmosopt_amd.storage.h5 import h5_load_raw
+
+    def obj(pp):
+        x = np.array([pp[k] for k in sorted(pp.keys())])
+        return np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+
+    fp = str(tmp_path / "se.h5")
+    params = {
+        "opt_id": "t_se",
+        "obj_fun": obj,
+        "problem_parameters": {},
+        "space": {f"x{i}": [0.0, 1.0] for i in range(3)},
+        "objective_names": ["f1", "f2"],
+        "population_size": 16,
+        "num_generations": 4,
+        "n_initial": 2,
+        # the reference's save gating (dmosopt.py:1450) writes these groups
+        # only for epochs that are > 0 AND followed by another epoch, so
+        # three epochs are the minimum that produces them
+        "n_epochs": 3,
+        "surrogate_method_name": "gpr",
+        "surrogate_method_kwargs": {"anisotropic": False, "optimizer": "sceua"},
+        "optimizer": "nsga2",
+        "random_seed": 3,
+        "file_path": fp,
+        "save": True,
+        "save_surrogate_evals": True,
+        "save_optimizer_params": True,
+    }
+    dmosopt_amd.run(params, verbose=False)
+    dmosopt_amd.sopt_dict.clear()
+
+    from dmosopt_amd.storage.h5 import _h5core
+
+    h5 = _h5core()
+    f = h5.H5File(fp, "r")
+    try:
+        assert f.has("t_se/surrogate_evals/objectives")
+        assert f.has("t_se/surrogate_evals/parameters")
+        assert f.has("t_se/surrogate_evals/epochs")
+        assert f.has("t_se/optimizer_params/1/optimizer_name")
+    finally:
+        f.close()
