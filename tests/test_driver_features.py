@@ -221,3 +221,30 @@ def test_get_best_and_print_best_feasibility_filters(capsys):
 
     dopt.print_best(feasible=True)  # must not raise
     assert len(objs) == 2
+
+
+def test_zero_epochs_evaluates_initial_only():
+    """n_epochs <= 0 evaluates the initial design and stops (reference
+    run_epoch(completed_epoch=True) path)."""
+    params = _base("t_zero", n_epochs=0)
+    best = dmosopt_amd.run(params, verbose=False)
+    assert best is not None
+    x, y = dmosopt_amd.sopt_dict["t_zero"].optimizer_dict[0].get_evals()
+    assert x.shape[0] > 0 and y.shape[1] == 2
+
+
+def test_callable_initial_method():
+    """initial_method may be a callable (reference MOASMO.xinit:180-181):
+    called as method(Ninit, nInput, local_random) returning unit-box
+    samples, scaled to the bounds by xinit."""
+    seen = {}
+
+    def my_sampler(n, d, local_random):
+        seen["shape"] = (n, d)
+        return local_random.random((n, d))
+
+    params = _base("t_init_call", initial_method=my_sampler, n_epochs=1,
+                   surrogate_method_name=None, num_generations=3)
+    best = dmosopt_amd.run(params, verbose=False)
+    assert best is not None
+    assert "shape" in seen and seen["shape"][1] == 4
