@@ -98,3 +98,36 @@ def test_optimizer_cycling_mixed():
         n_epochs=2,
     )
     assert best is not None
+
+
+def test_seeded_runs_are_bit_identical():
+    """Two runs from the same seed must produce BIT-identical archives —
+    the multi-GPU weak-scaling scheme replicates the MOEA control flow
+    across ranks from identical seeds, so any nondeterminism here would
+    desynchronize ranks (reference seed discipline, dmosopt.py:629-636)."""
+    def run_once(tag):
+        params = {
+            "opt_id": tag,
+            "obj_fun": _zdt1_objfun,
+            "problem_parameters": {},
+            "space": {f"x{i:02d}": [0.0, 1.0] for i in range(6)},
+            "objective_names": ["f1", "f2"],
+            "population_size": 24,
+            "num_generations": 5,
+            "n_initial": 2,
+            "initial_maxiter": 2,
+            "n_epochs": 2,
+            "surrogate_method_name": "gpr",
+            "surrogate_method_kwargs": {"anisotropic": False, "optimizer": "sceua"},
+            "optimizer": "nsga2",
+            "random_seed": 77,
+        }
+        dmosopt_amd.run(params, verbose=False)
+        x, y = dmosopt_amd.sopt_dict[tag].optimizer_dict[0].get_evals()
+        return x.copy(), y.copy()
+
+    x1, y1 = run_once("t_det_a")
+    x2, y2 = run_once("t_det_b")
+    assert x1.shape == x2.shape
+    assert np.array_equal(x1, x2), "parameter archives diverged"
+    assert np.array_equal(y1, y2), "objective archives diverged"
