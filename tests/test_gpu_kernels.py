@@ -443,3 +443,18 @@ def test_gp_predict_mean_fused_matches_oracle(dev):
     got = gp.evaluate_tensor(xq).cpu().numpy()          # fused affine path
     want, _ = gp.predict(xq.cpu().numpy())               # host oracle path
     assert np.allclose(got, want, rtol=1e-3, atol=1e-3)
+
+
+@pytest.mark.gpu
+def test_gpu_epoch_bit_determinism(dev):
+    """Two same-seed epochs on the GPU must agree bitwise — the multi-GPU
+    weak-scaling bench replicates the MOEA control flow across ranks, so
+    every kernel feeding control flow must be run-to-run deterministic
+    (no float atomics, fixed-order reductions, counter-based RNG)."""
+    from bench import make_archive, one_epoch
+
+    X, Y = make_archive(seed=9)
+    r1 = one_epoch(X, Y, pop=64, rank=0, world=1, device=dev, seed=5, n_gen=12)
+    r2 = one_epoch(X, Y, pop=64, rank=0, world=1, device=dev, seed=5, n_gen=12)
+    assert torch.equal(r1[0].cpu(), r2[0].cpu()), "resample params diverged"
+    assert torch.equal(r1[1].cpu(), r2[1].cpu()), "resample objectives diverged"
