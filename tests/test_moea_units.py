@@ -106,3 +106,28 @@ def test_smpso_velocity_constriction_bounded():
         v = opt.state.velocity
         v = v.cpu().numpy() if isinstance(v, torch.Tensor) else v
         assert np.isfinite(v).all()
+
+
+def test_batched_cholesky_update_high_psucc_branch():
+    """psucc >= pthresh: pc decays without the z term and alpha gains the
+    ccov*cc*(2-cc) compensation (reference CMAES.py:500-511); A A^T must
+    still track (alpha C + ccov pc pc^T) and Ainv A = I."""
+    torch.manual_seed(1)
+    K, d = 3, 4
+    M = torch.randn(K, d, d, dtype=torch.float64) * 0.2
+    C0 = M @ M.transpose(-1, -2) + torch.eye(d, dtype=torch.float64)
+    A = torch.linalg.cholesky(C0)
+    Ainv = torch.linalg.inv(A)
+    pc = torch.randn(K, d, dtype=torch.float64)
+    z = torch.randn(K, d, dtype=torch.float64)
+    psucc = torch.ones(K, dtype=torch.float64)  # above pthresh
+    cc, ccov, pthresh = 0.2, 0.3, 0.44
+
+    A2, Ainv2, pc2 = batched_cholesky_update(A, Ainv, pc, z, psucc, cc, ccov, pthresh)
+    pc_want = (1 - cc) * pc
+    assert torch.allclose(pc2, pc_want, atol=1e-12)
+    alpha = (1 - ccov) + ccov * cc * (2 - cc)
+    C_want = alpha * C0 + ccov * pc_want[:, :, None] @ pc_want[:, None, :]
+    assert torch.allclose(A2 @ A2.transpose(-1, -2), C_want, atol=1e-8)
+    eye = torch.eye(d, dtype=torch.float64).expand(K, d, d)
+    assert torch.allclose(Ainv2 @ A2, eye, atol=1e-6)
