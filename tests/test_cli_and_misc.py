@@ -1,5 +1,7 @@
 """CLI, constrained sampling, and discrepancy tests."""
 
+import os
+
 import numpy as np
 import pytest
 
@@ -156,3 +158,33 @@ def test_normalization_roundtrip():
 
     p = PreNormalization(zero_to_one=True, ideal=xl, nadir=xu)
     assert p.ideal is not None
+
+
+def test_bench_json_contract():
+    """bench.py must emit the driver-contract JSON line: required fields,
+    whole-job value, weak scaling declaration, BASELINE metric name."""
+    import json
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run(
+        [_sys.executable, "bench.py", "--steps", "1", "--warmup", "0", "--gens", "2"],
+        capture_output=True, text=True, timeout=600,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, key
+    assert d["metric"].startswith("MO-ASMO epoch time")
+    assert d["n_gpus"] == 1 and d["steps"] == 1
+    assert d["higher_is_better"] is False and d["scaling"] == "weak"
+    assert d["value"] == d["ms_per_step"] > 0
+    cfg = d["config"]
+    for key in ("model", "global_batch", "parallelism", "population_size",
+                "num_generations", "archive_size", "final_hypervolume_ref11"):
+        assert key in cfg, key
+    assert "synthetic" in d["data"]
