@@ -179,7 +179,7 @@ class JointFTTransformer(nn.Module):
                 )
                 loss.backward()
                 opt.step()
-                total += float(loss) * len(idx)
+                total += loss.item() * len(idx)
             if verbose and (ep % 50 == 0):
                 logger.info(f"transformer epoch {ep}: loss {total / n:.5f}")
         self.eval()
