@@ -16,7 +16,6 @@ from __future__ import annotations
 
 import datetime
 import os
-import pickle
 import time
 from typing import Callable, Dict, List, Optional, Tuple
 
