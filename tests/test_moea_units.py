@@ -131,3 +131,33 @@ def test_batched_cholesky_update_high_psucc_branch():
     assert torch.allclose(A2 @ A2.transpose(-1, -2), C_want, atol=1e-8)
     eye = torch.eye(d, dtype=torch.float64).expand(K, d, d)
     assert torch.allclose(Ainv2 @ A2, eye, atol=1e-6)
+
+
+def test_population_diversity_and_adaptive_popsize():
+    """PopulationDiversity (front-0 fraction + crowding CV) and the NSGA2
+    adaptive-population-size rule driven by it."""
+    from dmosopt_amd.hv.indicators import PopulationDiversity
+    from dmosopt_amd.moea.nsga2 import NSGA2Optimizer
+
+    pd_ind = PopulationDiversity()
+    rank = np.array([0, 0, 1, 1, 2])
+    Y = np.random.default_rng(0).random((5, 2))
+    diversity, cd_spread = pd_ind.do(rank, Y)
+    assert diversity == pytest.approx(2 / 5)
+    assert np.isfinite(cd_spread)
+
+    rng = np.random.default_rng(1)
+    opt = NSGA2Optimizer(popsize=16, nInput=4, nOutput=2,
+                         adaptive_population_size=True)
+    bounds = np.column_stack([np.zeros(4), np.ones(4)])
+    x = rng.random((16, 4))
+    y = np.column_stack([x.sum(1), (1 - x).sum(1)])
+    opt.initialize_strategy(x, y, bounds, rng)
+    for _ in range(4):
+        x_gen, gs = opt.generate()
+        xg = x_gen.cpu().numpy() if hasattr(x_gen, "cpu") else x_gen
+        y_gen = np.column_stack([xg.sum(1), (1 - xg).sum(1)])
+        opt.update(xg, y_gen, gs)
+    p = opt.opt_params
+    assert p.min_population_size <= p.popsize <= p.max_population_size
+    assert p.poolsize == int(round(p.popsize / 2.0))
