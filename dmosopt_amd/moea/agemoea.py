@@ -292,7 +292,7 @@ class AGEMOEAOptimizer(MOEA):
         di_c = torch.as_tensor(p.di_crossover, dtype=pool.dtype, device=pool.device)
         di_m = torch.as_tensor(p.di_mutation, dtype=pool.dtype, device=pool.device)
         x_gen, _, _ = event_stream_variation(
-            pool, rng, p.popsize, p.poolsize, p.crossover_prob, p.mutation_prob,
+            pool, rng, p.popsize, pool.shape[0], p.crossover_prob, p.mutation_prob,
             p.mutation_rate, di_c, di_m, xlb, xub, torch_random=self.torch_random,
         )
         return x_gen, {}

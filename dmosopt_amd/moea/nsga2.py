@@ -123,6 +123,7 @@ class NSGA2Optimizer(MOEA):
         population = self.state.population_parm
         rank = self.state.rank
 
+        poolsize = min(poolsize, population.shape[0])
         pool = None
         if population.device.type == "cuda" and ops.native_available():
             from dmosopt_amd import _hipops
@@ -144,7 +145,7 @@ class NSGA2Optimizer(MOEA):
         from dmosopt_amd.moea.variation import event_stream_variation
 
         x_gen, crossover_indices, mutation_indices = event_stream_variation(
-            pool, rng, popsize, poolsize, p.crossover_prob, p.mutation_prob,
+            pool, rng, popsize, pool.shape[0], p.crossover_prob, p.mutation_prob,
             p.mutation_rate, di_c, di_m, xlb, xub, torch_random=self.torch_random,
         )
         self.state.total_crossovers += int(crossover_indices.shape[0]) // 2

@@ -230,6 +230,11 @@ def tournament_selection(
     replacement: argmax_k of log(p_i) + Gumbel noise); otherwise the host
     numpy Generator draws, matching the reference's control stream.
     """
+    # adaptive population sizing can briefly set poolsize above the CURRENT
+    # population (the reference's update_population_size jumps straight to
+    # min_population_size, NSGA2.py:268-270, and would crash its own
+    # replace=False draw) — clamp to what exists
+    poolsize = min(poolsize, pop)
     dev_metrics = [m if isinstance(m, torch.Tensor) else torch.as_tensor(m) for m in metrics]
     sorted_candidates = lexsort(dev_metrics)
     dev = sorted_candidates.device
