@@ -161,3 +161,28 @@ def test_population_diversity_and_adaptive_popsize():
     p = opt.opt_params
     assert p.min_population_size <= p.popsize <= p.max_population_size
     assert p.poolsize == int(round(p.popsize / 2.0))
+
+
+@pytest.mark.parametrize("name", ["nsga2", "age", "smpso", "cmaes", "trs"])
+@pytest.mark.parametrize("m", [3, 5])
+def test_many_objectives_duplicate_heavy(name, m):
+    """Every optimizer steps with 3/5 objectives and duplicate-heavy
+    populations (degenerate fronts, zero crowding spans)."""
+    from dmosopt_amd.config import optimizer_registry, resolve
+
+    rng = np.random.default_rng(0)
+    cls = resolve(optimizer_registry, name)
+    d, pop = 4, 12
+    opt = cls(popsize=pop, nInput=d, nOutput=m)
+    bounds = np.column_stack([np.zeros(d), np.ones(d)])
+    x = rng.random((pop, d))
+    x[: pop // 2] = x[0]
+    y = np.column_stack([x.sum(1) + j * 0.1 for j in range(m)])
+    opt.initialize_strategy(x, y, bounds, np.random.default_rng(1))
+    for _ in range(2):
+        xg, gs = opt.generate()
+        xgn = xg.cpu().numpy() if hasattr(xg, "cpu") else xg
+        yg = np.column_stack([xgn.sum(1) + j * 0.1 for j in range(m)])
+        opt.update(xgn, yg, gs)
+    px, py = opt.population_objectives
+    assert np.isfinite(py.cpu().numpy() if hasattr(py, "cpu") else py).all()
