@@ -387,8 +387,12 @@ class DistOptStrategy:
     # --------------------------------------------------------------- epochs
     def _epoch_kwargs(self, optimizer_index: int) -> Dict:
         optimizer_kwargs = {}
-        if self.optimizer_kwargs[optimizer_index] is not None:
-            optimizer_kwargs.update(self.optimizer_kwargs[optimizer_index])
+        # a single kwargs dict broadcasts across a cycled optimizer list
+        # (the reference indexes 1:1 and crashes on the mismatch,
+        # dmosopt.py:96-100 + epoch call site)
+        kw = self.optimizer_kwargs[optimizer_index % len(self.optimizer_kwargs)]
+        if kw is not None:
+            optimizer_kwargs.update(kw)
         if self.distance_metric is not None:
             optimizer_kwargs["distance_metric"] = self.distance_metric
         return dict(

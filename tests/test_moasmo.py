@@ -131,3 +131,18 @@ def test_seeded_runs_are_bit_identical():
     assert x1.shape == x2.shape
     assert np.array_equal(x1, x2), "parameter archives diverged"
     assert np.array_equal(y1, y2), "objective archives diverged"
+
+
+def test_optimizer_cycling_single_kwargs_broadcast():
+    """A cycled optimizer list with ONE kwargs dict must broadcast instead
+    of crashing (the reference indexes kwargs 1:1 and IndexErrors)."""
+    best = _run(
+        "t_cyc_bcast",
+        optimizer_name=["nsga2", "trs"],
+        optimizer_kwargs={"crossover_prob": 0.9, "mutation_prob": 0.1},
+        n_epochs=3,
+        num_generations=3,
+    )
+    bestx, besty = best
+    y = np.column_stack([v for _, v in besty])
+    assert np.isfinite(y).all()
