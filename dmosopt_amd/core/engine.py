@@ -199,10 +199,11 @@ def train(
     kwargs = dict(surrogate_method_kwargs)
     if device is not None:
         kwargs.setdefault("device", device)
-    return surrogate_cls(
-        x, y, nInput, nOutput, xlb, xub,
-        logger=logger, return_mean_variance=surrogate_return_mean_variance, **kwargs,
-    )
+    # a user-provided return_mean_variance in the kwargs wins over the
+    # engine-level flag (passing both is a reference API footgun:
+    # "got multiple values for keyword argument")
+    kwargs.setdefault("return_mean_variance", surrogate_return_mean_variance)
+    return surrogate_cls(x, y, nInput, nOutput, xlb, xub, logger=logger, **kwargs)
 
 
 def analyze_sensitivity(
