@@ -248,3 +248,26 @@ def test_callable_initial_method():
     best = dmosopt_amd.run(params, verbose=False)
     assert best is not None
     assert "shape" in seen and seen["shape"][1] == 4
+
+
+def test_termination_conditions_callable_factory():
+    """termination_conditions may be a callable(problem) -> Termination
+    (reference dmosopt.py:120-129); generation budget stops early."""
+    from dmosopt_amd.termination.basic import MaximumGenerationTermination
+
+    calls = [0]
+
+    def obj(pp):
+        calls[0] += 1
+        x = np.array([pp[k] for k in sorted(pp.keys())])
+        return np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+
+    params = _base(
+        "t_term_call", obj_fun=obj, num_generations=50, n_epochs=1,
+        surrogate_method_name=None,
+        termination_conditions=lambda prob: MaximumGenerationTermination(prob, 5),
+    )
+    dmosopt_amd.run(params, verbose=False)
+    # init (n_initial * dim = 8) + at most ~6 generations of <= popsize
+    assert calls[0] < 8 + 7 * 21, calls[0]
+    assert calls[0] > 8
