@@ -350,9 +350,10 @@ def run_epoch(
     stats["model_init_end"] = time.time()
     stats.update(mdl.get_stats())
 
+    optimizer_kwargs_.setdefault("distance_metric", None)
     optimizer = optimizer_cls(
         nInput=nInput, nOutput=nOutput, popsize=pop, model=mdl,
-        distance_metric=None, optimize_mean_variance=optimize_mean_variance,
+        optimize_mean_variance=optimize_mean_variance,
         **optimizer_kwargs_,
     )
     if device is not None:

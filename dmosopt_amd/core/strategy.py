@@ -77,6 +77,7 @@ class EpochRunner:
 
         optimizer_kwargs_ = {"sampling_method": "slh", "mutation_rate": None, "nchildren": 1}
         optimizer_kwargs_.update(kw.get("optimizer_kwargs") or {})
+        optimizer_kwargs_.setdefault("distance_metric", None)
         optimizer_cls = cfg.resolve(cfg.optimizer_registry, kw.get("optimizer_name", "nsga2"))
         from dmosopt_amd.models.model import Model
 
@@ -88,7 +89,6 @@ class EpochRunner:
             nOutput=nOutput,
             popsize=kw["pop"],
             model=self._mdl,
-            distance_metric=None,
             optimize_mean_variance=kw.get("optimize_mean_variance", False),
             **optimizer_kwargs_,
         )

