@@ -77,3 +77,18 @@ def test_constrained_with_feasibility_model():
     _, by = best
     y = np.column_stack([v for _, v in by])
     assert np.isfinite(y).all()
+
+
+@pytest.mark.parametrize("optimizer", ["age", "cmaes", "smpso", "trs"])
+def test_feasibility_model_with_every_optimizer(optimizer):
+    """The logreg feasibility model's rank() feeds each optimizer's
+    x-distance path without error."""
+    p = _params(
+        f"m_feas_{optimizer}",
+        obj_fun=_constrained,
+        constraint_names=["c1"],
+        feasibility_method_name="logreg",
+        optimizer=optimizer,
+        num_generations=3,
+    )
+    assert dmosopt_amd.run(p, verbose=False) is not None
