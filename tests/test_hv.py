@@ -163,3 +163,14 @@ def test_engine_get_feasible(rng):
     uniq_rank, rank_idx, rnk_cnt = rnk_arrs
     assert sum(rnk_cnt) == perm_x.shape[0]
     assert rnk_epc.shape == (len(uniq_rank), len(epc_arrs[0]))
+
+
+def test_box_decomposition_d6_vs_mc(rng):
+    """Six-objective exact HV (Lacour box decomposition) cross-checked
+    against a tight FPRAS Monte Carlo estimate."""
+    pts = rng.random((10, 6)) * 0.8
+    ref = np.ones(6) * 1.05
+    hv_box = HyperVolumeBoxDecomposition(ref).compute_hypervolume(pts)
+    est = hv_fpras(pts, ref, eps=0.02, seed=3, device="cpu")
+    assert hv_box > 0
+    assert est == pytest.approx(hv_box, rel=0.05)
