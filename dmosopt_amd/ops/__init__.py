@@ -291,12 +291,13 @@ def matern_cross_kernel(Xq, X, theta, nu, anisotropic):
 def chol_factor_batched(K):
     """Factor K (B,N,N) in place -> (L, logdet (B,), info (B,)).
 
-    Routing: the one-workgroup-per-matrix gfx950 kernel owns the batched
-    small/medium regime (and is the ONLY working path at N ~ 300, where
-    ROCm 7.2's batched rocSOLVER f32 cholesky raises launch failures); for
-    single large factorizations (N >= 1200, where rocSOLVER's multi-CU
-    decomposition wins ~10x and was verified working) torch dispatches to
-    rocSOLVER."""
+    Routing: the native launcher picks between the multi-launch
+    right-looking path (small batches; panel + MFMA SYRK tile launches)
+    and the one-workgroup-per-matrix kernel (B >= ~48) — either way the
+    ONLY working path at N ~ 300, where ROCm 7.2's batched rocSOLVER f32
+    cholesky raises launch failures. Single large factorizations
+    (N >= 1200, where rocSOLVER's multi-CU decomposition wins ~10x and
+    was verified working) dispatch to torch/rocSOLVER."""
     if _use_native(K) and K.dtype == torch.float32:
         if K.shape[1] < 1200:
             logdet, info = _native.cholesky_batched_(K)
