@@ -74,4 +74,10 @@ def resolve(registry: dict, name_or_path):
     if callable(name_or_path):
         return name_or_path
     path = registry.get(name_or_path, name_or_path)
-    return import_object_by_path(path)
+    try:
+        return import_object_by_path(path)
+    except (ValueError, ImportError, AttributeError) as e:
+        raise ValueError(
+            f"Cannot resolve {name_or_path!r}: not a registered shorthand "
+            f"({sorted(registry)}) and not an importable dotted path ({e})"
+        ) from e
