@@ -108,3 +108,22 @@ def test_two_rank_farm():
     # controller gets a best set; workers return None
     assert results[0] is True
     assert results[1] is False
+
+
+def test_four_rank_farm_oversubscribed():
+    """Four oversubscribed ranks on one box (the reference's mpi test runs
+    `mpirun --oversubscribe -n 4`): controller + 3 workers."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, 4, 29651, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, has_best = q.get(timeout=600)
+        results[rank] = has_best
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert results[0] is True
+    assert all(results[r] is False for r in (1, 2, 3))
