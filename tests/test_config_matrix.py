@@ -92,3 +92,10 @@ def test_feasibility_model_with_every_optimizer(optimizer):
         num_generations=3,
     )
     assert dmosopt_amd.run(p, verbose=False) is not None
+
+
+@pytest.mark.parametrize("method", ["glp", "slh", "lh", "mc", "sobol"])
+def test_every_initial_method_through_run(method):
+    p = _params(f"m_init_{method}", surrogate_method_name=None,
+                initial_method=method, num_generations=2)
+    assert dmosopt_amd.run(p, verbose=False) is not None
