@@ -53,13 +53,16 @@ def test_every_optimizer_with_gpr_surrogate(optimizer):
     assert np.isfinite(y).all()
 
 
-@pytest.mark.parametrize("surrogate", ["egp", "vgp", None])
+@pytest.mark.parametrize(
+    "surrogate",
+    ["egp", "vgp", "megp", "svgp", "spv", "siv", "crv", "mdgp", "mdspp", None],
+)
 def test_surrogate_variants_end_to_end(surrogate):
     over = {"surrogate_method_name": surrogate}
-    if surrogate == "egp":
-        over["surrogate_method_kwargs"] = {"n_iter": 30}
-    elif surrogate == "vgp":
-        over["surrogate_method_kwargs"] = {"n_iter": 30, "num_inducing": 16}
+    if surrogate == "vgp":
+        over["surrogate_method_kwargs"] = {"n_iter": 25, "num_inducing": 16}
+    elif surrogate is not None:
+        over["surrogate_method_kwargs"] = {"n_iter": 25}
     best = dmosopt_amd.run(_params(f"m_s_{surrogate}", **over), verbose=False)
     assert best is not None
 
