@@ -102,3 +102,11 @@ def test_every_initial_method_through_run(method):
     p = _params(f"m_init_{method}", surrogate_method_name=None,
                 initial_method=method, num_generations=2)
     assert dmosopt_amd.run(p, verbose=False) is not None
+
+
+@pytest.mark.parametrize("sa", ["dgsm", "fast"])
+def test_sensitivity_methods_through_run(sa):
+    """SA on the surrogate sets per-dimension di vectors for the inner
+    optimizer (reference MOASMO.py:329-361 wiring)."""
+    p = _params(f"m_sa_{sa}", sensitivity_method_name=sa, num_generations=3)
+    assert dmosopt_amd.run(p, verbose=False) is not None
