@@ -110,3 +110,19 @@ def test_sensitivity_methods_through_run(sa):
     optimizer (reference MOASMO.py:329-361 wiring)."""
     p = _params(f"m_sa_{sa}", sensitivity_method_name=sa, num_generations=3)
     assert dmosopt_amd.run(p, verbose=False) is not None
+
+
+def test_joint_transformer_custom_training_through_run():
+    """surrogate_custom_training='...transformer.joint' wires the
+    FT-Transformer in as objective + feasibility + sensitivity provider
+    (reference model_transformer.py:1112-1233 / MOASMO.py:267-295)."""
+    p = _params(
+        "m_joint",
+        obj_fun=_constrained,
+        constraint_names=["c1"],
+        surrogate_method_name=None,
+        surrogate_custom_training="dmosopt_amd.models.transformer.joint",
+        surrogate_custom_training_kwargs={"epochs": 30},
+        num_generations=3,
+    )
+    assert dmosopt_amd.run(p, verbose=False) is not None
