@@ -54,3 +54,16 @@ def test_cd2_known_property(rng):
     mc = sampling.mc(64, 4, np.random.default_rng(1))
     lh = sampling.lh(64, 4, np.random.default_rng(1))
     assert sampling.cd2(lh) < sampling.cd2(mc)
+
+
+def test_samplers_deterministic_under_seed():
+    """Same seed -> bit-identical designs for every sampler (required by
+    the replicated-rank scheme: all ranks draw the same initial design)."""
+    from dmosopt_amd import sampling
+
+    for name in ("glp", "slh", "lh", "mc", "sobol"):
+        fn = getattr(sampling, name)
+        a = fn(33, 5, np.random.default_rng(42))
+        b = fn(33, 5, np.random.default_rng(42))
+        assert np.array_equal(a, b), name
+        assert a.shape == (33, 5) and a.min() >= 0 and a.max() <= 1, name
