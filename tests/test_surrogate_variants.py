@@ -52,3 +52,17 @@ def test_mean_variance_mode():
     m = cls(X, Y, 5, 2, np.zeros(5), np.ones(5), seed=1, return_mean_variance=True)
     out = m.evaluate(X[:5])
     assert isinstance(out, tuple) and len(out) == 2
+
+
+def test_feasibility_model_discriminates():
+    """logreg feasibility: P(feasible) separates clearly on a linearly
+    separable constraint (reference feasibility.py:14-67 pipeline role)."""
+    from dmosopt_amd.models.feasibility import LogisticFeasibilityModel
+
+    rng = np.random.default_rng(0)
+    X = rng.random((120, 4))
+    C = np.column_stack([X[:, 0] - 0.5])
+    fsbm = LogisticFeasibilityModel(X, C)
+    Xhi = rng.random((20, 4)); Xhi[:, 0] = 0.9
+    Xlo = rng.random((20, 4)); Xlo[:, 0] = 0.1
+    assert float(np.mean(fsbm.rank(Xhi))) > float(np.mean(fsbm.rank(Xlo))) + 0.5
