@@ -66,3 +66,19 @@ def test_feasibility_model_discriminates():
     Xhi = rng.random((20, 4)); Xhi[:, 0] = 0.9
     Xlo = rng.random((20, 4)); Xlo[:, 0] = 0.1
     assert float(np.mean(fsbm.rank(Xhi))) > float(np.mean(fsbm.rank(Xlo))) + 0.5
+
+
+def test_sensitivity_methods_identify_influential_dims():
+    """DGSM and FAST attribute sensitivity to the right inputs on a
+    separable model (reference sa.py role: S1 per objective)."""
+    from dmosopt_amd.models.sa import SA_DGSM, SA_FAST
+
+    class M:
+        def evaluate(self, x):
+            x = np.atleast_2d(x)
+            return np.column_stack([3.0 * x[:, 0], 2.0 * x[:, 1]])
+
+    for cls in (SA_DGSM, SA_FAST):
+        sa = cls(np.zeros(4), np.ones(4), [f"x{i}" for i in range(4)], ["f1", "f2"])
+        S1 = sa.analyze(M(), 2048)["S1"]
+        assert np.argmax(S1["f1"]) == 0 and np.argmax(S1["f2"]) == 1, (cls, S1)
