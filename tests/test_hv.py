@@ -174,3 +174,19 @@ def test_box_decomposition_d6_vs_mc(rng):
     est = hv_fpras(pts, ref, eps=0.02, seed=3, device="cpu")
     assert hv_box > 0
     assert est == pytest.approx(hv_box, rel=0.05)
+
+
+def test_igd_and_hv_indicator_values():
+    """IGD = mean min-distance front->pf (0 on the pf itself, sqrt(2)*shift
+    under a diagonal shift); Hypervolume indicator matches the exact value."""
+    from dmosopt_amd.hv.indicators import IGD, Hypervolume
+
+    pf = np.array([[0.0, 1.0], [0.5, 0.5], [1.0, 0.0]])
+    ind = IGD(pf)
+    assert ind.do(pf) == pytest.approx(0.0)
+    assert ind.do(pf + 0.1) == pytest.approx(0.1 * np.sqrt(2))
+    assert ind.do(pf + 0.3) == pytest.approx(0.3 * np.sqrt(2))
+
+    hv = Hypervolume(ref_point=np.array([2.0, 2.0]))
+    # staircase: 1x2 + 1.5x1.5 strip + 2x1 union = 3.25
+    assert hv.do(pf) == pytest.approx(3.25)
