@@ -188,3 +188,17 @@ def test_bench_json_contract():
                 "num_generations", "archive_size", "final_hypervolume_ref11"):
         assert key in cfg, key
     assert "synthetic" in d["data"]
+
+
+@pytest.mark.parametrize("mod", ["analyze", "train", "onestep"])
+def test_cli_help_runs(mod):
+    import subprocess
+    import sys as _sys
+
+    out = subprocess.run(
+        [_sys.executable, "-m", f"dmosopt_amd.cli.{mod}", "--help"],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    )
+    assert out.returncode == 0
+    assert "--file-path" in out.stdout or "-p" in out.stdout
