@@ -9,7 +9,7 @@ import os
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from setuptools import setup
+from setuptools import find_packages, setup
 from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
@@ -51,7 +51,11 @@ h5ext = CppExtension(
 setup(
     name="dmosopt_amd",
     version="0.1.0",
-    packages=["dmosopt_amd"],
+    description=(
+        "MI355X-native multi-objective adaptive surrogate optimization "
+        "(MO-ASMO) with gfx950 HIP kernels"
+    ),
+    packages=find_packages(include=["dmosopt_amd", "dmosopt_amd.*"]),
     ext_modules=[ext, h5ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
     entry_points={
