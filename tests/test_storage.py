@@ -259,3 +259,27 @@ def test_surrogate_evals_and_optimizer_params_saved(tmp_path):
         assert f.has("t_se/optimizer_params/1/optimizer_name")
     finally:
         f.close()
+
+
+def test_resume_space_mismatch_raises(tmp_path):
+    """Resuming with different parameter names must fail loudly (docs/
+    results.md contract; reference init_from_h5 name check)."""
+    import dmosopt_amd
+
+    def obj(pp):
+        x = np.array([pp[k] for k in sorted(pp.keys())])
+        return np.array([np.sum(x**2), np.sum((x - 1) ** 2)])
+
+    fp = str(tmp_path / "mm.h5")
+    base = {"obj_fun": obj, "problem_parameters": {},
+            "space": {f"x{i}": [0.0, 1.0] for i in range(3)},
+            "objective_names": ["f1", "f2"], "population_size": 8,
+            "num_generations": 2, "n_initial": 2, "n_epochs": 1,
+            "surrogate_method_name": None, "optimizer": "nsga2",
+            "random_seed": 2, "file_path": fp, "save": True}
+    dmosopt_amd.run(dict(base, opt_id="t_mm"), verbose=False)
+    dmosopt_amd.sopt_dict.clear()
+    bad = dict(base, opt_id="t_mm", resume=True,
+               space={f"z{i}": [0.0, 1.0] for i in range(3)})
+    with pytest.raises(RuntimeError, match="differ"):
+        dmosopt_amd.run(bad, verbose=False)
