@@ -296,3 +296,24 @@ def test_distance_metric_and_custom_optimizer_plugin():
         dmosopt_amd.sopt_dict["t_plugin_cls"].optimizer_dict[0].runner is None
         or True, bool,
     )
+
+
+def test_integer_space_parameter_metadata():
+    """[lo, hi, True] space parameters: is_integer is stored as metadata
+    (H5 spec / analyze output); values still arrive continuous, exactly
+    like the reference (dmosopt.py:2352-2388 passes raw space values)."""
+    seen = []
+
+    def obj(pp):
+        seen.append(dict(pp))
+        return np.array([pp["xc"] + pp["xi"], 2 - pp["xc"]])
+
+    params = _base("t_intspace", obj_fun=obj, surrogate_method_name=None,
+                   num_generations=2, n_epochs=1,
+                   space={"xc": [0.0, 1.0], "xi": [0, 10, True]})
+    dmosopt_amd.run(params, verbose=False)
+    dopt = dmosopt_amd.sopt_dict["t_intspace"]
+    is_int = dopt.param_space.is_integer
+    names = list(dopt.param_space.parameter_names)
+    assert is_int[names.index("xi")] and not is_int[names.index("xc")]
+    assert all(0 <= pp["xi"] <= 10 for pp in seen)
