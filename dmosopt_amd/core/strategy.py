@@ -293,8 +293,17 @@ class DistOptStrategy:
         return self.reqs.pop(0) if self.reqs else None
 
     def complete_request(self, x, y, epoch=None, f=None, c=None, pred=None, time=-1.0):
-        assert x.shape[0] == self.prob.dim
-        assert y.shape[0] == self.prob.n_objectives
+        if x.shape[0] != self.prob.dim:
+            raise ValueError(
+                f"evaluation returned {x.shape[0]} parameters; the space has "
+                f"{self.prob.dim}"
+            )
+        if y.shape[0] != self.prob.n_objectives:
+            raise ValueError(
+                f"objective function returned {y.shape[0]} values but "
+                f"{self.prob.n_objectives} objective_names are declared "
+                f"({self.prob.objective_names})"
+            )
         if self.optimize_mean_variance and pred is not None:
             if pred.shape[0] == self.prob.n_objectives:
                 pred = np.column_stack((pred, np.zeros_like(pred)))
