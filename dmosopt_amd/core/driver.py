@@ -374,48 +374,74 @@ class DistOptimizer:
             self.print_best()
 
     # ------------------------------------------------------------- requests
+    def _complete_one(self, problem_id, req, res):
+        """Apply reduce_fun, slice this problem's entry out of the shared
+        result dict, and complete the strategy request."""
+        strategy = self.optimizer_dict[problem_id]
+        if self.reduce_fun is not None:
+            args = self.reduce_fun_args or ()
+            res = self.reduce_fun([res], *args)
+        entry = res[problem_id] if isinstance(res, dict) else res
+        t_eval = res.get("time", -1.0) if isinstance(res, dict) else -1.0
+        y, f, c = _split_result(
+            entry, self.objective_names, self.feature_names, self.constraint_names
+        )
+        strategy.complete_request(
+            req.parameters, y, epoch=req.epoch, f=f, c=c,
+            pred=req.prediction, time=t_eval,
+        )
+        self.storage_dict[problem_id].append(strategy.completed[-1])
+        self.eval_count += 1
+        if self.verbose:
+            self.logger.info(f"problem {problem_id}: eval {self.eval_count}: y = {y}")
+
     def _process_requests(self):
         """Drain all strategies' request queues through the farm as batched
-        collective evaluations; complete the requests with the results."""
-        for problem_id in self.problem_ids:
-            strategy = self.optimizer_dict[problem_id]
+        collective evaluations; complete the requests with the results.
+
+        Multi-problem dispatch zips ONE request per problem id into a single
+        farm point ({pid: its own parameters}) so the user objective runs
+        once per zipped point, completing each problem's request from its
+        slice of the shared result (reference dmosopt.py:1291-1313). Any
+        ragged tail (a problem with more queued requests than its siblings)
+        is dispatched per-problem with its own parameters in every slot —
+        progress is guaranteed, at the cost of redundant sibling entries for
+        those tail points only."""
+        if not self.has_problem_ids:
+            strategy = self.optimizer_dict[0]
             reqs = []
             while strategy.has_requests():
                 reqs.append(strategy.get_next_request())
-            if not reqs:
-                continue
-            points = [
-                {pid: r.parameters for pid in self.problem_ids}
-                if self.has_problem_ids
-                else {0: r.parameters}
-                for r in reqs
-            ]
-            results = self.farm.evaluate(self.opt_id, points)
-            for req, res in zip(reqs, results):
-                if self.reduce_fun is not None:
-                    args = self.reduce_fun_args or ()
-                    res = self.reduce_fun([res], *args)
-                entry = res[problem_id] if isinstance(res, dict) else res
-                t_eval = res.get("time", -1.0) if isinstance(res, dict) else -1.0
-                y, f, c = _split_result(
-                    entry, self.objective_names, self.feature_names, self.constraint_names
-                )
-                strategy.complete_request(
-                    req.parameters, y, epoch=req.epoch, f=f, c=c,
-                    pred=req.prediction, time=t_eval,
-                )
-                self.storage_dict[problem_id].append(strategy.completed[-1])
-                self.eval_count += 1
-                if self.verbose:
-                    self.logger.info(
-                        f"problem {problem_id}: eval {self.eval_count}: y = {y}"
-                    )
-            if (
-                self.save
-                and (self.eval_count - self.saved_eval_count) >= self.save_eval
-            ):
-                self.save_evals()
-                self.saved_eval_count = self.eval_count
+            if reqs:
+                points = [{0: r.parameters} for r in reqs]
+                results = self.farm.evaluate(self.opt_id, points)
+                for req, res in zip(reqs, results):
+                    self._complete_one(0, req, res)
+        else:
+            pids = list(self.problem_ids)
+            strategies = {pid: self.optimizer_dict[pid] for pid in pids}
+            zipped = []
+            while all(s.has_requests() for s in strategies.values()):
+                zipped.append({pid: strategies[pid].get_next_request() for pid in pids})
+            if zipped:
+                points = [{pid: rd[pid].parameters for pid in pids} for rd in zipped]
+                results = self.farm.evaluate(self.opt_id, points)
+                for rd, res in zip(zipped, results):
+                    for pid in pids:
+                        self._complete_one(pid, rd[pid], res)
+            for pid in pids:
+                strategy = strategies[pid]
+                tail = []
+                while strategy.has_requests():
+                    tail.append(strategy.get_next_request())
+                if tail:
+                    points = [{q: r.parameters for q in pids} for r in tail]
+                    results = self.farm.evaluate(self.opt_id, points)
+                    for req, res in zip(tail, results):
+                        self._complete_one(pid, req, res)
+        if self.save and (self.eval_count - self.saved_eval_count) >= self.save_eval:
+            self.save_evals()
+            self.saved_eval_count = self.eval_count
         # flush the tail below the cadence (reference dmosopt.py:1329-1335:
         # nothing unsaved survives the end of a request pump)
         if self.save and 0 < self.saved_eval_count < self.eval_count or (
@@ -474,14 +500,21 @@ class DistOptimizer:
             strategy.initialize_epoch(epoch)
         self.stats["init_sampling_end"] = time.time()
 
-        while not completed_epoch:
+        # per-problem completion tracking: with per-problem termination one
+        # strategy can finish its epoch before its siblings; a completed
+        # strategy must not see another update_epoch call (its runner is
+        # already cleared by _finish_epoch)
+        done = set(self.problem_ids) if completed_epoch else set()
+        while len(done) < len(self.problem_ids):
             self._process_requests()
             for problem_id in self.problem_ids:
+                if problem_id in done:
+                    continue
                 state, value, completed_evals = self.optimizer_dict[problem_id].update_epoch(
                     resample=advance_epoch
                 )
-                completed_epoch = state == StrategyState.CompletedEpoch
-                if completed_epoch:
+                if state == StrategyState.CompletedEpoch:
+                    done.add(problem_id)
                     res = value
                     if completed_evals is not None and epoch > 1:
                         self._report_surrogate_accuracy(problem_id, epoch, completed_evals)

@@ -128,6 +128,26 @@ def euclidean_distance_metric(Y: torch.Tensor) -> torch.Tensor:
     return torch_ref.euclidean_distance_metric(Y)
 
 
+def fused_rank_metric_perm(rank: torch.Tensor, metric: torch.Tensor) -> torch.Tensor:
+    """Fused sort key for the common (rank, -metric) case: one stable argsort
+    on (rank << 32 | descending_monotone_bits(metric)) instead of two radix
+    sorts + gathers per generation. The IEEE sign-flip mapping
+    (b | 0x80000000 for b >= 0, ~b for b < 0) is a total-order bijection
+    float32 -> uint32 that handles NEGATIVE metric values too (user-supplied
+    callables may return them); ties fall back to index order exactly like
+    np.lexsort."""
+    d = torch.nan_to_num(
+        metric.float(),
+        nan=0.0,
+        posinf=float(torch.finfo(torch.float32).max),
+        neginf=float(torch.finfo(torch.float32).min),
+    )
+    b = d.view(torch.int32).to(torch.int64) & 0xFFFFFFFF  # raw bits as u32
+    asc = torch.where(b >= 0x80000000, (~b) & 0xFFFFFFFF, b | 0x80000000)
+    key = (rank.to(torch.int64) << 32) | ((0xFFFFFFFF - asc) & 0xFFFFFFFF)
+    return torch.argsort(key, stable=True)
+
+
 def order_mo(
     x: torch.Tensor,
     y: torch.Tensor,
@@ -153,17 +173,7 @@ def order_mo(
         and len(y_dist_vals) == 1
         and y.device.type == "cuda"
     ):
-        # fused sort key for the common (rank, -crowding) case: one stable
-        # argsort on (rank << 32 | ~monotone_float32_bits(crowding))
-        # instead of two radix sorts + gathers per generation. Non-negative
-        # IEEE floats compare like their bit patterns, so descending
-        # crowding == ascending complemented bits; ties fall back to index
-        # order exactly like np.lexsort.
-        d = y_dist_vals[0].float().clamp_min(0.0)
-        d = torch.nan_to_num(d, nan=0.0, posinf=float(torch.finfo(torch.float32).max))
-        bits = d.view(torch.int32).to(torch.int64)
-        key = (rank.to(torch.int64) << 32) | ((0x7FFFFFFF - bits) & 0xFFFFFFFF)
-        perm = torch.argsort(key, stable=True)
+        perm = fused_rank_metric_perm(rank, y_dist_vals[0])
         return perm, rank[perm], (y_dist_vals[0][perm],)
     keys = [-d for d in x_dist_vals] + [-d for d in y_dist_vals] + [rank.to(y.dtype)]
     perm = lexsort(keys)

@@ -287,8 +287,10 @@ def h5_init_opt_group(
     pids = np.asarray(sorted(problem_ids) if has_problem_ids else [0], dtype=np.int32)
     f.write_simple(f"{g}/problem_ids", "i4", pids.tobytes(), [len(pids)])
     if random_seed is not None:
-        rs = np.asarray([random_seed], dtype=np.int32)
-        f.write_simple(f"{g}/random_seed", "i4", rs.tobytes(), [1])
+        # int64: h5py stores the full Python int; an i4 would overflow for
+        # seeds >= 2**31 and break resume reproducibility
+        rs = np.asarray([random_seed], dtype=np.int64)
+        f.write_simple(f"{g}/random_seed", "i8", rs.tobytes(), [1])
 
 
 def init_h5(
@@ -545,7 +547,9 @@ def h5_load_raw(input_file, opt_id):
 
         random_seed = None
         if f.has(f"{g}/random_seed"):
-            random_seed = int(np.frombuffer(f.read_rows(f"{g}/random_seed"), dtype=np.int32)[0])
+            raw = f.read_rows(f"{g}/random_seed")
+            dt = np.int64 if len(raw) == 8 else np.int32  # i4 = pre-0.2 files
+            random_seed = int(np.frombuffer(raw, dtype=dt)[0])
 
         info = {
             "random_seed": random_seed,

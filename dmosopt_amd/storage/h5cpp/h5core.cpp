@@ -216,6 +216,11 @@ struct H5File {
     H5Sselect_hyperslab(fspace, H5S_SELECT_SET, start, nullptr, count, nullptr);
     hid_t mspace = H5Screate_simple(1, count, nullptr);
     std::string buf = data;
+    if (buf.size() != (size_t)n_rows * H5Tget_size(t))
+      throw std::runtime_error(
+          "append_rows: buffer size " + std::to_string(buf.size()) +
+          " != n_rows * type_size " +
+          std::to_string((size_t)n_rows * H5Tget_size(t)) + " for " + path);
     check(H5Dwrite(d, t, mspace, fspace, H5P_DEFAULT, buf.data()), "dwrite");
     H5Sclose(mspace);
     H5Sclose(fspace);
@@ -232,6 +237,11 @@ struct H5File {
     hid_t fspace = H5Dget_space(d);
     hid_t mspace = H5Screate_simple(1, dims, nullptr);
     std::string buf = data;
+    if (buf.size() != (size_t)n_rows * H5Tget_size(t))
+      throw std::runtime_error(
+          "write_rows: buffer size " + std::to_string(buf.size()) +
+          " != n_rows * type_size " +
+          std::to_string((size_t)n_rows * H5Tget_size(t)) + " for " + path);
     check(H5Dwrite(d, t, mspace, fspace, H5P_DEFAULT, buf.data()), "dwrite");
     H5Sclose(mspace);
     H5Sclose(fspace);
