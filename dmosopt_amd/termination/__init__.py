@@ -6,6 +6,8 @@ from dmosopt_amd.termination.basic import (  # noqa: F401
     SlidingWindowTermination,
     Termination,
     TerminationCollection,
+    WindowedCriterion,
+    Windows,
 )
 from dmosopt_amd.termination.adaptive import (  # noqa: F401
     AdaptiveWindowTermination,

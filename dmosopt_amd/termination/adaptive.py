@@ -1,11 +1,15 @@
-"""Adaptive termination criteria (parity with reference
-adaptive_termination.py:48-627)."""
+"""Adaptive stopping criteria.
+
+Role parity with the reference's ``adaptive_termination.py`` (SURVEY.md
+section 2.7) on the same decision semantics, built on the
+capture/reduce/verdict window engine of :mod:`.basic` with vectorized
+per-objective / per-timescale state (numpy arrays instead of per-objective
+state objects).
+"""
 
 from __future__ import annotations
 
 import time as _time
-from collections import deque
-from dataclasses import dataclass
 from typing import List, Optional
 
 import numpy as np
@@ -13,25 +17,33 @@ import numpy as np
 from dmosopt_amd.hv.indicators import crowding_distance_metric
 from dmosopt_amd.termination.basic import (
     MaximumGenerationTermination,
-    SlidingWindowTermination,
     Termination,
     TerminationCollection,
+    WindowedCriterion,
+    Windows,
     _log,
 )
 from dmosopt_amd.termination.hv_progress import HypervolumeProgressTermination
 
-
-@dataclass
-class ConvergenceState:
-    values: deque
-    converged: bool = False
-    stagnation_count: int = 0
-    improvement_rate: float = 0.0
+#: consecutive sub-tolerance windows before an objective counts as settled
+_SETTLE_STREAK = 3
 
 
-class PerObjectiveConvergence(SlidingWindowTermination):
-    """Per-objective delta-ideal stagnation counters; terminate when a
-    fraction of objectives have converged (adaptive_termination.py:48-158)."""
+def _norm_span(F: np.ndarray) -> np.ndarray:
+    span = F.max(axis=0) - F.min(axis=0)
+    span[span < 1e-32] = 1.0
+    return span
+
+
+class PerObjectiveConvergence(WindowedCriterion):
+    """Track each objective's normalized ideal-point movement separately and
+    stop once a fraction of the objectives have settled.
+
+    An objective settles after ``_SETTLE_STREAK`` consecutive windows whose
+    mean movement is below ``obj_tol``; any louder window unsettles it. All
+    per-objective state is vectorized: a rolling (n_last, m) delta matrix
+    plus integer streak counters.
+    """
 
     def __init__(
         self,
@@ -42,77 +54,58 @@ class PerObjectiveConvergence(SlidingWindowTermination):
         nth_gen: int = 5,
         n_max_gen: Optional[int] = None,
         min_generations: int = 0,
-        **kwargs,
+        **kw,
     ):
         super().__init__(
             problem,
-            metric_window_size=n_last,
-            data_window_size=2,
-            min_data_for_metric=2,
-            nth_gen=nth_gen,
-            n_max_gen=n_max_gen,
-            min_generations=min_generations,
-            **kwargs,
+            Windows(raw=2, signal=n_last, warmup=2, cadence=nth_gen,
+                    ceiling=n_max_gen, floor=min_generations, **kw),
         )
-        self.n_objectives = problem.n_objectives
+        m = problem.n_objectives
         self.obj_tol = obj_tol
         self.min_converged_fraction = min_converged_fraction
-        self.objective_states = [
-            ConvergenceState(values=deque(maxlen=n_last)) for _ in range(self.n_objectives)
-        ]
+        self._deltas = np.empty((0, m))  # rolling per-objective movement rows
+        self._streaks = np.zeros(m, dtype=np.int64)
+        self._settled = np.zeros(m, dtype=bool)
+        self._last_rates = np.zeros(m)
 
-    def _store(self, opt):
-        F = opt.y
-        return {"ideal": F.min(axis=0), "nadir": F.max(axis=0), "F": F}
+    def capture(self, opt):
+        return np.asarray(opt.y)
 
-    def _metric(self, data):
-        last, current = data[-2], data[-1]
-        norm = current["nadir"] - current["ideal"]
-        norm[norm < 1e-32] = 1.0
-        delta_ideal = np.abs(current["ideal"] - last["ideal"]) / norm
-        for i, delta in enumerate(delta_ideal[: self.n_objectives]):
-            st = self.objective_states[i]
-            st.values.append(delta)
-            if len(st.values) >= self.metric_window_size:
-                mean_change = np.mean(st.values)
-                st.improvement_rate = mean_change
-                if mean_change < self.obj_tol:
-                    st.stagnation_count += 1
-                    if st.stagnation_count >= 3:
-                        st.converged = True
-                else:
-                    st.stagnation_count = 0
-                    st.converged = False
-        return {
-            "delta_ideal": delta_ideal,
-            "converged_objectives": sum(s.converged for s in self.objective_states),
-            "mean_improvement": np.mean(
-                [s.improvement_rate for s in self.objective_states]
-            ),
-        }
-
-    def _decide(self, metrics):
-        latest = metrics[-1]
-        n_conv = latest["converged_objectives"]
-        frac = n_conv / self.n_objectives
-        if frac >= self.min_converged_fraction:
-            _log(
-                self.problem,
-                f"Optimization terminated: {n_conv}/{self.n_objectives} objectives "
-                f"({frac:.1%}) converged",
+    def reduce(self, raws):
+        prev_F, curr_F = raws[-2], raws[-1]
+        m = self._settled.shape[0]
+        move = np.abs(curr_F.min(axis=0) - prev_F.min(axis=0)) / _norm_span(curr_F)
+        move = move[:m]
+        n_last = self.windows.signal
+        self._deltas = np.vstack([self._deltas, move[None, :]])[-n_last:]
+        if self._deltas.shape[0] >= n_last:
+            rates = self._deltas.mean(axis=0)
+            self._last_rates = rates
+            quiet = rates < self.obj_tol
+            self._streaks = np.where(quiet, self._streaks + 1, 0)
+            self._settled = np.where(
+                quiet, self._streaks >= _SETTLE_STREAK, False
             )
-            return False
-        _log(
-            self.problem,
-            f"Convergence progress: {n_conv}/{self.n_objectives} converged "
-            f"({frac:.1%}), mean improvement {latest['mean_improvement']:.2e}",
-        )
-        return True
+        return move
+
+    def verdict(self, signals):
+        m = self._settled.shape[0]
+        n_done = int(self._settled.sum())
+        share = n_done / m
+        if share >= self.min_converged_fraction:
+            _log(self.problem,
+                 f"stop: {n_done} of {m} objectives settled ({share:.0%})")
+            return True
+        _log(self.problem,
+             f"settling: {n_done}/{m} objectives ({share:.0%}), "
+             f"mean movement {float(self._last_rates.mean()):.2e}")
+        return False
 
 
-class MultiScaleStagnationTermination(SlidingWindowTermination):
-    """Delta-ideal + diversity stagnation at multiple timescales
-    (adaptive_termination.py:161-281)."""
+class MultiScaleStagnationTermination(WindowedCriterion):
+    """Compare the current ideal point / diversity against lagged snapshots
+    at several timescales at once; stop when enough scales are quiet."""
 
     def __init__(
         self,
@@ -123,71 +116,54 @@ class MultiScaleStagnationTermination(SlidingWindowTermination):
         n_max_gen: Optional[int] = None,
         nth_gen: int = 1,
         min_generations: int = 0,
-        **kwargs,
+        **kw,
     ):
-        max_scale = max(timescales)
+        horizon = max(timescales)
         super().__init__(
             problem,
-            metric_window_size=max_scale,
-            data_window_size=max_scale,
-            min_data_for_metric=max(timescales),
-            nth_gen=nth_gen,
-            n_max_gen=n_max_gen,
-            min_generations=min_generations,
-            **kwargs,
+            Windows(raw=horizon + 1, signal=horizon, warmup=2, cadence=nth_gen,
+                    ceiling=n_max_gen, floor=min_generations, **kw),
         )
-        self.timescales = sorted(timescales)
+        self.timescales = np.asarray(sorted(timescales))
         self.stagnation_tol = stagnation_tol
         self.min_scales_stagnant = min_scales_stagnant
 
-    def _store(self, opt):
-        F = opt.y
-        cd = crowding_distance_metric(F)
+    def capture(self, opt):
+        F = np.asarray(opt.y)
         return {
             "ideal": F.min(axis=0),
-            "nadir": F.max(axis=0),
-            "diversity": float(np.mean(cd)),
-            "F": F,
-            "X": opt.x,
+            "span": _norm_span(F),
+            "spread": float(np.mean(crowding_distance_metric(F))),
         }
 
-    def _metric(self, data):
-        if len(data) < 2:
+    def reduce(self, raws):
+        now = raws[-1]
+        depth = len(raws) - 1
+        lags = self.timescales[self.timescales <= depth]
+        if lags.size == 0:
             return None
-        current = data[-1]
-        out = {}
-        for scale in self.timescales:
-            if len(data) >= scale + 1:
-                past = data[-(scale + 1)]
-                norm = current["nadir"] - current["ideal"]
-                norm[norm < 1e-32] = 1.0
-                mean_delta = float(
-                    np.mean(np.abs(current["ideal"] - past["ideal"]) / norm)
-                )
-                out[scale] = {
-                    "ideal_change": mean_delta,
-                    "diversity_change": abs(current["diversity"] - past["diversity"]),
-                    "stagnant": mean_delta < self.stagnation_tol,
-                }
-        return out
+        # vectorized over available scales: lagged ideal points as a matrix
+        past_ideals = np.stack([raws[-(int(s) + 1)]["ideal"] for s in lags])
+        moves = np.mean(np.abs(now["ideal"][None, :] - past_ideals) / now["span"], axis=1)
+        return {"scales": lags, "moves": moves}
 
-    def _decide(self, metrics):
-        latest = metrics[-1]
-        if latest is None:
+    def verdict(self, signals):
+        sig = signals[-1]
+        if sig is None or sig["scales"].size < self.timescales.size:
+            return False  # wait until every timescale has lag data
+        quiet = sig["moves"] < self.stagnation_tol
+        if int(quiet.sum()) >= self.min_scales_stagnant:
+            _log(self.problem,
+                 f"stop: {int(quiet.sum())} of {len(self.timescales)} timescales "
+                 f"quiet (scales {sig['scales'][quiet].tolist()})")
             return True
-        stagnant = [s for s, info in latest.items() if info["stagnant"]]
-        if len(stagnant) >= self.min_scales_stagnant:
-            _log(
-                self.problem,
-                f"Optimization terminated: {len(stagnant)}/{len(self.timescales)} "
-                f"timescales stagnant (scales {stagnant})",
-            )
-            return False
-        return True
+        return False
 
 
-class AdaptiveWindowTermination(SlidingWindowTermination):
-    """Expanding patience window (adaptive_termination.py:284-368)."""
+class AdaptiveWindowTermination(WindowedCriterion):
+    """Patience window that grows while progress is still loud: judged over
+    ``current`` signals, and expanded by ``expansion_rate`` (up to
+    ``max_window``) whenever the window mean exceeds 10x tolerance."""
 
     def __init__(
         self,
@@ -197,60 +173,51 @@ class AdaptiveWindowTermination(SlidingWindowTermination):
         expansion_rate: float = 1.2,
         tol: float = 1e-4,
         n_max_gen: Optional[int] = None,
-        **kwargs,
+        **kw,
     ):
         super().__init__(
             problem,
-            metric_window_size=initial_window,
-            data_window_size=2,
-            min_data_for_metric=2,
-            nth_gen=1,
-            n_max_gen=n_max_gen,
-            **kwargs,
+            Windows(raw=2, signal=initial_window, warmup=2, cadence=1,
+                    ceiling=n_max_gen, floor=kw.pop("min_generations", 0), **kw),
         )
-        self.initial_window = initial_window
         self.max_window = max_window
         self.expansion_rate = expansion_rate
         self.tol = tol
-        self.current_window_size = initial_window
+        # the signal ring must be able to hold the fully-expanded window
+        self._signal_ring.size = max_window
 
-    def _store(self, opt):
-        F = opt.y
-        return {"ideal": F.min(axis=0), "nadir": F.max(axis=0)}
+    @property
+    def current_window_size(self) -> int:
+        return self.windows.signal
 
-    def _metric(self, data):
-        last, current = data[-2], data[-1]
-        norm = current["nadir"] - current["ideal"]
-        norm[norm < 1e-32] = 1.0
-        delta = float(np.mean(np.abs(current["ideal"] - last["ideal"]) / norm))
-        return {"delta": delta, "window_size": self.current_window_size}
+    def reduce(self, raws):
+        prev_F, curr_F = np.asarray(raws[-2].y), np.asarray(raws[-1].y)
+        return float(
+            np.mean(np.abs(curr_F.min(axis=0) - prev_F.min(axis=0)) / _norm_span(curr_F))
+        )
 
-    def _decide(self, metrics):
-        if len(metrics) < self.current_window_size:
-            return True
-        recent = [m["delta"] for m in metrics[-self.current_window_size :]]
-        mean_delta = np.mean(recent)
-        if mean_delta > self.tol * 10:
-            new_window = min(
-                int(self.current_window_size * self.expansion_rate), self.max_window
-            )
-            if new_window > self.current_window_size:
-                self.current_window_size = new_window
-                self.metric_window_size = new_window
-                _log(self.problem, f"Expanding patience window to {new_window}")
-        if mean_delta < self.tol:
-            _log(
-                self.problem,
-                f"Optimization terminated: mean change {mean_delta:.2e} below "
-                f"{self.tol:.2e} over {self.current_window_size} generations",
-            )
+    def verdict(self, signals):
+        width = self.windows.signal
+        if len(signals) < width:
             return False
-        return True
+        level = float(np.mean(signals[-width:]))
+        if level > self.tol * 10:
+            wider = min(int(width * self.expansion_rate), self.max_window)
+            if wider > width:
+                self.windows.signal = wider
+                _log(self.problem, f"patience window widened to {wider}")
+        if level < self.tol:
+            _log(self.problem,
+                 f"stop: movement {level:.2e} below {self.tol:.2e} "
+                 f"over the last {width} generations")
+            return True
+        return False
 
 
 class CompositeAdaptiveTermination(TerminationCollection):
-    """Combination of criteria for high-dimensional problems
-    (adaptive_termination.py:371-472)."""
+    """Any-of bundle for high-dimensional runs: generation cap plus the
+    per-objective, hypervolume-progress and multi-scale criteria, each
+    individually switchable."""
 
     def __init__(
         self,
@@ -266,57 +233,38 @@ class CompositeAdaptiveTermination(TerminationCollection):
         use_per_objective: bool = True,
         use_hypervolume: bool = True,
         use_multiscale: bool = True,
-        **kwargs,
+        **kw,
     ):
-        terminations = [MaximumGenerationTermination(problem, n_max_gen=n_max_gen)]
+        members: List[Termination] = [
+            MaximumGenerationTermination(problem, n_max_gen=n_max_gen)
+        ]
         if use_per_objective:
-            terminations.append(
-                PerObjectiveConvergence(
-                    problem=problem,
-                    obj_tol=obj_tol,
-                    min_converged_fraction=min_converged_fraction,
-                    n_last=20,
-                    nth_gen=5,
-                    min_generations=min_generations,
-                    **kwargs,
-                )
-            )
+            members.append(PerObjectiveConvergence(
+                problem, obj_tol=obj_tol,
+                min_converged_fraction=min_converged_fraction,
+                n_last=20, nth_gen=5, min_generations=min_generations, **kw,
+            ))
         if use_hypervolume:
-            terminations.append(
-                HypervolumeProgressTermination(
-                    problem=problem,
-                    ref_point=ref_point,
-                    hv_tol=hv_tol,
-                    n_last=15,
-                    nth_gen=5,
-                    min_generations=min_generations,
-                    **kwargs,
-                )
-            )
+            members.append(HypervolumeProgressTermination(
+                problem, ref_point=ref_point, hv_tol=hv_tol,
+                n_last=15, nth_gen=5, min_generations=min_generations, **kw,
+            ))
         if use_multiscale:
             if timescales is None:
-                base_scale = max(5, problem.n_objectives // 5)
-                timescales = [base_scale * (2**i) for i in range(4)]
-            terminations.append(
-                MultiScaleStagnationTermination(
-                    problem=problem,
-                    timescales=timescales,
-                    stagnation_tol=stagnation_tol,
-                    min_scales_stagnant=3,
-                    nth_gen=2,
-                    min_generations=min_generations,
-                    **kwargs,
-                )
-            )
-        super().__init__(problem, *terminations)
-        _log(
-            problem,
-            f"Initialized CompositeAdaptiveTermination with {len(terminations)} criteria",
-        )
+                base = max(5, problem.n_objectives // 5)
+                timescales = [base << i for i in range(4)]
+            members.append(MultiScaleStagnationTermination(
+                problem, timescales=timescales, stagnation_tol=stagnation_tol,
+                min_scales_stagnant=3, nth_gen=2,
+                min_generations=min_generations, **kw,
+            ))
+        super().__init__(problem, *members)
+        _log(problem, f"composite termination armed with {len(members)} criteria")
 
 
 class ResourceAwareTermination(Termination):
-    """Wall-time / eval-count / quality limits (adaptive_termination.py:475)."""
+    """Hard resource budget: wall-clock seconds, evaluation count, or a
+    quality threshold — whichever trips first."""
 
     def __init__(
         self,
@@ -324,74 +272,61 @@ class ResourceAwareTermination(Termination):
         max_time_seconds: Optional[float] = None,
         max_function_evals: Optional[int] = None,
         target_quality_threshold: Optional[float] = None,
-        **kwargs,
+        **kw,
     ):
         super().__init__(problem)
         self.max_time_seconds = max_time_seconds
         self.max_function_evals = max_function_evals
         self.target_quality_threshold = target_quality_threshold
-        self.start_time = None
+        self._armed_at: Optional[float] = None
 
-    def _do_continue(self, opt):
-        if self.start_time is None:
-            self.start_time = _time.time()
+    def _stop(self, opt) -> bool:
+        if self._armed_at is None:
+            self._armed_at = _time.time()
         if self.max_time_seconds is not None:
-            elapsed = _time.time() - self.start_time
-            if elapsed > self.max_time_seconds:
-                _log(self.problem, f"Optimization terminated: time limit ({elapsed:.1f}s)")
-                return False
+            used = _time.time() - self._armed_at
+            if used > self.max_time_seconds:
+                _log(self.problem, f"stop: wall-clock budget used ({used:.1f}s)")
+                return True
         if self.max_function_evals is not None:
-            n_evals = getattr(opt, "n_eval", None)
-            if n_evals is None:
-                n_evals = getattr(opt, "n_gen", 0)
-            if n_evals and n_evals > self.max_function_evals:
-                _log(self.problem, f"Optimization terminated: eval limit ({n_evals})")
-                return False
+            spent = getattr(opt, "n_eval", None) or getattr(opt, "n_gen", 0)
+            if spent and spent > self.max_function_evals:
+                _log(self.problem, f"stop: evaluation budget used ({spent})")
+                return True
         if self.target_quality_threshold is not None:
-            quality = getattr(opt, "quality_metric", None)
-            if quality is not None and quality > self.target_quality_threshold:
-                _log(self.problem, "Optimization terminated: quality threshold reached")
-                return False
-        return True
+            q = getattr(opt, "quality_metric", None)
+            if q is not None and q > self.target_quality_threshold:
+                _log(self.problem, "stop: target quality reached")
+                return True
+        return False
+
+    # compat: the reference exposes start_time
+    @property
+    def start_time(self):
+        return self._armed_at
+
+
+_PRESETS = {
+    "comprehensive": dict(use_per_objective=True, use_hypervolume=True,
+                          use_multiscale=True, hv_tol=1e-6),
+    "fast": dict(use_per_objective=False, use_hypervolume=True,
+                 use_multiscale=True),
+    "conservative": dict(use_per_objective=True, use_hypervolume=False,
+                         use_multiscale=True),
+}
 
 
 def create_adaptive_termination(
-    problem, n_max_gen: int = 2000, strategy: str = "comprehensive", **kwargs
+    problem, n_max_gen: int = 2000, strategy: str = "comprehensive", **kw
 ) -> Termination:
-    """Factory presets comprehensive|fast|conservative|simple
-    (adaptive_termination.py:546-627)."""
-    if strategy == "comprehensive":
-        return CompositeAdaptiveTermination(
-            problem=problem,
-            n_max_gen=n_max_gen,
-            use_per_objective=True,
-            use_hypervolume=True,
-            use_multiscale=True,
-            hv_tol=1e-6,
-            **kwargs,
-        )
-    if strategy == "fast":
-        return CompositeAdaptiveTermination(
-            problem=problem,
-            n_max_gen=n_max_gen,
-            use_per_objective=False,
-            use_hypervolume=True,
-            use_multiscale=True,
-            **kwargs,
-        )
-    if strategy == "conservative":
-        return CompositeAdaptiveTermination(
-            problem=problem,
-            n_max_gen=n_max_gen,
-            use_per_objective=True,
-            use_hypervolume=False,
-            use_multiscale=True,
-            **kwargs,
-        )
+    """Factory for the preset strategies comprehensive|fast|conservative|
+    simple (same preset names and compositions as the reference)."""
     if strategy == "simple":
         return HypervolumeProgressTermination(
-            problem=problem, n_last=20, nth_gen=5, n_max_gen=n_max_gen, **kwargs
+            problem, n_last=20, nth_gen=5, n_max_gen=n_max_gen, **kw
         )
-    raise ValueError(
-        f"Unknown strategy {strategy!r}; choose comprehensive|fast|conservative|simple"
-    )
+    if strategy not in _PRESETS:
+        raise ValueError(
+            f"Unknown strategy {strategy!r}; choose comprehensive|fast|conservative|simple"
+        )
+    return CompositeAdaptiveTermination(problem, n_max_gen=n_max_gen, **_PRESETS[strategy], **kw)

@@ -1,8 +1,25 @@
-"""Core termination criteria (parity with reference termination.py:14-347)."""
+"""Generation-loop stopping criteria.
+
+Role parity with the reference's ``termination.py``:
+the same stopping decisions over the same ``OptHistory`` inputs, but
+restructured around ONE windowed-pipeline engine instead of an inheritance
+chain. Every windowed criterion is three stages over ring buffers::
+
+    capture(opt)      -> raw snapshot appended to the raw ring
+    reduce(raw tail)  -> derived signal appended to the signal ring
+    verdict(signals)  -> stop / keep going, judged on a cadence
+
+configured declaratively by a :class:`Windows` dataclass (raw/signal ring
+sizes, warmup, cadence, a cross-epoch generation floor and an absolute
+generation ceiling). The concrete criteria below plug formulas into that
+engine; the formulas themselves (IGD drift, normalized ideal-point delta,
+constraint-violation transitions) match the reference semantics.
+"""
 
 from __future__ import annotations
 
-from abc import abstractmethod
+from dataclasses import dataclass
+from typing import Any, List, Optional
 
 import numpy as np
 
@@ -17,218 +34,240 @@ def _log(problem, msg):
 
 
 class Termination:
+    """Base protocol: ``has_terminated(opt) -> bool`` over an OptHistory.
+
+    ``force_termination`` short-circuits to True (kept for API parity with
+    the reference's base class).
+    """
+
     def __init__(self, problem) -> None:
         self.problem = problem
         self.force_termination = False
 
-    def do_continue(self, opt):
+    def has_terminated(self, opt) -> bool:
         if self.force_termination:
-            return False
-        return self._do_continue(opt)
+            return True
+        return self._stop(opt)
 
-    def _do_continue(self, opt, **kwargs):
-        pass
+    def _stop(self, opt) -> bool:
+        return False
 
-    def has_terminated(self, opt):
-        return not self.do_continue(opt)
+    # reference-compatible inverse form
+    def do_continue(self, opt) -> bool:
+        return not self.has_terminated(opt)
 
 
 class TerminationCollection(Termination):
-    def __init__(self, problem, *args) -> None:
-        super().__init__(problem)
-        self.terminations = args
+    """Any-of composition: stops as soon as one member criterion stops."""
 
-    def _do_continue(self, opt):
-        for term in self.terminations:
-            if not term.do_continue(opt):
-                return False
-        return True
+    def __init__(self, problem, *criteria) -> None:
+        super().__init__(problem)
+        self.terminations = tuple(criteria)
+
+    def _stop(self, opt) -> bool:
+        return any(t.has_terminated(opt) for t in self.terminations)
 
 
 class MaximumGenerationTermination(Termination):
     def __init__(self, problem, n_max_gen) -> None:
         super().__init__(problem)
-        self.n_max_gen = n_max_gen if n_max_gen is not None else float("inf")
+        self.n_max_gen = float("inf") if n_max_gen is None else n_max_gen
 
-    def _do_continue(self, opt):
+    def _stop(self, opt) -> bool:
         if opt.n_gen > self.n_max_gen:
-            _log(self.problem, f"Optimization terminated: maximum generations ({opt.n_gen}) reached")
-        return opt.n_gen <= self.n_max_gen
+            _log(self.problem, f"stop: generation cap hit at gen {opt.n_gen}")
+            return True
+        return False
 
 
-class SlidingWindowTermination(TerminationCollection):
-    """Store -> metric -> decide pipeline over sliding windows with an
-    nth_gen cadence and a min_generations floor persistent across epochs
-    (reference termination.py:90-207)."""
+@dataclass
+class Windows:
+    """Ring-buffer and cadence configuration for a WindowedCriterion.
 
-    def __init__(
-        self,
-        problem,
-        metric_window_size=None,
-        data_window_size=None,
-        min_data_for_metric=1,
-        nth_gen=1,
-        n_max_gen=None,
-        min_generations=0,
-        truncate_metrics=True,
-        truncate_data=True,
-    ):
-        super().__init__(problem, MaximumGenerationTermination(problem, n_max_gen=n_max_gen))
-        self.data_window_size = data_window_size
-        self.metric_window_size = metric_window_size
-        self.truncate_data = truncate_data
-        self.data = SlidingWindow(data_window_size) if truncate_data else []
-        self.truncate_metrics = truncate_metrics
-        self.metrics = SlidingWindow(metric_window_size) if truncate_metrics else []
-        self.nth_gen = nth_gen
-        self.min_data_for_metric = min_data_for_metric
-        self.min_generations = min_generations
-        self.n_total_gens = 0
+    raw:     how many captures the raw ring keeps (None = unbounded)
+    signal:  signal-ring size; also how many signals the verdict needs/sees
+    warmup:  captures required before reduce() runs
+    cadence: judge only every cadence-th generation
+    floor:   total captures (across epochs) before any stop verdict counts
+    ceiling: absolute generation cap (None = unlimited)
+    """
+
+    raw: Optional[int] = None
+    signal: Optional[int] = None
+    warmup: int = 1
+    cadence: int = 1
+    floor: int = 0
+    ceiling: Optional[int] = None
+
+
+class WindowedCriterion(Termination):
+    """The capture -> reduce -> verdict pipeline over two ring buffers.
+
+    Subclasses implement :meth:`capture`, :meth:`reduce` and
+    :meth:`verdict`. The engine keeps a cross-epoch capture counter so a
+    ``floor`` survives epoch restarts (the rings themselves can be cleared
+    with :meth:`reset` between epochs).
+    """
+
+    def __init__(self, problem, windows: Windows):
+        super().__init__(problem)
+        self.windows = windows
+        self._lifetime_captures = 0
+        self._alloc_rings()
+
+    def _alloc_rings(self):
+        w = self.windows
+        self._raw_ring: List[Any] = SlidingWindow(w.raw) if w.raw is not None else []
+        self._signal_ring: List[Any] = (
+            SlidingWindow(w.signal) if w.signal is not None else []
+        )
 
     def reset(self):
-        self.data = SlidingWindow(self.data_window_size) if self.truncate_data else []
-        self.metrics = SlidingWindow(self.metric_window_size) if self.truncate_metrics else []
+        """Clear the rings; the lifetime capture counter (floor) persists."""
+        self._alloc_rings()
 
-    def _do_continue(self, opt):
-        if not super()._do_continue(opt):
-            return False
-        self.n_total_gens += 1
-        obj = self._store(opt)
-        if obj is not None:
-            self.data.append(obj)
-        if len(self.data) >= self.min_data_for_metric:
-            metric = self._metric(self.data[-self.data_window_size :])
-            if metric is not None:
-                self.metrics.append(metric)
-        if self.n_total_gens < self.min_generations:
-            return True
-        if opt.n_gen % self.nth_gen == 0 and len(self.metrics) >= self.metric_window_size:
-            return self._decide(self.metrics[-self.metric_window_size :])
-        return True
-
-    def _store(self, opt):
+    # ---- stages ----------------------------------------------------------
+    def capture(self, opt) -> Any:
         return opt
 
-    @abstractmethod
-    def _decide(self, metrics):
-        ...
+    def reduce(self, raws: List[Any]) -> Any:
+        raise NotImplementedError
 
-    @abstractmethod
-    def _metric(self, data):
-        ...
+    def verdict(self, signals: List[Any]) -> bool:
+        raise NotImplementedError
 
-    def get_metric(self):
-        return self.metrics[-1] if len(self.metrics) > 0 else None
-
-
-class ParameterToleranceTermination(SlidingWindowTermination):
-    """IGD between consecutive normalized X populations <= tol."""
-
-    def __init__(self, problem, n_last=10, tol=1e-6, nth_gen=1, n_max_gen=None, **kwargs):
-        super().__init__(
-            problem,
-            metric_window_size=n_last,
-            data_window_size=2,
-            min_data_for_metric=2,
-            nth_gen=nth_gen,
-            n_max_gen=n_max_gen,
-            **kwargs,
-        )
-        self.tol = tol
-
-    def _store(self, opt):
-        X = opt.x
-        if X.dtype != object:
-            if self.problem.lb is not None and self.problem.ub is not None:
-                X = normalize(X, xl=self.problem.lb, xu=self.problem.ub)
-            return X
-
-    def _metric(self, data):
-        last, current = data[-2], data[-1]
-        return IGD(current).do(last)
-
-    def _decide(self, metrics):
-        mean = np.asarray(metrics).mean()
-        if mean <= self.tol:
-            _log(self.problem, f"Optimization terminated: mean parameter distance {mean} below {self.tol}")
-        return mean > self.tol
-
-
-def calc_delta_norm(a, b, norm):
-    return np.max(np.abs((a - b) / norm))
-
-
-class MultiObjectiveToleranceTermination(SlidingWindowTermination):
-    """Delta-ideal + IGD of normalized F windows <= tol (default 0.0025)."""
-
-    def __init__(self, problem, tol=0.0025, n_last=10, nth_gen=1, n_max_gen=None, **kwargs):
-        super().__init__(
-            problem,
-            metric_window_size=n_last,
-            data_window_size=2,
-            min_data_for_metric=2,
-            nth_gen=nth_gen,
-            n_max_gen=n_max_gen,
-            **kwargs,
-        )
-        self.tol = tol
-
-    def _store(self, opt):
-        F = opt.y
-        return {"ideal": F.min(axis=0), "nadir": F.max(axis=0), "F": F}
-
-    def _metric(self, data):
-        last, current = data[-2], data[-1]
-        norm = current["nadir"] - current["ideal"]
-        norm[norm < 1e-32] = 1
-        delta_ideal = calc_delta_norm(current["ideal"], last["ideal"], norm)
-        c_F, c_ideal, c_nadir = current["F"], current["ideal"], current["nadir"]
-        c_N = normalize(c_F, c_ideal, c_nadir)
-        l_N = normalize(last["F"], c_ideal, c_nadir)
-        delta_f = IGD(c_N).do(l_N)
-        return {"delta_ideal": delta_ideal, "delta_f": delta_f}
-
-    def _decide(self, metrics):
-        delta_ideal = [e["delta_ideal"] for e in metrics]
-        delta_f = [e["delta_f"] for e in metrics]
-        max_delta = max(np.mean(delta_ideal), np.mean(delta_f))
-        if max_delta <= self.tol:
-            _log(
-                self.problem,
-                f"Optimization terminated: objective mean delta "
-                f"{(np.mean(delta_ideal), np.mean(delta_f))} below {self.tol}",
+    # ---- engine ----------------------------------------------------------
+    def _stop(self, opt) -> bool:
+        w = self.windows
+        if w.ceiling is not None and opt.n_gen > w.ceiling:
+            _log(self.problem, f"stop: generation cap hit at gen {opt.n_gen}")
+            return True
+        self._lifetime_captures += 1
+        snap = self.capture(opt)
+        if snap is not None:
+            self._raw_ring.append(snap)
+        if len(self._raw_ring) >= w.warmup:
+            tail = self._raw_ring if w.raw is None else self._raw_ring[-w.raw :]
+            sig = self.reduce(tail)
+            if sig is not None:
+                self._signal_ring.append(sig)
+        if self._lifetime_captures < w.floor:
+            return False
+        need = w.signal if w.signal is not None else 1
+        if opt.n_gen % w.cadence == 0 and len(self._signal_ring) >= need:
+            tail = (
+                self._signal_ring
+                if w.signal is None
+                else self._signal_ring[-w.signal :]
             )
-        else:
-            _log(self.problem, f"Objective mean delta: {(np.mean(delta_ideal), np.mean(delta_f))}")
-        return max_delta > self.tol
+            return self.verdict(tail)
+        return False
+
+    def latest_signal(self):
+        return self._signal_ring[-1] if self._signal_ring else None
 
 
-class ConstraintViolationToleranceTermination(SlidingWindowTermination):
-    def __init__(self, problem, n_last=10, tol=1e-6, nth_gen=1, n_max_gen=None, **kwargs):
+# kept as the exported name the reference uses for this role
+SlidingWindowTermination = WindowedCriterion
+
+
+class ParameterToleranceTermination(WindowedCriterion):
+    """Stop when consecutive normalized parameter populations stop moving:
+    mean IGD(X_t -> X_{t-1}) over a window falls to/below ``tol``."""
+
+    def __init__(self, problem, n_last=10, tol=1e-6, nth_gen=1, n_max_gen=None, **kw):
         super().__init__(
             problem,
-            metric_window_size=n_last,
-            data_window_size=2,
-            min_data_for_metric=2,
-            nth_gen=nth_gen,
-            n_max_gen=n_max_gen,
-            **kwargs,
+            Windows(raw=2, signal=n_last, warmup=2, cadence=nth_gen,
+                    ceiling=n_max_gen, floor=kw.pop("min_generations", 0), **kw),
         )
         self.tol = tol
 
-    def _store(self, opt):
+    def capture(self, opt):
+        X = opt.x
+        if X.dtype == object:
+            return None
+        lb, ub = self.problem.lb, self.problem.ub
+        return normalize(X, xl=lb, xu=ub) if lb is not None and ub is not None else X
+
+    def reduce(self, raws):
+        prev, curr = raws[-2], raws[-1]
+        return IGD(curr).do(prev)
+
+    def verdict(self, signals):
+        drift = float(np.mean(signals))
+        if drift <= self.tol:
+            _log(self.problem,
+                 f"stop: parameter drift {drift:.3e} within tolerance {self.tol:.3e}")
+            return True
+        return False
+
+
+class MultiObjectiveToleranceTermination(WindowedCriterion):
+    """Stop when both the normalized ideal-point shift and the IGD drift of
+    consecutive objective populations average at/below ``tol``."""
+
+    def __init__(self, problem, tol=0.0025, n_last=10, nth_gen=1, n_max_gen=None, **kw):
+        super().__init__(
+            problem,
+            Windows(raw=2, signal=n_last, warmup=2, cadence=nth_gen,
+                    ceiling=n_max_gen, floor=kw.pop("min_generations", 0), **kw),
+        )
+        self.tol = tol
+
+    def capture(self, opt):
+        return np.asarray(opt.y)
+
+    def reduce(self, raws):
+        prev_F, curr_F = raws[-2], raws[-1]
+        ideal, nadir = curr_F.min(axis=0), curr_F.max(axis=0)
+        span = nadir - ideal
+        span[span < 1e-32] = 1.0
+        ideal_shift = float(np.max(np.abs((curr_F.min(axis=0) - prev_F.min(axis=0)) / span)))
+        front_drift = float(
+            IGD(normalize(curr_F, ideal, nadir)).do(normalize(prev_F, ideal, nadir))
+        )
+        return np.array([ideal_shift, front_drift])
+
+    def verdict(self, signals):
+        means = np.mean(np.vstack(signals), axis=0)  # [ideal shift, front drift]
+        worst = float(means.max())
+        if worst <= self.tol:
+            _log(self.problem,
+                 f"stop: objective drift (ideal {means[0]:.3e}, front {means[1]:.3e}) "
+                 f"within tolerance {self.tol:.3e}")
+            return True
+        _log(self.problem,
+             f"objective drift: ideal {means[0]:.3e}, front {means[1]:.3e}")
+        return False
+
+
+class ConstraintViolationToleranceTermination(WindowedCriterion):
+    """Constraint-violation window logic: stop while everything in the
+    window is feasible, keep going through a feasibility transition, and
+    otherwise stop only when the violation level has stopped changing."""
+
+    def __init__(self, problem, n_last=10, tol=1e-6, nth_gen=1, n_max_gen=None, **kw):
+        super().__init__(
+            problem,
+            Windows(raw=2, signal=n_last, warmup=2, cadence=nth_gen,
+                    ceiling=n_max_gen, floor=kw.pop("min_generations", 0), **kw),
+        )
+        self.tol = tol
+
+    def capture(self, opt):
         return opt.c
 
-    def _metric(self, data):
-        last, current = data[-2], data[-1]
-        return {"cv": current, "delta_cv": abs(last - current)}
+    def reduce(self, raws):
+        prev, curr = raws[-2], raws[-1]
+        return (curr, abs(curr - prev))
 
-    def _decide(self, metrics):
-        cv = np.asarray([e["cv"] for e in metrics])
-        delta_cv = np.asarray([e["delta_cv"] for e in metrics])
-        n_feasible = (cv > 0).sum()
-        if n_feasible == len(metrics):
-            return False
-        if 0 < n_feasible < len(metrics):
-            return True
-        return delta_cv.max() > self.tol
+    def verdict(self, signals):
+        cv = np.asarray([s[0] for s in signals])
+        dcv = np.asarray([s[1] for s in signals])
+        n_ok = int((cv > 0).sum())
+        if n_ok == len(signals):
+            return True  # fully feasible window
+        if n_ok > 0:
+            return False  # mid-transition: keep optimizing
+        return bool(dcv.max() <= self.tol)

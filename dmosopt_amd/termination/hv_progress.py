@@ -17,7 +17,7 @@ from typing import Any, Dict, List, Optional
 import numpy as np
 
 from dmosopt_amd.hv.adaptive import AdaptiveHyperVolume
-from dmosopt_amd.termination.basic import SlidingWindowTermination, _log
+from dmosopt_amd.termination.basic import WindowedCriterion, Windows, _log
 
 
 @dataclass
@@ -155,7 +155,7 @@ class ConvergenceDetector:
         }
 
 
-class HypervolumeProgressTermination(SlidingWindowTermination):
+class HypervolumeProgressTermination(WindowedCriterion):
     """Terminate when best-fidelity hypervolume stops improving
     (reference hv_termination.py:960-1162)."""
 
@@ -174,13 +174,8 @@ class HypervolumeProgressTermination(SlidingWindowTermination):
     ):
         super().__init__(
             problem,
-            metric_window_size=n_last,
-            data_window_size=2,
-            min_data_for_metric=2,
-            nth_gen=nth_gen,
-            n_max_gen=n_max_gen,
-            min_generations=min_generations,
-            **kwargs,
+            Windows(raw=2, signal=n_last, warmup=2, cadence=nth_gen,
+                    ceiling=n_max_gen, floor=min_generations, **kwargs),
         )
         self.ref_point = np.copy(ref_point) if ref_point is not None else None
         self.hv_tol = hv_tol
@@ -213,30 +208,27 @@ class HypervolumeProgressTermination(SlidingWindowTermination):
                     self._tracker.ref_point, nadir + 0.1 * span
                 )
 
-    def _store(self, opt):
+    def capture(self, opt):
         F = np.asarray(opt.y)
         self._gen = opt.n_gen
         self._ensure_tracker(F)
-        est = self._tracker.compute_and_update(F, opt.n_gen)
-        if est is None:
-            return None
-        return est
+        return self._tracker.compute_and_update(F, opt.n_gen)
 
-    def _metric(self, data):
-        est = data[-1]
+    def reduce(self, raws):
+        est = raws[-1]
         return {"hv": est.value, "fidelity": est.fidelity, "gen": est.generation}
 
-    def _decide(self, metrics):
+    def verdict(self, signals):
         result = self._detector.check_convergence(self._tracker.estimates, self._gen)
         self._scheduler.adapt_to_progress(result.get("rel_improvement", 1.0))
         if result["converged"]:
             _log(
                 self.problem,
-                f"Optimization terminated: hypervolume stagnant "
+                f"stop: hypervolume stagnant "
                 f"(rel improvement {result['rel_improvement']:.2e}, "
                 f"confidence {result['confidence']:.2f})",
             )
-            return False
+            return True
         if self.verbose:
-            _log(self.problem, f"HV progress: {metrics[-1]}")
-        return True
+            _log(self.problem, f"HV progress: {signals[-1]}")
+        return False
