@@ -36,7 +36,6 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from dmosopt_amd.benchmarks.problems import zdt1
 from dmosopt_amd.core import engine
 from dmosopt_amd.hv.exact import hv_2d
-from dmosopt_amd.models.gp import GPRMatern
 from dmosopt_amd.models.model import Model
 from dmosopt_amd.parallel import comm
 
@@ -47,61 +46,6 @@ N_GEN = 200
 ARCHIVE_N = 300
 
 
-class ShardedGPObjective:
-    """GP posterior evaluate() with rank-sharded prediction.
-
-    Queries arrive replicated on every rank; each rank predicts rows
-    [rank::world] on its GPU and the full result is re-assembled with ONE
-    all_gather per generation (a single large collective over xGMI, per the
-    comm design of SURVEY.md section 2.10).
-    """
-
-    def __init__(self, gp: GPRMatern, rank: int, world: int, device):
-        self.gp = gp
-        self.rank = rank
-        self.world = world
-        self.device = device
-
-    def evaluate_tensor(self, x):
-        if self.world == 1:
-            return self.gp.evaluate_tensor(x)
-        import torch.distributed as dist
-
-        # device-resident shard -> predict -> ONE all_gather over RCCL/xGMI
-        # (no host round trip inside the generation loop)
-        P = x.shape[0]
-        pad = (self.world - P % self.world) % self.world
-        if pad:
-            x = torch.cat([x, x[-1:].expand(pad, -1)], dim=0)
-        shard = x[self.rank :: self.world]
-        xq = self.gp.normalize_query(shard.to(self.gp.device, self.gp.dtype))
-        mean, _ = self.gp._fitted.predict(xq, return_var=False)
-        mean = mean.to(torch.float32).contiguous()
-        out = [torch.empty_like(mean) for _ in range(self.world)]
-        dist.all_gather(out, mean)
-        full = torch.stack(out, dim=1).reshape(-1, mean.shape[1])
-        return full[:P].to(dtype=x.dtype, device=x.device)
-
-    def evaluate(self, x):
-        import torch.distributed as dist
-
-        x = np.asarray(x, dtype=np.float64)
-        if self.world == 1:
-            mean, _ = self.gp.predict(x)
-            return mean
-        P = x.shape[0]
-        pad = (self.world - P % self.world) % self.world
-        if pad:
-            x = np.vstack([x, np.repeat(x[-1:], pad, axis=0)])
-        shard = x[self.rank :: self.world]
-        mean, _ = self.gp.predict(shard)  # (P/world, m)
-        mean_t = torch.as_tensor(mean, dtype=torch.float32, device=self.device)
-        out = [torch.empty_like(mean_t) for _ in range(self.world)]
-        dist.all_gather(out, mean_t)
-        full = torch.stack(out, dim=1).reshape(-1, mean_t.shape[1])  # interleave
-        return full[:P].cpu().numpy()
-
-
 def make_archive(seed: int) -> tuple:
     rng = np.random.default_rng(seed)
     X = rng.random((ARCHIVE_N, D_IN))
@@ -110,12 +54,26 @@ def make_archive(seed: int) -> tuple:
 
 
 def one_epoch(X, Y, pop, rank, world, device, seed, n_gen=N_GEN):
-    """One MO-ASMO epoch; returns (resample_x, predicted_y, hv)."""
-    gp = GPRMatern(
-        X, Y, D_IN, N_OBJ, np.zeros(D_IN), np.ones(D_IN),
-        optimizer="sceua", seed=seed, device=device,
+    """One MO-ASMO epoch; returns (resample_x, predicted_y, hv).
+
+    Multi-rank path = the PACKAGE path: rank 0 fits the GP and broadcasts
+    theta (engine.train), predictions are rank-sharded and all-gathered
+    each generation (parallel.sharded.ShardedObjective)."""
+    gp = engine.train(
+        D_IN, N_OBJ, np.zeros(D_IN), np.ones(D_IN), X, Y, None,
+        surrogate_method_name="gpr",
+        surrogate_method_kwargs={
+            "anisotropic": False, "optimizer": "sceua", "seed": seed,
+        },
+        logger=None, device=device,
     )
-    mdl = Model(objective=ShardedGPObjective(gp, rank, world, device))
+    if world > 1:
+        from dmosopt_amd.parallel.context import get_context
+        from dmosopt_amd.parallel.sharded import ShardedObjective
+
+        mdl = Model(objective=ShardedObjective(gp, get_context()))
+    else:
+        mdl = Model(objective=gp)
 
     from dmosopt_amd.moea.nsga2 import NSGA2Optimizer
 

@@ -11,6 +11,8 @@ from __future__ import annotations
 import logging
 from typing import Dict, Optional
 
+import numpy as np
+
 from dmosopt_amd.config import import_object_by_path
 from dmosopt_amd.core.driver import DistOptimizer
 from dmosopt_amd.parallel import comm
@@ -72,42 +74,72 @@ def run(
     ctrl_init_fun_name = params.pop("controller_init_fun_name", None)
     ctrl_init_fun_args = params.pop("controller_init_fun_args", {})
 
-    if rank == 0:
-        if ctrl_init_fun_name is not None:
-            import_object_by_path(ctrl_init_fun_name)(**ctrl_init_fun_args)
-        dopt = DistOptimizer(**params, verbose=verbose)
-        farm = comm.make_farm({opt_id: dopt.eval_fun})
-        dopt.farm = farm
-        dopt.initialize_strategy()
-        dopt_dict[opt_id] = dopt
-        dmosopt_amd.sopt_dict[opt_id] = dopt
-        logger.info(f"Optimizing for {dopt.n_epochs} epochs...")
-        import time as _time
+    # Replicated control flow (parallel/context.py): EVERY rank builds the
+    # full driver with identical seeds and runs the same epoch loop; the
+    # data plane (objective results, surrogate predictions, GP
+    # hyperparameters) moves as tensor collectives. Rank 0 alone touches
+    # the H5 file for writing, logs, and returns the best set (API parity:
+    # non-root ranks return None like the reference's workers).
+    from dmosopt_amd.parallel.context import get_context
 
-        t_start = _time.time()
-        try:
-            if dopt.n_epochs <= 0:
-                dopt.run_epoch(completed_epoch=True)
-            else:
-                while dopt.epoch_count < dopt.n_epochs:
-                    if time_limit is not None and (_time.time() - t_start) > time_limit:
-                        logger.info("Time limit reached; stopping.")
-                        break
-                    dopt.run_epoch()
-        finally:
-            farm.shutdown()
-        dopt.print_best()
-        return dopt.get_best(
-            feasible=feasible,
-            return_features=return_features,
-            return_constraints=return_constraints,
+    ctx = get_context()
+    if rank == 0 and ctrl_init_fun_name is not None:
+        import_object_by_path(ctrl_init_fun_name)(**ctrl_init_fun_args)
+    if rank != 0:
+        params["save"] = False  # restore-only: read the H5, never write it
+    if (
+        world > 1
+        and params.get("random_seed") is None
+        and params.get("local_random") is None
+    ):
+        # replicated control flow requires a shared seed: an unseeded run
+        # would silently diverge, so rank 0 draws one and broadcasts it
+        seed = int(np.random.default_rng().integers(2**31)) if rank == 0 else 0
+        params["random_seed"] = ctx.bcast_int(seed, src=0)
+    dopt = DistOptimizer(**params, verbose=verbose and rank == 0)
+    if world > 1:
+        spec = comm.ResultSpec(
+            problem_ids=tuple(dopt.problem_ids),
+            n_objectives=len(dopt.objective_names),
+            n_constraints=len(dopt.constraint_names) if dopt.constraint_names else 0,
+            feature_dtype=(
+                np.dtype(dopt.feature_dtypes) if dopt.feature_dtypes else None
+            ),
         )
+        farm = comm.make_farm({opt_id: dopt.eval_fun}, spec)
     else:
-        # worker rank: build eval_fun identically, serve the farm
-        params.pop("file_path", None)
-        params.pop("save", None)
-        dopt = DistOptimizer(**params, verbose=False)
         farm = comm.make_farm({opt_id: dopt.eval_fun})
-        dopt_dict[opt_id] = dopt
-        farm.worker_loop()
+    dopt.farm = farm
+    dopt.initialize_strategy()
+    dopt_dict[opt_id] = dopt
+    dmosopt_amd.sopt_dict[opt_id] = dopt
+    if rank == 0:
+        logger.info(f"Optimizing for {dopt.n_epochs} epochs...")
+    import time as _time
+
+    t_start = _time.time()
+    try:
+        if dopt.n_epochs <= 0:
+            dopt.run_epoch(completed_epoch=True)
+        else:
+            while dopt.epoch_count < dopt.n_epochs:
+                # rank 0 owns the wall-clock decision; replicated ranks
+                # must agree, so the stop flag is broadcast
+                over = time_limit is not None and (_time.time() - t_start) > time_limit
+                if ctx is not None and ctx.world > 1:
+                    over = ctx.bcast_flag(over, src=0)
+                if over:
+                    if rank == 0:
+                        logger.info("Time limit reached; stopping.")
+                    break
+                dopt.run_epoch()
+    finally:
+        farm.shutdown()
+    if rank != 0:
         return None
+    dopt.print_best()
+    return dopt.get_best(
+        feasible=feasible,
+        return_features=return_features,
+        return_constraints=return_constraints,
+    )

@@ -531,6 +531,18 @@ class DistOptimizer:
         if self.save:
             for problem_id in self.problem_ids:
                 self.save_stats(problem_id, epoch)
+        # replicated-mode divergence guard: the per-problem archives must be
+        # bit-identical on every rank after each epoch — fail loudly if any
+        # nondeterminism crept into the replicated control flow
+        from dmosopt_amd.parallel.context import get_context
+
+        ctx = get_context()
+        if ctx is not None and ctx.world > 1:
+            for problem_id in self.problem_ids:
+                s = self.optimizer_dict[problem_id]
+                ctx.assert_synchronized(
+                    [s.x, s.y, s.c], tag=f"epoch{epoch}/problem{problem_id}"
+                )
         self.epoch_count += 1
         return self.epoch_count
 

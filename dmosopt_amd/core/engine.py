@@ -203,6 +203,30 @@ def train(
     # engine-level flag (passing both is a reference API footgun:
     # "got multiple values for keyword argument")
     kwargs.setdefault("return_mean_variance", surrogate_return_mean_variance)
+
+    # Multi-rank hyperparameter broadcast ("broadcast
+    # of surrogate hyperparameters"): surrogate classes whose fitted state is
+    # fully determined by (archive, theta) opt in via supports_theta_broadcast.
+    # Rank 0 runs the hyperparameter search; everyone else receives theta in
+    # ONE tensor broadcast and rebuilds the posterior locally — correct by
+    # construction even if the search itself were nondeterministic.
+    from dmosopt_amd.parallel.context import get_context
+
+    ctx = get_context()
+    if (
+        ctx is not None
+        and ctx.world > 1
+        and getattr(surrogate_cls, "supports_theta_broadcast", False)
+    ):
+        if ctx.is_root:
+            sm = surrogate_cls(x, y, nInput, nOutput, xlb, xub, logger=logger, **kwargs)
+            ctx.bcast_payload(sm.theta.detach().to("cpu", torch.float64), src=0)
+            return sm
+        theta = ctx.bcast_payload(None, src=0).cpu().numpy()
+        return surrogate_cls(
+            x, y, nInput, nOutput, xlb, xub, logger=logger,
+            theta_override=theta, **kwargs,
+        )
     return surrogate_cls(x, y, nInput, nOutput, xlb, xub, logger=logger, **kwargs)
 
 
@@ -324,6 +348,17 @@ def run_epoch(
             surrogate_return_mean_variance=optimize_mean_variance,
             logger=logger, file_path=file_path, device=device,
         )
+
+    # Multi-rank: shard every surrogate prediction (the per-generation hot
+    # path AND the sensitivity sweep below) across ranks; results reassemble
+    # via one all_gather so every replicated rank sees identical values.
+    from dmosopt_amd.parallel.context import get_context as _get_ctx
+
+    _ctx = _get_ctx()
+    if _ctx is not None and _ctx.world > 1 and mdl.objective is not None:
+        from dmosopt_amd.parallel.sharded import ShardedObjective
+
+        mdl.objective = ShardedObjective(mdl.objective, _ctx)
 
     if sensitivity_method_name is not None and mdl.sensitivity is None:
         class _S:

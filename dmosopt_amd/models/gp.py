@@ -39,6 +39,10 @@ def _top_k_mo(x: np.ndarray, y: np.ndarray, top_k):
 
 class _GPRBase:
     nu: float = 2.5
+    #: fitted state is fully determined by (archive, theta): the engine may
+    #: run the hyperparameter search on rank 0 only and broadcast theta
+    #: (core/engine.py train())
+    supports_theta_broadcast = True
 
     def __init__(
         self,
@@ -63,6 +67,7 @@ class _GPRBase:
         device=None,
         dtype=None,
         logger=None,
+        theta_override=None,
         **kwargs,
     ):
         self.nInput = nInput
@@ -115,6 +120,20 @@ class _GPRBase:
             ]
         )
         Yn = (Y - y_mean[None, :]) / y_std[None, :]
+
+        if theta_override is not None:
+            # broadcast-received hyperparameters: skip the search, rebuild
+            # the posterior (Cholesky + alpha) locally from the archive
+            theta = torch.as_tensor(
+                np.asarray(theta_override), dtype=self.dtype, device=self.device
+            )
+            assert theta.shape == (m, nopt), (theta.shape, (m, nopt))
+            self._fitted = FittedGP(
+                X, Y, theta, y_mean, y_std, nu=self.nu, anisotropic=anisotropic,
+                jitter=1e-10 if self.dtype == torch.float64 else 1e-6,
+            )
+            self.theta = theta
+            return
 
         if optimizer == "dlib" and logger is not None:
             logger.warning(

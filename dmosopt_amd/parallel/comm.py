@@ -1,15 +1,16 @@
 """Distributed evaluation farm over torch.distributed (RCCL on ROCm).
 
-Replaces the reference's distwq MPI task farm: one
-process per GPU, rank 0 is the controller/driver. Instead of point-to-point
-task dispatch, candidate batches are BROADCAST to all ranks, each rank
-evaluates an even shard (rank strided), and results return via a single
-gather — the collective pattern sized for xGMI (few large messages over the
-7 p2p links, not thousands of small sends).
+Replaces the reference's distwq MPI task farm.
+Execution model: REPLICATED control flow — every rank runs the same driver
+code with identical seeds and derives the same candidate batches — so task
+dispatch needs NO control messages at all. Each rank evaluates the
+rank-strided shard of every batch and the results return as ONE tensor
+all_gather of packed (y[,c], t) rows (+ one uint8 all_gather when feature
+records ride along): a single large collective over the xGMI links instead
+of per-point pickle traffic.
 
 Backend: "cpu:gloo,cuda:nccl" when CUDA is present (nccl IS RCCL on ROCm),
-plain gloo otherwise. Control-plane objects ride the gloo lane; bulk tensors
-ride RCCL.
+plain gloo otherwise.
 """
 
 from __future__ import annotations
@@ -17,11 +18,14 @@ from __future__ import annotations
 import datetime
 import os
 import time
+from dataclasses import dataclass
 from typing import Callable, Dict, List, Optional, Tuple
 
 import numpy as np
 import torch
 import torch.distributed as dist
+
+from dmosopt_amd.parallel.context import ParallelContext, get_context, set_context
 
 
 def dist_is_initialized() -> bool:
@@ -29,11 +33,14 @@ def dist_is_initialized() -> bool:
 
 
 def init_from_env(timeout_s: int = 1800) -> Tuple[int, int]:
-    """Initialize the process group from torchrun env vars if present.
+    """Initialize the process group from torchrun env vars if present and
+    install the global ParallelContext.
 
     Returns (rank, world_size); (0, 1) when not launched distributed.
     """
     if dist_is_initialized():
+        if get_context() is None:
+            _install_context()
         return dist.get_rank(), dist.get_world_size()
     if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
         return 0, 1
@@ -47,11 +54,52 @@ def init_from_env(timeout_s: int = 1800) -> Tuple[int, int]:
     else:
         backend = "gloo"
     dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
+    _install_context()
     return dist.get_rank(), dist.get_world_size()
+
+
+def _install_context():
+    device = (
+        torch.device("cuda", torch.cuda.current_device())
+        if torch.cuda.is_available()
+        else torch.device("cpu")
+    )
+    set_context(ParallelContext(dist.get_rank(), dist.get_world_size(), device))
 
 
 def is_controller() -> bool:
     return (not dist_is_initialized()) or dist.get_rank() == 0
+
+
+@dataclass
+class ResultSpec:
+    """Fixed per-point result layout the tensor farm packs/unpacks.
+
+    Derived from the driver's configuration (objective/constraint names,
+    feature dtypes) so every rank agrees on the wire format without any
+    negotiation round."""
+
+    problem_ids: Tuple[int, ...]
+    n_objectives: int
+    n_constraints: int = 0
+    feature_dtype: Optional[np.dtype] = None
+
+    @property
+    def floats_per_problem(self) -> int:
+        return self.n_objectives + self.n_constraints
+
+    @property
+    def float_width(self) -> int:
+        # per point: per-problem (y | c) blocks + one shared eval time
+        return len(self.problem_ids) * self.floats_per_problem + 1
+
+    @property
+    def feature_itemsize(self) -> int:
+        return 0 if self.feature_dtype is None else np.dtype(self.feature_dtype).itemsize
+
+    @property
+    def byte_width(self) -> int:
+        return len(self.problem_ids) * self.feature_itemsize
 
 
 class _FarmStats:
@@ -92,7 +140,7 @@ class _FarmStats:
 
 
 class LocalFarm:
-    """Single-process evaluation (controller evaluates everything)."""
+    """Single-process evaluation (the world-size-1 path)."""
 
     def __init__(self, eval_funs: Dict[str, Callable]):
         self.eval_funs = eval_funs
@@ -114,42 +162,142 @@ class LocalFarm:
         pass
 
 
-class TorchDistFarm:
-    """Collective evaluation farm: rank 0 drives, every rank evaluates.
+class CollectiveFarm:
+    """Symmetric tensor-collective evaluation farm.
 
-    Protocol per batch (all ranks participate):
-      1. broadcast_object_list([("eval", opt_id, x_batch)])  (gloo lane)
-      2. each rank r evaluates rows r, r+W, r+2W, ...
-      3. gather_object(shard_results) to rank 0
-    Rank 0 calls evaluate(); other ranks sit in worker_loop() until the
-    controller broadcasts ("stop",).
+    Every rank calls :meth:`evaluate` with the IDENTICAL points list (the
+    replicated-control-flow invariant). Each rank runs the user objective on
+    rows [rank::world], packs each result into a fixed-layout float64 row
+    (per-problem objectives, constraints, then the eval time) plus an
+    optional uint8 feature-record row, and the full batch is reassembled
+    from ONE float all_gather (+ one byte all_gather when features are
+    configured). The returned result dicts — {pid: y | (y,f) | (y,f,c) |
+    (y,c), "time": t} — are bit-identical on every rank.
     """
 
-    def __init__(self, eval_funs: Dict[str, Callable]):
-        assert dist_is_initialized()
+    def __init__(self, eval_funs: Dict[str, Callable], spec: ResultSpec):
+        ctx = get_context()
+        assert ctx is not None and ctx.world > 1
+        self.ctx = ctx
         self.eval_funs = eval_funs
-        self.rank = dist.get_rank()
-        self.world = dist.get_world_size()
-        self._stats = _FarmStats(self.world)
+        self.spec = spec
+        self._stats = _FarmStats(ctx.world)
 
-    # ------------------------------------------------------------ controller
+    # ---------------------------------------------------------------- pack
+    def _pack_one(self, res: Dict, frow: np.ndarray, brow: Optional[np.ndarray]):
+        spec = self.spec
+        col = 0
+        bcol = 0
+        for pid in spec.problem_ids:
+            entry = res[pid]
+            f = c = None
+            if isinstance(entry, tuple):
+                if len(entry) == 3:
+                    y, f, c = entry
+                elif len(entry) == 2:
+                    if spec.feature_dtype is not None:
+                        y, f = entry
+                    else:
+                        y, c = entry
+                else:
+                    y = entry[0]
+            else:
+                y = entry
+            y = np.asarray(y, dtype=np.float64).ravel()
+            if y.shape[0] != spec.n_objectives:
+                raise ValueError(
+                    f"objective returned {y.shape[0]} values for problem {pid}; "
+                    f"expected {spec.n_objectives}"
+                )
+            frow[col : col + spec.n_objectives] = y
+            col += spec.n_objectives
+            if spec.n_constraints:
+                cv = np.asarray(c, dtype=np.float64).ravel()
+                if cv.shape[0] != spec.n_constraints:
+                    raise ValueError(
+                        f"objective returned {cv.shape[0]} constraints for "
+                        f"problem {pid}; expected {spec.n_constraints}"
+                    )
+                frow[col : col + spec.n_constraints] = cv
+                col += spec.n_constraints
+            if spec.feature_dtype is not None:
+                fa = np.asarray(f)
+                raw = fa.tobytes()
+                if len(raw) != spec.feature_itemsize:
+                    raise ValueError(
+                        "multi-rank feature transport requires one fixed-size "
+                        f"record per evaluation: got {len(raw)} bytes, dtype "
+                        f"itemsize is {spec.feature_itemsize} (problem {pid})"
+                    )
+                brow[bcol : bcol + spec.feature_itemsize] = np.frombuffer(raw, np.uint8)
+                bcol += spec.feature_itemsize
+        frow[col] = float(res.get("time", -1.0))
+
+    def _unpack_one(self, frow: np.ndarray, brow: Optional[np.ndarray]) -> Dict:
+        spec = self.spec
+        out: Dict = {}
+        col = 0
+        bcol = 0
+        for pid in spec.problem_ids:
+            y = frow[col : col + spec.n_objectives].copy()
+            col += spec.n_objectives
+            c = None
+            if spec.n_constraints:
+                c = frow[col : col + spec.n_constraints].copy()
+                col += spec.n_constraints
+            f = None
+            if spec.feature_dtype is not None:
+                f = np.frombuffer(
+                    brow[bcol : bcol + spec.feature_itemsize].tobytes(),
+                    dtype=spec.feature_dtype,
+                )
+                bcol += spec.feature_itemsize
+            if f is not None and c is not None:
+                out[pid] = (y, f, c)
+            elif f is not None:
+                out[pid] = (y, f)
+            elif c is not None:
+                out[pid] = (y, c)
+            else:
+                out[pid] = y
+        out["time"] = float(frow[-1])
+        return out
+
+    # ------------------------------------------------------------ evaluate
     def evaluate(self, opt_id: str, points: List) -> List:
-        assert self.rank == 0
+        ctx, spec = self.ctx, self.spec
+        fn = self.eval_funs[opt_id]
+        P = len(points)
         t_call = time.time()
-        cmd = [("eval", opt_id, points)]
-        dist.broadcast_object_list(cmd, src=0)
-        my_results = self._eval_shard(opt_id, points)
-        gathered: List = [None] * self.world
-        dist.gather_object(my_results, gathered, dst=0)
-        # interleave shards back into original order
-        out: List = [None] * len(points)
-        for r, shard in enumerate(gathered):
-            shard_t = 0.0
-            for j, res in enumerate(shard):
-                out[r + j * self.world] = res
-                if isinstance(res, dict) and "time" in res:
-                    shard_t += res["time"]
-            self._stats.record(r, len(shard), shard_t)
+        my_idx = list(ctx.shard_indices(P))
+        max_shard = ctx.max_shard_size(P)
+        fbuf = np.zeros((max_shard, spec.float_width), dtype=np.float64)
+        bbuf = (
+            np.zeros((max_shard, spec.byte_width), dtype=np.uint8)
+            if spec.byte_width
+            else None
+        )
+        for j, i in enumerate(my_idx):
+            res = fn(points[i])
+            self._pack_one(res, fbuf[j], bbuf[j] if bbuf is not None else None)
+
+        full_f = ctx.all_gather_interleaved(torch.from_numpy(fbuf), P).cpu().numpy()
+        full_b = None
+        if bbuf is not None:
+            full_b = ctx.all_gather_interleaved(torch.from_numpy(bbuf), P).cpu().numpy()
+
+        out: List = []
+        for i in range(P):
+            out.append(
+                self._unpack_one(full_f[i], full_b[i] if full_b is not None else None)
+            )
+        # per-rank accounting from the gathered time column (identical on
+        # every rank, so the stats written to H5 by rank 0 cover everyone)
+        for r in range(ctx.world):
+            rows = range(r, P, ctx.world)
+            self._stats.record(
+                r, len(rows), float(sum(max(full_f[i][-1], 0.0) for i in rows))
+            )
         self._stats.call_times.append(time.time() - t_call)
         return out
 
@@ -157,30 +305,11 @@ class TorchDistFarm:
         return self._stats.summary()
 
     def shutdown(self):
-        if self.rank == 0:
-            dist.broadcast_object_list([("stop",)], src=0)
-
-    # --------------------------------------------------------------- worker
-    def worker_loop(self):
-        assert self.rank != 0
-        while True:
-            cmd = [None]
-            dist.broadcast_object_list(cmd, src=0)
-            tag = cmd[0][0]
-            if tag == "stop":
-                break
-            if tag == "eval":
-                _, opt_id, points = cmd[0]
-                my_results = self._eval_shard(opt_id, points)
-                dist.gather_object(my_results, None, dst=0)
-
-    # --------------------------------------------------------------- shared
-    def _eval_shard(self, opt_id: str, points: List) -> List:
-        fn = self.eval_funs[opt_id]
-        return [fn(points[i]) for i in range(self.rank, len(points), self.world)]
+        pass
 
 
-def make_farm(eval_funs: Dict[str, Callable]):
+def make_farm(eval_funs: Dict[str, Callable], spec: Optional[ResultSpec] = None):
     if dist_is_initialized() and dist.get_world_size() > 1:
-        return TorchDistFarm(eval_funs)
+        assert spec is not None, "multi-rank farm needs a ResultSpec"
+        return CollectiveFarm(eval_funs, spec)
     return LocalFarm(eval_funs)
