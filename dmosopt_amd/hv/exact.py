@@ -18,16 +18,73 @@ import math
 from typing import List, Optional, Tuple
 
 import numpy as np
+import torch
 from scipy.stats import norm
 
 
+def _device_ready() -> bool:
+    from dmosopt_amd import ops
+
+    return torch.cuda.is_available() and ops.native_available()
+
+
+def _as_dev64(a) -> torch.Tensor:
+    if isinstance(a, torch.Tensor):
+        return a.to("cuda", torch.float64).contiguous()
+    return torch.as_tensor(np.asarray(a, dtype=np.float64), device="cuda")
+
+
 # ------------------------------------------------------------------ 2D / 3D
-def hv_2d(points: np.ndarray, ref_point: np.ndarray) -> float:
+def hv_2d_device(points, ref_point) -> float:
+    """2D hypervolume on the GPU (ops/hip/hv_exact.hip hv2d_kernel): LDS
+    bitonic sort + prefix-min staircase integral; accepts device tensors
+    directly so device-resident fronts never round-trip to the host."""
+    from dmosopt_amd import _hipops
+
+    P = _as_dev64(points)
+    if P.numel() == 0:
+        return 0.0
+    return float(_hipops.hv2d(P, _as_dev64(ref_point)))
+
+
+def hv_3d_device(points, ref_point) -> float:
+    """3D hypervolume on the GPU: device sorts (torch) + one slice-parallel
+    sweep kernel (ops/hip/hv_exact.hip hv3d_slices_kernel)."""
+    from dmosopt_amd import _hipops
+
+    P = _as_dev64(points)
+    ref = _as_dev64(ref_point)
+    P = P[(P < ref[None, :]).all(dim=1)]
+    n = P.shape[0]
+    if n == 0:
+        return 0.0
+    z_order = torch.argsort(P[:, 2], stable=True)
+    Pz = P[z_order]
+    z_thr = Pz[:, 2].contiguous()
+    dz = torch.empty_like(z_thr)
+    dz[:-1] = z_thr[1:] - z_thr[:-1]
+    dz[-1] = ref[2] - z_thr[-1]
+    x_order = torch.argsort(P[:, 0], stable=True)
+    Px = P[x_order].contiguous()
+    out = _hipops.hv3d_slices(Px, z_thr, dz, ref)
+    return float(out.sum().item())
+
+
+def hv_2d(points, ref_point) -> float:
     """Vectorized 2D hypervolume: sort by (f0, f1), prefix-min prune, swept
     area. Note: this matches the moocore/exact semantics; the reference's
     pure-Python fallback (hv_box_decomposition.py:44-78) prunes with an
     inverted scan and under-counts dominated staircases — its test suite
     gates correctness against moocore, which we reproduce here."""
+    if isinstance(points, torch.Tensor) and points.is_cuda:
+        return hv_2d_device(points, ref_point)
+    points = np.asarray(points, dtype=np.float64)
+    ref_point = np.asarray(ref_point, dtype=np.float64)
+    if len(points) >= 64 and _device_ready():
+        # H2D of an (n,2) block + one kernel beats the host lexsort well
+        # before n=8192 (the kernel's LDS capacity); larger fronts stay host
+        if len(points) <= 8192:
+            return hv_2d_device(points, ref_point)
     pts = points[np.all(points < ref_point, axis=1)]
     if len(pts) == 0:
         return 0.0
@@ -49,8 +106,14 @@ def hv_2d(points: np.ndarray, ref_point: np.ndarray) -> float:
     return float(np.dot(x_next - pts[:, 0], ref_point[1] - pts[:, 1]))
 
 
-def hv_3d(points: np.ndarray, ref_point: np.ndarray) -> float:
+def hv_3d(points, ref_point) -> float:
     """3D hypervolume: z-sorted plane sweep, 2D HV per slice."""
+    if isinstance(points, torch.Tensor) and points.is_cuda:
+        return hv_3d_device(points, ref_point)
+    points = np.asarray(points, dtype=np.float64)
+    ref_point = np.asarray(ref_point, dtype=np.float64)
+    if len(points) >= 48 and _device_ready():
+        return hv_3d_device(points, ref_point)
     pts = points[np.all(points < ref_point, axis=1)]
     if len(pts) == 0:
         return 0.0
@@ -169,6 +232,96 @@ class _FlatUBSet:
         return np.where(ok, vol, 0.0)
 
 
+class _DeviceUBSet:
+    """Device-resident Lacour local-upper-bound set (the stream-compaction
+    variant of _FlatUBSet): per insertion, a flag kernel marks dominated UBs
+    and admissible step-3 candidates, torch cumsums assign deterministic
+    output slots in the exact numpy emission order, and a scatter kernel
+    writes the compacted successor set (ops/hip/hv_exact.hip lacour_*)."""
+
+    def __init__(self, ref_point: np.ndarray, points: np.ndarray):
+        from dmosopt_amd import _hipops  # noqa: F401  (require native)
+
+        self.ref = np.asarray(ref_point, dtype=np.float64)
+        d = self.d = len(self.ref)
+        n = self.n = len(points)
+        dummies = np.zeros((d, d))
+        dummies[np.arange(d), np.arange(d)] = self.ref
+        self.pts_aug = torch.as_tensor(
+            np.vstack([points, dummies]), device="cuda"
+        ).contiguous()
+        self.ref_t = torch.as_tensor(self.ref, device="cuda")
+        self.coords = self.ref_t[None, :].clone().contiguous()
+        self.defs = torch.arange(
+            n, n + d, dtype=torch.int64, device="cuda"
+        )[None, :].contiguous()
+
+    def insert(self, point_idx: int) -> None:
+        from dmosopt_amd import _hipops
+
+        U, d = self.coords.shape
+        if U == 0:
+            return
+        z = self.pts_aug[point_idx].contiguous()
+        dominated, okj = _hipops.lacour_flags(
+            self.coords, self.defs, self.pts_aug, z
+        )
+        dom64 = dominated.to(torch.int64)
+        ok64 = okj.to(torch.int64)
+        counts = torch.cat([dom64.sum()[None], ok64.sum(dim=0)]).cpu()  # 1 sync
+        nA = int(counts[0])
+        if nA == 0:
+            return
+        nBj = counts[1 : d]  # per-j counts, j < d-1 (col d-1 is always 0)
+        slotA = torch.cumsum(dom64, 0) - dom64  # exclusive
+        slotBj = torch.cumsum(ok64, 0) - ok64  # per-column exclusive
+        baseBj_np = np.zeros(d, dtype=np.int64)
+        run = nA
+        for j in range(d - 1):
+            baseBj_np[j] = run
+            run += int(nBj[j])
+        baseK = run
+        total = baseK + (U - nA)
+        out_coords = torch.empty((total, d), dtype=torch.float64, device="cuda")
+        out_defs = torch.empty((total, d), dtype=torch.int64, device="cuda")
+        baseBj = torch.as_tensor(baseBj_np, device="cuda")
+        _hipops.lacour_scatter(
+            self.coords, self.defs, z, dominated, okj,
+            slotA.contiguous(), slotBj.contiguous(), baseBj,
+            int(baseK), int(point_idx), out_coords, out_defs,
+        )
+        # dedupe by rounded coordinate rows, keep first occurrence (same
+        # semantics as the numpy np.unique(..., return_index) path)
+        key = torch.round(out_coords, decimals=15)
+        _, inverse = torch.unique(key, dim=0, return_inverse=True)
+        n_uniq = int(inverse.max().item()) + 1 if total > 0 else 0
+        first = torch.full((n_uniq,), total, dtype=torch.int64, device="cuda")
+        first.scatter_reduce_(
+            0, inverse, torch.arange(total, device="cuda"), reduce="amin"
+        )
+        first, _ = torch.sort(first)
+        self.coords = out_coords[first].contiguous()
+        self.defs = out_defs[first].contiguous()
+
+    def volume(self) -> float:
+        from dmosopt_amd import _hipops
+
+        if self.coords.shape[0] == 0:
+            return 0.0
+        vol = _hipops.lacour_volumes(
+            self.coords, self.defs, self.pts_aug, self.ref_t
+        )
+        return float(vol.sum().item())
+
+
+def lacour_hv_device(points: np.ndarray, ref_point: np.ndarray) -> float:
+    """Full d>=4 hypervolume via the device UB-set."""
+    ubset = _DeviceUBSet(ref_point, points)
+    for i in range(len(points)):
+        ubset.insert(i)
+    return ubset.volume()
+
+
 class HyperVolumeBoxDecomposition:
     """Drop-in equivalent of the reference class (hv_box_decomposition.py:155).
 
@@ -196,6 +349,8 @@ class HyperVolumeBoxDecomposition:
         if len(pts) == 0:
             return 0.0
         pts = pts[np.argsort(pts[:, -1], kind="stable")]
+        if len(pts) >= 48 and _device_ready():
+            return lacour_hv_device(pts, self.ref_point)
         ubset = _FlatUBSet(self.ref_point, pts)
         for i in range(len(pts)):
             ubset.insert(i)
@@ -226,6 +381,8 @@ class HyperVolumeBoxDecomposition:
         dims, summed over boxes. Large batches route to the torch (GPU when
         available) implementation.
         """
+        if _device_ready() and means.shape[0] * lowers.shape[0] >= 2048:
+            return self._batch_ehvi_device(lowers, uppers, means, variances)
         if means.shape[0] * lowers.shape[0] * lowers.shape[1] > 200_000:
             return self._batch_ehvi_torch(lowers, uppers, means, variances)
         std = np.sqrt(variances)[:, None, :]  # (B, 1, d)
@@ -241,6 +398,17 @@ class HyperVolumeBoxDecomposition:
         phi_u = np.where(np.isinf(U), 0.0, norm.pdf(zu))
         partial = std * (phi_l - phi_u) + mu * (Phi_u - Phi_l)
         return partial.prod(axis=2).sum(axis=1)
+
+    def _batch_ehvi_device(self, lowers, uppers, means, variances) -> np.ndarray:
+        """EHVI via the fp64 HIP kernel (ops/hip/hv_exact.hip ehvi_kernel):
+        one thread per candidate, boxes streamed from L2."""
+        from dmosopt_amd import _hipops
+
+        L = _as_dev64(lowers)
+        U = _as_dev64(uppers)
+        mu = _as_dev64(means)
+        var = _as_dev64(variances)
+        return _hipops.ehvi_batch(L, U, mu, var).cpu().numpy()
 
     def _batch_ehvi_torch(self, lowers, uppers, means, variances) -> np.ndarray:
         """Device-capable EHVI: same math with torch (erf-based normal cdf),

@@ -51,6 +51,21 @@ void launch_variation_slots(const float*, const long long*, const long long*, co
 void launch_gather3(const float*, const float*, const long long*, const long long*, float*, float*, long long*, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
+void launch_hv2d(const double*, const double*, double*, int, hipStream_t);
+void launch_hv3d_slices(const double*, const double*, const double*,
+                        const double*, double*, int, hipStream_t);
+void launch_ehvi(const double*, const double*, const double*, const double*,
+                 double*, int, int, int, hipStream_t);
+void launch_lacour_flags(const double*, const long long*, const double*,
+                         const double*, unsigned char*, unsigned char*, int,
+                         int, hipStream_t);
+void launch_lacour_scatter(const double*, const long long*, const double*,
+                           const unsigned char*, const unsigned char*,
+                           const long long*, const long long*,
+                           const long long*, long long, long long, double*,
+                           long long*, int, int, hipStream_t);
+void launch_lacour_volumes(const double*, const long long*, const double*,
+                           const double*, double*, int, int, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -474,6 +489,104 @@ void cmaes_update_(torch::Tensor A, torch::Tensor Ainv, torch::Tensor pc,
                       (float)pthresh, cur_stream());
 }
 
+// ------------------------------------------------------- exact hypervolume
+double hv2d(torch::Tensor P, torch::Tensor ref) {
+  CHECK_GPU(P);
+  CHECK_GPU(ref);
+  TORCH_CHECK(P.scalar_type() == torch::kFloat64, "hv2d expects float64");
+  const int n = P.size(0);
+  if (n == 0) return 0.0;
+  TORCH_CHECK(n <= 8192, "hv2d kernel supports n <= 8192");
+  auto out = torch::zeros({1}, P.options());
+  launch_hv2d(P.data_ptr<double>(), ref.data_ptr<double>(),
+              out.data_ptr<double>(), n, cur_stream());
+  return out.item<double>();
+}
+
+torch::Tensor hv3d_slices(torch::Tensor Px, torch::Tensor z_thr,
+                          torch::Tensor dz, torch::Tensor ref) {
+  CHECK_GPU(Px);
+  CHECK_GPU(z_thr);
+  CHECK_GPU(dz);
+  CHECK_GPU(ref);
+  TORCH_CHECK(Px.scalar_type() == torch::kFloat64, "hv3d expects float64");
+  const int n = Px.size(0);
+  auto out = torch::zeros({std::max(n, 1)}, Px.options());
+  if (n > 0)
+    launch_hv3d_slices(Px.data_ptr<double>(), z_thr.data_ptr<double>(),
+                       dz.data_ptr<double>(), ref.data_ptr<double>(),
+                       out.data_ptr<double>(), n, cur_stream());
+  return out;
+}
+
+torch::Tensor ehvi_batch(torch::Tensor L, torch::Tensor U, torch::Tensor mu,
+                         torch::Tensor var) {
+  CHECK_GPU(L);
+  CHECK_GPU(U);
+  CHECK_GPU(mu);
+  CHECK_GPU(var);
+  TORCH_CHECK(mu.scalar_type() == torch::kFloat64, "ehvi expects float64");
+  const int B = mu.size(0), nb = L.size(0), d = mu.size(1);
+  TORCH_CHECK(d <= 16, "ehvi kernel supports d <= 16");
+  auto out = torch::zeros({B}, mu.options());
+  if (B > 0 && nb > 0)
+    launch_ehvi(L.data_ptr<double>(), U.data_ptr<double>(),
+                mu.data_ptr<double>(), var.data_ptr<double>(),
+                out.data_ptr<double>(), B, nb, d, cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> lacour_flags(torch::Tensor coords,
+                                        torch::Tensor defs,
+                                        torch::Tensor pts_aug,
+                                        torch::Tensor z) {
+  CHECK_GPU(coords);
+  CHECK_GPU(defs);
+  CHECK_GPU(pts_aug);
+  CHECK_GPU(z);
+  const int U = coords.size(0), d = coords.size(1);
+  auto opts_u8 = coords.options().dtype(torch::kUInt8);
+  auto dominated = torch::zeros({U}, opts_u8);
+  auto okj = torch::zeros({U, d}, opts_u8);
+  launch_lacour_flags(coords.data_ptr<double>(),
+                      (const long long*)defs.data_ptr<int64_t>(),
+                      pts_aug.data_ptr<double>(), z.data_ptr<double>(),
+                      dominated.data_ptr<uint8_t>(), okj.data_ptr<uint8_t>(),
+                      U, d, cur_stream());
+  return {dominated, okj};
+}
+
+void lacour_scatter(torch::Tensor coords, torch::Tensor defs, torch::Tensor z,
+                    torch::Tensor dominated, torch::Tensor okj,
+                    torch::Tensor slotA, torch::Tensor slotBj,
+                    torch::Tensor baseBj, int64_t baseK, int64_t point_idx,
+                    torch::Tensor out_coords, torch::Tensor out_defs) {
+  CHECK_GPU(coords);
+  CHECK_GPU(out_coords);
+  const int U = coords.size(0), d = coords.size(1);
+  launch_lacour_scatter(
+      coords.data_ptr<double>(), (const long long*)defs.data_ptr<int64_t>(),
+      z.data_ptr<double>(), dominated.data_ptr<uint8_t>(),
+      okj.data_ptr<uint8_t>(), (const long long*)slotA.data_ptr<int64_t>(),
+      (const long long*)slotBj.data_ptr<int64_t>(),
+      (const long long*)baseBj.data_ptr<int64_t>(), baseK, point_idx,
+      out_coords.data_ptr<double>(), (long long*)out_defs.data_ptr<int64_t>(),
+      U, d, cur_stream());
+}
+
+torch::Tensor lacour_volumes(torch::Tensor coords, torch::Tensor defs,
+                             torch::Tensor pts_aug, torch::Tensor ref) {
+  CHECK_GPU(coords);
+  const int U = coords.size(0), d = coords.size(1);
+  auto vol = torch::zeros({std::max(U, 1)}, coords.options());
+  if (U > 0)
+    launch_lacour_volumes(coords.data_ptr<double>(),
+                          (const long long*)defs.data_ptr<int64_t>(),
+                          pts_aug.data_ptr<double>(), ref.data_ptr<double>(),
+                          vol.data_ptr<double>(), U, d, cur_stream());
+  return vol.narrow(0, 0, U);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
@@ -500,6 +613,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hv_mc_uniform_hits", &hv_mc_uniform_hits);
   m.def("hv_fpras_hits", &hv_fpras_hits);
   m.def("get_duplicates", &get_duplicates);
+  m.def("hv2d", &hv2d, "Exact 2D hypervolume (LDS bitonic staircase sweep)");
+  m.def("hv3d_slices", &hv3d_slices,
+        "Per-slice 2D sweep terms of the 3D hypervolume");
+  m.def("ehvi_batch", &ehvi_batch,
+        "Batched EHVI over the dominated-space box decomposition");
+  m.def("lacour_flags", &lacour_flags);
+  m.def("lacour_scatter", &lacour_scatter);
+  m.def("lacour_volumes", &lacour_volumes);
   m.def("cmaes_update_", &cmaes_update_,
         "In-place batched MO-CMA-ES rank-1 Cholesky update");
 }
