@@ -100,8 +100,11 @@ def one_epoch(X, Y, pop, rank, world, device, seed, n_gen=N_GEN):
     x_res = best_x[idx]
     y_res = zdt1(x_res.double())  # on-device synthetic objective evaluation
     # final hypervolume of the REAL-evaluated resample batch (the quality
-    # half of the headline metric)
-    hv = hv_2d(y_res.cpu().numpy(), np.array([11.0, 11.0]))
+    # half of the headline metric); device-resident fronts stay on device
+    if y_res.is_cuda:
+        hv = hv_2d(y_res, np.array([11.0, 11.0]))
+    else:
+        hv = hv_2d(y_res.cpu().numpy(), np.array([11.0, 11.0]))
     return x_res, y_res, hv
 
 

@@ -529,9 +529,21 @@ def test_ehvi_kernel_matches_oracle(dev):
         L, U = box._decompose_dominated_space(front)
         mu = rng.random((300, d)) * 1.5
         var = rng.random((300, d)) * 0.2 + 0.01
-        want = box._batch_ehvi_torch(L, U, mu, var)  # fp64 torch oracle (CPU path semantics)
+        # explicit fp64 scipy oracle (the _batch_ehvi numpy small-path math)
+        from scipy.stats import norm
+
+        std = np.sqrt(var)[:, None, :]
+        mub = mu[:, None, :]
+        Lb, Ub = L[None, :, :], U[None, :, :]
+        with np.errstate(invalid="ignore"):
+            zl, zu = (Lb - mub) / std, (Ub - mub) / std
+        Phi_l = np.where(np.isinf(Lb), 0.0, norm.cdf(zl))
+        Phi_u = np.where(np.isinf(Ub), 1.0, norm.cdf(zu))
+        phi_l = np.where(np.isinf(Lb), 0.0, norm.pdf(zl))
+        phi_u = np.where(np.isinf(Ub), 0.0, norm.pdf(zu))
+        want = (std * (phi_l - phi_u) + mub * (Phi_u - Phi_l)).prod(axis=2).sum(axis=1)
         got = box._batch_ehvi_device(L, U, mu, var)
-        np.testing.assert_allclose(got, want, rtol=1e-6, atol=1e-10)
+        np.testing.assert_allclose(got, want, rtol=1e-9, atol=1e-12)
 
 
 def test_lacour_device_matches_numpy(dev):
