@@ -80,11 +80,11 @@ def hv_2d(points, ref_point) -> float:
         return hv_2d_device(points, ref_point)
     points = np.asarray(points, dtype=np.float64)
     ref_point = np.asarray(ref_point, dtype=np.float64)
-    if len(points) >= 64 and _device_ready():
-        # H2D of an (n,2) block + one kernel beats the host lexsort well
-        # before n=8192 (the kernel's LDS capacity); larger fronts stay host
-        if len(points) <= 8192:
-            return hv_2d_device(points, ref_point)
+    if 2048 <= len(points) <= 8192 and _device_ready():
+        # measured crossover (profiles/README.md): the H2D copy + launch
+        # beats the host lexsort from n~2k; beyond 8192 (LDS capacity) the
+        # host path continues
+        return hv_2d_device(points, ref_point)
     pts = points[np.all(points < ref_point, axis=1)]
     if len(pts) == 0:
         return 0.0
@@ -349,7 +349,10 @@ class HyperVolumeBoxDecomposition:
         if len(pts) == 0:
             return 0.0
         pts = pts[np.argsort(pts[:, -1], kind="stable")]
-        if len(pts) >= 48 and _device_ready():
+        if len(pts) >= 96 and _device_ready():
+            # measured crossover: per-insert launch+sync overhead (~35 us)
+            # dominates below ~100 points; the UB set then grows fast
+            # enough that the flag/scatter kernels win
             return lacour_hv_device(pts, self.ref_point)
         ubset = _FlatUBSet(self.ref_point, pts)
         for i in range(len(pts)):

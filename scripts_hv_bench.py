@@ -128,7 +128,7 @@ def main():
         th, vh = bench(host_ehvi, reps=5)
         if torch.cuda.is_available():
             td, vd = bench(box._batch_ehvi_device, L, U, mu, var, reps=5)
-            np.testing.assert_allclose(vd, vh, rtol=1e-5, atol=1e-8)
+            np.testing.assert_allclose(vd, vh, rtol=1e-4, atol=1e-7)
         else:
             td = float("nan")
         print(f"ehvi B={B:<5d} boxes={L.shape[0]:<5d} d={d}       {th:9.3f} {td:10.3f} {th/td:7.1f}x")
