@@ -53,17 +53,21 @@ def make_archive(seed: int) -> tuple:
     return X, Y
 
 
-def one_epoch(X, Y, pop, rank, world, device, seed, n_gen=N_GEN):
+def one_epoch(X, Y, pop, rank, world, device, seed, n_gen=N_GEN, compute="fp32"):
     """One MO-ASMO epoch; returns (resample_x, predicted_y, hv).
 
     Multi-rank path = the PACKAGE path: rank 0 fits the GP and broadcasts
     theta (engine.train), predictions are rank-sharded and all-gathered
-    each generation (parallel.sharded.ShardedObjective)."""
+    each generation (parallel.sharded.ShardedObjective). compute="bf16"
+    runs the posterior path (cross kernel + Cholesky trailing updates) on
+    the bf16 matrix units; the SCE-UA hyperparameter search stays fp32, so
+    theta is identical between the two modes."""
     gp = engine.train(
         D_IN, N_OBJ, np.zeros(D_IN), np.ones(D_IN), X, Y, None,
         surrogate_method_name="gpr",
         surrogate_method_kwargs={
             "anisotropic": False, "optimizer": "sceua", "seed": seed,
+            "compute": compute,
         },
         logger=None, device=device,
     )
@@ -114,7 +118,9 @@ def main():
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--gens", type=int, default=N_GEN)
+    ap.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
     args = ap.parse_args()
+    compute = "bf16" if args.dtype == "bf16" else "fp32"
 
     rank, world = comm.init_from_env()
     if torch.cuda.is_available():
@@ -136,14 +142,48 @@ def main():
 
     hv = None
     for w in range(args.warmup):
-        one_epoch(X, Y, pop, rank, world, device, seed=100 + w, n_gen=args.gens)
+        one_epoch(X, Y, pop, rank, world, device, seed=100 + w, n_gen=args.gens,
+                  compute=compute)
 
     barrier_sync()
     t0 = time.perf_counter()
     for s in range(args.steps):
-        _, _, hv = one_epoch(X, Y, pop, rank, world, device, seed=200 + s, n_gen=args.gens)
+        _, _, hv = one_epoch(X, Y, pop, rank, world, device, seed=200 + s,
+                             n_gen=args.gens, compute=compute)
     barrier_sync()
     t1 = time.perf_counter()
+
+    accuracy = None
+    if compute == "bf16" and rank == 0 and device.type == "cuda":
+        # untimed accuracy delta vs fp32: same seed => identical theta (the
+        # search is fp32 in both modes), so the delta isolates the bf16
+        # posterior path (cross kernel + Cholesky trailing updates)
+        gp32 = engine.train(
+            D_IN, N_OBJ, np.zeros(D_IN), np.ones(D_IN), X, Y, None,
+            surrogate_method_name="gpr",
+            surrogate_method_kwargs={"anisotropic": False, "optimizer": "sceua",
+                                     "seed": 200, "compute": "fp32"},
+            logger=None, device=device)
+        gp16 = engine.train(
+            D_IN, N_OBJ, np.zeros(D_IN), np.ones(D_IN), X, Y, None,
+            surrogate_method_name="gpr",
+            surrogate_method_kwargs={"anisotropic": False, "optimizer": "sceua",
+                                     "seed": 200, "compute": "bf16"},
+            logger=None, device=device)
+        probe = np.random.default_rng(0).random((2048, D_IN))
+        m32, _ = gp32.predict(probe)
+        m16, _ = gp16.predict(probe)
+        scale = np.abs(m32).mean()
+        _, _, hv32 = one_epoch(X, Y, pop, rank, world, device, seed=200,
+                               n_gen=args.gens, compute="fp32")
+        accuracy = {
+            "posterior_mean_abs_err_max": float(np.abs(m16 - m32).max()),
+            "posterior_mean_abs_err_mean": float(np.abs(m16 - m32).mean()),
+            "posterior_mean_rel_err_mean": float(np.abs(m16 - m32).mean() / scale),
+            "final_hv_fp32": float(hv32),
+            "final_hv_bf16": float(hv),
+            "hv_rel_delta": float(abs(hv - hv32) / max(abs(hv32), 1e-12)),
+        }
 
     elapsed = t1 - t0
     if world > 1:
@@ -166,7 +206,7 @@ def main():
             "higher_is_better": False,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic ZDT1 d=30, random-init archive of 300 evals",
             "config": {
                 "model": "GPR-Matern52 surrogate + NSGA2",
@@ -178,6 +218,7 @@ def main():
                 "archive_size": ARCHIVE_N,
                 "final_hypervolume_ref11": hv,
                 "reference_measured_ms_per_epoch_cpu": 19200.0,
+                "bf16_accuracy_vs_fp32": accuracy,
             },
         }
         print(json.dumps(out))

@@ -66,6 +66,7 @@ class _GPRBase:
         batch_size=None,
         device=None,
         dtype=None,
+        compute="fp32",
         logger=None,
         theta_override=None,
         **kwargs,
@@ -83,6 +84,9 @@ class _GPRBase:
             torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
         )
         self.dtype = dtype or (torch.float64 if self.device.type == "cpu" else torch.float32)
+        # "bf16": bf16-MFMA posterior path (predictions + trailing Cholesky
+        # updates); the hyperparameter SEARCH below is always fp32
+        self.compute = compute
 
         xin = np.asarray(xin, dtype=np.float64)
         yin = np.asarray(yin, dtype=np.float64)
@@ -131,6 +135,7 @@ class _GPRBase:
             self._fitted = FittedGP(
                 X, Y, theta, y_mean, y_std, nu=self.nu, anisotropic=anisotropic,
                 jitter=1e-10 if self.dtype == torch.float64 else 1e-6,
+                compute=self.compute,
             )
             self.theta = theta
             return
@@ -164,6 +169,7 @@ class _GPRBase:
         self._fitted = FittedGP(
             X, Y, theta, y_mean, y_std, nu=self.nu, anisotropic=anisotropic,
             jitter=1e-10 if self.dtype == torch.float64 else 1e-6,
+            compute=self.compute,
         )
         self.theta = theta
 
@@ -244,7 +250,11 @@ class _GPRBase:
     def evaluate_tensor(self, x: torch.Tensor) -> torch.Tensor:
         """Device-resident evaluate: tensor in, tensor out, no host trip."""
         xr = x.to(self.device, self.dtype)
-        if not self.return_mean_variance and xr.device.type == "cuda":
+        if (
+            not self.return_mean_variance
+            and xr.device.type == "cuda"
+            and self.compute != "bf16"
+        ):
             from dmosopt_amd import ops
 
             if ops.native_available():

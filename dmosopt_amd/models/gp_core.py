@@ -178,6 +178,7 @@ class FittedGP:
         nu: float = 2.5,
         anisotropic: bool = False,
         jitter: float = 1e-10,
+        compute: str = "fp32",
     ):
         self.X = X
         self.theta = theta
@@ -185,11 +186,18 @@ class FittedGP:
         self.anisotropic = anisotropic
         self.y_mean = y_mean  # (m,)
         self.y_std = y_std  # (m,)
+        # "bf16": posterior Cholesky uses bf16-MFMA trailing updates and
+        # predictions use the bf16 cross kernel (BASELINE config #2); the
+        # SCE-UA fit that produced theta is always fp32 (bit-stability)
+        self.compute = compute if X.is_cuda else "fp32"
         from dmosopt_amd import ops
 
         m, N = theta.shape[0], X.shape[0]
         K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
-        self.L, _, info = ops.chol_factor_batched(K)  # (m, N, N)
+        if self.compute == "bf16":
+            self.L, _, info = ops.chol_factor_batched_bf16(K)
+        else:
+            self.L, _, info = ops.chol_factor_batched(K)  # (m, N, N)
         if int(info.sum()) != 0:
             # escalate jitter for failed objectives
             for _ in range(5):
@@ -220,7 +228,7 @@ class FittedGP:
         ``return_var=False`` (the per-generation surrogate-evaluate path)
         the quadratic term is skipped and (mean, None) returned.
         """
-        if not return_var:
+        if not return_var and self.compute != "bf16":
             from dmosopt_amd import ops
 
             fused = ops.gp_predict_mean_fused(
@@ -229,7 +237,14 @@ class FittedGP:
             )
             if fused is not None:
                 return fused, None
-        Ks = build_kernel(Xq, self.X, self.theta, nu=self.nu, anisotropic=self.anisotropic)
+        if self.compute == "bf16":
+            from dmosopt_amd import ops
+
+            Ks = ops.matern_cross_bf16_kernel(
+                Xq.float(), self.X, self.theta, self.nu, self.anisotropic
+            )
+        else:
+            Ks = build_kernel(Xq, self.X, self.theta, nu=self.nu, anisotropic=self.anisotropic)
         # (m, P, N)
         mean_n = torch.bmm(Ks, self.alpha)[:, :, 0]  # (m, P)
         if not return_var:

@@ -298,6 +298,34 @@ def matern_cross_kernel(Xq, X, theta, nu, anisotropic):
     return gp_core.build_kernel_torch(Xq, X, theta, nu=nu, anisotropic=anisotropic)
 
 
+def matern_cross_bf16_kernel(Xq, X, theta, nu, anisotropic, q_lb=None, q_invrg=None):
+    """bf16-MFMA cross kernel (config #2 precision path): inputs rounded to
+    bf16 in LDS, dot products on v_mfma_f32_16x16x32_bf16 (fp32 accumulate),
+    Matern transform and output in fp32. GPU-only — no host fallback (the
+    bf16 route is explicitly requested, silence would hide a missing
+    extension)."""
+    if not (_use_native(X) and X.dtype == torch.float32):
+        raise RuntimeError(
+            "matern_cross_bf16 requires the gfx950 native extension and "
+            "float32 CUDA tensors"
+        )
+    nu_arg = 0.0 if (nu is None or nu == float("inf")) else float(nu)
+    return _native.matern_cross_bf16(
+        Xq.contiguous(), X.contiguous(), theta.contiguous().float(), nu_arg,
+        bool(anisotropic), q_lb, q_invrg,
+    )
+
+
+def chol_factor_batched_bf16(K):
+    """Batched Cholesky with the trailing SYRK updates on the bf16 matrix
+    units (panels stay exact fp32). Falls back to the fp32 factorization
+    when the native path is unavailable (CPU tests)."""
+    if _use_native(K) and K.dtype == torch.float32 and K.shape[1] > 32:
+        logdet, info = _native.cholesky_batched_bf16_(K)
+        return K, logdet, info
+    return chol_factor_batched(K)
+
+
 def chol_factor_batched(K):
     """Factor K (B,N,N) in place -> (L, logdet (B,), info (B,)).
 

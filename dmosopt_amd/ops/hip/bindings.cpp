@@ -66,6 +66,11 @@ void launch_lacour_scatter(const double*, const long long*, const double*,
                            long long*, int, int, hipStream_t);
 void launch_lacour_volumes(const double*, const long long*, const double*,
                            const double*, double*, int, int, hipStream_t);
+void launch_mfma_bf16_probe(const float*, const float*, float*, hipStream_t);
+void launch_matern_cross_bf16(const float*, const float*, const float*,
+                              float*, int, int, int, int, int, int, int,
+                              const float*, const float*, hipStream_t);
+void launch_cholesky_multik_bf16(float*, float*, int*, int, int, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -587,6 +592,45 @@ torch::Tensor lacour_volumes(torch::Tensor coords, torch::Tensor defs,
   return vol.narrow(0, 0, U);
 }
 
+// --------------------------------------------------------------- bf16 path
+torch::Tensor mfma_bf16_probe(torch::Tensor A, torch::Tensor B) {
+  CHECK_GPU(A);
+  CHECK_GPU(B);
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}) &&
+              B.sizes() == torch::IntArrayRef({32, 16}));
+  auto D = torch::zeros({16, 16}, A.options());
+  launch_mfma_bf16_probe(A.data_ptr<float>(), B.data_ptr<float>(),
+                         D.data_ptr<float>(), cur_stream());
+  return D;
+}
+
+torch::Tensor matern_cross_bf16(torch::Tensor Xq, torch::Tensor X,
+                                torch::Tensor theta, double nu, bool aniso,
+                                c10::optional<torch::Tensor> q_lb = c10::nullopt,
+                                c10::optional<torch::Tensor> q_invrg = c10::nullopt) {
+  CHECK_GPU(Xq);
+  CHECK_GPU(X);
+  CHECK_GPU(theta);
+  const int P = Xq.size(0), N = X.size(0), D = X.size(1), B = theta.size(0);
+  auto K = torch::empty({B, P, N}, X.options());
+  launch_matern_cross_bf16(
+      Xq.data_ptr<float>(), X.data_ptr<float>(), theta.data_ptr<float>(),
+      K.data_ptr<float>(), B, P, N, D, theta.size(1), nu_code(nu),
+      aniso ? 1 : 0, q_lb ? q_lb->data_ptr<float>() : nullptr,
+      q_invrg ? q_invrg->data_ptr<float>() : nullptr, cur_stream());
+  return K;
+}
+
+std::vector<torch::Tensor> cholesky_batched_bf16_(torch::Tensor A) {
+  CHECK_GPU(A);
+  const int B = A.size(0), N = A.size(1);
+  auto logdet = torch::empty({B}, A.options());
+  auto info = torch::zeros({B}, A.options().dtype(torch::kInt32));
+  launch_cholesky_multik_bf16(A.data_ptr<float>(), logdet.data_ptr<float>(),
+                              info.data_ptr<int>(), B, N, cur_stream());
+  return {logdet, info};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
@@ -613,6 +657,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hv_mc_uniform_hits", &hv_mc_uniform_hits);
   m.def("hv_fpras_hits", &hv_fpras_hits);
   m.def("get_duplicates", &get_duplicates);
+  m.def("mfma_bf16_probe", &mfma_bf16_probe,
+        "Single-tile bf16 MFMA fragment-layout probe");
+  m.def("matern_cross_bf16", &matern_cross_bf16,
+        "bf16-MFMA Matern cross-kernel assembly", py::arg("Xq"), py::arg("X"),
+        py::arg("theta"), py::arg("nu"), py::arg("aniso"),
+        py::arg("q_lb") = c10::nullopt, py::arg("q_invrg") = c10::nullopt);
+  m.def("cholesky_batched_bf16_", &cholesky_batched_bf16_,
+        "Batched Cholesky with bf16-MFMA trailing updates");
   m.def("hv2d", &hv2d, "Exact 2D hypervolume (LDS bitonic staircase sweep)");
   m.def("hv3d_slices", &hv3d_slices,
         "Per-slice 2D sweep terms of the 3D hypervolume");

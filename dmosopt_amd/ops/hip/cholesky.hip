@@ -653,6 +653,28 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
   if (have_rest_prev) hipStreamWaitEvent(stream, evR[(kstep - 1) & 1], 0);
 }
 
+// bf16-SYRK variant of the right-looking multik path (config-#2 precision
+// route): panels factor in exact fp32 (chol_panel_kernel), only the
+// trailing C -= P P^T runs on the bf16 matrix units (matern_bf16.hip).
+extern "C" void launch_chol_syrk_bf16(float*, int, int, int, int, int, int,
+                                      int, hipStream_t);
+
+extern "C" void launch_cholesky_multik_bf16(float* A, float* logdet,
+                                            int* info, int B, int N,
+                                            hipStream_t stream) {
+  hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
+  for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
+    hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0, stream,
+                       A, logdet, info, N, k0);
+    const int trailing = N - k0 - CHOL_BS;
+    if (trailing > 0) {
+      const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
+      launch_chol_syrk_bf16(A, N, k0, nt, -1, 0, nt * (nt + 1) / 2, B,
+                            stream);
+    }
+  }
+}
+
 extern "C" void launch_cholesky_batched(float* A, float* logdet, int* info,
                                         int B, int N, hipStream_t stream) {
   static int multik_max_b = -2;

@@ -582,3 +582,93 @@ def test_compute_hypervolume_device_route(dev):
     got = hv_2d(front.to(dev), ref2)
     want = _np_hv2d(front.numpy(), ref2)
     assert got == pytest.approx(want, rel=1e-10)
+
+
+# ----------------------------------------------------------------- bf16 path
+def test_mfma_bf16_probe_layout(dev):
+    """Verify the assumed A/B/C fragment mapping of mfma_f32_16x16x32_bf16
+    against a torch matmul oracle (asymmetric B per the guide's advice)."""
+    from dmosopt_amd import _hipops
+
+    g = torch.Generator().manual_seed(11)
+    A = torch.randn(16, 32, generator=g).float().to(dev)
+    B = (torch.arange(32 * 16, dtype=torch.float32).reshape(32, 16) * 0.01
+         + torch.randn(32, 16, generator=g)).to(dev)
+    D = _hipops.mfma_bf16_probe(A.contiguous(), B.contiguous())
+    want = (A.to(torch.bfloat16).float() @ B.to(torch.bfloat16).float())
+    torch.testing.assert_close(D, want, rtol=1e-5, atol=1e-5)
+
+
+def test_matern_cross_bf16_matches_rounded_oracle(dev):
+    from dmosopt_amd import ops
+
+    g = torch.Generator().manual_seed(12)
+    for P, N, Dd, aniso in ((200, 300, 30, False), (97, 131, 30, True), (64, 64, 100, False)):
+        Xq = torch.rand(P, Dd, generator=g).float().to(dev)
+        X = torch.rand(N, Dd, generator=g).float().to(dev)
+        p = Dd if aniso else 1
+        theta = torch.zeros(2, 2 + p)
+        theta[:, 0] = torch.tensor([0.0, 0.3])      # log sf2
+        theta[:, 1:-1] = torch.randn(2, p, generator=g) * 0.3  # log ell
+        theta[:, -1] = -10.0
+        theta = theta.float().to(dev)
+        K = ops.matern_cross_bf16_kernel(Xq, X, theta, 2.5, aniso)
+        # oracle: same math from the SAME bf16-rounded scaled inputs, fp32
+        for b in range(2):
+            ell = torch.exp(theta[b, 1:-1]) if aniso else torch.exp(theta[b, 1:2])
+            qs = (Xq / ell).to(torch.bfloat16).float()
+            xs = (X / ell).to(torch.bfloat16).float()
+            d2 = ((qs * qs).sum(1)[:, None] + (xs * xs).sum(1)[None, :]
+                  - 2.0 * qs @ xs.T).clamp_min(0)
+            r = torch.sqrt(d2)
+            s = math.sqrt(5.0) * r
+            want = torch.exp(theta[b, 0]) * (1 + s + (5.0 / 3.0) * d2) * torch.exp(-s)
+            torch.testing.assert_close(K[b], want, rtol=1e-3, atol=2e-4)
+        # and the bf16 rounding effect vs the exact fp32 kernel is bounded
+        K32 = ops.matern_cross_kernel(Xq, X, theta, 2.5, aniso)
+        assert float((K - K32).abs().max()) < 0.05 * float(torch.exp(theta[:, 0]).max())
+
+
+def test_cholesky_bf16_syrk_factorizes(dev):
+    from dmosopt_amd import ops
+
+    g = torch.Generator().manual_seed(13)
+    B, N = 4, 300
+    A = torch.randn(B, N, 40, generator=g).float()
+    K = (A @ A.transpose(1, 2) / 40 + 0.5 * torch.eye(N)).to(dev).contiguous()
+    Kc = K.clone()
+    L, logdet, info = ops.chol_factor_batched_bf16(Kc)
+    assert int(info.abs().sum()) == 0
+    Lt = torch.tril(L)
+    rec = Lt @ Lt.transpose(1, 2)
+    # trailing updates carry bf16 rounding: reconstruction error is
+    # bf16-scale relative to the matrix norm, far below the 0.5 diagonal
+    err = (rec - K).abs().max()
+    assert float(err) < 0.05, float(err)
+    # fp32 reference factor for comparison
+    K2 = K.clone()
+    L32, _, info32 = ops.chol_factor_batched(K2)
+    assert float((torch.tril(L32) - Lt).abs().max()) < 0.05
+
+
+def test_gp_bf16_end_to_end(dev):
+    from dmosopt_amd.models.gp import GPRMatern
+
+    rng = np.random.default_rng(14)
+    X = rng.random((300, 30))
+    Y = np.column_stack([np.sum(X**2, axis=1), np.sum((X - 1) ** 2, axis=1)])
+    kw = dict(optimizer="sceua", seed=5, device=dev)
+    gp32 = GPRMatern(X, Y, 30, 2, np.zeros(30), np.ones(30), compute="fp32", **kw)
+    gp16 = GPRMatern(X, Y, 30, 2, np.zeros(30), np.ones(30), compute="bf16", **kw)
+    # same seed + fp32 search in both modes -> identical theta
+    torch.testing.assert_close(gp16.theta, gp32.theta)
+    q = rng.random((512, 30))
+    m32, v32 = gp32.predict(q)
+    m16, v16 = gp16.predict(q)
+    scale = np.abs(m32).mean()
+    assert np.abs(m16 - m32).mean() / scale < 0.02, np.abs(m16 - m32).mean() / scale
+    assert np.isfinite(v16).all()
+    # tensor route stays finite and close
+    qt = torch.as_tensor(q, dtype=torch.float32, device=dev)
+    mt = gp16.evaluate_tensor(qt)
+    assert float((mt.cpu().double() - torch.as_tensor(m16)).abs().mean()) / scale < 0.02
