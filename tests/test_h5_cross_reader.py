@@ -169,7 +169,8 @@ def test_values_via_independent_reader_match(tmp_path):
     ours = np.array(
         [e.objectives for e in old_evals[0]], dtype=np.float32
     )  # (n, 2)
-    dump = _dump(fp, "-d", "/xh5v/0/objectives", "-O")
+    # %.9g: 9 significant digits round-trip any float32 exactly
+    dump = _dump(fp, "-d", "/xh5v/0/objectives", "-O", "-m", "%.9g")
     # h5dump compound rows print as "(i): {\n v1,\n v2\n }"
     rows = re.findall(r"\(\d+\):\s*{\s*([-\d.e+]+),\s*([-\d.e+]+)\s*}", dump, re.S)
     theirs = np.array([[float(a), float(b)] for a, b in rows], dtype=np.float32)
