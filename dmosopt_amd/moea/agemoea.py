@@ -119,6 +119,26 @@ def survival_score(y: np.ndarray, front: np.ndarray, ideal_point: np.ndarray):
     selected = np.zeros(m, dtype=bool)
     selected[extreme] = True
 
+    # device route: the WHOLE greedy loop runs in one single-workgroup
+    # kernel (ops/hip/agemoea_survival.hip) over a device-built distance
+    # matrix — removes the last hot-path host-side loop (PARITY 2.9)
+    if (
+        m >= 128
+        and m <= 8192
+        and np.isfinite(p)
+        and torch.cuda.is_available()
+        and ops.native_available()
+    ):
+        from dmosopt_amd import _hipops
+
+        At = torch.as_tensor(ynfront, dtype=torch.float32, device="cuda")
+        nn_t = torch.linalg.vector_norm(At.abs(), ord=float(p), dim=1)
+        Dt = (torch.cdist(At, At, p=float(p)) / nn_t[:, None]).contiguous()
+        pre = torch.zeros(m, dtype=torch.uint8, device="cuda")
+        pre[torch.as_tensor(extreme, dtype=torch.int64, device="cuda")] = 1
+        crowd = _hipops.agemoea_survival(Dt, pre).cpu().numpy().astype(np.float64)
+        return normalization, p, crowd
+
     with np.errstate(divide="ignore", invalid="ignore"):
         nn = np.power(np.power(np.abs(ynfront), p).sum(axis=1), 1.0 / p)
         distances = minkowski_matrix(ynfront, ynfront, p)

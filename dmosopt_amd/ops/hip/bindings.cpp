@@ -71,6 +71,8 @@ void launch_matern_cross_bf16(const float*, const float*, const float*,
                               float*, int, int, int, int, int, int, int,
                               const float*, const float*, hipStream_t);
 void launch_cholesky_multik_bf16(float*, float*, int*, int, int, hipStream_t);
+void launch_agemoea_survival(const float*, const unsigned char*, float*, int,
+                             hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -592,6 +594,19 @@ torch::Tensor lacour_volumes(torch::Tensor coords, torch::Tensor defs,
   return vol.narrow(0, 0, U);
 }
 
+torch::Tensor agemoea_survival(torch::Tensor D, torch::Tensor preselected) {
+  CHECK_GPU(D);
+  CHECK_GPU(preselected);
+  const int m = D.size(0);
+  TORCH_CHECK(m <= 8192, "agemoea_survival supports m <= 8192 (LDS)");
+  TORCH_CHECK(D.scalar_type() == torch::kFloat32);
+  auto crowd = torch::zeros({m}, D.options());
+  launch_agemoea_survival(D.data_ptr<float>(),
+                          preselected.data_ptr<uint8_t>(),
+                          crowd.data_ptr<float>(), m, cur_stream());
+  return crowd;
+}
+
 // --------------------------------------------------------------- bf16 path
 torch::Tensor mfma_bf16_probe(torch::Tensor A, torch::Tensor B) {
   CHECK_GPU(A);
@@ -657,6 +672,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hv_mc_uniform_hits", &hv_mc_uniform_hits);
   m.def("hv_fpras_hits", &hv_fpras_hits);
   m.def("get_duplicates", &get_duplicates);
+  m.def("agemoea_survival", &agemoea_survival,
+        "Greedy 2-NN AGE-MOEA survival scores (single-workgroup loop)");
   m.def("mfma_bf16_probe", &mfma_bf16_probe,
         "Single-tile bf16 MFMA fragment-layout probe");
   m.def("matern_cross_bf16", &matern_cross_bf16,

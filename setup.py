@@ -32,6 +32,7 @@ ext = CUDAExtension(
         os.path.join("dmosopt_amd", "ops", "hip", "moea_ops.hip"),
         os.path.join("dmosopt_amd", "ops", "hip", "hv_exact.hip"),
         os.path.join("dmosopt_amd", "ops", "hip", "matern_bf16.hip"),
+        os.path.join("dmosopt_amd", "ops", "hip", "agemoea_survival.hip"),
     ],
     include_dirs=[HIP_DIR],
     extra_compile_args={
