@@ -362,6 +362,206 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
   }
 }
 
+// ------------------------------------------------------------ 64-wide panel
+// PANEL64 variant: one launch factors a 64x64 diagonal block (two wave-
+// parallel 32-factors + an in-LDS TRSM + SYRK between them) and solves the
+// whole 64-column panel below it. Halves the panel-launch count of the
+// right-looking loop (N=300: 5 launches instead of 10) — the multik path
+// is launch-gap bound at ~10 us fixed cost per launch (profiles/README.md).
+#define PANEL64 64
+
+// Factor the bs_ x bs_ (<=32) diagonal sub-block of S at offset (o, o):
+// wave-parallel, lane i owns row i in registers (same scheme as
+// chol_panel_kernel's diagonal factor). Caller must __syncthreads() after.
+__device__ __forceinline__ void factor32_at(
+    float S[PANEL64][PANEL64 + 1], float* colbuf, int o, int bs_,
+    float* logdet, int* info, int b, int gk0, int tid) {
+  if (tid < 64) {
+    const int lane = tid;
+    float r[CHOL_BS];
+#pragma unroll
+    for (int t = 0; t < CHOL_BS; ++t)
+      r[t] = (lane < bs_ && t < bs_) ? S[o + lane][o + t] : 0.0f;
+    float mylog = 0.0f;
+    int bad = 0;
+#pragma unroll
+    for (int j = 0; j < CHOL_BS; ++j) {
+      if (j >= bs_) continue;
+      float d = __shfl(r[j], j);
+      if (d <= 0.0f || !isfinite(d)) {
+        bad = bad ? bad : (gk0 + j + 1);
+        d = 1e-30f;
+      }
+      d = sqrtf(d);
+      if (lane == j) {
+        r[j] = d;
+        mylog = logf(d);
+      } else if (lane > j) {
+        r[j] /= d;
+      }
+      if (lane < bs_) colbuf[lane] = r[j];
+      __threadfence_block();
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t) {
+        if (t <= j || t >= bs_) continue;
+        if (lane >= t) r[t] = fmaf(-r[j], colbuf[t], r[t]);
+      }
+    }
+    if (lane < bs_) {
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t) {
+        if (t >= bs_) continue;
+        S[o + lane][o + t] = r[t];
+      }
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      mylog += __shfl_down(mylog, off);
+      const int ob = __shfl_down(bad, off);
+      bad = bad ? bad : ob;
+    }
+    if (lane == 0) {
+      logdet[b] += mylog;
+      if (bad && info[b] == 0) info[b] = bad;
+    }
+  }
+}
+
+#define P64_CHUNK 128  // panel rows staged per LDS buffer
+
+__global__ __launch_bounds__(CHOLP_TPB) void chol_panel64_kernel(
+    float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
+    int N, int k0) {
+  __shared__ float S[PANEL64][PANEL64 + 1];
+  __shared__ float colbuf[CHOL_BS];
+  __shared__ float P[P64_CHUNK][PANEL64 + 1];
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  float* Ab = A + (long long)b * N * N;
+  const int bs = min(PANEL64, N - k0);
+  const int bs1 = min(CHOL_BS, bs);
+  const int bs2 = bs - bs1;
+
+  for (int idx = tid; idx < bs * bs; idx += blockDim.x)
+    S[idx / bs][idx % bs] = Ab[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
+  __syncthreads();
+
+  factor32_at(S, colbuf, 0, bs1, logdet, info, b, k0, tid);
+  __syncthreads();
+
+  if (bs2 > 0) {
+    // TRSM: rows [32, bs) of columns [0, 32) against the factored block
+    if (tid < bs2) {
+      const int r = CHOL_BS + tid;
+#pragma unroll 4
+      for (int j = 0; j < CHOL_BS; ++j) {
+        float v = S[r][j];
+        for (int t = 0; t < j; ++t) v = fmaf(-S[r][t], S[j][t], v);
+        S[r][j] = v / S[j][j];
+      }
+    }
+    __syncthreads();
+    // SYRK the lower-right bs2 x bs2 block (lower triangle only: the
+    // factor routine never reads above the diagonal)
+    for (int idx = tid; idx < bs2 * bs2; idx += blockDim.x) {
+      const int i = idx / bs2, j = idx % bs2;
+      if (j > i) continue;
+      float acc = 0.0f;
+#pragma unroll 8
+      for (int k = 0; k < CHOL_BS; ++k)
+        acc = fmaf(S[CHOL_BS + i][k], S[CHOL_BS + j][k], acc);
+      S[CHOL_BS + i][CHOL_BS + j] -= acc;
+    }
+    __syncthreads();
+    factor32_at(S, colbuf, CHOL_BS, bs2, logdet, info, b, k0 + CHOL_BS, tid);
+    __syncthreads();
+  }
+
+  for (int idx = tid; idx < bs * bs; idx += blockDim.x) {
+    const int i = idx / bs, t = idx % bs;
+    if (t <= i) Ab[(long long)(k0 + i) * N + k0 + t] = S[i][t];
+  }
+
+  // 64-column panel solve below, staged through LDS in coalesced chunks
+  for (int c0 = k0 + PANEL64; c0 < N; c0 += P64_CHUNK) {
+    const int rows = min(P64_CHUNK, N - c0);
+    for (int idx = tid; idx < rows * PANEL64; idx += CHOLP_TPB) {
+      const int r = idx / PANEL64, c = idx % PANEL64;
+      P[r][c] = Ab[(long long)(c0 + r) * N + k0 + c];
+    }
+    __syncthreads();
+    if (tid < rows) {
+#pragma unroll 4
+      for (int j = 0; j < PANEL64; ++j) {
+        float v = P[tid][j];
+        for (int t = 0; t < j; ++t) v = fmaf(-P[tid][t], S[j][t], v);
+        P[tid][j] = v / S[j][j];
+      }
+    }
+    __syncthreads();
+    for (int idx = tid; idx < rows * PANEL64; idx += CHOLP_TPB) {
+      const int r = idx / PANEL64, c = idx % PANEL64;
+      Ab[(long long)(c0 + r) * N + k0 + c] = P[r][c];
+    }
+    __syncthreads();
+  }
+}
+
+// 64-deep trailing update: C -= Pi Pj^T with K = 64 (the PANEL64 step).
+// Same tile-pair enumeration as chol_syrk_kernel; k-loop runs the full 64
+// panel columns, so HALF the launches touch the same trailing bytes.
+__global__ __launch_bounds__(CHOLP_TPB) void chol_syrk64_kernel(
+    float* __restrict__ A, int N, int k0, int nt, int tj_fixed, int off) {
+  __shared__ float Pi[SYRK_TS][PANEL64];
+  __shared__ float Pj[SYRK_TS][PANEL64];
+  const int b = blockIdx.x;
+  float* Ab = A + (long long)b * N * N;
+  const int r0 = k0 + PANEL64;
+  int ti, tj;
+  if (tj_fixed >= 0) {
+    tj = tj_fixed;
+    ti = blockIdx.y + tj_fixed;
+  } else {
+    int p = blockIdx.y;
+    ti = 0;
+    while (p > ti) { p -= ti + 1; ++ti; }
+    tj = p + off;
+    ti += off;
+  }
+  const int i0 = r0 + ti * SYRK_TS, j0 = r0 + tj * SYRK_TS;
+  const int tid = threadIdx.x;
+
+  for (int idx = tid; idx < SYRK_TS * PANEL64; idx += blockDim.x) {
+    const int r = idx / PANEL64, c = idx % PANEL64;
+    const int cs = c ^ ((r & 7) << 2);  // bank-spread swizzle
+    Pi[r][cs] = (i0 + r < N) ? Ab[(long long)(i0 + r) * N + k0 + c] : 0.0f;
+    Pj[r][cs] = (j0 + r < N) ? Ab[(long long)(j0 + r) * N + k0 + c] : 0.0f;
+  }
+  __syncthreads();
+
+  typedef __attribute__((ext_vector_type(4))) float f32x4;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int lr = lane & 15, lk = lane >> 4;
+#pragma unroll
+  for (int sIdx = 0; sIdx < 4; ++sIdx) {
+    const int sub = wave * 4 + sIdx;
+    const int r16 = (sub >> 2) * 16, c16 = (sub & 3) * 16;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int k = 0; k < PANEL64; k += 4) {
+      const int ra = r16 + lr, rb = c16 + lr;
+      const float a = Pi[ra][(k + lk) ^ ((ra & 7) << 2)];
+      const float bv = Pj[rb][(k + lk) ^ ((rb & 7) << 2)];
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int i = i0 + r16 + lk * 4 + r;
+      const int j = j0 + c16 + lr;
+      if (i < N && j < N && j <= i) Ab[(long long)i * N + j] -= acc[r];
+    }
+  }
+}
+
 // Trailing update A[ti,tj] -= P_i P_j^T over 64x64 tiles of the submatrix
 // below/right of the panel; blockIdx.y enumerates lower-triangular tile
 // pairs, each 256-thread block computes a 4x4 register tile per thread.
@@ -611,6 +811,26 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
     }
   }
   hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
+  static int panel64 = -1;
+  if (panel64 < 0) {
+    // PANEL64 halves the launch count of the launch-gap-bound multik loop
+    // (measured A/B in profiles/README.md); DMOSOPT_CHOL_PANEL=32 reverts
+    const char* env = getenv("DMOSOPT_CHOL_PANEL");
+    panel64 = (env && env[0] == '3') ? 0 : 1;
+  }
+  if (panel64 && !overlap) {
+    for (int k0 = 0; k0 < N; k0 += PANEL64) {
+      hipLaunchKernelGGL(chol_panel64_kernel, dim3(B), dim3(CHOLP_TPB), 0,
+                         stream, A, logdet, info, N, k0);
+      const int trailing = N - k0 - PANEL64;
+      if (trailing > 0) {
+        const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
+        hipLaunchKernelGGL(chol_syrk64_kernel, dim3(B, nt * (nt + 1) / 2),
+                           dim3(CHOLP_TPB), 0, stream, A, N, k0, nt, -1, 0);
+      }
+    }
+    return;
+  }
   if (!overlap) {
     for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
       hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0,
