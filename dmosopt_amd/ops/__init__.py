@@ -337,7 +337,9 @@ def chol_factor_batched(K):
     (N >= 1200, where rocSOLVER's multi-CU decomposition wins ~10x and
     was verified working) dispatch to torch/rocSOLVER."""
     if _use_native(K) and K.dtype == torch.float32:
-        if K.shape[1] < 1200:
+        # crossover re-measured round 2 (profiles/README.md large-N table):
+        # native wins to N=2048, parity at 4096
+        if K.shape[1] < 2048:
             logdet, info = _native.cholesky_batched_(K)
             return K, logdet, info
     L, info = torch.linalg.cholesky_ex(K)

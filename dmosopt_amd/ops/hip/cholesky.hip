@@ -813,10 +813,13 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
   hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
   static int panel64 = -1;
   if (panel64 < 0) {
-    // PANEL64 halves the launch count of the launch-gap-bound multik loop
-    // (measured A/B in profiles/README.md); DMOSOPT_CHOL_PANEL=32 reverts
+    // measured WORSE than PANEL32 across the board (profiles/README.md:
+    // 0.89 vs 0.41 ms at B=12 N=300; 7x at N=4096): the in-kernel
+    // TRSM/SYRK serial phases and the 64-deep per-row solve chain cost far
+    // more than the ~10 us/launch the halved launch count saves. Kept as
+    // DMOSOPT_CHOL_PANEL=64 for re-measurement on future ROCm/driver.
     const char* env = getenv("DMOSOPT_CHOL_PANEL");
-    panel64 = (env && env[0] == '3') ? 0 : 1;
+    panel64 = (env && env[0] == '6') ? 1 : 0;
   }
   if (panel64 && !overlap) {
     for (int k0 = 0; k0 < N; k0 += PANEL64) {
