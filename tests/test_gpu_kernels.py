@@ -741,3 +741,34 @@ def test_agemoea_gpu_selection_e2e(dev):
     )
     assert xs.shape == (700, d) and ys.shape == (700, m)
     assert np.isfinite(ys).all()
+
+
+def test_nmll_graph_matches_eager(dev, monkeypatch):
+    """The hipGraph-captured NMLL pipeline must be BITWISE identical to the
+    eager launch sequence (same kernels, same order), across replays with
+    changing theta."""
+    from dmosopt_amd.models import gp_core
+
+    g = torch.Generator().manual_seed(31)
+    N, D, B = 200, 30, 18
+    X = torch.rand(N, D, generator=g).float().to(dev)
+    y = torch.randn(B, N, generator=g).float().to(dev)
+    gp_core._nmll_graphs.clear()
+
+    thetas = [
+        torch.cat([
+            torch.randn(B, 1, generator=g) * 0.3,
+            torch.randn(B, 1, generator=g) * 0.5,
+            torch.full((B, 1), -8.0) + torch.rand(B, 1, generator=g),
+        ], dim=1).float().to(dev)
+        for _ in range(4)
+    ]
+    # eager reference
+    monkeypatch.setenv("DMOSOPT_NMLL_GRAPH", "0")
+    eager = [gp_core.batched_nmll(X, y, t, nu=2.5, anisotropic=False) for t in thetas]
+    # graph path (fresh capture + replays)
+    monkeypatch.setenv("DMOSOPT_NMLL_GRAPH", "1")
+    graphed = [gp_core.batched_nmll(X, y, t, nu=2.5, anisotropic=False) for t in thetas]
+    gp_core._nmll_graphs.clear()
+    for e, gr in zip(eager, graphed):
+        assert torch.equal(e, gr), float((e - gr).abs().max())
