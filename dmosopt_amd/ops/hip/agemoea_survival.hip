@@ -5,9 +5,10 @@
 // inherently serial (each round picks the remaining point with the largest
 // sum of its two smallest distances TO THE SELECTED SET, then that point
 // joins the set), but every round's argmax and distance-merge are parallel
-// over m. One workgroup runs the whole loop with d1/d2/alive in LDS and the
-// (m x m) distance matrix streamed row-wise from L2 (D is symmetric, so
-// column j == row j and the reads coalesce).
+// over m. One workgroup runs the whole loop with d1/d2 in LDS; the merge
+// reads COLUMN sel of the (m x m) matrix (D is row-normalized by each
+// point's own norm, so NOT symmetric) — strided reads served from L2
+// after the first pass.
 //
 // Semantics match the numpy path bit-for-bit in selection ORDER except for
 // argmax ties (numpy picks the lowest index; the LDS tree reduction below
@@ -93,11 +94,13 @@ extern "C" __global__ __launch_bounds__(AGES_TPB) void agemoea_survival_kernel(
       d1[sel] = -HUGE_VALF;  // consume
     }
     __syncthreads();
-    // merge the new member's distances into everyone's two smallest
-    const float* Drow = D + (long long)sel * m;  // symmetric: row == column
+    // merge the new member's distances into everyone's two smallest.
+    // D is row-normalized (D[i][j] = d(i,j)/nn[i]) and therefore NOT
+    // symmetric: point i's distance to the new member sel is column sel
+    // of ROW i (matching the host path's distances[remaining, best]).
     for (int i = tid; i < m; i += AGES_TPB) {
       if (d1[i] == -HUGE_VALF || crowd[i] == HUGE_VALF) continue;
-      const float dn = Drow[i];
+      const float dn = D[(long long)i * m + sel];
       if (dn < d2[i]) d2[i] = dn;
       if (d2[i] < d1[i]) { const float t = d1[i]; d1[i] = d2[i]; d2[i] = t; }
     }
