@@ -254,6 +254,7 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     int N, int k0) {
   __shared__ float S[CHOL_BS][CHOL_BS + 1];
   __shared__ float colbuf[CHOL_BS];
+  __shared__ float colbuf2[CHOL_BS];
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   float* Ab = A + (long long)b * N * N;
@@ -278,8 +279,14 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
       r[t] = (lane < bs && t < bs) ? S[lane][t] : 0.0f;
     float mylog = 0.0f;
     int bad = 0;
+    // TWO columns per round: factor j, apply its rank-1 update to column
+    // j+1 only, factor j+1, then ONE fused rank-2 pass over the remaining
+    // columns. Halves the broadcast/fence rounds on the serial pivot chain
+    // (the dominant latency of this kernel at ~30 us mean); the fmaf
+    // sequence per element is UNCHANGED, so the factor stays bitwise equal
+    // to the one-column version (SCE-UA accept decisions are bit-stable).
 #pragma unroll
-    for (int j = 0; j < CHOL_BS; ++j) {
+    for (int j = 0; j < CHOL_BS; j += 2) {
       if (j >= bs) continue;
       float d = __shfl(r[j], j);
       if (d <= 0.0f || !isfinite(d)) {
@@ -289,19 +296,50 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
       d = sqrtf(d);
       if (lane == j) {
         r[j] = d;
-        mylog = logf(d);
+        mylog += logf(d);
       } else if (lane > j) {
         r[j] /= d;
       }
       // broadcast column j through LDS with ONE workgroup fence: the
-      // (31-j) reads then pipeline freely, unlike a per-t __shfl chain
+      // reads then pipeline freely, unlike a per-t __shfl chain
       // (ds_bpermute each) or a volatile pointer (per-access ordering)
       if (lane < bs) colbuf[lane] = r[j];
       __threadfence_block();
+      const int j1 = j + 1;
+      if (j1 >= bs) {
+        // odd tail: plain rank-1 update of the remainder
+#pragma unroll
+        for (int t = 0; t < CHOL_BS; ++t) {
+          if (t <= j || t >= bs) continue;
+          if (lane >= t) r[t] = fmaf(-r[j], colbuf[t], r[t]);
+        }
+        continue;
+      }
+      // column j+1: apply col j's update, then factor it
+      if (lane >= j1) r[j1] = fmaf(-r[j], colbuf[j1], r[j1]);
+      float d1 = __shfl(r[j1], j1);
+      if (d1 <= 0.0f || !isfinite(d1)) {
+        bad = bad ? bad : (k0 + j1 + 1);
+        d1 = 1e-30f;
+      }
+      d1 = sqrtf(d1);
+      if (lane == j1) {
+        r[j1] = d1;
+        mylog += logf(d1);
+      } else if (lane > j1) {
+        r[j1] /= d1;
+      }
+      if (lane < bs) colbuf2[lane] = r[j1];
+      __threadfence_block();
+      // fused rank-2 update of the remaining columns (same fmaf order per
+      // element as two sequential rank-1 passes)
 #pragma unroll
       for (int t = 0; t < CHOL_BS; ++t) {
-        if (t <= j || t >= bs) continue;
-        if (lane >= t) r[t] = fmaf(-r[j], colbuf[t], r[t]);
+        if (t <= j1 || t >= bs) continue;
+        if (lane >= t) {
+          r[t] = fmaf(-r[j], colbuf[t], r[t]);
+          r[t] = fmaf(-r[j1], colbuf2[t], r[t]);
+        }
       }
     }
     // stage the factored block back to LDS (global writeback below is
