@@ -132,8 +132,7 @@ def survival_score(y: np.ndarray, front: np.ndarray, ideal_point: np.ndarray):
         from dmosopt_amd import _hipops
 
         At = torch.as_tensor(ynfront, dtype=torch.float32, device="cuda")
-        nn_t = torch.linalg.vector_norm(At.abs(), ord=float(p), dim=1)
-        Dt = (torch.cdist(At, At, p=float(p)) / nn_t[:, None]).contiguous()
+        Dt = _hipops.minkowski_norm_matrix(At.contiguous(), float(p))
         pre = torch.zeros(m, dtype=torch.uint8, device="cuda")
         pre[torch.as_tensor(extreme, dtype=torch.int64, device="cuda")] = 1
         crowd = _hipops.agemoea_survival(Dt, pre).cpu().numpy().astype(np.float64)

@@ -73,6 +73,8 @@ void launch_matern_cross_bf16(const float*, const float*, const float*,
 void launch_cholesky_multik_bf16(float*, float*, int*, int, int, hipStream_t);
 void launch_agemoea_survival(const float*, const unsigned char*, float*, int,
                              hipStream_t);
+void launch_minkowski_norm_matrix(const float*, float*, int, int, float,
+                                  hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -594,6 +596,16 @@ torch::Tensor lacour_volumes(torch::Tensor coords, torch::Tensor defs,
   return vol.narrow(0, 0, U);
 }
 
+torch::Tensor minkowski_norm_matrix(torch::Tensor Y, double p) {
+  CHECK_GPU(Y);
+  const int m = Y.size(0), d = Y.size(1);
+  TORCH_CHECK(d <= 16, "minkowski_norm_matrix supports d <= 16");
+  auto D = torch::empty({m, m}, Y.options());
+  launch_minkowski_norm_matrix(Y.data_ptr<float>(), D.data_ptr<float>(), m, d,
+                               (float)p, cur_stream());
+  return D;
+}
+
 torch::Tensor agemoea_survival(torch::Tensor D, torch::Tensor preselected) {
   CHECK_GPU(D);
   CHECK_GPU(preselected);
@@ -672,6 +684,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hv_mc_uniform_hits", &hv_mc_uniform_hits);
   m.def("hv_fpras_hits", &hv_fpras_hits);
   m.def("get_duplicates", &get_duplicates);
+  m.def("minkowski_norm_matrix", &minkowski_norm_matrix,
+        "Row-normalized Minkowski-p distance matrix in one pass");
   m.def("agemoea_survival", &agemoea_survival,
         "Greedy 2-NN AGE-MOEA survival scores (single-workgroup loop)");
   m.def("mfma_bf16_probe", &mfma_bf16_probe,

@@ -685,8 +685,11 @@ def test_agemoea_survival_kernel_matches_host_loop(dev):
     for m, n_ext, p in ((150, 2, 2.0), (600, 3, 1.3), (1024, 2, 0.7)):
         Y = rng.random((m, 3)).astype(np.float32)
         At = torch.as_tensor(Y, device=dev)
+        Dt = _hipops.minkowski_norm_matrix(At.contiguous(), float(p))
+        # the fused matrix kernel must match the torch route closely
         nn_t = torch.linalg.vector_norm(At.abs(), ord=float(p), dim=1)
-        Dt = (torch.cdist(At, At, p=float(p)) / nn_t[:, None]).contiguous()
+        Dt_ref = torch.cdist(At.double(), At.double(), p=float(p)) / nn_t.double()[:, None]
+        torch.testing.assert_close(Dt.double(), Dt_ref, rtol=2e-4, atol=2e-5)
         extreme = np.arange(n_ext)
         pre = torch.zeros(m, dtype=torch.uint8, device=dev)
         pre[:n_ext] = 1
