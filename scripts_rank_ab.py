@@ -4,6 +4,7 @@ import os, sys, time
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import numpy as np, torch
 from dmosopt_amd import ops, _hipops
+assert ops.native_available()
 
 dev = torch.device("cuda", 0)
 w = torch.randn(512, 512, device=dev); (w @ w).sum().item()
