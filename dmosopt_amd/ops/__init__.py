@@ -72,13 +72,14 @@ def _use_native(*tensors: torch.Tensor) -> bool:
 def pareto_rank(Y: torch.Tensor) -> torch.Tensor:
     if _use_native(Y):
         # small N: bit-matrix ranking (grid-wide packed dominator build +
-        # one-block popcount peel); larger N: matvec dominator-count peel.
-        # Crossover re-measured round 2 (scripts_rank_ab.py, N=2048):
-        # matvec 1.17 ms vs single-block 2.6-5.0 ms in the few-front
-        # (converged-population) regime that dominates a run; single-block
-        # only wins on many-front random clouds, which occur for a handful
-        # of early generations.
-        if Y.shape[0] <= 1024:
+        # one-block popcount peel); large N: matvec dominator-count peel.
+        # NOTE (round-2 A/B, scripts_rank_ab.py): in ISOLATION the matvec
+        # path wins at N=2048 in the few-front regime (1.17 vs 2.6 ms), but
+        # routing it there made config #3 10x SLOWER end-to-end — its
+        # .item() sync every 16 fronts stalls the asynchronous generation
+        # pipeline, which costs far more than the kernel time it saves.
+        # The sync-free single-block peel keeps the host running ahead.
+        if Y.shape[0] <= 2048:
             return _native.pareto_rank(Y.contiguous().float())
         return _pareto_rank_gpu(Y)
     return torch_ref.pareto_rank(Y)
