@@ -140,6 +140,11 @@ class _NmllGraph:
             self.out = ops.gp_nmll_fused(
                 self.Xb, self.tb, self.yb, nu, anisotropic, jitter
             )
+        # priming replay: the very first replay after capture was observed
+        # to differ from all subsequent replays on ROCm 7.2 (bit-level;
+        # scripts_det_debug2.py) — burn it here so every user-visible call
+        # is a steady-state replay and runs stay bit-deterministic
+        self.graph.replay()
 
     def run(self, X, y, theta):
         self.Xb.copy_(X)
