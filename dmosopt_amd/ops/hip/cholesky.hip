@@ -823,6 +823,11 @@ extern "C" void launch_nmll_reduce(const float* Z, const float* half_logdet,
                      half_logdet, info, out, N, c);
 }
 
+__global__ void zero_f32_kernel(float* __restrict__ p, int n) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = 0.0f;
+}
+
 extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
                                        int B, int N, hipStream_t stream) {
   // Cross-stream software pipeline: panel(k+1) only depends on the FIRST
@@ -848,7 +853,13 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
       }
     }
   }
-  hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
+  // explicit zero KERNEL instead of hipMemsetAsync: under hipGraph stream
+  // capture (models/gp_core.py _NmllGraph) the captured memset node was
+  // observed to race with the first panel kernel's logdet accumulation on
+  // replay (bit-nondeterministic outputs with bit-identical inputs,
+  // scripts_det_debug6.py); a kernel node orders correctly
+  hipLaunchKernelGGL(zero_f32_kernel, dim3((B + 255) / 256), dim3(256), 0,
+                     stream, logdet, B);
   static int panel64 = -1;
   if (panel64 < 0) {
     // measured WORSE than PANEL32 across the board (profiles/README.md:
@@ -923,7 +934,8 @@ extern "C" void launch_chol_syrk_bf16(float*, int, int, int, int, int, int,
 extern "C" void launch_cholesky_multik_bf16(float* A, float* logdet,
                                             int* info, int B, int N,
                                             hipStream_t stream) {
-  hipMemsetAsync(logdet, 0, (size_t)B * sizeof(float), stream);
+  hipLaunchKernelGGL(zero_f32_kernel, dim3((B + 255) / 256), dim3(256), 0,
+                     stream, logdet, B);
   for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
     hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0, stream,
                        A, logdet, info, N, k0);
