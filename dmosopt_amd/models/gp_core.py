@@ -140,11 +140,12 @@ class _NmllGraph:
             self.out = ops.gp_nmll_fused(
                 self.Xb, self.tb, self.yb, nu, anisotropic, jitter
             )
-        # priming replay: the very first replay after capture was observed
-        # to differ from all subsequent replays on ROCm 7.2 (bit-level;
-        # scripts_det_debug2.py) — burn it here so every user-visible call
-        # is a steady-state replay and runs stay bit-deterministic
-        self.graph.replay()
+        # NOTE: a hipMemsetAsync enqueued inside the captured region raced
+        # with the following kernel's accumulation ON REPLAY (ROCm 7.2;
+        # bit-nondeterministic outputs with bit-identical inputs,
+        # scripts_det_debug6.py) — the multik Cholesky therefore zeroes
+        # logdet with an explicit kernel (cholesky.hip zero_f32_kernel)
+        # instead of a memset. Keep memsets out of captured pipelines.
 
     def run(self, X, y, theta):
         self.Xb.copy_(X)
