@@ -71,15 +71,16 @@ def _use_native(*tensors: torch.Tensor) -> bool:
 # ------------------------------------------------------------------ ranking
 def pareto_rank(Y: torch.Tensor) -> torch.Tensor:
     if _use_native(Y):
-        # small N: bit-matrix ranking (grid-wide packed dominator build +
-        # one-block popcount peel); large N: matvec dominator-count peel.
+        # SYNC-FREE single-block peels up to their LDS capacity
+        # (N*(4m+8) <= 144 KB: N<=9215 at m=2, 5266 at m=5); only beyond
+        # that the chased matvec peel.
         # NOTE (round-2 A/B, scripts_rank_ab.py): in ISOLATION the matvec
         # path wins at N=2048 in the few-front regime (1.17 vs 2.6 ms), but
         # routing it there made config #3 10x SLOWER end-to-end — its
         # .item() sync every 16 fronts stalls the asynchronous generation
         # pipeline, which costs far more than the kernel time it saves.
-        # The sync-free single-block peel keeps the host running ahead.
-        if Y.shape[0] <= 2048:
+        n, m = Y.shape
+        if n * (4 * m + 8) + 8 <= 144 * 1024:
             return _native.pareto_rank(Y.contiguous().float())
         return _pareto_rank_gpu(Y)
     return torch_ref.pareto_rank(Y)

@@ -308,7 +308,11 @@ torch::Tensor dominance_degree_matrix(torch::Tensor Y) {
 torch::Tensor pareto_rank(torch::Tensor Y) {
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
-  if (N <= 2048) {
+  // prefer the SYNC-FREE single-block peels up to their LDS capacity
+  // (N*(4m+8) bytes for peel_from_y: N<=9215 at m=2, 5266 at m=5): the
+  // chased matvec path below syncs per 16 fronts, which stalls pipelined
+  // generation loops far beyond its kernel-time advantage (NOTES.md)
+  if ((size_t)N * (4 * m + 8) + 8 <= 144 * 1024) {
     auto Yc = Y.contiguous().to(torch::kFloat32);
     auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));
     // bit-matrix path: grid-wide packed dominator build + popcount peel
