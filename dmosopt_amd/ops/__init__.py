@@ -71,18 +71,13 @@ def _use_native(*tensors: torch.Tensor) -> bool:
 # ------------------------------------------------------------------ ranking
 def pareto_rank(Y: torch.Tensor) -> torch.Tensor:
     if _use_native(Y):
-        # SYNC-FREE single-block peels up to their LDS capacity
-        # (N*(4m+8) <= 144 KB: N<=9215 at m=2, 5266 at m=5); only beyond
-        # that the chased matvec peel.
-        # NOTE (round-2 A/B, scripts_rank_ab.py): in ISOLATION the matvec
-        # path wins at N=2048 in the few-front regime (1.17 vs 2.6 ms), but
-        # routing it there made config #3 10x SLOWER end-to-end — its
-        # .item() sync every 16 fronts stalls the asynchronous generation
-        # pipeline, which costs far more than the kernel time it saves.
-        n, m = Y.shape
-        if n * (4 * m + 8) + 8 <= 144 * 1024:
-            return _native.pareto_rank(Y.contiguous().float())
-        return _pareto_rank_gpu(Y)
+        # the native binding routes internally: N <= 2048 one-workgroup
+        # bit/LDS peels; N > 2048 the grid-wide COOPERATIVE peel (sync-free
+        # from the host, all CUs; falls back to the chased matvec peel only
+        # if cooperative launch is unavailable). Host syncs in ranking stall
+        # pipelined generation loops far beyond their kernel-time cost
+        # (round-2 lesson, NOTES.md).
+        return _native.pareto_rank(Y.contiguous().float())
     return torch_ref.pareto_rank(Y)
 
 

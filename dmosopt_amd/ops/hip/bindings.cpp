@@ -50,6 +50,8 @@ void launch_pack_rank_crowd(const long long*, const float*, long long*, int, hip
 void launch_variation_slots(const float*, const long long*, const long long*, const long long*, const long long*, const float*, const float*, const float*, const float*, float*, int, int, int, float, unsigned long long, unsigned long long, hipStream_t);
 void launch_gather3(const float*, const float*, const long long*, const long long*, float*, float*, long long*, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
+int launch_coop_peel(const float*, unsigned int*, unsigned int*, int*, int*,
+                     int*, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
 void launch_hv2d(const double*, const double*, double*, int, hipStream_t);
 void launch_hv3d_slices(const double*, const double*, const double*,
@@ -308,11 +310,27 @@ torch::Tensor dominance_degree_matrix(torch::Tensor Y) {
 torch::Tensor pareto_rank(torch::Tensor Y) {
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
-  // prefer the SYNC-FREE single-block peels up to their LDS capacity
-  // (N*(4m+8) bytes for peel_from_y: N<=9215 at m=2, 5266 at m=5): the
-  // chased matvec path below syncs per 16 fronts, which stalls pipelined
-  // generation loops far beyond its kernel-time advantage (NOTES.md)
-  if ((size_t)N * (4 * m + 8) + 8 <= 144 * 1024) {
+  // N > 2048: grid-wide COOPERATIVE peel — sync-free from the host (the
+  // chased matvec path's readback every 16 fronts stalls pipelined
+  // generation loops) and parallel across all CUs (the one-workgroup
+  // peels serialize on one CU: 13.8 ms at N=8192 vs ~0.1-1 ms here)
+  if (N > 2048) {
+    const int W = (N + 31) / 32;
+    auto Yc = Y.contiguous().to(torch::kFloat32);
+    auto opts_i = Y.options().dtype(torch::kInt32);
+    auto Dbits = torch::empty({N, W}, opts_i);
+    auto fmask = torch::empty({W}, opts_i);
+    auto n_dom = torch::empty({N}, opts_i);
+    auto ctrl = torch::empty({2}, opts_i);
+    auto rank = torch::zeros({N}, opts_i);
+    if (launch_coop_peel(Yc.data_ptr<float>(),
+                         (unsigned int*)Dbits.data_ptr<int>(),
+                         (unsigned int*)fmask.data_ptr<int>(),
+                         n_dom.data_ptr<int>(), ctrl.data_ptr<int>(),
+                         rank.data_ptr<int>(), N, m, cur_stream()) == 0)
+      return rank.to(torch::kLong);
+  }
+  if (N <= 2048) {
     auto Yc = Y.contiguous().to(torch::kFloat32);
     auto rank = torch::zeros({N}, Y.options().dtype(torch::kInt32));
     // bit-matrix path: grid-wide packed dominator build + popcount peel

@@ -323,6 +323,101 @@ extern "C" int launch_peel_bits(const float* Y, unsigned int* Dbits_scratch,
   return 0;
 }
 
+// ---------------------------------------------------- cooperative peel
+// Grid-wide SYNC-FREE-FROM-HOST ranking for N beyond the one-workgroup
+// kernels' capacity (and their one-CU serialization): the packed dominator
+// bit-matrix lives in GLOBAL memory (L2-resident: N=8192 -> 8 MB), every
+// CU participates, and the front-peel loop synchronizes with grid.sync()
+// instead of host readbacks (the chased matvec path's .item() every 16
+// fronts stalls pipelined generation loops — NOTES.md).
+#include <hip/hip_cooperative_groups.h>
+
+__global__ __launch_bounds__(256) void coop_peel_bits_kernel(
+    const unsigned int* __restrict__ Dbits,  // (N, W) i-dominates-j bits
+    unsigned int* __restrict__ fmask,        // (W,) scratch
+    int* __restrict__ n_dom,                 // (N,) scratch
+    int* __restrict__ ctrl,                  // (2,) scratch
+    int* __restrict__ rank,                  // (N,) out
+    int N, int W) {
+  cooperative_groups::grid_group grid = cooperative_groups::this_grid();
+  const int gtid = blockIdx.x * blockDim.x + threadIdx.x;
+  const int gsize = gridDim.x * blockDim.x;
+
+  for (int j = gtid; j < N; j += gsize) {
+    int c = 0;
+    const unsigned int* row = Dbits + (size_t)j * W;
+    for (int w = 0; w < W; ++w) c += __popc(row[w]);
+    n_dom[j] = c;
+    rank[j] = 0;
+  }
+  if (gtid == 0) ctrl[1] = N;
+  grid.sync();
+
+  for (int k = 0; k <= N; ++k) {
+    for (int w = gtid; w < W; w += gsize) fmask[w] = 0u;
+    if (gtid == 0) ctrl[0] = 0;
+    grid.sync();
+    for (int j = gtid; j < N; j += gsize) {
+      if (n_dom[j] == 0) {
+        rank[j] = k;
+        n_dom[j] = -1;
+        atomicOr(&fmask[j >> 5], 1u << (j & 31));
+        atomicAdd(&ctrl[0], 1);
+      }
+    }
+    grid.sync();
+    const int fs = ctrl[0];  // uniform: written before the sync
+    if (fs == 0) break;
+    for (int j = gtid; j < N; j += gsize) {
+      if (n_dom[j] <= 0) continue;
+      int dec = 0;
+      const unsigned int* row = Dbits + (size_t)j * W;
+      for (int w = 0; w < W; ++w) dec += __popc(row[w] & fmask[w]);
+      n_dom[j] -= dec;
+    }
+    if (gtid == 0) ctrl[1] -= fs;
+    grid.sync();
+    if (ctrl[1] <= 0) break;  // uniform: written pre-sync by thread 0
+  }
+}
+
+extern "C" int launch_coop_peel(const float* Y, unsigned int* Dbits,
+                                unsigned int* fmask, int* n_dom, int* ctrl,
+                                int* rank, int N, int m, hipStream_t stream) {
+  const int W = (N + 31) / 32;
+  static int coop_ok = -1;
+  if (coop_ok < 0) {
+    int dev = 0, attr = 0;
+    hipGetDevice(&dev);
+    hipDeviceGetAttribute(&attr, hipDeviceAttributeCooperativeLaunch, dev);
+    coop_ok = attr ? 1 : 0;
+  }
+  if (!coop_ok) return -1;
+  static int max_blocks = -1;
+  if (max_blocks < 0) {
+    int per_cu = 0;
+    hipOccupancyMaxActiveBlocksPerMultiprocessor(
+        &per_cu, (const void*)coop_peel_bits_kernel, 256, 0);
+    hipDeviceProp_t prop;
+    int dev = 0;
+    hipGetDevice(&dev);
+    hipGetDeviceProperties(&prop, dev);
+    max_blocks = per_cu * prop.multiProcessorCount;
+    if (max_blocks < 1) max_blocks = 1;
+  }
+  const long long total = (long long)N * W;
+  hipLaunchKernelGGL(dom_bits_kernel, dim3((int)((total + 255) / 256)),
+                     dim3(256), 0, stream, Y, Dbits, N, m, W);
+  int blocks = (N + 255) / 256;
+  if (blocks > max_blocks) blocks = max_blocks;
+  void* args[] = {(void*)&Dbits, (void*)&fmask, (void*)&n_dom,
+                  (void*)&ctrl, (void*)&rank, (void*)&N, (void*)&W};
+  const hipError_t err = hipLaunchCooperativeKernel(
+      (const void*)coop_peel_bits_kernel, dim3(blocks), dim3(256),
+      args, 0, stream);
+  return err == hipSuccess ? 0 : -1;
+}
+
 extern "C" int launch_peel_from_y(const float* Y, int* rank, int N, int m,
                                   hipStream_t stream) {
   const size_t lds = (size_t)N * m * sizeof(float) + (2 * N + 2) * sizeof(int);

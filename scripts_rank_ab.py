@@ -31,3 +31,14 @@ for name, Y in cases.items():
     same = torch.equal(r1.cpu(), r2.cpu())
     nf = int(r1.max().item()) + 1
     print(f"{name}: fronts={nf} single-block {t1:.3f} ms  matvec {t2:.3f} ms  agree={same}")
+
+# cooperative-path timings at large N
+print("--- large-N (cooperative route) ---")
+for n, m in ((3200, 2), (8192, 2), (4096, 5)):
+    Yr = torch.rand(n, m, generator=g).float().to(dev)
+    t3, r3 = bench(lambda y: ops.pareto_rank(y), Yr, reps=20)
+    from dmosopt_amd.ops import torch_ref
+    r_ref = torch_ref.pareto_rank(Yr.cpu().double())
+    print(f"N={n} m={m}: dispatched {t3:.3f} ms  fronts={int(r3.max())+1}  correct={torch.equal(r3.cpu(), r_ref)}")
+    tb, rb = bench(lambda y: ops._pareto_rank_gpu(y), Yr, reps=10)
+    print(f"          matvec {tb:.3f} ms")
