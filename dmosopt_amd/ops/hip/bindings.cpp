@@ -307,6 +307,11 @@ torch::Tensor dominance_degree_matrix(torch::Tensor Y) {
   return D;
 }
 
+static int COOP_MIN_N = []() {
+  const char* e = getenv("DMOSOPT_COOP_MIN_N");
+  return e ? atoi(e) : 2048;
+}();
+
 torch::Tensor pareto_rank(torch::Tensor Y) {
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
@@ -314,7 +319,7 @@ torch::Tensor pareto_rank(torch::Tensor Y) {
   // chased matvec path's readback every 16 fronts stalls pipelined
   // generation loops) and parallel across all CUs (the one-workgroup
   // peels serialize on one CU: 13.8 ms at N=8192 vs ~0.1-1 ms here)
-  if (N > 2048) {
+  if (N > COOP_MIN_N) {
     const int W = (N + 31) / 32;
     auto Yc = Y.contiguous().to(torch::kFloat32);
     auto opts_i = Y.options().dtype(torch::kInt32);

@@ -34,7 +34,7 @@ for name, Y in cases.items():
 
 # cooperative-path timings at large N
 print("--- large-N (cooperative route) ---")
-for n, m in ((3200, 2), (8192, 2), (4096, 5)):
+for n, m in ((1024, 2), (1536, 2), (2048, 2), (3200, 2), (8192, 2), (4096, 5)):
     Yr = torch.rand(n, m, generator=g).float().to(dev)
     t3, r3 = bench(lambda y: ops.pareto_rank(y), Yr, reps=20)
     from dmosopt_amd.ops import torch_ref
