@@ -309,7 +309,7 @@ torch::Tensor dominance_degree_matrix(torch::Tensor Y) {
 
 static int COOP_MIN_N = []() {
   const char* e = getenv("DMOSOPT_COOP_MIN_N");
-  return e ? atoi(e) : 2048;
+  return e ? atoi(e) : 1024;
 }();
 
 torch::Tensor pareto_rank(torch::Tensor Y) {
