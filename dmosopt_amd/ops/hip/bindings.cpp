@@ -421,13 +421,27 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   auto rank = pareto_rank(obj);                    // (N,) long
   auto crowd = crowding_distance(obj).to(torch::kFloat32);  // (N,)
   const int N = obj.size(0), d = parm.size(1), m = obj.size(1);
-  auto key = torch::empty({N}, rank.options());
-  launch_pack_rank_crowd((long long*)rank.data_ptr<int64_t>(), crowd.data_ptr<float>(),
-                         (long long*)key.data_ptr<int64_t>(), N, cur_stream());
-  auto perm = torch::argsort(key, /*stable=*/true, /*dim=*/-1,
-                             /*descending=*/false)
-                  .slice(0, 0, pop)
-                  .contiguous();
+  const int P0 = (int)std::min<int64_t>(pop, N);
+  torch::Tensor perm;
+  // single-block bitonic (key, idx) sort replaces pack + radix argsort
+  // (3-4 launches -> 1); comparator (key asc, idx asc) == stable argsort
+  auto perm_f = torch::empty({P0}, rank.options());
+  if (launch_rank_crowd_sort((long long*)rank.data_ptr<int64_t>(),
+                             crowd.data_ptr<float>(),
+                             (long long*)perm_f.data_ptr<int64_t>(), N, P0,
+                             cur_stream()) == 0) {
+    perm = perm_f;
+  } else {
+    auto key = torch::empty({N}, rank.options());
+    launch_pack_rank_crowd((long long*)rank.data_ptr<int64_t>(),
+                           crowd.data_ptr<float>(),
+                           (long long*)key.data_ptr<int64_t>(), N,
+                           cur_stream());
+    perm = torch::argsort(key, /*stable=*/true, /*dim=*/-1,
+                          /*descending=*/false)
+               .slice(0, 0, pop)
+               .contiguous();
+  }
   const int P = perm.size(0);
   auto parm_o = torch::empty({P, d}, parm.options());
   auto obj_o = torch::empty({P, m}, obj.options());

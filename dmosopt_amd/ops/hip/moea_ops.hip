@@ -203,3 +203,79 @@ extern "C" void launch_gather3(const float* parm, const float* obj,
                      dim3(256), 0, stream, parm, obj, rank, perm, parm_o,
                      obj_o, rank_o, pop, d, m);
 }
+
+// ---------------------------------------------- fused rank/crowd argsort
+// (key, idx) pairs bitonic-sorted in LDS by ONE workgroup: replaces
+// pack_rank_crowd + torch's radix argsort (3-4 launches) with one. The
+// comparator is (key asc, idx asc) — exactly torch's stable argsort of the
+// packed key, so the selection permutation is bitwise unchanged.
+#define RCS_TPB 1024
+
+__global__ __launch_bounds__(RCS_TPB) void rank_crowd_sort_kernel(
+    const long long* __restrict__ rank, const float* __restrict__ crowd,
+    long long* __restrict__ perm, int N, int pop) {
+  extern __shared__ char sh[];
+  long long* keys = (long long*)sh;  // M
+  int* idxs = (int*)(keys + 0);      // placed after keys below
+  int M = 1;
+  while (M < N) M <<= 1;
+  idxs = (int*)(keys + M);
+  const int tid = threadIdx.x;
+
+  for (int i = tid; i < M; i += RCS_TPB) {
+    if (i < N) {
+      // (rank << 32) | ~monotone_bits(crowd): descending crowding within
+      // ascending rank (same mapping as pack_rank_crowd_kernel /
+      // ops.fused_rank_metric_perm)
+      float d = crowd[i];
+      if (isnan(d)) d = 0.f;
+      unsigned int b = __float_as_uint(d);
+      unsigned int asc = (b & 0x80000000u) ? ~b : (b | 0x80000000u);
+      keys[i] = ((long long)rank[i] << 32) |
+                (long long)(0xFFFFFFFFu - asc);
+      idxs[i] = i;
+    } else {
+      keys[i] = 0x7FFFFFFFFFFFFFFFLL;  // pads sort to the end
+      idxs[i] = 0x7FFFFFFF;
+    }
+  }
+  __syncthreads();
+  for (int k = 2; k <= M; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int i = tid; i < M; i += RCS_TPB) {
+        const int ixj = i ^ j;
+        if (ixj > i) {
+          const bool up = ((i & k) == 0);
+          const long long ka = keys[i], kb = keys[ixj];
+          const int ia = idxs[i], ib = idxs[ixj];
+          const bool gt = (ka > kb) || (ka == kb && ia > ib);
+          if (gt == up) {
+            keys[i] = kb; keys[ixj] = ka;
+            idxs[i] = ib; idxs[ixj] = ia;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+  for (int i = tid; i < pop && i < N; i += RCS_TPB)
+    perm[i] = (long long)idxs[i];
+}
+
+extern "C" int launch_rank_crowd_sort(const long long* rank,
+                                      const float* crowd, long long* perm,
+                                      int N, int pop, hipStream_t stream) {
+  int M = 1;
+  while (M < N) M <<= 1;
+  const size_t lds = (size_t)M * (sizeof(long long) + sizeof(int));
+  if (lds > 144 * 1024) return -1;  // caller uses pack + torch argsort
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)rank_crowd_sort_kernel,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, 144 * 1024);
+    attr_set = true;
+  }
+  hipLaunchKernelGGL(rank_crowd_sort_kernel, dim3(1), dim3(RCS_TPB), lds,
+                     stream, rank, crowd, perm, N, pop);
+  return 0;
+}
