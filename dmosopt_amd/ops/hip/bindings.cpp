@@ -47,6 +47,8 @@ int launch_sceua_accept(float*, float*, const float*, const float*, const int*, 
 int launch_tournament(const float*, const long long*, float*, long long*, int, int, int, float, unsigned long long, hipStream_t);
 int launch_survivor_count(const long long*, const long long*, int, int, int, long long*, long long*, hipStream_t);
 void launch_pack_rank_crowd(const long long*, const float*, long long*, int, hipStream_t);
+int launch_rank_crowd_sort(const long long*, const float*, long long*, int,
+                           int, hipStream_t);
 void launch_variation_slots(const float*, const long long*, const long long*, const long long*, const long long*, const float*, const float*, const float*, const float*, float*, int, int, int, float, unsigned long long, unsigned long long, hipStream_t);
 void launch_gather3(const float*, const float*, const long long*, const long long*, float*, float*, long long*, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
