@@ -424,11 +424,16 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   auto crowd = crowding_distance(obj).to(torch::kFloat32);  // (N,)
   const int N = obj.size(0), d = parm.size(1), m = obj.size(1);
   const int P0 = (int)std::min<int64_t>(pop, N);
+  static int rcs_on = []() {
+    const char* e = getenv("DMOSOPT_RCS");
+    return e && e[0] == '0' ? 0 : 1;
+  }();
   torch::Tensor perm;
   // single-block bitonic (key, idx) sort replaces pack + radix argsort
   // (3-4 launches -> 1); comparator (key asc, idx asc) == stable argsort
   auto perm_f = torch::empty({P0}, rank.options());
-  if (launch_rank_crowd_sort((long long*)rank.data_ptr<int64_t>(),
+  if (rcs_on &&
+      launch_rank_crowd_sort((long long*)rank.data_ptr<int64_t>(),
                              crowd.data_ptr<float>(),
                              (long long*)perm_f.data_ptr<int64_t>(), N, P0,
                              cur_stream()) == 0) {
