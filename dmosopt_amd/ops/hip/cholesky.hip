@@ -253,8 +253,7 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
     int N, int k0) {
   __shared__ float S[CHOL_BS][CHOL_BS + 1];
-  __shared__ float colbuf[CHOL_BS];
-  __shared__ float colbuf2[CHOL_BS];
+  __shared__ float colbuf4[4][CHOL_BS];
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   float* Ab = A + (long long)b * N * N;
@@ -279,69 +278,63 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
       r[t] = (lane < bs && t < bs) ? S[lane][t] : 0.0f;
     float mylog = 0.0f;
     int bad = 0;
-    // TWO columns per round: factor j, apply its rank-1 update to column
-    // j+1 only, factor j+1, then ONE fused rank-2 pass over the remaining
-    // columns. Halves the broadcast/fence rounds on the serial pivot chain
-    // (the dominant latency of this kernel at ~30 us mean); the fmaf
-    // sequence per element is UNCHANGED, so the factor stays bitwise equal
-    // to the one-column version (SCE-UA accept decisions are bit-stable).
+    // GROUP_COLS columns per round: within the group each column first
+    // receives the updates of its in-group predecessors, is factored
+    // (shfl pivot + sqrt + div), and broadcast through LDS with ONE
+    // workgroup fence; then ONE fused rank-GROUP pass updates the
+    // remaining columns. Quarters the update-loop rounds on the serial
+    // pivot chain (the dominant latency of this kernel) while keeping the
+    // fmaf sequence PER ELEMENT identical to the one-column version —
+    // the factor is bitwise unchanged (SCE-UA accept decisions are
+    // bit-stable). Measured A/B in profiles/README.md.
+#define GROUP_COLS 4
 #pragma unroll
-    for (int j = 0; j < CHOL_BS; j += 2) {
-      if (j >= bs) continue;
-      float d = __shfl(r[j], j);
-      if (d <= 0.0f || !isfinite(d)) {
-        bad = bad ? bad : (k0 + j + 1);
-        d = 1e-30f;
-      }
-      d = sqrtf(d);
-      if (lane == j) {
-        r[j] = d;
-        mylog += logf(d);
-      } else if (lane > j) {
-        r[j] /= d;
-      }
-      // broadcast column j through LDS with ONE workgroup fence: the
-      // reads then pipeline freely, unlike a per-t __shfl chain
-      // (ds_bpermute each) or a volatile pointer (per-access ordering)
-      if (lane < bs) colbuf[lane] = r[j];
-      __threadfence_block();
-      const int j1 = j + 1;
-      if (j1 >= bs) {
-        // odd tail: plain rank-1 update of the remainder
+    for (int g = 0; g < CHOL_BS; g += GROUP_COLS) {
+      if (g >= bs) continue;
 #pragma unroll
-        for (int t = 0; t < CHOL_BS; ++t) {
-          if (t <= j || t >= bs) continue;
-          if (lane >= t) r[t] = fmaf(-r[j], colbuf[t], r[t]);
+      for (int q = 0; q < GROUP_COLS; ++q) {
+        const int j = g + q;
+        if (j >= bs) break;
+        // in-group predecessor updates for column j (same order as the
+        // sequential rank-1 passes)
+#pragma unroll
+        for (int p = 0; p < GROUP_COLS; ++p) {
+          if (p >= q) break;
+          if (lane >= j) r[j] = fmaf(-r[g + p], colbuf4[p][j], r[j]);
         }
-        continue;
+        float d = __shfl(r[j], j);
+        if (d <= 0.0f || !isfinite(d)) {
+          bad = bad ? bad : (k0 + j + 1);
+          d = 1e-30f;
+        }
+        d = sqrtf(d);
+        if (lane == j) {
+          r[j] = d;
+          mylog += logf(d);
+        } else if (lane > j) {
+          r[j] /= d;
+        }
+        // broadcast column j through LDS with ONE workgroup fence: the
+        // reads then pipeline freely, unlike a per-t __shfl chain
+        // (ds_bpermute each) or a volatile pointer (per-access ordering)
+        if (lane < bs) colbuf4[q][lane] = r[j];
+        __threadfence_block();
       }
-      // column j+1: apply col j's update, then factor it
-      if (lane >= j1) r[j1] = fmaf(-r[j], colbuf[j1], r[j1]);
-      float d1 = __shfl(r[j1], j1);
-      if (d1 <= 0.0f || !isfinite(d1)) {
-        bad = bad ? bad : (k0 + j1 + 1);
-        d1 = 1e-30f;
-      }
-      d1 = sqrtf(d1);
-      if (lane == j1) {
-        r[j1] = d1;
-        mylog += logf(d1);
-      } else if (lane > j1) {
-        r[j1] /= d1;
-      }
-      if (lane < bs) colbuf2[lane] = r[j1];
-      __threadfence_block();
-      // fused rank-2 update of the remaining columns (same fmaf order per
-      // element as two sequential rank-1 passes)
+      // fused rank-GROUP update of the columns beyond the group
+      const int gend = min(g + GROUP_COLS, bs);
 #pragma unroll
       for (int t = 0; t < CHOL_BS; ++t) {
-        if (t <= j1 || t >= bs) continue;
+        if (t < g + GROUP_COLS || t >= bs) continue;
         if (lane >= t) {
-          r[t] = fmaf(-r[j], colbuf[t], r[t]);
-          r[t] = fmaf(-r[j1], colbuf2[t], r[t]);
+#pragma unroll
+          for (int q = 0; q < GROUP_COLS; ++q) {
+            if (g + q >= gend) break;
+            r[t] = fmaf(-r[g + q], colbuf4[q][t], r[t]);
+          }
         }
       }
     }
+#undef GROUP_COLS
     // stage the factored block back to LDS (global writeback below is
     // done coalesced by the whole workgroup)
     if (lane < bs) {
@@ -672,11 +665,11 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
 }
 
 // Blocked forward substitution: solve L z = y for R right-hand sides.
-// One block per (batch, rhs). Per 32-column panel: thread 0 solves the
-// 32x32 diagonal block serially from LDS while the other threads wait,
-// then all threads apply the rank-32 update in parallel — 2 barriers per
-// panel instead of 2 per column (the former per-column loop spent ~600
-// barriers on N=300).
+// One block per (batch, rhs). Per 32-column panel: wave 0 solves the
+// 32x32 diagonal block wave-synchronously (lane j owns row j, solved
+// entries broadcast by shuffle), then all threads apply the rank-32
+// update in parallel — 2 barriers per panel instead of 2 per column (the
+// former per-column loop spent ~600 barriers on N=300).
 #define TRSV_BS 32
 
 __global__ void forward_solve_batched_kernel(const float* __restrict__ L,
