@@ -249,11 +249,12 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
 // Factor the 32x32 diagonal block at (k0,k0) and panel-solve the rows below
 // it. One block per matrix; also accumulates the step's logdet contribution
 // deterministically (no atomics: k-steps are stream-ordered).
+template <int GROUP_COLS>
 __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
     int N, int k0) {
   __shared__ float S[CHOL_BS][CHOL_BS + 1];
-  __shared__ float colbuf4[4][CHOL_BS];
+  __shared__ float colbuf4[GROUP_COLS][CHOL_BS];
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   float* Ab = A + (long long)b * N * N;
@@ -287,7 +288,6 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
     // fmaf sequence PER ELEMENT identical to the one-column version —
     // the factor is bitwise unchanged (SCE-UA accept decisions are
     // bit-stable). Measured A/B in profiles/README.md.
-#define GROUP_COLS 4
 #pragma unroll
     for (int g = 0; g < CHOL_BS; g += GROUP_COLS) {
       if (g >= bs) continue;
@@ -334,7 +334,6 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
         }
       }
     }
-#undef GROUP_COLS
     // stage the factored block back to LDS (global writeback below is
     // done coalesced by the whole workgroup)
     if (lane < bs) {
@@ -821,6 +820,18 @@ __global__ void zero_f32_kernel(float* __restrict__ p, int n) {
   if (i < n) p[i] = 0.0f;
 }
 
+// panel-group-size selection: 2- vs 4-column factor rounds, same-box A/B
+// via DMOSOPT_CHOL_GROUP (bitwise-identical results either way)
+static inline int panel_group_cols() {
+  static int g = -1;
+  if (g < 0) {
+    const char* e = getenv("DMOSOPT_CHOL_GROUP");
+    g = (e && e[0] == '2') ? 2 : 4;
+  }
+  return g;
+}
+#define LAUNCH_PANEL(B_, stream_, A_, logdet_, info_, N_, k0_)                 do {                                                                           if (panel_group_cols() == 2)                                                   hipLaunchKernelGGL((chol_panel_kernel<2>), dim3(B_), dim3(CHOLP_TPB),                           0, stream_, A_, logdet_, info_, N_, k0_);               else                                                                           hipLaunchKernelGGL((chol_panel_kernel<4>), dim3(B_), dim3(CHOLP_TPB),                           0, stream_, A_, logdet_, info_, N_, k0_);             } while (0)
+
 extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
                                        int B, int N, hipStream_t stream) {
   // Cross-stream software pipeline: panel(k+1) only depends on the FIRST
@@ -878,8 +889,7 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
   }
   if (!overlap) {
     for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
-      hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0,
-                         stream, A, logdet, info, N, k0);
+      LAUNCH_PANEL(B, stream, A, logdet, info, N, k0);
       const int trailing = N - k0 - CHOL_BS;
       if (trailing > 0) {
         const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
@@ -892,8 +902,7 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
   int kstep = 0;
   bool have_rest_prev = false;
   for (int k0 = 0; k0 < N; k0 += CHOL_BS, ++kstep) {
-    hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0, stream,
-                       A, logdet, info, N, k0);
+    LAUNCH_PANEL(B, stream, A, logdet, info, N, k0);
     const int trailing = N - k0 - CHOL_BS;
     if (trailing <= 0) continue;
     const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
@@ -918,6 +927,7 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
   if (have_rest_prev) hipStreamWaitEvent(stream, evR[(kstep - 1) & 1], 0);
 }
 
+
 // bf16-SYRK variant of the right-looking multik path (config-#2 precision
 // route): panels factor in exact fp32 (chol_panel_kernel), only the
 // trailing C -= P P^T runs on the bf16 matrix units (matern_bf16.hip).
@@ -930,8 +940,7 @@ extern "C" void launch_cholesky_multik_bf16(float* A, float* logdet,
   hipLaunchKernelGGL(zero_f32_kernel, dim3((B + 255) / 256), dim3(256), 0,
                      stream, logdet, B);
   for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
-    hipLaunchKernelGGL(chol_panel_kernel, dim3(B), dim3(CHOLP_TPB), 0, stream,
-                       A, logdet, info, N, k0);
+    LAUNCH_PANEL(B, stream, A, logdet, info, N, k0);
     const int trailing = N - k0 - CHOL_BS;
     if (trailing > 0) {
       const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
