@@ -269,6 +269,18 @@ class CollectiveFarm:
         fn = self.eval_funs[opt_id]
         P = len(points)
         t_call = time.time()
+        # replicated-flow guard: a rank with a different batch size would
+        # otherwise hang in the mismatched all_gather below — make it a
+        # loud error instead (one tiny collective per BATCH, not per point)
+        sizes = ctx.all_gather_stack(
+            torch.tensor([P], dtype=torch.int64, device=ctx.device)
+        ).flatten().tolist()
+        if len(set(sizes)) != 1:
+            raise RuntimeError(
+                f"CollectiveFarm: request-batch sizes diverged across ranks "
+                f"({sizes}); the replicated control flow is broken "
+                "(unseeded RNG or rank-dependent strategy state?)"
+            )
         my_idx = list(ctx.shard_indices(P))
         max_shard = ctx.max_shard_size(P)
         fbuf = np.zeros((max_shard, spec.float_width), dtype=np.float64)

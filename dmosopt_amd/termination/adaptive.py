@@ -285,7 +285,15 @@ class ResourceAwareTermination(Termination):
             self._armed_at = _time.time()
         if self.max_time_seconds is not None:
             used = _time.time() - self._armed_at
-            if used > self.max_time_seconds:
+            over = used > self.max_time_seconds
+            # multi-rank: local clocks diverge and a split decision would
+            # hang the replicated control flow — rank 0's clock decides
+            from dmosopt_amd.parallel.context import get_context
+
+            ctx = get_context()
+            if ctx is not None and ctx.world > 1:
+                over = ctx.bcast_flag(over, src=0)
+            if over:
                 _log(self.problem, f"stop: wall-clock budget used ({used:.1f}s)")
                 return True
         if self.max_function_evals is not None:
