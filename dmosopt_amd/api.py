@@ -1,9 +1,12 @@
 """Public API: dmosopt_amd.run(dopt_params) (reference dmosopt.py:2441-2596).
 
-Role split is by torch.distributed rank instead of MPI rank: rank 0 builds
-the driver and runs the epoch loop; other ranks (one per GPU over RCCL, or
-gloo on CPU) enter the farm worker loop. Single-process runs use a local
-farm — no process group needed.
+Multi-rank execution (one process per GPU over RCCL, or gloo on CPU) is
+REPLICATED CONTROL FLOW: every rank runs the full driver/epoch loop with
+identical seeds, while objective results, surrogate predictions and GP
+hyperparameters move as tensor collectives (parallel/context.py). Rank 0
+alone writes the H5 file, logs, and returns the best set; other ranks
+return None (API parity with the reference's MPI workers). Single-process
+runs use a local farm — no process group needed.
 """
 
 from __future__ import annotations
