@@ -37,6 +37,20 @@ def velocity_vector(rng, position, velocity, archive, crowding, xlb, xub):
             ind_1, ind_2 = ind_2, ind_1
     else:
         ind_1 = ind_2 = 0
+    if (
+        position.device.type == "cuda"
+        and position.dtype == torch.float32
+        and ops.native_available()
+    ):
+        # fused velocity + clamp in one launch (ops/hip/moea_ops.hip)
+        from dmosopt_amd import _hipops
+
+        return _hipops.smpso_velocity(
+            position.contiguous(), velocity.contiguous(),
+            archive[int(ind_1)].contiguous(), archive[int(ind_2)].contiguous(),
+            xlb.float().contiguous(), xub.float().contiguous(),
+            w, c1 * r1, c2 * r2, chi,
+        )
     delta = (xub - xlb) / 2.0
     out = (
         w * velocity

@@ -79,6 +79,9 @@ void launch_agemoea_survival(const float*, const unsigned char*, float*, int,
                              hipStream_t);
 void launch_minkowski_norm_matrix(const float*, float*, int, int, float,
                                   hipStream_t);
+void launch_smpso_velocity(const float*, const float*, const float*,
+                           const float*, const float*, const float*, float*,
+                           int, int, float, float, float, float, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -644,6 +647,22 @@ torch::Tensor lacour_volumes(torch::Tensor coords, torch::Tensor defs,
   return vol.narrow(0, 0, U);
 }
 
+torch::Tensor smpso_velocity(torch::Tensor position, torch::Tensor velocity,
+                             torch::Tensor leader1, torch::Tensor leader2,
+                             torch::Tensor xlb, torch::Tensor xub, double w,
+                             double a1, double a2, double chi) {
+  CHECK_GPU(position);
+  CHECK_GPU(velocity);
+  const int n = position.size(0), d = position.size(1);
+  auto out = torch::empty_like(position);
+  launch_smpso_velocity(position.data_ptr<float>(), velocity.data_ptr<float>(),
+                        leader1.data_ptr<float>(), leader2.data_ptr<float>(),
+                        xlb.data_ptr<float>(), xub.data_ptr<float>(),
+                        out.data_ptr<float>(), n, d, (float)w, (float)a1,
+                        (float)a2, (float)chi, cur_stream());
+  return out;
+}
+
 torch::Tensor minkowski_norm_matrix(torch::Tensor Y, double p) {
   CHECK_GPU(Y);
   const int m = Y.size(0), d = Y.size(1);
@@ -732,6 +751,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hv_mc_uniform_hits", &hv_mc_uniform_hits);
   m.def("hv_fpras_hits", &hv_fpras_hits);
   m.def("get_duplicates", &get_duplicates);
+  m.def("smpso_velocity", &smpso_velocity,
+        "Fused SMPSO constriction velocity + clamp");
   m.def("minkowski_norm_matrix", &minkowski_norm_matrix,
         "Row-normalized Minkowski-p distance matrix in one pass");
   m.def("agemoea_survival", &agemoea_survival,

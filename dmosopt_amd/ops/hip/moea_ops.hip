@@ -279,3 +279,40 @@ extern "C" int launch_rank_crowd_sort(const long long* rank,
                      stream, rank, crowd, perm, N, pop);
   return 0;
 }
+
+// ------------------------------------------------- SMPSO velocity update
+// Constriction-factor velocity + clamp in one launch (reference
+// SMPSO.py:316-348; replaces the ~6-op torch expression per swarm per
+// generation). Scalars (w, c1*r1, c2*r2, chi) are drawn host-side for RNG
+// parity with the reference's per-swarm scalar draws.
+__global__ void smpso_velocity_kernel(
+    const float* __restrict__ position,  // (n, d)
+    const float* __restrict__ velocity,  // (n, d)
+    const float* __restrict__ leader1,   // (d,)
+    const float* __restrict__ leader2,   // (d,)
+    const float* __restrict__ xlb, const float* __restrict__ xub,
+    float* __restrict__ out, int n, int d, float w, float a1, float a2,
+    float chi) {
+  const long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= (long long)n * d) return;
+  const int k = (int)(t % d);
+  const float p = position[t];
+  const float v = chi * (w * velocity[t] + a1 * (leader1[k] - p) +
+                         a2 * (leader2[k] - p));
+  const float delta = 0.5f * (xub[k] - xlb[k]);
+  out[t] = fminf(fmaxf(v, -delta), delta);
+}
+
+extern "C" void launch_smpso_velocity(const float* position,
+                                      const float* velocity,
+                                      const float* leader1,
+                                      const float* leader2, const float* xlb,
+                                      const float* xub, float* out, int n,
+                                      int d, float w, float a1, float a2,
+                                      float chi, hipStream_t stream) {
+  const long long total = (long long)n * d;
+  hipLaunchKernelGGL(smpso_velocity_kernel,
+                     dim3((unsigned)((total + 255) / 256)), dim3(256), 0,
+                     stream, position, velocity, leader1, leader2, xlb, xub,
+                     out, n, d, w, a1, a2, chi);
+}

@@ -830,3 +830,48 @@ def test_variational_gp_on_gpu(dev):
     mean, var = sm.predict(rng.random((16, 5)))
     assert mean.shape == (16, 2) and np.isfinite(mean).all()
     assert (np.asarray(var) >= 0).all()
+
+
+def test_smpso_velocity_kernel_matches_torch(dev):
+    from dmosopt_amd import _hipops
+
+    g = torch.Generator().manual_seed(71)
+    n, d = 200, 12
+    pos = torch.rand(n, d, generator=g).float().to(dev)
+    vel = (torch.rand(n, d, generator=g).float().to(dev) - 0.5) * 0.2
+    L1 = torch.rand(d, generator=g).float().to(dev)
+    L2 = torch.rand(d, generator=g).float().to(dev)
+    xlb = torch.zeros(d).float().to(dev)
+    xub = torch.ones(d).float().to(dev)
+    w, a1, a2, chi = 0.3, 1.9 * 0.7, 2.1 * 0.2, 0.73
+    got = _hipops.smpso_velocity(pos, vel, L1, L2, xlb, xub, w, a1, a2, chi)
+    delta = (xub - xlb) / 2.0
+    want = (
+        (w * vel + a1 * (L1[None, :] - pos) + a2 * (L2[None, :] - pos)) * chi
+    ).clamp(-delta, delta)
+    torch.testing.assert_close(got, want, rtol=1e-6, atol=1e-6)
+
+
+def test_smpso_gpu_e2e_uses_kernel(dev):
+    from dmosopt_amd.moea.smpso import SMPSOOptimizer
+    from dmosopt_amd.models.model import Model
+
+    rng = np.random.default_rng(72)
+
+    class Obj:
+        def evaluate(self, x):
+            x = np.asarray(x)
+            return np.column_stack([x.sum(1), (1 - x).sum(1)])
+
+    opt = SMPSOOptimizer(popsize=20, nInput=6, nOutput=2, model=Model(objective=Obj()))
+    opt.set_device(dev)
+    bounds = np.stack([np.zeros(6), np.ones(6)], axis=1)
+    x0 = opt.generate_initial(bounds, rng)
+    y0 = Obj().evaluate(x0)
+    opt.initialize_strategy(x0, y0, bounds, rng)
+    for _ in range(3):
+        xg, st = opt.generate()
+        yg = Obj().evaluate(xg.cpu().numpy() if isinstance(xg, torch.Tensor) else xg)
+        opt.update(xg, torch.as_tensor(yg, dtype=torch.float32, device=dev), st)
+    px, py = opt.population_objectives
+    assert torch.isfinite(torch.as_tensor(np.asarray(py) if not isinstance(py, torch.Tensor) else py.cpu().numpy())).all()
