@@ -76,3 +76,71 @@ def test_epoch_resample_quality_matches_reference():
     # ideal at this reference point is ~120.66)
     assert ours_hv == pytest.approx(ref_hv, abs=0.5)
     assert ours_hv > 118.0
+
+
+def _ref_modules():
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    import dmosopt.indicators as ref_ind
+    import dmosopt.dda as ref_dda
+
+    return ref_ind, ref_dda
+
+
+def test_crowding_distance_exact_parity():
+    """ops.crowding_distance == the reference's crowding_distance_metric
+    value-for-value (fp64)."""
+    import torch
+
+    from dmosopt_amd import ops
+
+    ref_ind, _ = _ref_modules()
+    rng = np.random.default_rng(7)
+    # distinct coordinate values: with TIED values the per-dim sort order
+    # is implementation-defined (the reference uses numpy's unstable
+    # quicksort) and crowding then differs legitimately
+    for n, m in ((16, 2), (100, 3), (257, 5)):
+        Y = rng.random((n, m))
+        ours = ops.crowding_distance(torch.as_tensor(Y, dtype=torch.float64)).numpy()
+        theirs = ref_ind.crowding_distance_metric(Y)
+        np.testing.assert_allclose(ours, theirs, rtol=1e-12, atol=1e-12)
+
+
+def test_pareto_rank_partition_parity():
+    """Front partitions equal the reference's dda_ens ranking."""
+    import torch
+
+    from dmosopt_amd import ops
+
+    _, ref_dda = _ref_modules()
+    rng = np.random.default_rng(8)
+    for n, m in ((60, 2), (200, 3), (500, 4)):
+        Y = rng.random((n, m))
+        Y[:5] = Y[10:15]
+        ours = ops.pareto_rank(torch.as_tensor(Y, dtype=torch.float64)).numpy()
+        theirs = ref_dda.dda_ens(Y)
+        np.testing.assert_array_equal(ours, np.asarray(theirs))
+
+
+def test_sortmo_permutation_parity():
+    """order_mo's (perm, rank) against the reference sortMO with the
+    crowding metric: identical survivor ordering."""
+    import torch
+
+    from dmosopt_amd import ops
+
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.MOEA import sortMO
+
+    rng = np.random.default_rng(9)
+    x = rng.random((120, 6))
+    y = rng.random((120, 3))
+    xs, ys, ranks, dists = sortMO(x, y, y_distance_metrics=["crowding"])
+    perm, rank_t, _ = ops.order_mo(
+        torch.as_tensor(x, dtype=torch.float64),
+        torch.as_tensor(y, dtype=torch.float64),
+        y_distance_metrics=["crowding"],
+    )
+    np.testing.assert_allclose(x[perm.numpy()], xs, rtol=0, atol=0)
+    np.testing.assert_array_equal(rank_t.numpy(), ranks)
