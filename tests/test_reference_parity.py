@@ -144,3 +144,103 @@ def test_sortmo_permutation_parity():
     )
     np.testing.assert_allclose(x[perm.numpy()], xs, rtol=0, atol=0)
     np.testing.assert_array_equal(rank_t.numpy(), ranks)
+
+
+def test_lacour_hv_parity_d4_d5():
+    """d>=4 hypervolume equals the reference's Lacour box decomposition."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.hv_box_decomposition import HyperVolumeBoxDecomposition as RefHV
+
+    from dmosopt_amd.hv.exact import HyperVolumeBoxDecomposition
+
+    rng = np.random.default_rng(11)
+    for d, n in ((4, 40), (5, 30)):
+        ref_pt = np.full(d, 1.2)
+        pts = rng.random((n, d))
+        ours = HyperVolumeBoxDecomposition(ref_pt).compute_hypervolume(pts)
+        theirs = RefHV(ref_pt).compute_hypervolume(pts)
+        assert ours == pytest.approx(theirs, rel=1e-9), (d, n)
+
+
+def test_ehvi_selection_parity():
+    """EHVI candidate selection picks the same candidates with matching
+    scores as the reference's batch EHVI."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.hv_box_decomposition import HyperVolumeBoxDecomposition as RefHV
+
+    from dmosopt_amd.hv.exact import HyperVolumeBoxDecomposition
+
+    rng = np.random.default_rng(12)
+    ref_pt = np.full(2, 2.0)
+    front = rng.random((25, 2))
+    means = rng.random((200, 2)) * 1.5
+    vars_ = rng.random((200, 2)) * 0.1 + 0.01
+    sel_o, sc_o = HyperVolumeBoxDecomposition(ref_pt).select_candidates(
+        front, means, vars_, n_select=10
+    )
+    sel_r, sc_r = RefHV(ref_pt).select_candidates(front, means, vars_, n_select=10)
+    np.testing.assert_array_equal(np.sort(sel_o), np.sort(sel_r))
+    np.testing.assert_allclose(np.sort(sc_o), np.sort(sc_r), rtol=1e-6)
+
+
+def _decision_sequence(term, X_seq, F_seq):
+    from dmosopt_amd.datatypes import OptHistory
+
+    out = []
+    for g, (X, F) in enumerate(zip(X_seq, F_seq), start=1):
+        out.append(bool(term.has_terminated(OptHistory(g, g * 10, X, F, None))))
+        if out[-1]:
+            break
+    return out
+
+
+class _P:
+    def __init__(self, d=4, m=2):
+        import logging
+
+        self.n_objectives = m
+        self.lb = np.zeros(d)
+        self.ub = np.ones(d)
+        # the reference's criteria log unconditionally
+        self.logger = logging.getLogger("parity_test")
+
+
+def test_termination_decision_parity():
+    """The rewritten windowed criteria make the SAME stop decisions, at the
+    SAME generations, as the reference classes on identical histories."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    import dmosopt.termination as rt
+
+    import dmosopt_amd.termination as ot
+
+    rng = np.random.default_rng(13)
+    d, m, T = 4, 2, 60
+
+    # history A: converging front (deltas decay geometrically)
+    X_seq, F_seq = [], []
+    F = rng.random((30, m)) + 1.0
+    X = rng.random((30, d))
+    for t in range(T):
+        F = 1.0 + (F - 1.0) * 0.7 + rng.normal(0, 1e-6, F.shape)
+        X = 0.5 + (X - 0.5) * 0.7
+        F_seq.append(F.copy())
+        X_seq.append(X.copy())
+    # history B: noisy, never converges
+    Xb_seq = [rng.random((30, d)) for _ in range(T)]
+    Fb_seq = [rng.random((30, m)) for _ in range(T)]
+
+    for mk_ref, mk_ours in (
+        (lambda: rt.MultiObjectiveToleranceTermination(_P(), tol=0.005, n_last=5),
+         lambda: ot.MultiObjectiveToleranceTermination(_P(), tol=0.005, n_last=5)),
+        (lambda: rt.ParameterToleranceTermination(_P(), tol=1e-3, n_last=5),
+         lambda: ot.ParameterToleranceTermination(_P(), tol=1e-3, n_last=5)),
+        (lambda: rt.MaximumGenerationTermination(_P(), 25),
+         lambda: ot.MaximumGenerationTermination(_P(), 25)),
+    ):
+        for Xs, Fs in ((X_seq, F_seq), (Xb_seq, Fb_seq)):
+            ref_seq = _decision_sequence(mk_ref(), Xs, Fs)
+            our_seq = _decision_sequence(mk_ours(), Xs, Fs)
+            assert ref_seq == our_seq, (mk_ref, len(ref_seq), len(our_seq))
