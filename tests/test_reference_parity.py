@@ -244,3 +244,25 @@ def test_termination_decision_parity():
             ref_seq = _decision_sequence(mk_ref(), Xs, Fs)
             our_seq = _decision_sequence(mk_ours(), Xs, Fs)
             assert ref_seq == our_seq, (mk_ref, len(ref_seq), len(our_seq))
+
+
+def test_agemoea_environmental_selection_parity():
+    """AGE-MOEA survivor selection picks the same survivors with the same
+    scores as the reference on identical populations."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.AGEMOEA import environmental_selection as ref_sel
+
+    from dmosopt_amd.moea.agemoea import environmental_selection as our_sel
+
+    rng = np.random.default_rng(14)
+    n, d, m, pop = 180, 5, 3, 90
+    X = rng.random((n, d))
+    Y = rng.random((n, m))
+    ox, oy, orank, ocrowd = our_sel(np.random.default_rng(0), X, Y, pop, d, m)
+    rx, ry, rrank, rcrowd = ref_sel(np.random.default_rng(0), X, Y, pop, d, m)
+    # same survivor SET (ordering may differ by tie-order inside fronts)
+    np.testing.assert_allclose(
+        np.sort(oy, axis=0), np.sort(np.asarray(ry), axis=0), rtol=1e-12
+    )
+    np.testing.assert_array_equal(np.sort(np.asarray(orank)), np.sort(np.asarray(rrank)))
