@@ -111,11 +111,15 @@ def glp_design(n: int, s: int, local_random, max_candidates: int = 64) -> np.nda
     best, best_score = None, np.inf
     i = np.arange(1, m)[:, None]  # (n, 1) — drop the final all-ones row later
     for g in designs:
-        x = ((i * g[None, :]) % m) / float(m)
-        x = x[:n]
-        score = cd2(x) if n <= 512 else _cd2_cheap(x)
-        if score < best_score:
-            best, best_score = x, score
+        u = (i * g[None, :]) % m
+        # two lattice-to-unit-cube maps compete on CD2: the plain u/m and
+        # the reference's CENTERED (2u-1)/(2m) (GLP.py glpmod + centering),
+        # which usually scores lower discrepancy
+        for x in (u / float(m), (2.0 * u - 1.0) / (2.0 * m)):
+            x = x[:n]
+            score = cd2(x) if n <= 512 else _cd2_cheap(x)
+            if score < best_score:
+                best, best_score = x, score
     return best
 
 

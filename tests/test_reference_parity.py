@@ -297,3 +297,33 @@ def test_gp_surrogate_prediction_parity():
     rmse_r = float(np.sqrt(((mr - Ytrue) ** 2).mean()))
     rmse_o = float(np.sqrt(((mo - Ytrue) ** 2).mean()))
     assert abs(rmse_r - rmse_o) < 5e-3, (rmse_r, rmse_o)
+
+
+def test_discrepancy_metrics_parity():
+    """CD2/MD2/WD2 match the reference's discrepancy implementations."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt import discrepancy as rd
+
+    from dmosopt_amd.sampling import discrepancy as od
+
+    x = np.random.default_rng(0).random((30, 4))
+    assert od.CD2(x) == pytest.approx(rd.CD2(x), rel=1e-10)
+    assert od.MD2(x) == pytest.approx(rd.MD2(x), rel=1e-10)
+    assert od.WD2(x) == pytest.approx(rd.WD2(x), rel=1e-10)
+
+
+def test_glp_design_quality_comparable():
+    """Our GLP designs score within 10% of the reference's CD2 (different
+    generating-vector searches, same family)."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt import discrepancy as rd
+    from dmosopt import sampling as rs
+
+    from dmosopt_amd import sampling as os_
+
+    for n, s in ((50, 4), (100, 6)):
+        a = rs.glp(n, s, local_random=np.random.default_rng(3))
+        b = os_.glp(n, s, local_random=np.random.default_rng(3))
+        assert rd.CD2(b) <= rd.CD2(a) * 1.10, (n, s)
