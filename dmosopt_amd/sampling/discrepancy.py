@@ -17,7 +17,19 @@ def _pair_terms(x: np.ndarray):
 
 
 def MD2(x: np.ndarray) -> float:
-    """Mixture discrepancy."""
+    """Modified L2-discrepancy (the reference's MD2, discrepancy.py:38-59),
+    vectorized: sqrt((4/3)^s - 2^(1-s)/n * sum_k prod_j (3 - x_kj^2)
+    + 1/n^2 * sum_kl prod_j (2 - max(x_kj, x_lj)))."""
+    n, s = x.shape
+    t1 = (4.0 / 3.0) ** s
+    t2 = (2.0 ** (1 - s) / n) * np.prod(3.0 - x**2, axis=1).sum()
+    xi, xj = _pair_terms(x)
+    t3 = np.prod(2.0 - np.maximum(xi, xj), axis=2).sum() / (n * n)
+    return float(np.sqrt(max(t1 - t2 + t3, 0.0)))
+
+
+def mixture_discrepancy(x: np.ndarray) -> float:
+    """Mixture discrepancy (Zhou et al.)."""
     n, s = x.shape
     d1 = np.abs(x - 0.5)
     t1 = (19.0 / 12.0) ** s
