@@ -207,86 +207,104 @@ def _wfg_shape_convex(t: torch.Tensor, m: int) -> torch.Tensor:
 
 
 def wfg1(x, n_obj: int = 3, k: Optional[int] = None) -> torch.Tensor:
-    """Simplified WFG1: bias + mixed shape (reference moo_benchmarks.py:286)."""
+    """WFG1 (value parity with reference moo_benchmarks.py wfg1, batched):
+    [0, 2i] normalization, ^0.02 bias on the tail, 0.35 + 0.65 scaling,
+    chunk-max shape vector, convex shape scaled by (1 + i)."""
     x = _as2d(x)
     n_var = x.shape[1]
     if k is None:
         k = n_obj - 1
-    z = x / torch.arange(
-        2, 2 * n_var + 1, 2, dtype=x.dtype, device=x.device
-    )  # normalize domain [0, 2i]
-    z = z.clamp(0.0, 1.0)
-    # b_flat on tail
-    zt = z.clone()
-    zt[:, k:] = torch.abs(z[:, k:] - 0.35) / 0.35
-    t = zt[:, :k]
-    g = zt[:, k:].mean(dim=1)
-    f = _wfg_shape_convex(torch.cat([t, torch.zeros_like(t[:, :0])], dim=1), n_obj)
-    return (1.0 + g[:, None]) * f + 2.0 * torch.arange(
-        1, n_obj + 1, dtype=x.dtype, device=x.device
-    )[None, :] * 0.0
+    ll = n_var - k
+    idx = torch.arange(1, n_var + 1, dtype=x.dtype, device=x.device)
+    y = x / (2.0 * idx)
+    t1 = y.clone()
+    t1[:, k:] = y[:, k:].clamp_min(0.0) ** 0.02
+    t2 = t1.clone()
+    t2[:, k:] = 0.35 + 0.65 * t1[:, k:]
+    xv = torch.empty(x.shape[0], n_obj, dtype=x.dtype, device=x.device)
+    for i in range(n_obj - 1):
+        xv[:, i] = t2[:, i * ll : (i + 1) * ll].max(dim=1).values
+    xv[:, -1] = t2[:, -ll:].mean(dim=1)
+    scale = torch.arange(2, n_obj + 2, dtype=x.dtype, device=x.device)
+    return _wfg_shape_convex(xv, n_obj) * scale[None, :]
 
 
 def wfg4(x, n_obj: int = 3, k: Optional[int] = None) -> torch.Tensor:
-    """Simplified WFG4: multi-modal shift + concave shape."""
+    """WFG4 (value parity with the reference, batched): multi-modal
+    transform y + 0.35 - 0.15 cos(10 pi y - 5), chunk-mean shape vector,
+    convex shape scaled by (1 + i)."""
     x = _as2d(x)
     n_var = x.shape[1]
     if k is None:
         k = n_obj - 1
-    z = x / torch.arange(2, 2 * n_var + 1, 2, dtype=x.dtype, device=x.device)
-    z = z.clamp(0.0, 1.0)
-    s = (
-        torch.sin(10.0 * PI * z + PI / 10.0).abs() * 0.1 + torch.abs(z - 0.35)
-    )  # multimodal transform
-    g = s[:, k:].mean(dim=1)
-    theta = z[:, :k]
-    return (1.0 + g[:, None]) * _dtlz_concave(theta, torch.zeros_like(g), n_obj)
+    ll = n_var - k
+    idx = torch.arange(1, n_var + 1, dtype=x.dtype, device=x.device)
+    y = x / (2.0 * idx)
+    t1 = y + 0.35 - 0.15 * torch.cos(10.0 * PI * y - 5.0)
+    xv = torch.empty(x.shape[0], n_obj, dtype=x.dtype, device=x.device)
+    for i in range(n_obj - 1):
+        xv[:, i] = t1[:, i * ll : (i + 1) * ll].mean(dim=1)
+    xv[:, -1] = t1[:, -ll:].mean(dim=1)
+    scale = torch.arange(2, n_obj + 2, dtype=x.dtype, device=x.device)
+    return _wfg_shape_convex(xv, n_obj) * scale[None, :]
 
 
 # ------------------------------------------------------------------- MaF
-def maf1(x, n_obj: int = 5) -> torch.Tensor:
-    """Inverted DTLZ1-like linear front."""
-    x = _as2d(x)
-    k = x.shape[1] - n_obj + 1
-    g = _dtlz_g2(x[:, -k:])
-    lin = _dtlz_linear(x[:, : n_obj - 1], torch.zeros_like(g), n_obj)
-    return (1.0 + g[:, None]) * (1.0 - lin)
-
-
-def maf2(x, n_obj: int = 5) -> torch.Tensor:
-    """DTLZ2 variant with per-objective g segments."""
-    x = _as2d(x)
-    n_var = x.shape[1]
-    k = n_var - n_obj + 1
-    n = x.shape[0]
-    f = torch.empty(n, n_obj, dtype=x.dtype, device=x.device)
-    seg = max(1, k // n_obj)
-    for i in range(n_obj):
-        lo = n_obj - 1 + i * seg
-        hi = min(n_var, lo + seg)
-        xi = x[:, lo:hi] if hi > lo else x[:, -1:]
-        g = (((xi / 2.0 + 0.25) - 0.5) ** 2).sum(dim=1)
-        theta = x[:, : n_obj - 1] / 2.0 + 0.25
-        fi = 1.0 + g
-        for j in range(n_obj - i - 1):
-            fi = fi * torch.cos(theta[:, j] * PI / 2.0)
+def _maf_products(x: torch.Tensor, g: torch.Tensor, n_obj: int,
+                  trig: bool) -> torch.Tensor:
+    """Common MaF objective products: f[i] = (1+g) * prod_{j<m-i-1} p_j *
+    (q_{m-i-1} if i > 0), with (p, q) = (cos(x pi/2), sin(x pi/2)) for the
+    concave family or (x, 1-x) for the linear family."""
+    n, m = x.shape[0], n_obj
+    if trig:
+        p = torch.cos(x[:, :m] * PI / 2.0)
+        q = torch.sin(x[:, :m] * PI / 2.0)
+    else:
+        p = x[:, :m]
+        q = 1.0 - x[:, :m]
+    f = torch.empty(n, m, dtype=x.dtype, device=x.device)
+    base = 1.0 + g
+    for i in range(m):
+        fi = base.clone()
+        for j in range(m - i - 1):
+            fi = fi * p[:, j]
         if i > 0:
-            fi = fi * torch.sin(theta[:, n_obj - i - 1] * PI / 2.0)
+            fi = fi * q[:, m - i - 1]
         f[:, i] = fi
     return f
 
 
-def maf4(x, n_obj: int = 5) -> torch.Tensor:
-    """Badly-scaled inverted concave front (scale 2^i)."""
+def maf1(x, n_obj: int = 5) -> torch.Tensor:
+    """MaF1 (value parity with the reference): linear products with a
+    Rastrigin-style distance term."""
     x = _as2d(x)
-    k = x.shape[1] - n_obj + 1
-    g = _dtlz_g1(x[:, -k:])
-    conc = _dtlz_concave(x[:, : n_obj - 1], torch.zeros_like(g), n_obj)
+    nd = x.shape[1] - n_obj + 1
+    xm = x[:, -nd:]
+    g = ((xm - 0.5) ** 2 - torch.cos(20.0 * PI * (xm - 0.5))).sum(dim=1)
+    return _maf_products(x, g, n_obj, trig=False)
+
+
+def maf2(x, n_obj: int = 5) -> torch.Tensor:
+    """MaF2 (value parity with the reference): concave products with a
+    spherical distance term."""
+    x = _as2d(x)
+    nd = x.shape[1] - n_obj + 1
+    g = ((x[:, -nd:] - 0.5) ** 2).sum(dim=1)
+    return _maf_products(x, g, n_obj, trig=True)
+
+
+def maf4(x, n_obj: int = 5) -> torch.Tensor:
+    """MaF4 (value parity with the reference): MaF2 objectives scaled by
+    10^(2i) — badly-scaled objective ranges."""
+    x = _as2d(x)
+    nd = x.shape[1] - n_obj + 1
+    g = ((x[:, -nd:] - 0.5) ** 2).sum(dim=1)
+    f = _maf_products(x, g, n_obj, trig=True)
     scale = torch.pow(
-        torch.tensor(2.0, dtype=x.dtype, device=x.device),
-        torch.arange(1, n_obj + 1, dtype=x.dtype, device=x.device),
+        torch.tensor(10.0, dtype=x.dtype, device=x.device),
+        2.0 * torch.arange(n_obj, dtype=x.dtype, device=x.device),
     )
-    return scale[None, :] * (1.0 + g[:, None]) * (1.0 - conc)
+    return f * scale[None, :]
 
 
 # ----------------------------------------------------- constrained problems

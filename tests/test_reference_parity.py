@@ -327,3 +327,26 @@ def test_glp_design_quality_comparable():
         a = rs.glp(n, s, local_random=np.random.default_rng(3))
         b = os_.glp(n, s, local_random=np.random.default_rng(3))
         assert rd.CD2(b) <= rd.CD2(a) * 1.10, (n, s)
+
+
+def test_benchmark_problem_values_parity():
+    """Every benchmark problem evaluates to the reference's values (machine
+    precision): the quality numbers users compare are directly commensurate."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.benchmarks import moo_benchmarks as rb
+
+    from dmosopt_amd.benchmarks import problems as ob
+
+    rng = np.random.default_rng(0)
+    x = rng.random((13, 12))
+    cases = [(n, 3) for n in ("dtlz1", "dtlz2", "dtlz3", "dtlz4", "dtlz5",
+                              "dtlz7", "wfg1", "wfg4")]
+    cases += [(n, 5) for n in ("maf1", "maf2", "maf4")]
+    for name, m in cases:
+        theirs = np.vstack([
+            np.atleast_2d(getattr(rb, name)(x[i], m)) for i in range(len(x))
+        ])
+        ours = getattr(ob, name)(x, n_obj=m).numpy()
+        np.testing.assert_allclose(ours, theirs, rtol=1e-7, atol=1e-10,
+                                   err_msg=name)
