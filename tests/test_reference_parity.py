@@ -350,3 +350,67 @@ def test_benchmark_problem_values_parity():
         ours = getattr(ob, name)(x, n_obj=m).numpy()
         np.testing.assert_allclose(ours, theirs, rtol=1e-7, atol=1e-10,
                                    err_msg=name)
+
+
+def test_epsilon_sort_archive_parity():
+    """Identical epsilon-box archives on identical insertion sequences."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.MOEA import EpsilonSort as RefES
+
+    from dmosopt_amd.moea.epsilon import EpsilonSort as OurES
+
+    rng = np.random.default_rng(1)
+    Y = rng.random((60, 3))
+    r = RefES([0.05, 0.05, 0.05])
+    o = OurES([0.05, 0.05, 0.05])
+    for i in range(60):
+        r.sortinto(Y[i], tagalong=i)
+        o.sortinto(Y[i], tagalong=i)
+    assert sorted(r.tagalongs) == sorted(o.tagalongs)
+
+
+def test_parameter_space_parity():
+    """Nested-space flatten/unflatten/bounds equal the reference's."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.datatypes import ParameterSpace as RefPS
+
+    from dmosopt_amd.datatypes import ParameterSpace as OurPS
+
+    spec = {"a": [0.0, 1.0], "grp": {"b": [1.0, 2.0], "c": [0, 5, True]}}
+    rp, op = RefPS.from_dict(spec), OurPS.from_dict(spec)
+    assert rp.parameter_names == op.parameter_names
+    np.testing.assert_array_equal(rp.bound1, op.bound1)
+    np.testing.assert_array_equal(rp.bound2, op.bound2)
+    np.testing.assert_array_equal(rp.is_integer, op.is_integer)
+    v = np.array([0.5, 1.5, 3.0])
+    assert rp.unflatten(v) == op.unflatten(v)
+
+
+def test_indicator_parity():
+    """IGD, euclidean metric and PopulationDiversity match the reference."""
+    import torch
+
+    from dmosopt_amd import ops
+
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.indicators import IGD as RefIGD
+    from dmosopt.indicators import PopulationDiversity as RefPD
+    from dmosopt.indicators import euclidean_distance_metric as ref_eu
+
+    from dmosopt_amd.hv.indicators import IGD as OurIGD
+    from dmosopt_amd.hv.indicators import PopulationDiversity as OurPD
+
+    rng = np.random.default_rng(2)
+    F = rng.random((40, 3))
+    P = rng.random((25, 3))
+    assert OurIGD(P).do(F) == pytest.approx(RefIGD(P).do(F), rel=1e-12)
+    np.testing.assert_allclose(
+        ops.euclidean_distance_metric(torch.as_tensor(F)).numpy(),
+        ref_eu(F), rtol=1e-12,
+    )
+    ranks = np.zeros(40, dtype=int)
+    ranks[20:] = 1
+    assert RefPD().do(ranks, F) == OurPD().do(ranks, F)
