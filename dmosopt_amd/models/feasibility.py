@@ -21,26 +21,55 @@ import numpy as np
 import torch
 
 
-class _TorchLogreg:
-    def __init__(self, X: torch.Tensor, c: torch.Tensor, l1: float, iters: int = 300):
+class _TorchLogregBatch:
+    """G independent L1-logistic fits as ONE batched Adam run: W is (G, d),
+    the loss is the sum of per-column BCEs (gradients stay independent), the
+    proximal soft-threshold applies per-column lambda. Replaces the former
+    per-candidate sequential loop (the whole regularization grid now costs
+    one optimization, with a plateau early stop)."""
+
+    def __init__(self, X: torch.Tensor, c: torch.Tensor, l1s, iters: int = 300):
         n, d = X.shape
-        self.w = torch.zeros(d, dtype=X.dtype, device=X.device)
-        self.b = torch.zeros((), dtype=X.dtype, device=X.device)
-        w = self.w.clone().requires_grad_(True)
-        b = self.b.clone().requires_grad_(True)
-        opt = torch.optim.Adam([w, b], lr=0.1)
-        lam = l1 / n
-        for _ in range(iters):
+        l1s = torch.as_tensor(np.asarray(l1s, dtype=np.float64), dtype=X.dtype,
+                              device=X.device)
+        G = l1s.shape[0]
+        W = torch.zeros(G, d, dtype=X.dtype, device=X.device, requires_grad=True)
+        b = torch.zeros(G, dtype=X.dtype, device=X.device, requires_grad=True)
+        opt = torch.optim.Adam([W, b], lr=0.1)
+        lam = (l1s / n)[:, None]
+        ct = c[:, None].expand(n, G)
+        prev = None
+        for it in range(iters):
             opt.zero_grad(set_to_none=True)
-            logits = X @ w + b
-            loss = torch.nn.functional.binary_cross_entropy_with_logits(logits, c)
+            logits = X @ W.T + b[None, :]
+            loss = torch.nn.functional.binary_cross_entropy_with_logits(
+                logits, ct, reduction="mean"
+            )
             loss.backward()
             opt.step()
             with torch.no_grad():  # proximal soft-threshold for L1
                 step = 0.1 * lam
-                w.copy_(torch.sign(w) * (w.abs() - step).clamp_min(0.0))
-        self.w = w.detach()
+                W.copy_(torch.sign(W) * (W.abs() - step).clamp_min(0.0))
+            if it % 25 == 24:
+                cur = float(loss.detach())
+                if prev is not None and abs(prev - cur) < 1e-7:
+                    break
+                prev = cur
+        self.W = W.detach()
         self.b = b.detach()
+
+    def proba_all(self, X: torch.Tensor) -> torch.Tensor:
+        """(n, G) feasibility probabilities per grid candidate."""
+        return torch.sigmoid(X @ self.W.T + self.b[None, :])
+
+    def select(self, g: int) -> "_TorchLogreg":
+        return _TorchLogreg(self.W[g], self.b[g])
+
+
+class _TorchLogreg:
+    def __init__(self, w: torch.Tensor, b: torch.Tensor):
+        self.w = w
+        self.b = b
 
     def proba(self, X: torch.Tensor) -> torch.Tensor:
         return torch.sigmoid(X @ self.w + self.b)
@@ -81,17 +110,15 @@ class LogisticFeasibilityModel:
             idx = rng.permutation(n)
             n_val = max(1, n // 5)
             val, tr = idx[:n_val], idx[n_val:]
-            best, best_loss = None, np.inf
-            for inv_C in np.logspace(-4, 4, 4):
-                clf = _TorchLogreg(Z[tr], c[tr], l1=1.0 / inv_C)
-                p = clf.proba(Z[val]).clamp(1e-7, 1 - 1e-7)
-                loss = float(
-                    torch.nn.functional.binary_cross_entropy(p, c[val])
-                )
-                if loss < best_loss:
-                    best, best_loss = clf, loss
-            return best
-        return _TorchLogreg(Z, c, l1=1.0)
+            l1s = 1.0 / np.logspace(-4, 4, 4)
+            batch = _TorchLogregBatch(Z[tr], c[tr], l1s)
+            p = batch.proba_all(Z[val]).clamp(1e-7, 1 - 1e-7)  # (n_val, G)
+            losses = torch.nn.functional.binary_cross_entropy(
+                p, c[val][:, None].expand_as(p), reduction="none"
+            ).mean(dim=0)
+            return batch.select(int(losses.argmin()))
+        batch = _TorchLogregBatch(Z, c, [1.0])
+        return batch.select(0)
 
     # ------------------------------------------------------------- interface
     def _transform(self, x) -> torch.Tensor:
