@@ -219,7 +219,15 @@ def train(
         and getattr(surrogate_cls, "supports_theta_broadcast", False)
     ):
         if ctx.is_root:
-            sm = surrogate_cls(x, y, nInput, nOutput, xlb, xub, logger=logger, **kwargs)
+            try:
+                sm = surrogate_cls(
+                    x, y, nInput, nOutput, xlb, xub, logger=logger, **kwargs
+                )
+            except BaseException:
+                # non-root ranks are waiting in bcast_payload — poison the
+                # header so they raise instead of hanging forever
+                ctx.bcast_poison(src=0)
+                raise
             ctx.bcast_payload(sm.theta.detach().to("cpu", torch.float64), src=0)
             return sm
         theta = ctx.bcast_payload(None, src=0).cpu().numpy()

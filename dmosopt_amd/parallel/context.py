@@ -72,7 +72,9 @@ class ParallelContext:
         """Broadcast a tensor whose shape only ``src`` knows.
 
         Two collectives: an i64 shape header (rank 8 max), then the payload
-        as float64. Non-src ranks pass None.
+        as float64. Non-src ranks pass None. A header with nd == -2 is the
+        POISON value sent by :meth:`bcast_poison` when the source failed to
+        produce the payload — receivers raise instead of hanging.
         """
         header = torch.full((9,), -1, dtype=torch.int64, device=self.device)
         if self.rank == src:
@@ -82,6 +84,11 @@ class ParallelContext:
                 header[1 + i] = s
         dist.broadcast(header, src=src)
         nd = int(header[0])
+        if nd == -2:
+            raise RuntimeError(
+                f"bcast_payload: rank {src} signalled failure while producing "
+                "the broadcast payload (see its traceback)"
+            )
         shape = [int(header[1 + i]) for i in range(nd)]
         if self.rank == src:
             payload = t.to(self.device, torch.float64).contiguous()
@@ -89,6 +96,12 @@ class ParallelContext:
             payload = torch.empty(shape, dtype=torch.float64, device=self.device)
         dist.broadcast(payload, src=src)
         return payload
+
+    def bcast_poison(self, src: int = 0):
+        """Tell bcast_payload receivers the payload will never come (the
+        producing rank hit an exception): they raise instead of hanging."""
+        header = torch.full((9,), -2, dtype=torch.int64, device=self.device)
+        dist.broadcast(header, src=src)
 
     def bcast_flag(self, value: bool, src: int = 0) -> bool:
         return bool(self.bcast_int(1 if value else 0, src=src))

@@ -436,3 +436,49 @@ def test_world2_mean_variance_mode():
     _, objs = pickle.loads(results[0])
     y = np.column_stack([v for _, v in objs])
     assert y.shape[1] == 2 and np.isfinite(y).all()
+
+
+def _poison_worker(rank, world_size, port, out_q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    sys.path.insert(0, _ROOT)
+    import numpy as np
+
+    from dmosopt_amd.core import engine
+    from dmosopt_amd.parallel import comm
+
+    comm.init_from_env()
+    rng = np.random.default_rng(0)
+    X = rng.random((20, 3))
+    Y = np.column_stack([X.sum(1), (1 - X).sum(1)])
+    raised = False
+    try:
+        # bogus surrogate kwargs: rank 0's fit raises; non-root ranks must
+        # RAISE too (poison header) instead of hanging in the broadcast
+        engine.train(
+            3, 2, np.zeros(3), np.ones(3), X, Y, None,
+            surrogate_method_name="gpr",
+            surrogate_method_kwargs={"optimizer": "definitely-not-real"},
+        )
+    except Exception:
+        raised = True
+    import torch.distributed as dist
+
+    dist.destroy_process_group()
+    out_q.put((rank, raised))
+
+
+def test_theta_broadcast_poison_on_fit_failure():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_poison_worker, args=(r, 2, 29863, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for _ in range(2):
+        rank, raised = q.get(timeout=240)
+        assert raised, f"rank {rank} did not raise"
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
