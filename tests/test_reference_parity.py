@@ -414,3 +414,71 @@ def test_indicator_parity():
     ranks = np.zeros(40, dtype=int)
     ranks[20:] = 1
     assert RefPD().do(ranks, F) == OurPD().do(ranks, F)
+
+
+def test_get_best_front_parity():
+    """get_best returns the same non-dominated feasible set as the
+    reference's MOASMO.get_best on identical archives."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt.MOASMO import get_best as ref_get_best
+
+    from dmosopt_amd.core.engine import get_best as our_get_best
+
+    rng = np.random.default_rng(3)
+    x = rng.random((80, 5))
+    y = rng.random((80, 3))
+    f = rng.random(80)
+    c = rng.random((80, 2)) - 0.3
+    ep = rng.integers(0, 4, 80)
+    rb = ref_get_best(x, y, f, c, 5, 3, epochs=ep, feasible=True)
+    ob = our_get_best(x, y, f, c, 5, 3, epochs=ep, feasible=True)
+    assert rb[0].shape == ob[0].shape
+    np.testing.assert_allclose(np.sort(rb[1], axis=0), np.sort(ob[1], axis=0))
+
+
+def test_sobol_design_bitwise_parity():
+    """Identical Sobol designs for identical seeds (both sides use
+    scipy.stats.qmc with the seeded generator)."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    from dmosopt import sampling as rs
+
+    from dmosopt_amd import sampling as osamp
+
+    a = rs.sobol(64, 4, local_random=np.random.default_rng(5))
+    b = osamp.sobol(64, 4, local_random=np.random.default_rng(5))
+    np.testing.assert_array_equal(a, b)
+
+
+def test_sceua_optimization_quality_parity():
+    """Our batched-speculative SCE-UA reaches the reference's optimum
+    quality on Rosenbrock within the same evaluation budget regime."""
+    if REF not in sys.path:
+        sys.path.insert(0, REF)
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        from dmosopt.model import sceua as ref_sceua
+
+        from dmosopt_amd.models.sceua import sceua_batched
+
+        def rosen_np(x):
+            return (float(np.sum(100.0 * (x[1:] - x[:-1] ** 2) ** 2
+                                 + (1 - x[:-1]) ** 2)),)
+
+        bl, bu = np.full(4, -2.0), np.full(4, 2.0)
+        out = ref_sceua(rosen_np, bl, bu, 4, 4, 3000, 10, 0.1, 0.001, seed=1)
+        ref_best = float(np.atleast_1d(out[1])[0])
+        ref_icall = int(np.atleast_1d(out[2])[0])
+
+        import torch
+
+        def rosen_batch(xb, stream):
+            x = xb.double()
+            return (100.0 * (x[:, 1:] - x[:, :-1] ** 2) ** 2
+                    + (1 - x[:, :-1]) ** 2).sum(dim=1)
+
+        _, bf, ic = sceua_batched(rosen_batch, bl, bu, 4, n_streams=1, seed=1)
+    # measured: ref 1.2e-7 @ 1939 evals; ours 1.5e-8 @ 2118 evals
+    assert float(bf[0]) <= max(ref_best * 10.0, 1e-5)
+    assert int(ic[0]) <= ref_icall * 2
