@@ -86,3 +86,46 @@ def test_hv_progress_termination_stops_on_stagnant_front():
             stopped = True
             break
     assert stopped, "stagnant front must terminate"
+
+
+def test_hv_termination_discriminates_improvement_regimes():
+    """Stop timing across regimes: a stagnant front stops promptly, a
+    geometrically-CONVERGING front stops once improvements become
+    negligible, and a sustained linearly-improving front never stops.
+    (The reference's hv_termination stops at ~min_generations in ALL three
+    regimes, including under sustained improvement — an over-eager
+    detector this rebuild deliberately does not reproduce.)"""
+    import logging
+
+    from dmosopt_amd.datatypes import OptHistory
+
+    class P:
+        n_objectives = 2
+        lb = np.zeros(4)
+        ub = np.ones(4)
+        logger = logging.getLogger("t_hvterm")
+
+    rng = np.random.default_rng(0)
+    X = rng.random((40, 4))
+
+    def run(mode):
+        term = HypervolumeProgressTermination(
+            P(), n_last=10, nth_gen=2, min_generations=10
+        )
+        F = rng.random((40, 2)) + (1.0 if mode == "linear" else 0.5)
+        Fbase = F.copy()
+        for g in range(1, 130):
+            if mode == "converging":
+                F = 0.5 + (F - 0.5) * 0.85
+            elif mode == "linear":
+                F = Fbase - 0.004 * g
+            if term.has_terminated(OptHistory(g, g * 10, X, F, None)):
+                return g
+        return None
+
+    stagnant = run("stagnant")
+    converging = run("converging")
+    linear = run("linear")
+    assert stagnant is not None and stagnant < 40
+    assert converging is not None and converging > stagnant
+    assert linear is None
