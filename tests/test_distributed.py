@@ -482,3 +482,87 @@ def test_theta_broadcast_poison_on_fit_failure():
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
+
+
+def _mp_obj(mpp):
+    out = {}
+    for pid, pp in mpp.items():
+        x = np.array([pp[k] for k in sorted(pp)])
+        out[pid] = np.array([float((x**2).sum()) + pid, float(((x - 1) ** 2).sum())])
+    return out
+
+
+def _run_multiproblem(rank, world_size, port, out_q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    sys.path.insert(0, _ROOT)
+    import dmosopt_amd
+
+    params = {
+        "opt_id": "t_mp_dist",
+        "obj_fun": _mp_obj,
+        "problem_parameters": {},
+        "space": {f"x{i}": [0.0, 1.0] for i in range(3)},
+        "objective_names": ["a", "b"],
+        "problem_ids": {1, 2},
+        "population_size": 8,
+        "num_generations": 2,
+        "surrogate_method_name": None,
+        "optimizer": "nsga2",
+        "n_initial": 2,
+        "n_epochs": 1,
+        "random_seed": 21,
+    }
+    best = dmosopt_amd.run(params, verbose=False)
+    out_q.put((rank, None if best is None else pickle.dumps(best)))
+    import torch.distributed as dist
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+def test_world2_multi_problem_zipped_collective():
+    """problem_ids at world 2: the collective farm packs per-problem
+    (y | c) blocks into one row per zipped request; results must equal the
+    single-process run bit-for-bit (exact per-point objectives)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_multiproblem, args=(r, 2, 29885, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, payload = q.get(timeout=600)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert results[0] is not None and results[1] is None
+    best2 = pickle.loads(results[0])
+    assert set(best2.keys()) == {1, 2}
+
+    # single-process reference
+    import dmosopt_amd
+
+    params = {
+        "opt_id": "t_mp_single",
+        "obj_fun": _mp_obj,
+        "problem_parameters": {},
+        "space": {f"x{i}": [0.0, 1.0] for i in range(3)},
+        "objective_names": ["a", "b"],
+        "problem_ids": {1, 2},
+        "population_size": 8,
+        "num_generations": 2,
+        "surrogate_method_name": None,
+        "optimizer": "nsga2",
+        "n_initial": 2,
+        "n_epochs": 1,
+        "random_seed": 21,
+    }
+    best1 = dmosopt_amd.run(params, verbose=False)
+    for pid in (1, 2):
+        y1 = np.column_stack([v for _, v in best1[pid][1]])
+        y2 = np.column_stack([v for _, v in best2[pid][1]])
+        np.testing.assert_array_equal(y1, y2)
