@@ -119,6 +119,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--gens", type=int, default=N_GEN)
     ap.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
+    ap.add_argument("--pop-per-gpu", type=int, default=POP_PER_GPU,
+                    help="diagnostic override of the per-GPU population")
     args = ap.parse_args()
     compute = "bf16" if args.dtype == "bf16" else "fp32"
 
@@ -129,7 +131,7 @@ def main():
     else:
         device = torch.device("cpu")
 
-    pop = POP_PER_GPU * world  # weak scaling: global population grows
+    pop = args.pop_per_gpu * world  # weak scaling: global population grows
     X, Y = make_archive(seed=1234)
 
     import torch.distributed as dist
