@@ -30,12 +30,20 @@ __global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
   float* gkey = (float*)(idx + npow2);        // npow2 (gumbel keys)
 
   const int tid = threadIdx.x;
-  // stage 1: stable sort by rank — composite integer key rank*N + index
+  __shared__ int unsorted;
+  if (tid == 0) unsorted = 0;
+  __syncthreads();
+  // stage 1: stable sort by rank — composite integer key rank*N + index.
+  // nsga2_select emits the population ALREADY rank-sorted, making this a
+  // provable identity permutation (stable sort of a non-decreasing key):
+  // detect and skip the whole bitonic pass in that (dominant) case.
   for (int i = tid; i < npow2; i += TOUR_TPB) {
     key[i] = (i < N) ? (rank[i] * (long long)N + i) : 0x7FFFFFFFFFFFFFFFLL;
     idx[i] = i;
+    if (i + 1 < N && rank[i] > rank[i + 1]) atomicOr(&unsorted, 1);
   }
   __syncthreads();
+  if (unsorted)
   for (int ks = 2; ks <= npow2; ks <<= 1) {
     for (int js = ks >> 1; js > 0; js >>= 1) {
       for (int i = tid; i < npow2; i += TOUR_TPB) {

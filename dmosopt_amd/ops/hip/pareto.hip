@@ -350,13 +350,16 @@ __global__ __launch_bounds__(256) void coop_peel_bits_kernel(
     n_dom[j] = c;
     rank[j] = 0;
   }
-  if (gtid == 0) ctrl[1] = N;
+  // fmask cleared up front; each round's update phase re-clears it for the
+  // next round, so the loop needs only TWO grid syncs per round
+  for (int w = gtid; w < W; w += gsize) fmask[w] = 0u;
+  if (gtid == 0) {
+    ctrl[0] = 0;
+    ctrl[1] = N;
+  }
   grid.sync();
 
   for (int k = 0; k <= N; ++k) {
-    for (int w = gtid; w < W; w += gsize) fmask[w] = 0u;
-    if (gtid == 0) ctrl[0] = 0;
-    grid.sync();
     for (int j = gtid; j < N; j += gsize) {
       if (n_dom[j] == 0) {
         rank[j] = k;
@@ -375,7 +378,12 @@ __global__ __launch_bounds__(256) void coop_peel_bits_kernel(
       for (int w = 0; w < W; ++w) dec += __popc(row[w] & fmask[w]);
       n_dom[j] -= dec;
     }
-    if (gtid == 0) ctrl[1] -= fs;
+    grid.sync();  // everyone has consumed fmask; safe to clear and count
+    for (int w = gtid; w < W; w += gsize) fmask[w] = 0u;
+    if (gtid == 0) {
+      ctrl[1] -= fs;
+      ctrl[0] = 0;
+    }
     grid.sync();
     if (ctrl[1] <= 0) break;  // uniform: written pre-sync by thread 0
   }
