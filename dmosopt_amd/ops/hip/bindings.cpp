@@ -53,7 +53,7 @@ void launch_variation_slots(const float*, const long long*, const long long*, co
 void launch_gather3(const float*, const float*, const long long*, const long long*, float*, float*, long long*, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
 int launch_coop_peel(const float*, unsigned int*, unsigned int*, int*, int*,
-                     int*, int, int, hipStream_t);
+                     int*, int, int, int, hipStream_t);
 void launch_peel_single_block(const int*, int*, int, int, hipStream_t);
 void launch_hv2d(const double*, const double*, double*, int, hipStream_t);
 void launch_hv3d_slices(const double*, const double*, const double*,
@@ -317,7 +317,13 @@ static int COOP_MIN_N = []() {
   return e ? atoi(e) : 1024;
 }();
 
-torch::Tensor pareto_rank(torch::Tensor Y) {
+torch::Tensor pareto_rank(torch::Tensor Y, int64_t stop = -1) {
+  // stop > 0: truncation-selection mode — fronts are peeled only until
+  // `stop` points are ranked (the straddling front always completes), the
+  // rest receive a sentinel rank larger than every true one. Exact for
+  // nsga2_select, which discards everything beyond the kept `pop`; the
+  // public op default (-1) ranks everything. Only the cooperative path
+  // exploits it — the small-N one-workgroup peels are already ~us-scale.
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
   // N > 2048: grid-wide COOPERATIVE peel — sync-free from the host (the
@@ -337,7 +343,8 @@ torch::Tensor pareto_rank(torch::Tensor Y) {
                          (unsigned int*)Dbits.data_ptr<int>(),
                          (unsigned int*)fmask.data_ptr<int>(),
                          n_dom.data_ptr<int>(), ctrl.data_ptr<int>(),
-                         rank.data_ptr<int>(), N, m, cur_stream()) == 0)
+                         rank.data_ptr<int>(), N, m, (int)stop,
+                         cur_stream()) == 0)
       return rank.to(torch::kLong);
   }
   if (N <= 2048) {
@@ -423,7 +430,11 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
               "nsga2_select: shape mismatch");
   auto parm = torch::cat({x_gen, pop_parm}, 0);
   auto obj = torch::cat({y_gen, pop_obj}, 0);
-  auto rank = pareto_rank(obj);                    // (N,) long
+  const int64_t keep =
+      std::min<int64_t>(pop, x_gen.size(0) + pop_parm.size(0));
+  // truncation selection only consumes fronts up to the one straddling
+  // `keep`; the early-stop peel skips the (typically ~half) deeper rounds
+  auto rank = pareto_rank(obj, keep);              // (N,) long
   auto crowd = crowding_distance(obj).to(torch::kFloat32);  // (N,)
   const int N = obj.size(0), d = parm.size(1), m = obj.size(1);
   const int P0 = (int)std::min<int64_t>(pop, N);
@@ -744,7 +755,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("forward_solve_", &forward_solve_, "In-place batched L z = y solve");
   m.def("backward_solve_", &backward_solve_, "In-place batched L^T x = z solve");
   m.def("dominance_degree_matrix", &dominance_degree_matrix);
-  m.def("pareto_rank", &pareto_rank);
+  m.def("pareto_rank", &pareto_rank, py::arg("Y"), py::arg("stop") = -1);
   m.def("crowding_distance", &crowding_distance);
   m.def("sbx_batch", &sbx_batch);
   m.def("mutation_batch", &mutation_batch);

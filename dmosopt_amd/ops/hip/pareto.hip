@@ -338,7 +338,14 @@ __global__ __launch_bounds__(256) void coop_peel_bits_kernel(
     int* __restrict__ n_dom,                 // (N,) scratch
     int* __restrict__ ctrl,                  // (2,) scratch
     int* __restrict__ rank,                  // (N,) out
-    int N, int W) {
+    int N, int W, int stop) {
+  // stop: peel until >= stop points are ranked, then stamp the remaining
+  // alive points with a SENTINEL rank (last completed front + 1) and exit.
+  // Exact for truncation selection (nsga2_select): the straddling front is
+  // always completed before stopping, so every point that can enter the
+  // kept top-`stop` carries its true rank; points beyond are discarded by
+  // the caller and only need a rank larger than any kept one. stop >= N
+  // peels everything (the public pareto_rank default).
   cooperative_groups::grid_group grid = cooperative_groups::this_grid();
   const int gtid = blockIdx.x * blockDim.x + threadIdx.x;
   const int gsize = gridDim.x * blockDim.x;
@@ -386,13 +393,20 @@ __global__ __launch_bounds__(256) void coop_peel_bits_kernel(
     }
     grid.sync();
     if (ctrl[1] <= 0) break;  // uniform: written pre-sync by thread 0
+    if (N - ctrl[1] >= stop) {  // uniform: enough ranked, front k complete
+      for (int j = gtid; j < N; j += gsize)
+        if (n_dom[j] >= 0) rank[j] = k + 1;  // sentinel > every true rank
+      break;
+    }
   }
 }
 
 extern "C" int launch_coop_peel(const float* Y, unsigned int* Dbits,
                                 unsigned int* fmask, int* n_dom, int* ctrl,
-                                int* rank, int N, int m, hipStream_t stream) {
+                                int* rank, int N, int m, int stop,
+                                hipStream_t stream) {
   const int W = (N + 31) / 32;
+  if (stop <= 0 || stop > N) stop = N;
   static int coop_ok = -1;
   if (coop_ok < 0) {
     int dev = 0, attr = 0;
@@ -419,7 +433,8 @@ extern "C" int launch_coop_peel(const float* Y, unsigned int* Dbits,
   int blocks = (N + 255) / 256;
   if (blocks > max_blocks) blocks = max_blocks;
   void* args[] = {(void*)&Dbits, (void*)&fmask, (void*)&n_dom,
-                  (void*)&ctrl, (void*)&rank, (void*)&N, (void*)&W};
+                  (void*)&ctrl, (void*)&rank, (void*)&N, (void*)&W,
+                  (void*)&stop};
   const hipError_t err = hipLaunchCooperativeKernel(
       (const void*)coop_peel_bits_kernel, dim3(blocks), dim3(256),
       args, 0, stream);

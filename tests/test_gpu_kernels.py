@@ -793,6 +793,31 @@ def test_pareto_rank_large_n_sync_free(dev):
         assert torch.equal(r_gpu, r_ref), (n, m)
 
 
+def test_pareto_rank_early_stop_truncation_exact(dev):
+    """Early-stop peel (pareto_rank stop=k, the nsga2_select route): every
+    point that can enter the kept top-k carries its TRUE rank, the sentinel
+    exceeds every kept rank, and the kept set (rank-sorted top-k) is
+    identical to the full ranking's."""
+    from dmosopt_amd.ops import _load_native
+    from dmosopt_amd.ops import torch_ref
+
+    native = _load_native()
+    g = torch.Generator().manual_seed(43)
+    for n, m, k in ((3200, 2, 1600), (4096, 3, 1024), (2500, 2, 2499)):
+        Y = torch.rand(n, m, generator=g)
+        Y[: n // 30] = Y[n // 3 : n // 3 + n // 30]  # duplicates
+        Yd = Y.float().to(dev)
+        r_full = torch_ref.pareto_rank(Y.double())
+        r_stop = native.pareto_rank(Yd, k).cpu()
+        # truncation selection keeps the k lowest ranks (stable by index)
+        keep_full = torch.argsort(r_full, stable=True)[:k]
+        keep_stop = torch.argsort(r_stop, stable=True)[:k]
+        assert torch.equal(keep_full, keep_stop), (n, m, k)
+        # every kept point carries its true rank; sentinel dominates the rest
+        assert torch.equal(r_stop[keep_stop], r_full[keep_full]), (n, m, k)
+        assert int(r_stop.max()) <= int(r_full.max()) + 1
+
+
 def test_transformer_joint_on_gpu(dev):
     """The FT-Transformer surrogate family (plain PyTorch) fits and
     predicts on the GPU: the joint() custom-training hook end-to-end."""
