@@ -249,8 +249,8 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
 // Factor the 32x32 diagonal block at (k0,k0) and panel-solve the rows below
 // it. One block per matrix; also accumulates the step's logdet contribution
 // deterministically (no atomics: k-steps are stream-ordered).
-template <int GROUP_COLS>
-__global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
+template <int GROUP_COLS, int TPB = CHOLP_TPB>
+__global__ __launch_bounds__(TPB) void chol_panel_kernel(
     float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
     int N, int k0) {
   __shared__ float S[CHOL_BS][CHOL_BS + 1];
@@ -366,10 +366,10 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
   // thread; the +1-padded stride keeps lanes on distinct banks), and
   // written back coalesced. Trailing rows exist only under a FULL panel
   // (k0 + bs < N implies bs == CHOL_BS): constant trip counts throughout.
-  __shared__ float P[CHOLP_TPB][CHOL_BS + 1];
-  for (int c0 = k0 + CHOL_BS; c0 < N; c0 += CHOLP_TPB) {
-    const int rows = min(CHOLP_TPB, N - c0);
-    for (int idx = tid; idx < rows * CHOL_BS; idx += CHOLP_TPB) {
+  __shared__ float P[TPB][CHOL_BS + 1];
+  for (int c0 = k0 + CHOL_BS; c0 < N; c0 += TPB) {
+    const int rows = min(TPB, N - c0);
+    for (int idx = tid; idx < rows * CHOL_BS; idx += TPB) {
       const int r = idx / CHOL_BS, c = idx % CHOL_BS;
       P[r][c] = Ab[(long long)(c0 + r) * N + k0 + c];
     }
@@ -384,7 +384,7 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_panel_kernel(
       }
     }
     __syncthreads();
-    for (int idx = tid; idx < rows * CHOL_BS; idx += CHOLP_TPB) {
+    for (int idx = tid; idx < rows * CHOL_BS; idx += TPB) {
       const int r = idx / CHOL_BS, c = idx % CHOL_BS;
       Ab[(long long)(c0 + r) * N + k0 + c] = P[r][c];
     }
@@ -830,7 +830,20 @@ static inline int panel_group_cols() {
   }
   return g;
 }
-#define LAUNCH_PANEL(B_, stream_, A_, logdet_, info_, N_, k0_)                 do {                                                                           if (panel_group_cols() == 2)                                                   hipLaunchKernelGGL((chol_panel_kernel<2>), dim3(B_), dim3(CHOLP_TPB),                           0, stream_, A_, logdet_, info_, N_, k0_);               else                                                                           hipLaunchKernelGGL((chol_panel_kernel<4>), dim3(B_), dim3(CHOLP_TPB),                           0, stream_, A_, logdet_, info_, N_, k0_);             } while (0)
+// panel workgroup width: 384 threads solve the whole N=300 trailing panel
+// in ONE LDS chunk (256 needs two sequential chunk rounds: stage + solve +
+// writeback + 3 barriers each); static LDS caps the width at 384
+// (P[384][33] + S + colbuf = 55 KB < 64 KB). Same-box A/B via
+// DMOSOPT_CHOL_TPB.
+static inline int panel_tpb() {
+  static int t = -1;
+  if (t < 0) {
+    const char* e = getenv("DMOSOPT_CHOL_TPB");
+    t = (e && atoi(e) == 256) ? 256 : 384;
+  }
+  return t;
+}
+#define LAUNCH_PANEL(B_, stream_, A_, logdet_, info_, N_, k0_)                 do {                                                                           const int tpb_ = panel_tpb();                                                if (panel_group_cols() == 2) {                                                 if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<2, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<2, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_);           } else {                                                                       if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<4, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<4, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_);           }                                                                          } while (0)
 
 extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
                                        int B, int N, hipStream_t stream) {
