@@ -252,7 +252,14 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
 template <int GROUP_COLS, int TPB = CHOLP_TPB>
 __global__ __launch_bounds__(TPB) void chol_panel_kernel(
     float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
-    int N, int k0) {
+    int N, int k0, float* __restrict__ rhs) {
+  // rhs != nullptr: FUSED FORWARD SOLVE (bordered-matrix scheme). rhs (B, N)
+  // starts as y and finishes as z = L^-1 y without a separate TRSV kernel:
+  // this panel solves entries [k0, k0+bs) against the factored diagonal
+  // block (the rhs is one extra TRSM row), and the trailing SYRK's diagonal
+  // tiles apply the rank-32 update to the remaining entries. Saves the
+  // ~61 us serial forward_solve_batched launch per NMLL (same serial-chain
+  // structure as the factorization it now rides on).
   __shared__ float S[CHOL_BS][CHOL_BS + 1];
   __shared__ float colbuf4[GROUP_COLS][CHOL_BS];
   const int b = blockIdx.x;
@@ -355,6 +362,21 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
     }
   }
   __syncthreads();
+  // fused-solve panel step: wave 0 solves the rhs segment against the
+  // factored diagonal block (wave-synchronous, lane j owns entry j; same
+  // shuffle scheme as forward_solve_batched_kernel's diagonal solve). The
+  // segment already carries the trailing updates of all previous steps.
+  if (rhs != nullptr && tid < 64) {
+    float* yb = rhs + (long long)b * N + k0;
+    const int j = tid;
+    float v = (j < bs) ? yb[j] : 0.0f;
+    for (int t = 0; t < bs; ++t) {
+      const float zt = __shfl(v, t) / S[t][t];
+      if (j == t) v = zt;
+      else if (j > t && j < bs) v = fmaf(-S[j][t], zt, v);
+    }
+    if (j < bs) yb[j] = v;
+  }
   for (int idx = tid; idx < bs * bs; idx += blockDim.x) {
     const int i = idx / bs, t = idx % bs;
     if (t <= i) Ab[(long long)(k0 + i) * N + k0 + t] = S[i][t];
@@ -596,7 +618,8 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk64_kernel(
 // below/right of the panel; blockIdx.y enumerates lower-triangular tile
 // pairs, each 256-thread block computes a 4x4 register tile per thread.
 __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
-    float* __restrict__ A, int N, int k0, int nt, int tj_fixed, int off) {
+    float* __restrict__ A, int N, int k0, int nt, int tj_fixed, int off,
+    float* __restrict__ rhs) {
   // stride 32 with an XOR swizzle on the k-column: the MFMA operand read
   // (lane -> [row + lr][k + lk]) hits banks (row+lk) mod 32 under a +1 pad,
   // an up-to-8-way conflict (PMC: 2.2 extra cycles per LDS instruction);
@@ -659,6 +682,22 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
       const int i = i0 + r16 + lk * 4 + r;
       const int j = j0 + c16 + lr;
       if (i < N && j < N && j <= i) Ab[(long long)i * N + j] -= acc[r];
+    }
+  }
+  // fused-solve trailing update: each DIAGONAL tile covers every trailing
+  // row exactly once, so its wave 0 applies the rank-32 rhs update for its
+  // 64 rows — rhs[i] -= dot(L[i, k0:k0+32], z_panel) — reusing the Pi tile
+  // already staged (and swizzled) in LDS. z_panel = rhs[k0:k0+32] was
+  // written by this step's panel kernel; read/write ranges are disjoint.
+  if (rhs != nullptr && ti == tj && tid < 64) {
+    const int i = i0 + tid;
+    if (i < N) {
+      const float* z = rhs + (long long)b * N + k0;
+      float acc = rhs[(long long)b * N + i];
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t)
+        acc = fmaf(-Pi[tid][t ^ ((tid & 7) << 2)], z[t], acc);
+      rhs[(long long)b * N + i] = acc;
     }
   }
 }
@@ -843,10 +882,11 @@ static inline int panel_tpb() {
   }
   return t;
 }
-#define LAUNCH_PANEL(B_, stream_, A_, logdet_, info_, N_, k0_)                 do {                                                                           const int tpb_ = panel_tpb();                                                if (panel_group_cols() == 2) {                                                 if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<2, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<2, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_);           } else {                                                                       if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<4, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<4, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_);           }                                                                          } while (0)
+#define LAUNCH_PANEL(B_, stream_, A_, logdet_, info_, N_, k0_, rhs_)                 do {                                                                           const int tpb_ = panel_tpb();                                                if (panel_group_cols() == 2) {                                                 if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<2, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<2, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);           } else {                                                                       if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<4, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<4, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);           }                                                                          } while (0)
 
 extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
-                                       int B, int N, hipStream_t stream) {
+                                       float* rhs, int B, int N,
+                                       hipStream_t stream) {
   // Cross-stream software pipeline: panel(k+1) only depends on the FIRST
   // tile column of step k's trailing update (its 64 columns cover the next
   // panel), so the remaining tile pairs run on a side stream concurrently
@@ -902,20 +942,23 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
   }
   if (!overlap) {
     for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
-      LAUNCH_PANEL(B, stream, A, logdet, info, N, k0);
+      LAUNCH_PANEL(B, stream, A, logdet, info, N, k0, rhs);
       const int trailing = N - k0 - CHOL_BS;
       if (trailing > 0) {
         const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
         hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, nt * (nt + 1) / 2),
-                           dim3(CHOLP_TPB), 0, stream, A, N, k0, nt, -1, 0);
+                           dim3(CHOLP_TPB), 0, stream, A, N, k0, nt, -1, 0,
+                           rhs);
       }
     }
     return;
   }
   int kstep = 0;
   bool have_rest_prev = false;
+  // the overlap path does not carry a fused rhs (launch_cholesky_fused_solve
+  // refuses it): the tj0/rest split would double-apply the diagonal tiles
   for (int k0 = 0; k0 < N; k0 += CHOL_BS, ++kstep) {
-    LAUNCH_PANEL(B, stream, A, logdet, info, N, k0);
+    LAUNCH_PANEL(B, stream, A, logdet, info, N, k0, nullptr);
     const int trailing = N - k0 - CHOL_BS;
     if (trailing <= 0) continue;
     const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
@@ -924,12 +967,13 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
     // tj0(k) waits for rest(k-1): their column regions overlap
     if (have_rest_prev) hipStreamWaitEvent(stream, evR[slot ^ 1], 0);
     hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, nt), dim3(CHOLP_TPB), 0,
-                       stream, A, N, k0, nt, 0, 0);
+                       stream, A, N, k0, nt, 0, 0, nullptr);
     if (nt > 1) {
       const int rest_pairs = (nt - 1) * nt / 2;
       hipStreamWaitEvent(s2, evP[slot], 0);
       hipLaunchKernelGGL(chol_syrk_kernel, dim3(B, rest_pairs),
-                         dim3(CHOLP_TPB), 0, s2, A, N, k0, nt, -1, 1);
+                         dim3(CHOLP_TPB), 0, s2, A, N, k0, nt, -1, 1,
+                         nullptr);
       hipEventRecord(evR[slot], s2);
       have_rest_prev = true;
     } else {
@@ -953,7 +997,7 @@ extern "C" void launch_cholesky_multik_bf16(float* A, float* logdet,
   hipLaunchKernelGGL(zero_f32_kernel, dim3((B + 255) / 256), dim3(256), 0,
                      stream, logdet, B);
   for (int k0 = 0; k0 < N; k0 += CHOL_BS) {
-    LAUNCH_PANEL(B, stream, A, logdet, info, N, k0);
+    LAUNCH_PANEL(B, stream, A, logdet, info, N, k0, nullptr);
     const int trailing = N - k0 - CHOL_BS;
     if (trailing > 0) {
       const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
@@ -961,6 +1005,25 @@ extern "C" void launch_cholesky_multik_bf16(float* A, float* logdet,
                             stream);
     }
   }
+}
+
+extern "C" int launch_cholesky_fused_solve(float* A, float* logdet,
+                                           int* info, float* rhs, int B,
+                                           int N, hipStream_t stream) {
+  static int ok = -2;
+  if (ok == -2) {
+    const char* m = getenv("DMOSOPT_CHOL_MODE");
+    const char* o = getenv("DMOSOPT_CHOL_OVERLAP");
+    const char* p = getenv("DMOSOPT_CHOL_PANEL");
+    const char* f = getenv("DMOSOPT_CHOL_FUSED_SOLVE");
+    ok = ((m && m[0] == 's') || (o && o[0] == '1') || (p && p[0] == '6') ||
+          (f && f[0] == '0'))
+             ? 0
+             : 1;
+  }
+  if (!ok || N <= CHOL_BS || B > 48) return -1;
+  launch_cholesky_multik(A, logdet, info, rhs, B, N, stream);
+  return 0;
 }
 
 extern "C" void launch_cholesky_batched(float* A, float* logdet, int* info,
@@ -976,7 +1039,7 @@ extern "C" void launch_cholesky_batched(float* A, float* logdet, int* info,
     else multik_max_b = 48;
   }
   if (B <= multik_max_b && N > CHOL_BS) {
-    launch_cholesky_multik(A, logdet, info, B, N, stream);
+    launch_cholesky_multik(A, logdet, info, nullptr, B, N, stream);
     return;
   }
   static int mode = -1;
