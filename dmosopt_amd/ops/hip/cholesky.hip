@@ -278,7 +278,21 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
   for (int idx = tid; idx < bs * bs; idx += blockDim.x)
     S[idx / bs][idx % bs] = Ab[(long long)(k0 + idx / bs) * N + k0 + idx % bs];
   __syncthreads();
-  if (tid < 64) {
+  // TRSM staging buffer, declared up front: waves 1+ stage the FIRST chunk
+  // of trailing panel rows into LDS concurrently with wave 0's serial
+  // diagonal factor (the staging reads last step's SYRK output — global
+  // memory untouched by the factor — so the ~34 KB coalesced load hides
+  // entirely under the factor's shuffle chain instead of serializing after
+  // it; disjoint LDS regions).
+  __shared__ float P[TPB][CHOL_BS + 1];
+  const int c0_first = k0 + CHOL_BS;
+  const int rows_first = min(TPB, N - c0_first);
+  if (tid >= 64) {
+    for (int idx = tid - 64; idx < rows_first * CHOL_BS; idx += TPB - 64) {
+      const int r = idx / CHOL_BS, c = idx % CHOL_BS;
+      P[r][c] = Ab[(long long)(c0_first + r) * N + k0 + c];
+    }
+  } else {
     const int lane = tid;
     float r[CHOL_BS];
 #pragma unroll
@@ -377,25 +391,31 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
     }
     if (j < bs) yb[j] = v;
   }
-  for (int idx = tid; idx < bs * bs; idx += blockDim.x) {
+  // factored-block writeback by waves 1+ (concurrent with wave 0's rhs
+  // solve above; both only READ S)
+  for (int idx = tid - 64; idx >= 0 && idx < bs * bs; idx += TPB - 64) {
     const int i = idx / bs, t = idx % bs;
     if (t <= i) Ab[(long long)(k0 + i) * N + k0 + t] = S[i][t];
   }
+  // no barrier: the first chunk's P and the factored S were published by
+  // the post-factor barrier; the writes above touch disjoint addresses
   // panel solve, chunked through LDS: per-thread direct row loads are
   // uncoalesced (64 lanes touch 64 cache lines per load instruction, a
-  // constant ~27 us regardless of row count), so each 256-row chunk is
+  // constant ~27 us regardless of row count), so each chunk is
   // staged with a COALESCED copy, solved entirely in LDS (one row per
   // thread; the +1-padded stride keeps lanes on distinct banks), and
-  // written back coalesced. Trailing rows exist only under a FULL panel
+  // written back coalesced. The FIRST chunk was pre-staged during the
+  // factor. Trailing rows exist only under a FULL panel
   // (k0 + bs < N implies bs == CHOL_BS): constant trip counts throughout.
-  __shared__ float P[TPB][CHOL_BS + 1];
-  for (int c0 = k0 + CHOL_BS; c0 < N; c0 += TPB) {
+  for (int c0 = c0_first; c0 < N; c0 += TPB) {
     const int rows = min(TPB, N - c0);
-    for (int idx = tid; idx < rows * CHOL_BS; idx += TPB) {
-      const int r = idx / CHOL_BS, c = idx % CHOL_BS;
-      P[r][c] = Ab[(long long)(c0 + r) * N + k0 + c];
+    if (c0 != c0_first) {
+      for (int idx = tid; idx < rows * CHOL_BS; idx += TPB) {
+        const int r = idx / CHOL_BS, c = idx % CHOL_BS;
+        P[r][c] = Ab[(long long)(c0 + r) * N + k0 + c];
+      }
+      __syncthreads();
     }
-    __syncthreads();
     if (tid < rows) {
 #pragma unroll
       for (int j = 0; j < CHOL_BS; ++j) {
