@@ -417,13 +417,22 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
       __syncthreads();
     }
     if (tid < rows) {
+      // own row promoted to registers: the j-loop's 512-FMA dependency
+      // chain then reads only registers and broadcast S rows (the LDS
+      // version serializes a ds_read into every fmaf of the chain);
+      // identical fmaf order — the solve is bitwise unchanged
+      float row[CHOL_BS];
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t) row[t] = P[tid][t];
 #pragma unroll
       for (int j = 0; j < CHOL_BS; ++j) {
-        float v = P[tid][j];
+        float v = row[j];
 #pragma unroll
-        for (int t = 0; t < j; ++t) v = fmaf(-P[tid][t], S[j][t], v);
-        P[tid][j] = v / S[j][j];
+        for (int t = 0; t < j; ++t) v = fmaf(-row[t], S[j][t], v);
+        row[j] = v / S[j][j];
       }
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t) P[tid][t] = row[t];
     }
     __syncthreads();
     for (int idx = tid; idx < rows * CHOL_BS; idx += TPB) {
