@@ -6,9 +6,13 @@ and mutations (1 child) until popsize-1 children exist. The stream here is
 drawn VECTORIZED on the host (one `rng.random((n, 2))` draw instead of a
 per-event Python loop), all index bookkeeping is assembled with numpy, a
 single pinned H2D transfer carries every index array, and on GPU ALL
-variation (SBX + mutation + event-order assembly) executes as ONE
-slot-decoded gfx950 launch (CPU fallback: split Philox SBX/mutation
-kernels + gather, bitwise-identical output). Pair selection is
+variation (SBX + mutation + event-order placement) executes as ONE
+event-decoded gfx950 launch — each event scatters its own child rows, so
+the host never builds an inverse slot map (CPU fallback: split Philox
+SBX/mutation kernels + row scatter, bitwise-identical output). The
+generation loop is HOST-dispatch-bound (rocprof: GPU 33% busy inside the
+loop window), so host numpy work here is sized in single-digit
+microseconds: one uniform draw covers all parent indices. Pair selection is
 uniform over ordered distinct pairs (statistically equivalent to the
 reference's `choice(poolsize, 2, replace=False)`).
 """
@@ -88,73 +92,67 @@ def event_stream_variation(
 
     C = int(c_ev.sum())
     M = int(m_ev.sum())
+    # ONE uniform draw covers every parent index (each numpy Generator call
+    # costs ~6 us of host time and the generation loop is host-bound;
+    # floor(u * n) is the same distribution as rng.integers, drawn from a
+    # different — equally deterministic and rank-replicated — stream)
+    u = rng.random(2 * C + M)
     if C:
-        i1 = rng.integers(0, poolsize, C)
-        i2 = rng.integers(0, poolsize - 1, C)
+        i1 = (u[:C] * poolsize).astype(np.int64)
+        i2 = (u[C : 2 * C] * (poolsize - 1)).astype(np.int64)
         i2 = i2 + (i2 >= i1)
     else:
         i1 = i2 = np.empty(0, dtype=np.int64)
-    im = rng.integers(0, poolsize, M) if M else np.empty(0, dtype=np.int64)
+    im = (u[2 * C :] * poolsize).astype(np.int64) if M else np.empty(0, dtype=np.int64)
+    seed_sbx, seed_mut = (int(s) for s in rng.integers(0, 2**62, 2))
 
-    # events in iteration order: within an iteration the crossover precedes
-    # the mutation (matches the loop body order)
-    ev_is_c = np.stack([c_ev, np.zeros_like(c_ev)], 1).ravel()[
-        np.stack([c_ev, m_ev], 1).ravel()
-    ]
-    sizes = np.where(ev_is_c, 2, 1)
-    starts = np.cumsum(sizes) - sizes  # slot offset of each event
-    total = int(sizes.sum())
-
-    crossover_indices = np.repeat(starts[ev_is_c], 2)
+    # slot layout in iteration order: a crossover's two children precede the
+    # same iteration's mutation child (matches the reference loop body)
+    it_sizes = 2 * c_ev + m_ev  # children per iteration
+    it_starts = np.cumsum(it_sizes) - it_sizes
+    total = 2 * C + M
+    crossover_indices = np.repeat(it_starts[c_ev], 2)
     crossover_indices[1::2] += 1
-    mutation_indices = starts[~ev_is_c]
-
-    src_rows = np.empty(total, dtype=np.int64)
-    src_rows[crossover_indices[0::2]] = np.arange(C)
-    src_rows[crossover_indices[1::2]] = C + np.arange(C)
-    src_rows[mutation_indices] = 2 * C + np.arange(M)
+    mutation_indices = (it_starts + 2 * c_ev)[m_ev]
 
     # single H2D transfer for every index array of the generation, staged
     # through a cached PINNED buffer with a non-blocking copy — a pageable
     # torch.as_tensor(...) H2D blocks the host until the stream drains,
     # serializing every generation against the previous one's GPU work
-    combined = np.concatenate([i1, i2, im, src_rows, crossover_indices, mutation_indices])
+    combined = np.concatenate([i1, i2, im, crossover_indices, mutation_indices])
     dev = _to_device_pinned(combined, pool.device)
     o = 0
     i1_t = dev[o : o + C]; o += C
     i2_t = dev[o : o + C]; o += C
     im_t = dev[o : o + M]; o += M
-    gather_idx = dev[o : o + total]; o += total
     c_idx_t = dev[o : o + 2 * C]; o += 2 * C
     m_idx_t = dev[o : o + M]
 
     if pool.device.type == "cuda" and ops.native_available():
-        # whole-generation variation in ONE launch (bitwise identical to the
-        # split sbx/mutation kernels + gather)
+        # whole-generation variation in ONE event-decoded launch: each event
+        # scatters its own child rows, so no host-built src_rows inverse map
+        # (bitwise identical values to the split sbx/mutation kernels)
         from dmosopt_amd import _hipops
 
-        x_gen = _hipops.variation_slots(
-            pool.float().contiguous(), gather_idx, i1_t, i2_t, im_t,
+        x_gen = _hipops.variation_events(
+            pool.float().contiguous(), c_idx_t, m_idx_t, i1_t, i2_t, im_t,
             di_crossover.float().contiguous(), di_mutation.float().contiguous(),
             xlb.float().contiguous(), xub.float().contiguous(),
-            float(mutation_rate), C,
-            int(rng.integers(0, 2**62)), int(rng.integers(0, 2**62)),
+            float(mutation_rate), seed_sbx, seed_mut,
         )
         return x_gen.to(pool.dtype), c_idx_t, m_idx_t
 
-    parts = []
+    out = torch.empty((total, pool.shape[1]), dtype=pool.dtype, device=pool.device)
     if C:
         c1, c2 = ops.sbx_from_pool(
             pool, i1_t, i2_t, di_crossover, xlb, xub,
-            seed=int(rng.integers(0, 2**62)), generator=torch_random,
+            seed=seed_sbx, generator=torch_random,
         )
-        parts += [c1, c2]
+        out[c_idx_t[0::2]] = c1
+        out[c_idx_t[1::2]] = c2
     if M:
-        parts.append(
-            ops.mutation_from_pool(
-                pool, im_t, di_mutation, xlb, xub, mutation_rate,
-                seed=int(rng.integers(0, 2**62)), generator=torch_random,
-            )
+        out[m_idx_t] = ops.mutation_from_pool(
+            pool, im_t, di_mutation, xlb, xub, mutation_rate,
+            seed=seed_mut, generator=torch_random,
         )
-    src = torch.cat(parts, dim=0) if len(parts) > 1 else parts[0]
-    return src[gather_idx], c_idx_t, m_idx_t
+    return out, c_idx_t, m_idx_t

@@ -52,6 +52,7 @@ void launch_pack_rank_crowd(const long long*, const float*, long long*, int, hip
 int launch_rank_crowd_sort(const long long*, const float*, long long*, int,
                            int, hipStream_t);
 void launch_variation_slots(const float*, const long long*, const long long*, const long long*, const long long*, const float*, const float*, const float*, const float*, float*, int, int, int, float, unsigned long long, unsigned long long, hipStream_t);
+void launch_variation_events(const float*, const long long*, const long long*, const long long*, const long long*, const long long*, const float*, const float*, const float*, const float*, float*, int, int, int, float, unsigned long long, unsigned long long, hipStream_t);
 void launch_gather3(const float*, const float*, const long long*, const long long*, float*, float*, long long*, int, int, int, hipStream_t);
 int launch_peel_bits(const float*, unsigned int*, int*, int, int, hipStream_t);
 int launch_coop_peel(const float*, unsigned int*, unsigned int*, int*, int*,
@@ -256,6 +257,37 @@ torch::Tensor variation_slots(torch::Tensor pool, torch::Tensor src_rows,
       out.data_ptr<float>(), total, (int)C, d, (float)mutation_rate,
       (unsigned long long)seed_sbx, (unsigned long long)seed_mut,
       cur_stream());
+  return out;
+}
+
+// Event-decoded whole-generation variation (variation_events_kernel):
+// same child bits as variation_slots, but addressed by the per-event slot
+// lists so the host never materializes the src_rows inverse map.
+torch::Tensor variation_events(torch::Tensor pool, torch::Tensor ci,
+                               torch::Tensor mi, torch::Tensor p1,
+                               torch::Tensor p2, torch::Tensor im,
+                               torch::Tensor di_c, torch::Tensor di_m,
+                               torch::Tensor lo, torch::Tensor hi,
+                               double mutation_rate, int64_t seed_sbx,
+                               int64_t seed_mut) {
+  CHECK_GPU(pool);
+  const int C = p1.size(0), M = im.size(0), d = pool.size(1);
+  TORCH_CHECK(pool.dtype() == torch::kFloat32 && ci.dtype() == torch::kLong &&
+                  mi.dtype() == torch::kLong && p2.size(0) == C &&
+                  ci.size(0) == 2 * (int64_t)C && mi.size(0) == M,
+              "variation_events: f32 pool, int64 slot/parent indices");
+  auto out = torch::empty({(long)(2 * C + M), d}, pool.options());
+  launch_variation_events(
+      pool.data_ptr<float>(),
+      C ? (long long*)ci.data_ptr<int64_t>() : nullptr,
+      M ? (long long*)mi.data_ptr<int64_t>() : nullptr,
+      C ? (long long*)p1.data_ptr<int64_t>() : nullptr,
+      C ? (long long*)p2.data_ptr<int64_t>() : nullptr,
+      M ? (long long*)im.data_ptr<int64_t>() : nullptr,
+      di_c.data_ptr<float>(), di_m.data_ptr<float>(), lo.data_ptr<float>(),
+      hi.data_ptr<float>(), out.data_ptr<float>(), C, M, d,
+      (float)mutation_rate, (unsigned long long)seed_sbx,
+      (unsigned long long)seed_mut, cur_stream());
   return out;
 }
 
@@ -750,6 +782,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("matern_train", &matern_train, "Batched Matern train-kernel assembly");
   m.def("matern_cross", &matern_cross, "Batched Matern cross-kernel assembly");
   m.def("variation_slots", &variation_slots);
+  m.def("variation_events", &variation_events);
   m.def("tournament_pool", &tournament_pool);
   m.def("survivor_count", &survivor_count);
   m.def("sceua_propose", &sceua_propose);

@@ -127,6 +127,68 @@ __global__ void variation_slots_kernel(
   out[idx] = fminf(fmaxf(v, lo[g]), hi[g]);
 }
 
+// Event-decoded variant: thread (event, gene) writes its own output slots
+// directly — crossover event k computes BOTH children (sharing one Philox
+// draw, like sbx_batch_kernel) and scatters them to rows ci[2k], ci[2k+1];
+// mutation event k scatters to mi[k]. The slot lists partition [0, 2C+M),
+// so every output row is written exactly once and the host never builds the
+// inverse src_rows map (3 numpy scatters + 2 aranges per generation in the
+// slot-decoded path). Philox counters match the split kernels: identical
+// output bits, only the addressing differs.
+__global__ void variation_events_kernel(
+    const float* __restrict__ pool, const long long* __restrict__ ci,
+    const long long* __restrict__ mi, const long long* __restrict__ p1,
+    const long long* __restrict__ p2, const long long* __restrict__ im,
+    const float* __restrict__ di_c, const float* __restrict__ di_m,
+    const float* __restrict__ lo, const float* __restrict__ hi,
+    float* __restrict__ out, int C, int M, int d, float mutation_rate,
+    unsigned long long seed_sbx, unsigned long long seed_mut) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long long)(C + M) * d) return;
+  const int ev = (int)(idx / d);
+  const int g = (int)(idx % d);
+  if (ev < C) {
+    const float a = pool[p1[ev] * d + g];
+    const float b = pool[p2[ev] * d + g];
+    const Philox4 r = philox4x32(seed_sbx, (unsigned long long)ev * d + g);
+    const float u = u01(r.c0);
+    const float e = 1.f / (di_c[g] + 1.f);
+    const float beta = (u <= 0.5f) ? __powf(2.f * u, e)
+                                   : __powf(1.f / (2.f * (1.f - u)), e);
+    const float c1 = 0.5f * ((1.f - beta) * a + (1.f + beta) * b);
+    const float c2 = 0.5f * ((1.f + beta) * a + (1.f - beta) * b);
+    out[ci[2 * ev] * d + g] = fminf(fmaxf(c1, lo[g]), hi[g]);
+    out[ci[2 * ev + 1] * d + g] = fminf(fmaxf(c2, lo[g]), hi[g]);
+  } else {
+    const int m = ev - C;
+    const float p = pool[im[m] * d + g];
+    const Philox4 ph =
+        philox4x32(seed_mut ^ 0x5deece66dULL, (unsigned long long)m * d + g);
+    const float u = u01(ph.c0);
+    const float e = 1.f / (di_m[g] + 1.f);
+    const float delta = (u < mutation_rate)
+                            ? __powf(2.f * u, e) - 1.f
+                            : 1.f - __powf(2.f * (1.f - u), e);
+    const float v = p + (hi[g] - lo[g]) * delta;
+    out[mi[m] * d + g] = fminf(fmaxf(v, lo[g]), hi[g]);
+  }
+}
+
+extern "C" void launch_variation_events(
+    const float* pool, const long long* ci, const long long* mi,
+    const long long* p1, const long long* p2, const long long* im,
+    const float* di_c, const float* di_m, const float* lo, const float* hi,
+    float* out, int C, int M, int d, float mutation_rate,
+    unsigned long long seed_sbx, unsigned long long seed_mut,
+    hipStream_t stream) {
+  long long n = (long long)(C + M) * d;
+  if (n <= 0) return;
+  hipLaunchKernelGGL(variation_events_kernel, dim3((int)((n + 255) / 256)),
+                     dim3(256), 0, stream, pool, ci, mi, p1, p2, im, di_c,
+                     di_m, lo, hi, out, C, M, d, mutation_rate, seed_sbx,
+                     seed_mut);
+}
+
 extern "C" void launch_variation_slots(
     const float* pool, const long long* src_rows, const long long* p1,
     const long long* p2, const long long* im, const float* di_c,

@@ -368,6 +368,19 @@ def test_variation_slots_matches_split_path(dev):
     )
     assert torch.equal(got, want)
 
+    # event-decoded variant (the event_stream_variation route): slot lists
+    # partition [0, total); children must land bit-identically
+    c_idx = torch.empty(2 * C, dtype=torch.long, device=dev)
+    inv = torch.argsort(src_rows)  # slot of virtual row r
+    c_idx[0::2] = inv[:C]
+    c_idx[1::2] = inv[C : 2 * C]
+    m_idx = inv[2 * C :].contiguous()
+    got2 = _hipops.variation_events(
+        pool.contiguous(), c_idx.contiguous(), m_idx, i1.contiguous(),
+        i2.contiguous(), im.contiguous(), di_c, di_m, lo, hi, 0.1, s1, s2,
+    )
+    assert torch.equal(got2, want)
+
 
 @pytest.mark.gpu
 def test_all_optimizers_one_epoch_gpu(dev):
