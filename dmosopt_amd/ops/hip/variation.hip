@@ -8,6 +8,27 @@
 #include "common.h"
 #include <math.h>
 
+
+// SBX / mutation child arithmetic with a PINNED fmaf order: under the
+// default -ffp-contract=fast the compiler may contract the blend
+// differently in different kernels (observed: sbx_batch_kernel vs
+// variation_events_kernel differed by ULPs for identical Philox draws),
+// which breaks the cross-path bitwise-identity guarantee the tests assert.
+__device__ __forceinline__ void sbx_children(float a, float b, float beta,
+                                             float lo, float hi, float* c1,
+                                             float* c2) {
+  const float x1 = 0.5f * fmaf(-beta, a, fmaf(beta, b, a + b));
+  const float x2 = 0.5f * fmaf(beta, a, fmaf(-beta, b, a + b));
+  *c1 = fminf(fmaxf(x1, lo), hi);
+  *c2 = fminf(fmaxf(x2, lo), hi);
+}
+
+__device__ __forceinline__ float mutated_child(float p, float delta, float lo,
+                                               float hi) {
+  const float v = fmaf(hi - lo, delta, p);
+  return fminf(fmaxf(v, lo), hi);
+}
+
 // children layout: C pairs first (child1 rows [0,C), child2 rows [C,2C))
 // — the host reorders into event order with an index gather.
 __global__ void sbx_batch_kernel(const float* __restrict__ pool,  // (K, d)
@@ -30,10 +51,10 @@ __global__ void sbx_batch_kernel(const float* __restrict__ pool,  // (K, d)
   const float e = 1.f / (di[g] + 1.f);
   const float beta = (u <= 0.5f) ? __powf(2.f * u, e)
                                  : __powf(1.f / (2.f * (1.f - u)), e);
-  const float c1 = 0.5f * ((1.f - beta) * a + (1.f + beta) * b);
-  const float c2 = 0.5f * ((1.f + beta) * a + (1.f - beta) * b);
-  out[(long long)c * d + g] = fminf(fmaxf(c1, lo[g]), hi[g]);
-  out[((long long)C + c) * d + g] = fminf(fmaxf(c2, lo[g]), hi[g]);
+  float c1, c2;
+  sbx_children(a, b, beta, lo[g], hi[g], &c1, &c2);
+  out[(long long)c * d + g] = c1;
+  out[((long long)C + c) * d + g] = c2;
 }
 
 __global__ void mutation_batch_kernel(const float* __restrict__ pool,  // (K, d)
@@ -55,8 +76,7 @@ __global__ void mutation_batch_kernel(const float* __restrict__ pool,  // (K, d)
   const float delta = (u < mutation_rate)
                           ? __powf(2.f * u, e) - 1.f
                           : 1.f - __powf(2.f * (1.f - u), e);
-  const float child = p + (hi[g] - lo[g]) * delta;
-  out[(long long)r * d + g] = fminf(fmaxf(child, lo[g]), hi[g]);
+  out[(long long)r * d + g] = mutated_child(p, delta, lo[g], hi[g]);
 }
 
 extern "C" void launch_sbx_batch(const float* pool, const int* p1,
@@ -100,7 +120,6 @@ __global__ void variation_slots_kernel(
   const int s = (int)(idx / d);
   const int g = (int)(idx % d);
   const long long row = src_rows[s];
-  float v;
   if (row < 2 * C) {
     const int c = (int)(row < C ? row : row - C);
     const float a = pool[p1[c] * d + g];
@@ -110,8 +129,10 @@ __global__ void variation_slots_kernel(
     const float e = 1.f / (di_c[g] + 1.f);
     const float beta = (u <= 0.5f) ? __powf(2.f * u, e)
                                    : __powf(1.f / (2.f * (1.f - u)), e);
-    v = (row < C) ? 0.5f * ((1.f - beta) * a + (1.f + beta) * b)
-                  : 0.5f * ((1.f + beta) * a + (1.f - beta) * b);
+    float c1, c2;
+    sbx_children(a, b, beta, lo[g], hi[g], &c1, &c2);
+    out[idx] = (row < C) ? c1 : c2;
+    return;
   } else {
     const int m = (int)(row - 2 * C);
     const float p = pool[im[m] * d + g];
@@ -122,9 +143,8 @@ __global__ void variation_slots_kernel(
     const float delta = (u < mutation_rate)
                             ? __powf(2.f * u, e) - 1.f
                             : 1.f - __powf(2.f * (1.f - u), e);
-    v = p + (hi[g] - lo[g]) * delta;
+    out[idx] = mutated_child(p, delta, lo[g], hi[g]);
   }
-  out[idx] = fminf(fmaxf(v, lo[g]), hi[g]);
 }
 
 // Event-decoded variant: thread (event, gene) writes its own output slots
@@ -155,10 +175,10 @@ __global__ void variation_events_kernel(
     const float e = 1.f / (di_c[g] + 1.f);
     const float beta = (u <= 0.5f) ? __powf(2.f * u, e)
                                    : __powf(1.f / (2.f * (1.f - u)), e);
-    const float c1 = 0.5f * ((1.f - beta) * a + (1.f + beta) * b);
-    const float c2 = 0.5f * ((1.f + beta) * a + (1.f - beta) * b);
-    out[ci[2 * ev] * d + g] = fminf(fmaxf(c1, lo[g]), hi[g]);
-    out[ci[2 * ev + 1] * d + g] = fminf(fmaxf(c2, lo[g]), hi[g]);
+    float c1, c2;
+    sbx_children(a, b, beta, lo[g], hi[g], &c1, &c2);
+    out[ci[2 * ev] * d + g] = c1;
+    out[ci[2 * ev + 1] * d + g] = c2;
   } else {
     const int m = ev - C;
     const float p = pool[im[m] * d + g];
@@ -169,8 +189,7 @@ __global__ void variation_events_kernel(
     const float delta = (u < mutation_rate)
                             ? __powf(2.f * u, e) - 1.f
                             : 1.f - __powf(2.f * (1.f - u), e);
-    const float v = p + (hi[g] - lo[g]) * delta;
-    out[mi[m] * d + g] = fminf(fmaxf(v, lo[g]), hi[g]);
+    out[mi[m] * d + g] = mutated_child(p, delta, lo[g], hi[g]);
   }
 }
 
