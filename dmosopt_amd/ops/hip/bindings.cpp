@@ -377,7 +377,7 @@ torch::Tensor pareto_rank(torch::Tensor Y, int64_t stop = -1) {
     auto Yc = Y.contiguous().to(torch::kFloat32);
     auto opts_i = Y.options().dtype(torch::kInt32);
     auto Dbits = torch::empty({N, W}, opts_i);
-    auto fmask = torch::empty({W}, opts_i);
+    auto fmask = torch::empty({3 * W}, opts_i);  // triple-buffered front mask
     auto n_dom = torch::empty({N}, opts_i);
     auto ctrl = torch::empty({2}, opts_i);
     auto rank = torch::zeros({N}, opts_i);

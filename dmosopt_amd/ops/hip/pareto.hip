@@ -334,11 +334,19 @@ extern "C" int launch_peel_bits(const float* Y, unsigned int* Dbits_scratch,
 
 __global__ __launch_bounds__(256) void coop_peel_bits_kernel(
     const unsigned int* __restrict__ Dbits,  // (N, W) i-dominates-j bits
-    unsigned int* __restrict__ fmask,        // (W,) scratch
+    unsigned int* __restrict__ fmask,        // (3*W,) triple-buffered scratch
     int* __restrict__ n_dom,                 // (N,) scratch
-    int* __restrict__ ctrl,                  // (2,) scratch
+    int* __restrict__ ctrl,                  // (2,) scratch (only [0] used)
     int* __restrict__ rank,                  // (N,) out
     int N, int W, int stop) {
+  // ONE grid.sync per front round (the peel is grid-barrier-latency bound):
+  // the detect step is fused into the update step — when a point's
+  // dominator count hits zero it immediately stamps its rank and flags
+  // itself in the NEXT round's front mask. The mask is TRIPLE-buffered:
+  // round k reads buf[k%3], writes buf[(k+1)%3], and clears buf[(k+2)%3]
+  // (== the buffer round k-1 read, dead since the last barrier), so no
+  // round ever waits on a separate clear phase.
+  //
   // stop: peel until >= stop points are ranked, then stamp the remaining
   // alive points with a SENTINEL rank (last completed front + 1) and exit.
   // Exact for truncation selection (nsga2_select): the straddling front is
@@ -357,47 +365,51 @@ __global__ __launch_bounds__(256) void coop_peel_bits_kernel(
     n_dom[j] = c;
     rank[j] = 0;
   }
-  // fmask cleared up front; each round's update phase re-clears it for the
-  // next round, so the loop needs only TWO grid syncs per round
-  for (int w = gtid; w < W; w += gsize) fmask[w] = 0u;
-  if (gtid == 0) {
-    ctrl[0] = 0;
-    ctrl[1] = N;
+  for (int w = gtid; w < 3 * W; w += gsize) fmask[w] = 0u;
+  if (gtid == 0) ctrl[0] = 0;
+  grid.sync();
+  // front 0 into buffer 0
+  for (int j = gtid; j < N; j += gsize) {
+    if (n_dom[j] == 0) {
+      n_dom[j] = -1;
+      atomicOr(&fmask[j >> 5], 1u << (j & 31));
+      atomicAdd(&ctrl[0], 1);
+    }
   }
   grid.sync();
 
-  for (int k = 0; k <= N; ++k) {
-    for (int j = gtid; j < N; j += gsize) {
-      if (n_dom[j] == 0) {
-        rank[j] = k;
-        n_dom[j] = -1;
-        atomicOr(&fmask[j >> 5], 1u << (j & 31));
-        atomicAdd(&ctrl[0], 1);
-      }
-    }
-    grid.sync();
-    const int fs = ctrl[0];  // uniform: written before the sync
-    if (fs == 0) break;
-    for (int j = gtid; j < N; j += gsize) {
-      if (n_dom[j] <= 0) continue;
-      int dec = 0;
-      const unsigned int* row = Dbits + (size_t)j * W;
-      for (int w = 0; w < W; ++w) dec += __popc(row[w] & fmask[w]);
-      n_dom[j] -= dec;
-    }
-    grid.sync();  // everyone has consumed fmask; safe to clear and count
-    for (int w = gtid; w < W; w += gsize) fmask[w] = 0u;
-    if (gtid == 0) {
-      ctrl[1] -= fs;
-      ctrl[0] = 0;
-    }
-    grid.sync();
-    if (ctrl[1] <= 0) break;  // uniform: written pre-sync by thread 0
-    if (N - ctrl[1] >= stop) {  // uniform: enough ranked, front k complete
+  int prev = -1;
+  for (int k = 0; k < N; ++k) {
+    const int total = ctrl[0];  // uniform: all writes pre-barrier
+    if (total >= N || total == prev) break;
+    if (total >= stop) {  // fronts 0..k are complete
       for (int j = gtid; j < N; j += gsize)
         if (n_dom[j] >= 0) rank[j] = k + 1;  // sentinel > every true rank
       break;
     }
+    prev = total;
+    const unsigned int* rd = fmask + (size_t)(k % 3) * W;
+    unsigned int* wr = fmask + (size_t)((k + 1) % 3) * W;
+    unsigned int* cl = fmask + (size_t)((k + 2) % 3) * W;
+    for (int w = gtid; w < W; w += gsize) cl[w] = 0u;
+    for (int j = gtid; j < N; j += gsize) {
+      if (n_dom[j] <= 0) continue;
+      int dec = 0;
+      const unsigned int* row = Dbits + (size_t)j * W;
+      for (int w = 0; w < W; ++w) dec += __popc(row[w] & rd[w]);
+      if (dec) {
+        const int nd = n_dom[j] - dec;
+        if (nd == 0) {
+          rank[j] = k + 1;
+          n_dom[j] = -1;
+          atomicOr(&wr[j >> 5], 1u << (j & 31));
+          atomicAdd(&ctrl[0], 1);
+        } else {
+          n_dom[j] = nd;
+        }
+      }
+    }
+    grid.sync();
   }
 }
 
