@@ -292,6 +292,59 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
       const int r = idx / CHOL_BS, c = idx % CHOL_BS;
       P[r][c] = Ab[(long long)(c0_first + r) * N + k0 + c];
     }
+  } else if (GROUP_COLS == 1) {
+    // pure-shuffle factor (no LDS broadcast, no fences): per column j the
+    // rank-1 update's multiplier L[c][j] moves by one __shfl per target
+    // column — 31 independent shuffles that pipeline freely, where the
+    // LDS-broadcast rounds serialize on two __threadfence_block each.
+    // fmaf order per element is ascending-j rank-1, identical to the
+    // grouped variant: the factor is bitwise unchanged.
+    const int lane = tid;
+    float r[CHOL_BS];
+#pragma unroll
+    for (int t = 0; t < CHOL_BS; ++t)
+      r[t] = (lane < bs && t < bs) ? S[lane][t] : 0.0f;
+    float mylog = 0.0f;
+    int bad = 0;
+#pragma unroll
+    for (int j = 0; j < CHOL_BS; ++j) {
+      if (j >= bs) break;
+      float d = __shfl(r[j], j);
+      if (d <= 0.0f || !isfinite(d)) {
+        bad = bad ? bad : (k0 + j + 1);
+        d = 1e-30f;
+      }
+      d = sqrtf(d);
+      if (lane == j) {
+        r[j] = d;
+        mylog += logf(d);
+      } else if (lane > j) {
+        r[j] /= d;
+      }
+      const float lij = r[j];
+#pragma unroll
+      for (int c = j + 1; c < CHOL_BS; ++c) {
+        if (c >= bs) break;
+        const float lcj = __shfl(lij, c);
+        if (lane >= c) r[c] = fmaf(-lij, lcj, r[c]);
+      }
+    }
+    if (lane < bs) {
+#pragma unroll
+      for (int t = 0; t < CHOL_BS; ++t) {
+        if (t >= bs) continue;
+        S[lane][t] = r[t];
+      }
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      mylog += __shfl_down(mylog, off);
+      const int ob = __shfl_down(bad, off);
+      bad = bad ? bad : ob;
+    }
+    if (lane == 0) {
+      logdet[b] += mylog;
+      if (bad && info[b] == 0) info[b] = bad;
+    }
   } else {
     const int lane = tid;
     float r[CHOL_BS];
@@ -894,7 +947,10 @@ static inline int panel_group_cols() {
   static int g = -1;
   if (g < 0) {
     const char* e = getenv("DMOSOPT_CHOL_GROUP");
-    g = (e && e[0] == '4') ? 4 : 2;  // same-box A/B: 2-col 0.364 vs 4-col 0.390 ms
+    // 1 = pure-shuffle factor (no LDS broadcast/fences); 2/4 = grouped LDS
+    // broadcast rounds. Same-box A/B in profiles/README.md.
+    g = e ? atoi(e) : 1;
+    if (g != 2 && g != 4) g = 1;
   }
   return g;
 }
@@ -911,7 +967,7 @@ static inline int panel_tpb() {
   }
   return t;
 }
-#define LAUNCH_PANEL(B_, stream_, A_, logdet_, info_, N_, k0_, rhs_)                 do {                                                                           const int tpb_ = panel_tpb();                                                if (panel_group_cols() == 2) {                                                 if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<2, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<2, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);           } else {                                                                       if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<4, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<4, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);           }                                                                          } while (0)
+#define LAUNCH_PANEL(B_, stream_, A_, logdet_, info_, N_, k0_, rhs_)                 do {                                                                           const int tpb_ = panel_tpb();                                                if (panel_group_cols() == 1) {                                                 if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<1, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<1, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);           } else if (panel_group_cols() == 2) {                                                 if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<2, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<2, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);           } else {                                                                       if (tpb_ == 384)                                                               hipLaunchKernelGGL((chol_panel_kernel<4, 384>), dim3(B_), dim3(384),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);             else                                                                           hipLaunchKernelGGL((chol_panel_kernel<4, 256>), dim3(B_), dim3(256),                            0, stream_, A_, logdet_, info_, N_, k0_, rhs_);           }                                                                          } while (0)
 
 extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
                                        float* rhs, int B, int N,
