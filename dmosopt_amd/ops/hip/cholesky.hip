@@ -947,10 +947,13 @@ static inline int panel_group_cols() {
   static int g = -1;
   if (g < 0) {
     const char* e = getenv("DMOSOPT_CHOL_GROUP");
-    // 1 = pure-shuffle factor (no LDS broadcast/fences); 2/4 = grouped LDS
-    // broadcast rounds. Same-box A/B in profiles/README.md.
-    g = e ? atoi(e) : 1;
-    if (g != 2 && g != 4) g = 1;
+    // 2 (default) / 4 = grouped LDS-broadcast factor rounds; 1 = pure-
+    // shuffle factor, measured 2.5x SLOWER (804 vs 324 us per B=12 chol,
+    // same box): __shfl is ds_bpermute on CDNA4 — a full LDS round trip
+    // per value — so 512 dependent shuffles lose to 16 fence-paired LDS
+    // column broadcasts. Kept selectable for re-measurement.
+    g = e ? atoi(e) : 2;
+    if (g != 1 && g != 4) g = 2;
   }
   return g;
 }
