@@ -83,7 +83,7 @@ def one_epoch(X, Y, pop, rank, world, device, seed, n_gen=N_GEN, compute="fp32")
 
     optimizer = NSGA2Optimizer(
         popsize=pop, nInput=D_IN, nOutput=N_OBJ, model=mdl,
-        distance_metric=None, sampling_method="slh", mutation_rate=None, nchildren=1,
+        distance_metric="crowding", sampling_method="slh", mutation_rate=None, nchildren=1,
     )
     if device.type == "cuda":
         optimizer.set_device(device)

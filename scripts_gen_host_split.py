@@ -31,7 +31,7 @@ mdl = Model(objective=gp)
 
 POP = int(os.environ.get("POP", "200"))
 opt = NSGA2Optimizer(popsize=POP, nInput=D_IN, nOutput=N_OBJ, model=mdl,
-                     distance_metric=None, sampling_method="slh",
+                     distance_metric="crowding", sampling_method="slh",
                      mutation_rate=None, nchildren=1)
 opt.set_device(dev)
 rng = np.random.default_rng(3)

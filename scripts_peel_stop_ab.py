@@ -32,7 +32,7 @@ for pop in (1600, 3200):
         logger=None, device=dev)
     mdl = Model(objective=gp)
     opt = NSGA2Optimizer(popsize=pop, nInput=D_IN, nOutput=N_OBJ, model=mdl,
-                         distance_metric=None, sampling_method="slh",
+                         distance_metric="crowding", sampling_method="slh",
                          mutation_rate=None, nchildren=1)
     opt.set_device(dev)
     res = engine.optimize_loop(

@@ -34,7 +34,7 @@ def genloop(gp, seed, n_gen=200):
     mdl = Model(objective=gp)
     optimizer = NSGA2Optimizer(
         popsize=200, nInput=D_IN, nOutput=N_OBJ, model=mdl,
-        distance_metric=None, sampling_method="slh", mutation_rate=None,
+        distance_metric="crowding", sampling_method="slh", mutation_rate=None,
         nchildren=1)
     optimizer.set_device(dev)
     res = engine.optimize_loop(
