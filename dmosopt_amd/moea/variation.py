@@ -69,6 +69,77 @@ def _draw_event_stream(rng, popsize: int, crossover_prob: float, mutation_prob: 
     return np.concatenate(c_parts), np.concatenate(m_parts)
 
 
+def _assemble_event_indices(rng, popsize, poolsize, crossover_prob, mutation_prob):
+    """Draw + assemble the generation's event indices (host numpy side).
+
+    Returns (combined int64 array [i1|i2|im|ci|mi], C, M, seed_sbx, seed_mut).
+    """
+    c_ev, m_ev = _draw_event_stream(rng, popsize, crossover_prob, mutation_prob)
+    C = int(c_ev.sum())
+    M = int(m_ev.sum())
+    u = rng.random(2 * C + M)
+    if C:
+        i1 = (u[:C] * poolsize).astype(np.int64)
+        i2 = (u[C : 2 * C] * (poolsize - 1)).astype(np.int64)
+        i2 = i2 + (i2 >= i1)
+    else:
+        i1 = i2 = np.empty(0, dtype=np.int64)
+    im = (u[2 * C :] * poolsize).astype(np.int64) if M else np.empty(0, dtype=np.int64)
+    seed_sbx, seed_mut = (int(s) for s in rng.integers(0, 2**62, 2))
+    it_sizes = 2 * c_ev + m_ev
+    it_starts = np.cumsum(it_sizes) - it_sizes
+    crossover_indices = np.repeat(it_starts[c_ev], 2)
+    crossover_indices[1::2] += 1
+    mutation_indices = (it_starts + 2 * c_ev)[m_ev]
+    combined = np.concatenate([i1, i2, im, crossover_indices, mutation_indices])
+    return combined, C, M, seed_sbx, seed_mut
+
+
+def spawn_generation_native(
+    population: torch.Tensor,
+    rank: torch.Tensor,
+    poolsize: int,
+    p_sel: float,
+    rng: np.random.Generator,
+    popsize: int,
+    crossover_prob: float,
+    mutation_prob: float,
+    mutation_rate: float,
+    di_crossover: torch.Tensor,
+    di_mutation: torch.Tensor,
+    xlb: torch.Tensor,
+    xub: torch.Tensor,
+):
+    """Tournament selection + whole-generation variation in ONE extension
+    call (generation_spawn binding): the host-dispatch-bound loop pays one
+    binding round trip instead of two, and the pool tensor stays in C++.
+    Returns (x_gen, crossover_slot_idx, mutation_slot_idx) or None when the
+    tournament kernel refuses the shape (caller uses the split path)."""
+    from dmosopt_amd import _hipops
+
+    seed_t = int(rng.integers(0, 2**62))
+    combined, C, M, seed_sbx, seed_mut = _assemble_event_indices(
+        rng, popsize, poolsize, crossover_prob, mutation_prob
+    )
+    dev = _to_device_pinned(combined, population.device)
+    o = 0
+    i1_t = dev[o : o + C]; o += C
+    i2_t = dev[o : o + C]; o += C
+    im_t = dev[o : o + M]; o += M
+    c_idx_t = dev[o : o + 2 * C]; o += 2 * C
+    m_idx_t = dev[o : o + M]
+    x_gen = _hipops.generation_spawn(
+        population.float().contiguous(), rank.long().contiguous(), poolsize,
+        float(p_sel), seed_t, c_idx_t, m_idx_t, i1_t, i2_t, im_t,
+        di_crossover.float().contiguous(), di_mutation.float().contiguous(),
+        xlb.float().contiguous(), xub.float().contiguous(),
+        float(mutation_rate), seed_sbx, seed_mut,
+    )
+    if x_gen is None:
+        return None
+    return x_gen.to(population.dtype), c_idx_t, m_idx_t
+
+
 def event_stream_variation(
     pool: torch.Tensor,
     rng: np.random.Generator,
@@ -88,38 +159,16 @@ def event_stream_variation(
     The slot-index outputs are int64 tensors on ``pool.device`` so callers
     can track operator success without a device->host sync.
     """
-    c_ev, m_ev = _draw_event_stream(rng, popsize, crossover_prob, mutation_prob)
-
-    C = int(c_ev.sum())
-    M = int(m_ev.sum())
-    # ONE uniform draw covers every parent index (each numpy Generator call
-    # costs ~6 us of host time and the generation loop is host-bound;
-    # floor(u * n) is the same distribution as rng.integers, drawn from a
-    # different — equally deterministic and rank-replicated — stream)
-    u = rng.random(2 * C + M)
-    if C:
-        i1 = (u[:C] * poolsize).astype(np.int64)
-        i2 = (u[C : 2 * C] * (poolsize - 1)).astype(np.int64)
-        i2 = i2 + (i2 >= i1)
-    else:
-        i1 = i2 = np.empty(0, dtype=np.int64)
-    im = (u[2 * C :] * poolsize).astype(np.int64) if M else np.empty(0, dtype=np.int64)
-    seed_sbx, seed_mut = (int(s) for s in rng.integers(0, 2**62, 2))
-
-    # slot layout in iteration order: a crossover's two children precede the
-    # same iteration's mutation child (matches the reference loop body)
-    it_sizes = 2 * c_ev + m_ev  # children per iteration
-    it_starts = np.cumsum(it_sizes) - it_sizes
+    # ONE uniform draw covers every parent index; slot layout in iteration
+    # order (a crossover's two children precede the same iteration's
+    # mutation child, matching the reference loop body). Single H2D
+    # transfer for every index array, staged through a PINNED buffer with a
+    # non-blocking copy — a pageable torch.as_tensor(...) H2D blocks the
+    # host until the stream drains.
+    combined, C, M, seed_sbx, seed_mut = _assemble_event_indices(
+        rng, popsize, poolsize, crossover_prob, mutation_prob
+    )
     total = 2 * C + M
-    crossover_indices = np.repeat(it_starts[c_ev], 2)
-    crossover_indices[1::2] += 1
-    mutation_indices = (it_starts + 2 * c_ev)[m_ev]
-
-    # single H2D transfer for every index array of the generation, staged
-    # through a cached PINNED buffer with a non-blocking copy — a pageable
-    # torch.as_tensor(...) H2D blocks the host until the stream drains,
-    # serializing every generation against the previous one's GPU work
-    combined = np.concatenate([i1, i2, im, crossover_indices, mutation_indices])
     dev = _to_device_pinned(combined, pool.device)
     o = 0
     i1_t = dev[o : o + C]; o += C

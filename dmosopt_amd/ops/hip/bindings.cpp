@@ -516,6 +516,52 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   return {parm_o, obj_o, rank_o, perm};
 }
 
+// nsga2_select + device-side operator-success accounting in ONE python
+// call (the generation loop is host-dispatch-bound; every binding round
+// trip costs ~10 us of host time)
+std::vector<torch::Tensor> nsga2_select_acc(
+    torch::Tensor x_gen, torch::Tensor y_gen, torch::Tensor pop_parm,
+    torch::Tensor pop_obj, int64_t pop, torch::Tensor c_idx,
+    torch::Tensor succ_cross, torch::Tensor succ_mut) {
+  auto r = nsga2_select(x_gen, y_gen, pop_parm, pop_obj, pop);
+  auto& perm = r[3];
+  launch_survivor_count(
+      (long long*)perm.data_ptr<int64_t>(),
+      (long long*)c_idx.data_ptr<int64_t>(), perm.size(0), c_idx.size(0),
+      (int)x_gen.size(0), (long long*)succ_cross.data_ptr<int64_t>(),
+      (long long*)succ_mut.data_ptr<int64_t>(), cur_stream());
+  return r;
+}
+
+// Tournament pool + event-decoded variation chained inside one binding
+// call: the pool tensor never surfaces to python.
+torch::Tensor generation_spawn(
+    torch::Tensor population, torch::Tensor rank, int64_t poolsize,
+    double p_sel, int64_t seed_t, torch::Tensor ci, torch::Tensor mi,
+    torch::Tensor p1, torch::Tensor p2, torch::Tensor im, torch::Tensor di_c,
+    torch::Tensor di_m, torch::Tensor lo, torch::Tensor hi,
+    double mutation_rate, int64_t seed_sbx, int64_t seed_mut) {
+  CHECK_GPU(population);
+  TORCH_CHECK(population.dtype() == torch::kFloat32 &&
+                  rank.dtype() == torch::kLong &&
+                  poolsize <= population.size(0),
+              "generation_spawn: f32 population, int64 rank");
+  const int N = population.size(0), d = population.size(1);
+  auto pool = torch::empty({poolsize, d}, population.options());
+  auto pool_idx =
+      torch::empty({poolsize}, population.options().dtype(torch::kLong));
+  const float log1mp = logf(1.0f - (float)p_sel);
+  if (launch_tournament(population.data_ptr<float>(),
+                        (long long*)rank.data_ptr<int64_t>(),
+                        pool.data_ptr<float>(),
+                        (long long*)pool_idx.data_ptr<int64_t>(), N, d,
+                        (int)poolsize, log1mp, (unsigned long long)seed_t,
+                        cur_stream()) != 0)
+    return torch::Tensor();  // caller falls back to the split path
+  return variation_events(pool, ci, mi, p1, p2, im, di_c, di_m, lo, hi,
+                          mutation_rate, seed_sbx, seed_mut);
+}
+
 std::vector<torch::Tensor> sbx_batch(torch::Tensor pool, torch::Tensor p1,
                                      torch::Tensor p2, torch::Tensor di,
                                      torch::Tensor lo, torch::Tensor hi,
@@ -788,6 +834,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sceua_propose", &sceua_propose);
   m.def("sceua_accept", &sceua_accept);
   m.def("nsga2_select", &nsga2_select, "Fused survivor selection (cat+rank+crowding+sort+gather)");
+  m.def("nsga2_select_acc", &nsga2_select_acc);
+  m.def("generation_spawn", &generation_spawn);
   m.def("gp_predict_mean", &gp_predict_mean, "Fused cross-kernel + posterior mean",
         py::arg("Xq"), py::arg("X"), py::arg("theta"), py::arg("alpha"),
         py::arg("y_mean"), py::arg("y_std"), py::arg("nu"), py::arg("aniso"),
