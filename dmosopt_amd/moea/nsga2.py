@@ -125,6 +125,17 @@ class NSGA2Optimizer(MOEA):
 
         poolsize = min(poolsize, population.shape[0])
         di_c, di_m = self._di_tensors(population)
+        # bounds columns are stride-2 views of the (d,2) bounds tensor: the
+        # float32-contiguous copies the kernels need are cached (two copy
+        # kernels + allocs per generation otherwise)
+        bc = getattr(self, "_bounds_cache", None)
+        if bc is None or bc[0] != population.device:
+            bc = self._bounds_cache = (
+                population.device,
+                xlb.to(population.device, torch.float32).contiguous(),
+                xub.to(population.device, torch.float32).contiguous(),
+            )
+        xlb_f, xub_f = bc[1], bc[2]
         if population.device.type == "cuda" and ops.native_available():
             # tournament + event-decoded variation chained in ONE binding
             # call (the loop is host-dispatch-bound; the pool tensor never
@@ -134,7 +145,7 @@ class NSGA2Optimizer(MOEA):
             res = spawn_generation_native(
                 population, rank, poolsize, 0.5, rng, popsize,
                 p.crossover_prob, p.mutation_prob, p.mutation_rate,
-                di_c, di_m, xlb, xub,
+                di_c, di_m, xlb_f, xub_f,
             )
             if res is not None:
                 x_gen, crossover_indices, mutation_indices = res
@@ -143,6 +154,7 @@ class NSGA2Optimizer(MOEA):
                 return x_gen, {
                     "crossover_indices": crossover_indices,
                     "mutation_indices": mutation_indices,
+                    "clamped": True,  # variation kernels clamp to [xlb,xub]
                 }
         pool_idx = ops.tournament_selection(
             population.shape[0], poolsize, [rank], rng,
