@@ -308,8 +308,17 @@ class AGEMOEAOptimizer(MOEA):
             generator=self.torch_random,
         )
         pool = population[pool_idx]
-        di_c = torch.as_tensor(p.di_crossover, dtype=pool.dtype, device=pool.device)
-        di_m = torch.as_tensor(p.di_mutation, dtype=pool.dtype, device=pool.device)
+        # cached: re-uploading two pageable numpy arrays per generation
+        # costs two blocking H2D copies (they only change under adaptive
+        # operator rates, which AGEMOEA does not use)
+        cache = getattr(self, "_di_cache", None)
+        if cache is None or cache[0].dtype != pool.dtype or cache[0].device != pool.device:
+            cache = (
+                torch.as_tensor(p.di_crossover, dtype=pool.dtype, device=pool.device),
+                torch.as_tensor(p.di_mutation, dtype=pool.dtype, device=pool.device),
+            )
+            self._di_cache = cache
+        di_c, di_m = cache
         x_gen, _, _ = event_stream_variation(
             pool, rng, p.popsize, pool.shape[0], p.crossover_prob, p.mutation_prob,
             p.mutation_rate, di_c, di_m, xlb, xub, torch_random=self.torch_random,
