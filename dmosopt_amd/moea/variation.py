@@ -134,6 +134,7 @@ def spawn_generation_native(
         di_crossover.float().contiguous(), di_mutation.float().contiguous(),
         xlb.float().contiguous(), xub.float().contiguous(),
         float(mutation_rate), seed_sbx, seed_mut,
+        rank_sorted=True,  # NSGA2 state.rank comes out of nsga2_select sorted
     )
     if x_gen is None:
         return None

@@ -47,6 +47,7 @@ void launch_nmll_reduce(const float*, const float*, const int*, float*, int, int
 void launch_sceua_propose(const float*, const int*, const float*, const float*, float*, int, int, int, int, int, unsigned long long, hipStream_t);
 int launch_sceua_accept(float*, float*, const float*, const float*, const int*, const int*, int*, int, int, int, int, int, hipStream_t);
 int launch_tournament(const float*, const long long*, float*, long long*, int, int, int, float, unsigned long long, hipStream_t);
+void launch_tournament_keys(float*, int, float, unsigned long long, hipStream_t);
 int launch_survivor_count(const long long*, const long long*, int, int, int, long long*, long long*, hipStream_t);
 void launch_pack_rank_crowd(const long long*, const float*, long long*, int, hipStream_t);
 int launch_rank_crowd_sort(const long long*, const float*, long long*, int,
@@ -195,6 +196,31 @@ torch::Tensor gp_predict_mean(torch::Tensor Xq, torch::Tensor X,
 // Fused tournament selection: stable rank sort + Gumbel top-k weighted
 // sampling without replacement + pool gather in ONE launch. Returns
 // (pool, pool_idx); empty tensors when N exceeds the LDS sort bound.
+// Multi-block tournament (N > the one-workgroup bitonic's sweet spot):
+// Gumbel keys from the SAME Philox stream as tournament_kernel stage 2,
+// sorted with the radix argsort. rank_sorted=true skips the stage-1
+// stable rank sort (nsga2_select emits rank-sorted populations).
+static std::vector<torch::Tensor> tournament_torch(torch::Tensor population,
+                                                   torch::Tensor rank,
+                                                   int64_t poolsize,
+                                                   float log1mp, int64_t seed,
+                                                   bool rank_sorted) {
+  const int N = population.size(0);
+  auto gk = torch::empty({N}, population.options());
+  launch_tournament_keys(gk.data_ptr<float>(), N, log1mp,
+                         (unsigned long long)seed, cur_stream());
+  auto top = torch::argsort(gk, /*dim=*/-1, /*descending=*/true)
+                 .slice(0, 0, poolsize)
+                 .contiguous();
+  torch::Tensor sel = top;
+  if (!rank_sorted) {
+    auto order = torch::argsort(rank, /*stable=*/true, /*dim=*/-1,
+                                /*descending=*/false);
+    sel = order.index_select(0, top).contiguous();
+  }
+  return {population.index_select(0, sel), sel};
+}
+
 std::vector<torch::Tensor> tournament_pool(torch::Tensor population,
                                            torch::Tensor rank,
                                            int64_t poolsize, double p_sel,
@@ -211,11 +237,13 @@ std::vector<torch::Tensor> tournament_pool(torch::Tensor population,
   auto pool_idx =
       torch::empty({poolsize}, population.options().dtype(torch::kLong));
   const float log1mp = logf(1.0f - (float)p_sel);
+  if (N > 1024)
+    return tournament_torch(population, rank, poolsize, log1mp, seed, false);
   if (launch_tournament(population.data_ptr<float>(),
                         (long long*)rank.data_ptr<int64_t>(), pool.data_ptr<float>(),
                         (long long*)pool_idx.data_ptr<int64_t>(), N, d, (int)poolsize,
                         log1mp, (unsigned long long)seed, cur_stream()) != 0)
-    return {torch::Tensor(), torch::Tensor()};
+    return tournament_torch(population, rank, poolsize, log1mp, seed, false);
   return {pool, pool_idx};
 }
 
@@ -446,6 +474,33 @@ torch::Tensor crowding_distance(torch::Tensor Y) {
   CHECK_GPU(Y);
   const int N = Y.size(0), m = Y.size(1);
   if (N == 1) return torch::ones({1}, Y.options());
+  static int torch_min_n = []() {
+    const char* e = getenv("DMOSOPT_CROWD_TORCH_MIN_N");
+    return e ? atoi(e) : 1024;
+  }();
+  if (N > torch_min_n) {
+    // the single-workgroup-per-dim LDS bitonic degrades ~N log^2 N on one
+    // CU (208 us at N=3200, m=2); above the gate the multi-block radix
+    // route wins (same arithmetic as ops/torch_ref.crowding_distance)
+    auto lb = std::get<0>(Y.min(0, /*keepdim=*/true));
+    auto ub = std::get<0>(Y.max(0, /*keepdim=*/true));
+    auto span = (ub - lb).clamp_min(0.0);
+    span = torch::where(span == 0, torch::ones_like(span), span);
+    auto U = (Y - lb) / span;
+    auto idx = torch::argsort(U, /*dim=*/0, /*descending=*/false);
+    auto US = torch::gather(U, 0, idx);
+    auto DS = torch::empty_like(US);
+    DS.slice(0, 0, 1).fill_(1.0);
+    DS.slice(0, N - 1, N).fill_(1.0);
+    if (N > 2)
+      DS.slice(0, 1, N - 1) = US.slice(0, 2, N) - US.slice(0, 0, N - 2);
+    // per-column scatter (each column of idx is a permutation) + fixed-
+    // order row sum — scatter_ADD on CUDA floats resolves atomics in
+    // arbitrary order and would break run-to-run bit determinism
+    auto Dm = torch::zeros({N, m}, Y.options());
+    Dm.scatter_(0, idx, DS);
+    return torch::nan_to_num(Dm.sum(1), 0.0);
+  }
   TORCH_CHECK(N <= 16384, "crowding_distance HIP kernel supports N <= 16384");
   auto per_dim = torch::empty({m, N}, Y.options());
   launch_crowding(Y.data_ptr<float>(), per_dim.data_ptr<float>(), N, m,
@@ -486,9 +541,11 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   }();
   torch::Tensor perm;
   // single-block bitonic (key, idx) sort replaces pack + radix argsort
-  // (3-4 launches -> 1); comparator (key asc, idx asc) == stable argsort
+  // (3-4 launches -> 1); comparator (key asc, idx asc) == stable argsort.
+  // Gated by N: the one-workgroup bitonic degrades ~N log^2 N on one CU
+  // (122 us at N=3200) while the multi-block radix path stays ~40 us.
   auto perm_f = torch::empty({P0}, rank.options());
-  if (rcs_on &&
+  if (rcs_on && N <= 1024 &&
       launch_rank_crowd_sort((long long*)rank.data_ptr<int64_t>(),
                              crowd.data_ptr<float>(),
                              (long long*)perm_f.data_ptr<int64_t>(), N, P0,
@@ -540,7 +597,8 @@ torch::Tensor generation_spawn(
     double p_sel, int64_t seed_t, torch::Tensor ci, torch::Tensor mi,
     torch::Tensor p1, torch::Tensor p2, torch::Tensor im, torch::Tensor di_c,
     torch::Tensor di_m, torch::Tensor lo, torch::Tensor hi,
-    double mutation_rate, int64_t seed_sbx, int64_t seed_mut) {
+    double mutation_rate, int64_t seed_sbx, int64_t seed_mut,
+    bool rank_sorted) {
   CHECK_GPU(population);
   TORCH_CHECK(population.dtype() == torch::kFloat32 &&
                   rank.dtype() == torch::kLong &&
@@ -551,6 +609,12 @@ torch::Tensor generation_spawn(
   auto pool_idx =
       torch::empty({poolsize}, population.options().dtype(torch::kLong));
   const float log1mp = logf(1.0f - (float)p_sel);
+  if (N > 1024) {
+    auto r = tournament_torch(population, rank, poolsize, log1mp, seed_t,
+                              rank_sorted);
+    return variation_events(r[0].contiguous(), ci, mi, p1, p2, im, di_c,
+                            di_m, lo, hi, mutation_rate, seed_sbx, seed_mut);
+  }
   if (launch_tournament(population.data_ptr<float>(),
                         (long long*)rank.data_ptr<int64_t>(),
                         pool.data_ptr<float>(),
@@ -835,7 +899,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sceua_accept", &sceua_accept);
   m.def("nsga2_select", &nsga2_select, "Fused survivor selection (cat+rank+crowding+sort+gather)");
   m.def("nsga2_select_acc", &nsga2_select_acc);
-  m.def("generation_spawn", &generation_spawn);
+  m.def("generation_spawn", &generation_spawn, py::arg("population"), py::arg("rank"), py::arg("poolsize"), py::arg("p_sel"), py::arg("seed_t"), py::arg("ci"), py::arg("mi"), py::arg("p1"), py::arg("p2"), py::arg("im"), py::arg("di_c"), py::arg("di_m"), py::arg("lo"), py::arg("hi"), py::arg("mutation_rate"), py::arg("seed_sbx"), py::arg("seed_mut"), py::arg("rank_sorted") = false);
   m.def("gp_predict_mean", &gp_predict_mean, "Fused cross-kernel + posterior mean",
         py::arg("Xq"), py::arg("X"), py::arg("theta"), py::arg("alpha"),
         py::arg("y_mean"), py::arg("y_std"), py::arg("nu"), py::arg("aniso"),

@@ -96,6 +96,28 @@ __global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
   }
 }
 
+// Multi-block route for the Gumbel tournament keys (same Philox counters
+// and formula as tournament_kernel's stage 2): for N beyond the one-
+// workgroup bitonic's sweet spot the caller sorts these with a radix
+// argsort instead. Continuous keys -> tie order never matters, so both
+// routes produce the same pool for the same seed.
+__global__ void tournament_keys_kernel(float* __restrict__ gkey, int N,
+                                       float log1mp,
+                                       unsigned long long seed) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= N) return;
+  Philox4 r = philox4x32(seed, (unsigned long long)i);
+  const float u = u01(r.c0);
+  gkey[i] = (float)i * log1mp - logf(-logf(u));
+}
+
+extern "C" void launch_tournament_keys(float* gkey, int N, float log1mp,
+                                       unsigned long long seed,
+                                       hipStream_t stream) {
+  hipLaunchKernelGGL(tournament_keys_kernel, dim3((N + 255) / 256), dim3(256),
+                     0, stream, gkey, N, log1mp, seed);
+}
+
 extern "C" int launch_tournament(const float* population, const long long* rank,
                                  float* pool, long long* pool_idx, int N,
                                  int d, int poolsize, float log1mp,
