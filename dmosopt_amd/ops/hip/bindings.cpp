@@ -237,8 +237,6 @@ std::vector<torch::Tensor> tournament_pool(torch::Tensor population,
   auto pool_idx =
       torch::empty({poolsize}, population.options().dtype(torch::kLong));
   const float log1mp = logf(1.0f - (float)p_sel);
-  if (N > 1024)
-    return tournament_torch(population, rank, poolsize, log1mp, seed, false);
   if (launch_tournament(population.data_ptr<float>(),
                         (long long*)rank.data_ptr<int64_t>(), pool.data_ptr<float>(),
                         (long long*)pool_idx.data_ptr<int64_t>(), N, d, (int)poolsize,
@@ -475,8 +473,12 @@ torch::Tensor crowding_distance(torch::Tensor Y) {
   const int N = Y.size(0), m = Y.size(1);
   if (N == 1) return torch::ones({1}, Y.options());
   static int torch_min_n = []() {
+    // default DISABLED: the 10-dispatch torch chain measured 35 ms/epoch
+    // SLOWER end-to-end at pop=1600 than the 208-us single-CU kernel it
+    // replaced (same box) — dispatch cost beats kernel time in the
+    // pipelined loop. Kept selectable for bigger-N re-measurement.
     const char* e = getenv("DMOSOPT_CROWD_TORCH_MIN_N");
-    return e ? atoi(e) : 1024;
+    return e ? atoi(e) : (1 << 30);
   }();
   if (N > torch_min_n) {
     // the single-workgroup-per-dim LDS bitonic degrades ~N log^2 N on one
@@ -544,8 +546,12 @@ std::vector<torch::Tensor> nsga2_select(torch::Tensor x_gen,
   // (3-4 launches -> 1); comparator (key asc, idx asc) == stable argsort.
   // Gated by N: the one-workgroup bitonic degrades ~N log^2 N on one CU
   // (122 us at N=3200) while the multi-block radix path stays ~40 us.
+  static int rcs_max_n = []() {
+    const char* e = getenv("DMOSOPT_RCS_MAX_N");
+    return e ? atoi(e) : (1 << 30);
+  }();
   auto perm_f = torch::empty({P0}, rank.options());
-  if (rcs_on && N <= 1024 &&
+  if (rcs_on && N <= rcs_max_n &&
       launch_rank_crowd_sort((long long*)rank.data_ptr<int64_t>(),
                              crowd.data_ptr<float>(),
                              (long long*)perm_f.data_ptr<int64_t>(), N, P0,
@@ -609,7 +615,11 @@ torch::Tensor generation_spawn(
   auto pool_idx =
       torch::empty({poolsize}, population.options().dtype(torch::kLong));
   const float log1mp = logf(1.0f - (float)p_sel);
-  if (N > 1024) {
+  static int tour_torch_min_n = []() {
+    const char* e = getenv("DMOSOPT_TOUR_TORCH_MIN_N");
+    return e ? atoi(e) : (1 << 30);
+  }();
+  if (N > tour_torch_min_n) {
     auto r = tournament_torch(population, rank, poolsize, log1mp, seed_t,
                               rank_sorted);
     return variation_events(r[0].contiguous(), ci, mi, p1, p2, im, di_c,
