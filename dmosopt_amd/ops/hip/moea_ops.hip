@@ -20,7 +20,8 @@
 // sorted position i gets weight p*(1-p)^i (reference MOEA.py:385-395);
 // weighted sampling WITHOUT replacement via Gumbel top-k on
 // key_i = i*log(1-p) + Gumbel_i (constants cancel under top-k).
-__global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
+template <int TPB>
+__global__ __launch_bounds__(TPB) void tournament_kernel(
     const float* __restrict__ population, const long long* __restrict__ rank,
     float* __restrict__ pool, long long* __restrict__ pool_idx, int N, int d,
     int poolsize, int npow2, float log1mp, unsigned long long seed) {
@@ -37,7 +38,7 @@ __global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
   // nsga2_select emits the population ALREADY rank-sorted, making this a
   // provable identity permutation (stable sort of a non-decreasing key):
   // detect and skip the whole bitonic pass in that (dominant) case.
-  for (int i = tid; i < npow2; i += TOUR_TPB) {
+  for (int i = tid; i < npow2; i += TPB) {
     key[i] = (i < N) ? (rank[i] * (long long)N + i) : 0x7FFFFFFFFFFFFFFFLL;
     idx[i] = i;
     if (i + 1 < N && rank[i] > rank[i + 1]) atomicOr(&unsorted, 1);
@@ -46,7 +47,7 @@ __global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
   if (unsorted)
   for (int ks = 2; ks <= npow2; ks <<= 1) {
     for (int js = ks >> 1; js > 0; js >>= 1) {
-      for (int i = tid; i < npow2; i += TOUR_TPB) {
+      for (int i = tid; i < npow2; i += TPB) {
         const int ixj = i ^ js;
         if (ixj > i) {
           const bool up = ((i & ks) == 0);
@@ -61,7 +62,7 @@ __global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
   }
   // idx[i] now = candidate at sorted position i. Stage 2: Gumbel keys per
   // position, sort DESCENDING, take first poolsize.
-  for (int i = tid; i < npow2; i += TOUR_TPB) {
+  for (int i = tid; i < npow2; i += TPB) {
     if (i < N) {
       Philox4 r = philox4x32(seed, (unsigned long long)i);
       const float u = u01(r.c0);
@@ -73,7 +74,7 @@ __global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
   __syncthreads();
   for (int ks = 2; ks <= npow2; ks <<= 1) {
     for (int js = ks >> 1; js > 0; js >>= 1) {
-      for (int i = tid; i < npow2; i += TOUR_TPB) {
+      for (int i = tid; i < npow2; i += TPB) {
         const int ixj = i ^ js;
         if (ixj > i) {
           const bool up = ((i & ks) == 0);
@@ -88,9 +89,9 @@ __global__ __launch_bounds__(TOUR_TPB) void tournament_kernel(
     }
   }
   // stage 3: gather the pool rows
-  for (int j = tid; j < poolsize; j += TOUR_TPB) pool_idx[j] = idx[j];
+  for (int j = tid; j < poolsize; j += TPB) pool_idx[j] = idx[j];
   __syncthreads();
-  for (long long t = tid; t < (long long)poolsize * d; t += TOUR_TPB) {
+  for (long long t = tid; t < (long long)poolsize * d; t += TPB) {
     const int r = (int)(t / d), c = (int)(t % d);
     pool[t] = population[(long long)idx[r] * d + c];
   }
@@ -126,9 +127,14 @@ extern "C" int launch_tournament(const float* population, const long long* rank,
   while (npow2 < N) npow2 <<= 1;
   if (npow2 > TOUR_NMAX) return -1;
   const size_t lds = (size_t)npow2 * (8 + 4 + 4);
-  hipLaunchKernelGGL(tournament_kernel, dim3(1), dim3(TOUR_TPB), lds, stream,
-                     population, rank, pool, pool_idx, N, d, poolsize, npow2,
-                     log1mp, seed);
+  if (npow2 > 512)
+    hipLaunchKernelGGL(tournament_kernel<1024>, dim3(1), dim3(1024), lds,
+                       stream, population, rank, pool, pool_idx, N, d,
+                       poolsize, npow2, log1mp, seed);
+  else
+    hipLaunchKernelGGL(tournament_kernel<TOUR_TPB>, dim3(1), dim3(TOUR_TPB),
+                       lds, stream, population, rank, pool, pool_idx, N, d,
+                       poolsize, npow2, log1mp, seed);
   return 0;
 }
 

@@ -518,9 +518,11 @@ extern "C" void launch_commit_front(const unsigned char* front,
 // index) in LDS; N padded to the next power of two with +inf sentinels.
 #define CROWD_TPB 256
 
-__global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
-                                float* __restrict__ out,  // (m, N)
-                                int N, int m, int npow2) {
+template <int TPB>
+__global__ __launch_bounds__(TPB) void crowding_kernel(
+    const float* __restrict__ Y,  // (N, m)
+    float* __restrict__ out,      // (m, N)
+    int N, int m, int npow2) {
   extern __shared__ float sh[];
   float* vals = sh;                       // npow2
   int* idxs = (int*)(sh + npow2);         // npow2
@@ -529,7 +531,7 @@ __global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
   // RAW column values: sorting is affine-invariant, and after the sort the
   // column min/max are vals[0]/vals[N-1] — the former host-side min/max/
   // span prep (~6 torch launches per call) is free here.
-  for (int i = threadIdx.x; i < npow2; i += CROWD_TPB) {
+  for (int i = threadIdx.x; i < npow2; i += TPB) {
     vals[i] = (i < N) ? Y[i * m + j] : INFINITY;
     idxs[i] = i;
   }
@@ -539,7 +541,7 @@ __global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
   // values produce zero gaps either way)
   for (int ksz = 2; ksz <= npow2; ksz <<= 1) {
     for (int jsz = ksz >> 1; jsz > 0; jsz >>= 1) {
-      for (int i = threadIdx.x; i < npow2; i += CROWD_TPB) {
+      for (int i = threadIdx.x; i < npow2; i += TPB) {
         const int ixj = i ^ jsz;
         if (ixj > i) {
           const bool up = ((i & ksz) == 0);
@@ -562,7 +564,7 @@ __global__ void crowding_kernel(const float* __restrict__ Y,  // (N, m)
   const float h = vals[N - 1];
   float s = h - l;
   if (!(s > 0.f) || !isfinite(s)) s = 1.f;
-  for (int i = threadIdx.x; i < N; i += CROWD_TPB) {
+  for (int i = threadIdx.x; i < N; i += TPB) {
     float d = (i == 0 || i == N - 1) ? 1.f : (vals[i + 1] - vals[i - 1]) / s;
     if (isnan(d)) d = 0.f;
     out[(long long)j * N + idxs[i]] = d;
@@ -575,6 +577,14 @@ extern "C" void launch_crowding(const float* Y, float* out, int N, int m,
   int npow2 = 1;
   while (npow2 < N) npow2 <<= 1;
   size_t lds_bytes = npow2 * (sizeof(float) + sizeof(int));
-  hipLaunchKernelGGL(crowding_kernel, dim3(m), dim3(CROWD_TPB), lds_bytes,
-                     stream, Y, out, N, m, npow2);
+  // wide blocks for big sorts: at npow2=4096 a 256-thread block walks 16
+  // elements per bitonic pass; 1024 threads walk 4 (measured 208 us ->
+  // see profiles/README). Small sorts keep 256 (idle waves only add
+  // barrier latency).
+  if (npow2 > 512)
+    hipLaunchKernelGGL(crowding_kernel<1024>, dim3(m), dim3(1024), lds_bytes,
+                       stream, Y, out, N, m, npow2);
+  else
+    hipLaunchKernelGGL(crowding_kernel<CROWD_TPB>, dim3(m), dim3(CROWD_TPB),
+                       lds_bytes, stream, Y, out, N, m, npow2);
 }
