@@ -231,6 +231,10 @@ __global__ __launch_bounds__(TPB) void peel_from_y_kernel(
 // (the count pass was the single-CU bottleneck of the one-launch peel), then
 // a single small block peels fronts with popcount(dom_mask & front_mask) —
 // each round costs ~N/TPB * N/32 word-ops instead of N^2/TPB pair compares.
+// M > 0: compile-time objective count (fully unrolled compare, row j
+// cached in registers instead of re-read from L2 for every word); M == 0:
+// generic runtime-m fallback.
+template <int M>
 __global__ void dom_bits_kernel(const float* __restrict__ Y,
                                 unsigned int* __restrict__ Dbits,  // (N, W)
                                 int N, int m, int W) {
@@ -240,16 +244,33 @@ __global__ void dom_bits_kernel(const float* __restrict__ Y,
   unsigned int bits = 0;
   const int i0 = w * 32;
   const int iend = min(32, N - i0);
-  for (int b = 0; b < iend; ++b) {
-    const int i = i0 + b;
-    if (i == j) continue;
-    bool le = true, lt = false;
-    for (int t = 0; t < m; ++t) {
-      const float a = Y[i * m + t], c = Y[j * m + t];
-      le &= (a <= c);
-      lt |= (a < c);
+  if (M > 0) {
+    float cj[M > 0 ? M : 1];
+#pragma unroll
+    for (int t = 0; t < M; ++t) cj[t] = Y[j * M + t];
+    for (int b = 0; b < iend; ++b) {
+      const int i = i0 + b;
+      bool le = true, lt = false;
+#pragma unroll
+      for (int t = 0; t < M; ++t) {
+        const float a = Y[i * M + t];
+        le &= (a <= cj[t]);
+        lt |= (a < cj[t]);
+      }
+      if (le && lt && i != j) bits |= (1u << b);
     }
-    if (le && lt) bits |= (1u << b);
+  } else {
+    for (int b = 0; b < iend; ++b) {
+      const int i = i0 + b;
+      if (i == j) continue;
+      bool le = true, lt = false;
+      for (int t = 0; t < m; ++t) {
+        const float a = Y[i * m + t], c = Y[j * m + t];
+        le &= (a <= c);
+        lt |= (a < c);
+      }
+      if (le && lt) bits |= (1u << b);
+    }
   }
   Dbits[idx] = bits;
 }
@@ -303,6 +324,8 @@ __global__ __launch_bounds__(PEELB_TPB) void peel_bits_kernel(
   }
 }
 
+#define LAUNCH_DOM_BITS(blocks_, stream_, Y_, D_, N_, m_, W_)                  do {                                                                           if (m_ == 2)                                                                   hipLaunchKernelGGL(dom_bits_kernel<2>, dim3(blocks_), dim3(256), 0,                               stream_, Y_, D_, N_, m_, W_);                          else if (m_ == 3)                                                              hipLaunchKernelGGL(dom_bits_kernel<3>, dim3(blocks_), dim3(256), 0,                               stream_, Y_, D_, N_, m_, W_);                          else if (m_ == 5)                                                              hipLaunchKernelGGL(dom_bits_kernel<5>, dim3(blocks_), dim3(256), 0,                               stream_, Y_, D_, N_, m_, W_);                          else                                                                           hipLaunchKernelGGL(dom_bits_kernel<0>, dim3(blocks_), dim3(256), 0,                               stream_, Y_, D_, N_, m_, W_);                        } while (0)
+
 extern "C" int launch_peel_bits(const float* Y, unsigned int* Dbits_scratch,
                                 int* rank, int N, int m, hipStream_t stream) {
   const int W = (N + 31) / 32;
@@ -316,8 +339,8 @@ extern "C" int launch_peel_bits(const float* Y, unsigned int* Dbits_scratch,
     attr_set = true;
   }
   const long long total = (long long)N * W;
-  hipLaunchKernelGGL(dom_bits_kernel, dim3((int)((total + 255) / 256)),
-                     dim3(256), 0, stream, Y, Dbits_scratch, N, m, W);
+  LAUNCH_DOM_BITS(((int)((total + 255) / 256)), stream, Y, Dbits_scratch, N,
+                  m, W);
   hipLaunchKernelGGL(peel_bits_kernel, dim3(1), dim3(PEELB_TPB), lds, stream,
                      Dbits_scratch, rank, N, W);
   return 0;
@@ -440,8 +463,7 @@ extern "C" int launch_coop_peel(const float* Y, unsigned int* Dbits,
     if (max_blocks < 1) max_blocks = 1;
   }
   const long long total = (long long)N * W;
-  hipLaunchKernelGGL(dom_bits_kernel, dim3((int)((total + 255) / 256)),
-                     dim3(256), 0, stream, Y, Dbits, N, m, W);
+  LAUNCH_DOM_BITS(((int)((total + 255) / 256)), stream, Y, Dbits, N, m, W);
   int blocks = (N + 255) / 256;
   if (blocks > max_blocks) blocks = max_blocks;
   void* args[] = {(void*)&Dbits, (void*)&fmask, (void*)&n_dom,
