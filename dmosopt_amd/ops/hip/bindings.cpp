@@ -16,8 +16,8 @@ void launch_matern_assemble(const float*, const float*, const float*, float*,
                             int, int, int, int, int, float, int, int, int,
                             hipStream_t);
 void launch_cholesky_batched(float*, float*, int*, int, int, hipStream_t);
-int launch_cholesky_fused_solve(float*, float*, int*, float*, int, int,
-                                hipStream_t);
+int launch_cholesky_fused_solve(float*, float*, int*, float*, int*, int,
+                                int, hipStream_t);
 void launch_forward_solve_batched(const float*, float*, int, int, int,
                                   hipStream_t);
 void launch_backward_solve_batched(const float*, float*, int, int, int,
@@ -150,10 +150,12 @@ torch::Tensor gp_nmll(torch::Tensor X, torch::Tensor theta, torch::Tensor y,
   // fused factor+solve: the rhs rides the multik launch chain (panel solves
   // its 32-entry segment, the SYRK's diagonal tiles apply the trailing
   // update) — no separate ~61 us serial TRSV kernel per NMLL
+  auto ws = torch::empty({2 * B}, X.options().dtype(torch::kInt32));
   if (launch_cholesky_fused_solve(K.data_ptr<float>(),
                                   logdet.data_ptr<float>(),
                                   info.data_ptr<int>(), Z.data_ptr<float>(),
-                                  B, N, cur_stream()) != 0) {
+                                  ws.data_ptr<int>(), B, N,
+                                  cur_stream()) != 0) {
     launch_cholesky_batched(K.data_ptr<float>(), logdet.data_ptr<float>(),
                             info.data_ptr<int>(), B, N, cur_stream());
     launch_forward_solve_batched(K.data_ptr<float>(), Z.data_ptr<float>(), B,

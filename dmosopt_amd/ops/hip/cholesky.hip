@@ -249,10 +249,15 @@ __global__ __launch_bounds__(CHOL_TPB) void cholesky_batched_kernel(
 // Factor the 32x32 diagonal block at (k0,k0) and panel-solve the rows below
 // it. One block per matrix; also accumulates the step's logdet contribution
 // deterministically (no atomics: k-steps are stream-ordered).
-template <int GROUP_COLS, int TPB = CHOLP_TPB>
-__global__ __launch_bounds__(TPB) void chol_panel_kernel(
-    float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
-    int N, int k0, float* __restrict__ rhs) {
+// Device body shared by the per-step panel kernel and the persistent
+// whole-factorization kernel (chol_persist_kernel): identical arithmetic,
+// only the launch wrapper differs.
+template <int GROUP_COLS, int TPB>
+__device__ __forceinline__ void chol_panel_body(
+    float* __restrict__ Ab, float* __restrict__ logdet_b,
+    int* __restrict__ info_b, int N, int k0, float* __restrict__ rhs_b,
+    float (*S)[CHOL_BS + 1], float (*colbuf4)[CHOL_BS],
+    float (*P)[CHOL_BS + 1]) {
   // rhs != nullptr: FUSED FORWARD SOLVE (bordered-matrix scheme). rhs (B, N)
   // starts as y and finishes as z = L^-1 y without a separate TRSV kernel:
   // this panel solves entries [k0, k0+bs) against the factored diagonal
@@ -260,11 +265,7 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
   // tiles apply the rank-32 update to the remaining entries. Saves the
   // ~61 us serial forward_solve_batched launch per NMLL (same serial-chain
   // structure as the factorization it now rides on).
-  __shared__ float S[CHOL_BS][CHOL_BS + 1];
-  __shared__ float colbuf4[GROUP_COLS][CHOL_BS];
-  const int b = blockIdx.x;
   const int tid = threadIdx.x;
-  float* Ab = A + (long long)b * N * N;
   const int bs = min(CHOL_BS, N - k0);
 
   // Wave-parallel diagonal factor: lane i of the first wave owns row i of
@@ -284,7 +285,6 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
   // memory untouched by the factor — so the ~34 KB coalesced load hides
   // entirely under the factor's shuffle chain instead of serializing after
   // it; disjoint LDS regions).
-  __shared__ float P[TPB][CHOL_BS + 1];
   const int c0_first = k0 + CHOL_BS;
   const int rows_first = min(TPB, N - c0_first);
   if (tid >= 64) {
@@ -342,8 +342,8 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
       bad = bad ? bad : ob;
     }
     if (lane == 0) {
-      logdet[b] += mylog;
-      if (bad && info[b] == 0) info[b] = bad;
+      *logdet_b += mylog;
+      if (bad && *info_b == 0) *info_b = bad;
     }
   } else {
     const int lane = tid;
@@ -424,8 +424,8 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
       bad = bad ? bad : ob;
     }
     if (lane == 0) {
-      logdet[b] += mylog;
-      if (bad && info[b] == 0) info[b] = bad;
+      *logdet_b += mylog;
+      if (bad && *info_b == 0) *info_b = bad;
     }
   }
   __syncthreads();
@@ -433,8 +433,8 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
   // factored diagonal block (wave-synchronous, lane j owns entry j; same
   // shuffle scheme as forward_solve_batched_kernel's diagonal solve). The
   // segment already carries the trailing updates of all previous steps.
-  if (rhs != nullptr && tid < 64) {
-    float* yb = rhs + (long long)b * N + k0;
+  if (rhs_b != nullptr && tid < 64) {
+    float* yb = rhs_b + k0;
     const int j = tid;
     float v = (j < bs) ? yb[j] : 0.0f;
     for (int t = 0; t < bs; ++t) {
@@ -494,6 +494,20 @@ __global__ __launch_bounds__(TPB) void chol_panel_kernel(
     }
     __syncthreads();
   }
+}
+
+
+template <int GROUP_COLS, int TPB = CHOLP_TPB>
+__global__ __launch_bounds__(TPB) void chol_panel_kernel(
+    float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
+    int N, int k0, float* __restrict__ rhs) {
+  __shared__ float S[CHOL_BS][CHOL_BS + 1];
+  __shared__ float colbuf4[GROUP_COLS][CHOL_BS];
+  __shared__ float P[TPB][CHOL_BS + 1];
+  const int b = blockIdx.x;
+  chol_panel_body<GROUP_COLS, TPB>(
+      A + (long long)b * N * N, logdet + b, info + b, N, k0,
+      rhs ? rhs + (long long)b * N : nullptr, S, colbuf4, P);
 }
 
 // ------------------------------------------------------------ 64-wide panel
@@ -699,33 +713,10 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk64_kernel(
 // Trailing update A[ti,tj] -= P_i P_j^T over 64x64 tiles of the submatrix
 // below/right of the panel; blockIdx.y enumerates lower-triangular tile
 // pairs, each 256-thread block computes a 4x4 register tile per thread.
-__global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
-    float* __restrict__ A, int N, int k0, int nt, int tj_fixed, int off,
-    float* __restrict__ rhs) {
-  // stride 32 with an XOR swizzle on the k-column: the MFMA operand read
-  // (lane -> [row + lr][k + lk]) hits banks (row+lk) mod 32 under a +1 pad,
-  // an up-to-8-way conflict (PMC: 2.2 extra cycles per LDS instruction);
-  // c = k ^ ((row & 7) << 2) spreads (lr, lk) over all 32 banks.
-  __shared__ float Pi[SYRK_TS][CHOL_BS];
-  __shared__ float Pj[SYRK_TS][CHOL_BS];
-  const int b = blockIdx.x;
-  float* Ab = A + (long long)b * N * N;
+__device__ __forceinline__ void chol_syrk_tile_body(
+    float* __restrict__ Ab, int N, int k0, int ti, int tj,
+    float* __restrict__ rhs_b, float (*Pi)[CHOL_BS], float (*Pj)[CHOL_BS]) {
   const int r0 = k0 + CHOL_BS;  // first trailing row
-  // tile-pair selection: tj_fixed >= 0 enumerates one tile COLUMN
-  // (ti = blockIdx.y + tj_fixed); otherwise blockIdx.y walks the lower
-  // triangle, shifted by `off` columns/rows (used to split the first tile
-  // column from the rest for cross-stream overlap)
-  int ti, tj;
-  if (tj_fixed >= 0) {
-    tj = tj_fixed;
-    ti = blockIdx.y + tj_fixed;
-  } else {
-    int p = blockIdx.y;
-    ti = 0;
-    while (p > ti) { p -= ti + 1; ++ti; }
-    tj = p + off;
-    ti += off;
-  }
   const int i0 = r0 + ti * SYRK_TS, j0 = r0 + tj * SYRK_TS;
   const int tid = threadIdx.x;
 
@@ -737,19 +728,15 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
   }
   __syncthreads();
 
-  // MFMA tile core: C -= Pi * Pj^T on the matrix units.
-  // v_mfma_f32_16x16x4_f32 is EXACT fp32 (identical to an fmaf chain) at
-  // the f32 vector rate, but one instruction carries a whole 16x16x4 tile
-  // — 16x fewer issue slots than the 4x4-register-tile FMA version, which
-  // matters for this latency-bound kernel. Operand map (cdna4_isa §10):
-  // A: lane l -> A[l&15][l>>4]; B: lane l -> B[l>>4][l&15];
-  // C/D (f32x4): col = lane&15, row = (lane>>4)*4 + reg.
+  // MFMA tile core: C -= Pi * Pj^T on the matrix units (operand map and
+  // rationale in the original kernel comment below).
   typedef __attribute__((ext_vector_type(4))) float f32x4;
   const int wave = tid >> 6, lane = tid & 63;
   const int lr = lane & 15, lk = lane >> 4;
 #pragma unroll
   for (int sIdx = 0; sIdx < 4; ++sIdx) {
     const int sub = wave * 4 + sIdx;        // 4x4 grid of 16x16 subtiles
+    if (sub >= 16) break;  // blocks wider than 4 waves: extra waves idle
     const int r16 = (sub >> 2) * 16, c16 = (sub & 3) * 16;
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -766,22 +753,53 @@ __global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
       if (i < N && j < N && j <= i) Ab[(long long)i * N + j] -= acc[r];
     }
   }
-  // fused-solve trailing update: each DIAGONAL tile covers every trailing
-  // row exactly once, so its wave 0 applies the rank-32 rhs update for its
-  // 64 rows — rhs[i] -= dot(L[i, k0:k0+32], z_panel) — reusing the Pi tile
-  // already staged (and swizzled) in LDS. z_panel = rhs[k0:k0+32] was
-  // written by this step's panel kernel; read/write ranges are disjoint.
-  if (rhs != nullptr && ti == tj && tid < 64) {
+  // fused-solve trailing update (diagonal tiles only); see original
+  // kernel comment.
+  if (rhs_b != nullptr && ti == tj && tid < 64) {
     const int i = i0 + tid;
     if (i < N) {
-      const float* z = rhs + (long long)b * N + k0;
-      float acc = rhs[(long long)b * N + i];
+      const float* z = rhs_b + k0;
+      float acc = rhs_b[i];
 #pragma unroll
       for (int t = 0; t < CHOL_BS; ++t)
         acc = fmaf(-Pi[tid][t ^ ((tid & 7) << 2)], z[t], acc);
-      rhs[(long long)b * N + i] = acc;
+      rhs_b[i] = acc;
     }
   }
+}
+
+__global__ __launch_bounds__(CHOLP_TPB) void chol_syrk_kernel(
+    float* __restrict__ A, int N, int k0, int nt, int tj_fixed, int off,
+    float* __restrict__ rhs) {
+  // stride 32 with an XOR swizzle on the k-column: the MFMA operand read
+  // (lane -> [row + lr][k + lk]) hits banks (row+lk) mod 32 under a +1 pad,
+  // an up-to-8-way conflict (PMC: 2.2 extra cycles per LDS instruction);
+  // c = k ^ ((row & 7) << 2) spreads (lr, lk) over all 32 banks.
+  // v_mfma_f32_16x16x4_f32 is EXACT fp32 (identical to an fmaf chain) at
+  // the f32 vector rate; operand map (cdna4_isa section 10):
+  // A: lane l -> A[l&15][l>>4]; B: lane l -> B[l>>4][l&15];
+  // C/D (f32x4): col = lane&15, row = (lane>>4)*4 + reg.
+  __shared__ float Pi[SYRK_TS][CHOL_BS];
+  __shared__ float Pj[SYRK_TS][CHOL_BS];
+  const int b = blockIdx.x;
+  float* Ab = A + (long long)b * N * N;
+  // tile-pair selection: tj_fixed >= 0 enumerates one tile COLUMN
+  // (ti = blockIdx.y + tj_fixed); otherwise blockIdx.y walks the lower
+  // triangle, shifted by `off` columns/rows (used to split the first tile
+  // column from the rest for cross-stream overlap)
+  int ti, tj;
+  if (tj_fixed >= 0) {
+    tj = tj_fixed;
+    ti = blockIdx.y + tj_fixed;
+  } else {
+    int p = blockIdx.y;
+    ti = 0;
+    while (p > ti) { p -= ti + 1; ++ti; }
+    tj = p + off;
+    ti += off;
+  }
+  chol_syrk_tile_body(Ab, N, k0, ti, tj,
+                      rhs ? rhs + (long long)b * N : nullptr, Pi, Pj);
 }
 
 // Blocked forward substitution: solve L z = y for R right-hand sides.
@@ -936,6 +954,11 @@ extern "C" void launch_nmll_reduce(const float* Z, const float* half_logdet,
                      half_logdet, info, out, N, c);
 }
 
+__global__ void zero_i32_kernel(int* __restrict__ p, int n) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = 0;
+}
+
 __global__ void zero_f32_kernel(float* __restrict__ p, int n) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) p[i] = 0.0f;
@@ -1073,6 +1096,141 @@ extern "C" void launch_cholesky_multik(float* A, float* logdet, int* info,
 }
 
 
+
+// ------------------------------------------------ persistent whole-chol
+// ONE launch factorizes all B matrices: grid (B, 1 + T). Block (b, 0) is
+// matrix b's PANEL block — it loops the right-looking steps, running the
+// shared chol_panel_body per step; blocks (b, 1..T) are its SYRK TILE
+// blocks. Steps hand off through two per-matrix device-scope flags
+// (panel_flag counts completed panels, tiles_done counts completed tiles
+// cumulatively — monotone counters, so no reset races):
+//   tiles of step s spin until panel_flag >= s+1, then update their
+//   64x64 trailing tile; the panel of step s+1 spins until tiles_done ==
+//   cumulative pairs through step s. Replaces 2*ceil(N/32)-1 dispatches
+//   (each ~5-8 us of entry/exit on this stack even inside a hipGraph)
+//   with intra-kernel handoffs. DEADLOCK SAFETY: the launcher verifies
+//   full-grid residency via the occupancy API (all blocks co-resident by
+//   construction), and both spin loops are BOUNDED — on timeout they
+//   stamp info[b] and exit instead of hanging the device.
+#define PERSIST_TPB 384
+
+static __device__ __forceinline__ int persist_pairs(int N, int k0) {
+  const int trailing = N - k0 - CHOL_BS;
+  if (trailing <= 0) return 0;
+  const int nt = (trailing + SYRK_TS - 1) / SYRK_TS;
+  return nt * (nt + 1) / 2;
+}
+
+__global__ __launch_bounds__(PERSIST_TPB) void chol_persist_kernel(
+    float* __restrict__ A, float* __restrict__ logdet, int* __restrict__ info,
+    float* __restrict__ rhs, int* __restrict__ ws, int N) {
+  extern __shared__ char smem[];
+  const int b = blockIdx.x;
+  const int role = blockIdx.y;  // 0 = panel block, >=1 = tile block role-1
+  const int tid = threadIdx.x;
+  float* Ab = A + (long long)b * N * N;
+  float* rhs_b = rhs ? rhs + (long long)b * N : nullptr;
+  int* panel_flag = ws + 2 * b;
+  int* tiles_done = ws + 2 * b + 1;
+
+  if (role == 0) {
+    float(*S)[CHOL_BS + 1] = (float(*)[CHOL_BS + 1])smem;
+    float(*colbuf)[CHOL_BS] =
+        (float(*)[CHOL_BS])(smem + sizeof(float) * CHOL_BS * (CHOL_BS + 1));
+    float(*P)[CHOL_BS + 1] =
+        (float(*)[CHOL_BS + 1])(smem + sizeof(float) * (CHOL_BS * (CHOL_BS + 1) +
+                                                        2 * CHOL_BS));
+    int cum = 0;
+    for (int k0 = 0, step = 0; k0 < N; k0 += CHOL_BS, ++step) {
+      if (step > 0) {
+        if (tid == 0) {
+          long long it = 0;
+          while (atomicAdd(tiles_done, 0) < cum) {
+            __builtin_amdgcn_s_sleep(8);
+            if (++it > 5000000LL) {  // ~1 s at ~512 cycles per sleep(8)  // bounded: fail loudly, not a hang
+              if (atomicCAS(info + b, 0, -777) == 0) {}
+              break;
+            }
+          }
+        }
+        __syncthreads();
+        __threadfence();  // acquire: invalidate L1 before reading tiles' output
+        if (*(volatile int*)(info + b) < 0) return;  // timeout sentinel
+      }
+      chol_panel_body<2, PERSIST_TPB>(Ab, logdet + b, info + b, N, k0, rhs_b,
+                                      S, colbuf, P);
+      __threadfence();  // release: panel writes visible device-wide
+      __syncthreads();
+      if (tid == 0) atomicExch(panel_flag, step + 1);
+      cum += persist_pairs(N, k0);
+    }
+  } else {
+    float(*Pi)[CHOL_BS] = (float(*)[CHOL_BS])smem;
+    float(*Pj)[CHOL_BS] =
+        (float(*)[CHOL_BS])(smem + sizeof(float) * SYRK_TS * CHOL_BS);
+    const int t = role - 1;
+    for (int k0 = 0, step = 0; k0 < N; k0 += CHOL_BS, ++step) {
+      const int pairs = persist_pairs(N, k0);
+      if (t >= pairs) continue;  // no tile for this block at this step
+      if (tid == 0) {
+        long long it = 0;
+        while (atomicAdd(panel_flag, 0) < step + 1) {
+          __builtin_amdgcn_s_sleep(8);
+          if (++it > 5000000LL) {  // ~1 s at ~512 cycles per sleep(8)
+            if (atomicCAS(info + b, 0, -778) == 0) {}
+            break;
+          }
+        }
+      }
+      __syncthreads();
+      __threadfence();  // acquire before reading the panel output
+      if (*(volatile int*)(info + b) < 0) return;  // timeout sentinel
+      int p = t, ti = 0;
+      while (p > ti) { p -= ti + 1; ++ti; }
+      const int tj = p;
+      chol_syrk_tile_body(Ab, N, k0, ti, tj, rhs_b, Pi, Pj);
+      __threadfence();  // release tile writes
+      __syncthreads();
+      if (tid == 0) atomicAdd(tiles_done, 1);
+    }
+  }
+}
+
+extern "C" int launch_cholesky_persist(float* A, float* logdet, int* info,
+                                       float* rhs, int* ws, int B, int N,
+                                       hipStream_t stream) {
+  // only the default panel configuration is instantiated
+  if (panel_group_cols() != 2) return -1;
+  const int trailing0 = N - CHOL_BS;
+  if (trailing0 <= 0) return -1;
+  const int nt0 = (trailing0 + SYRK_TS - 1) / SYRK_TS;
+  const int T = nt0 * (nt0 + 1) / 2;
+  const size_t lds = sizeof(float) * (CHOL_BS * (CHOL_BS + 1) + 2 * CHOL_BS +
+                                      PERSIST_TPB * (CHOL_BS + 1));
+  static int max_resident = -1;
+  if (max_resident < 0) {
+    hipFuncSetAttribute((const void*)chol_persist_kernel,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    int per_cu = 0;
+    hipOccupancyMaxActiveBlocksPerMultiprocessor(
+        &per_cu, (const void*)chol_persist_kernel, PERSIST_TPB, lds);
+    hipDeviceProp_t prop;
+    int dev = 0;
+    hipGetDevice(&dev);
+    hipGetDeviceProperties(&prop, dev);
+    max_resident = per_cu * prop.multiProcessorCount;
+  }
+  // DEADLOCK GUARD: every block must be co-resident for the spin handoffs
+  if ((long long)B * (1 + T) > max_resident) return -1;
+  hipLaunchKernelGGL(zero_i32_kernel, dim3((2 * B + 255) / 256), dim3(256), 0,
+                     stream, ws, 2 * B);
+  hipLaunchKernelGGL(zero_f32_kernel, dim3((B + 255) / 256), dim3(256), 0,
+                     stream, logdet, B);
+  hipLaunchKernelGGL(chol_persist_kernel, dim3(B, 1 + T), dim3(PERSIST_TPB),
+                     lds, stream, A, logdet, info, rhs, ws, N);
+  return 0;
+}
+
 // bf16-SYRK variant of the right-looking multik path (config-#2 precision
 // route): panels factor in exact fp32 (chol_panel_kernel), only the
 // trailing C -= P P^T runs on the bf16 matrix units (matern_bf16.hip).
@@ -1095,10 +1253,14 @@ extern "C" void launch_cholesky_multik_bf16(float* A, float* logdet,
   }
 }
 
+extern "C" int launch_cholesky_persist(float*, float*, int*, float*, int*,
+                                       int, int, hipStream_t);
+
 extern "C" int launch_cholesky_fused_solve(float* A, float* logdet,
-                                           int* info, float* rhs, int B,
-                                           int N, hipStream_t stream) {
+                                           int* info, float* rhs, int* ws,
+                                           int B, int N, hipStream_t stream) {
   static int ok = -2;
+  static int persist_on = -1;
   if (ok == -2) {
     const char* m = getenv("DMOSOPT_CHOL_MODE");
     const char* o = getenv("DMOSOPT_CHOL_OVERLAP");
@@ -1108,8 +1270,13 @@ extern "C" int launch_cholesky_fused_solve(float* A, float* logdet,
           (f && f[0] == '0'))
              ? 0
              : 1;
+    const char* pe = getenv("DMOSOPT_CHOL_PERSIST");
+    persist_on = (pe && pe[0] == '0') ? 0 : 1;
   }
   if (!ok || N <= CHOL_BS || B > 48) return -1;
+  if (persist_on && ws != nullptr &&
+      launch_cholesky_persist(A, logdet, info, rhs, ws, B, N, stream) == 0)
+    return 0;
   launch_cholesky_multik(A, logdet, info, rhs, B, N, stream);
   return 0;
 }
