@@ -1270,8 +1270,11 @@ extern "C" int launch_cholesky_fused_solve(float* A, float* logdet,
           (f && f[0] == '0'))
              ? 0
              : 1;
+    // measured 205 vs 136 ms/epoch (same box, bit-identical HV): the
+    // intra-kernel handoffs + pointer-based LDS codegen cost far more
+    // than the ~16 saved dispatches. Kept as an opt-in experiment.
     const char* pe = getenv("DMOSOPT_CHOL_PERSIST");
-    persist_on = (pe && pe[0] == '0') ? 0 : 1;
+    persist_on = (pe && pe[0] == '1') ? 1 : 0;
   }
   if (!ok || N <= CHOL_BS || B > 48) return -1;
   if (persist_on && ws != nullptr &&
