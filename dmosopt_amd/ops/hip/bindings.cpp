@@ -190,7 +190,14 @@ torch::Tensor gp_predict_mean(torch::Tensor Xq, torch::Tensor X,
       aniso ? 1 : 0, 0, q_lb ? q_lb->data_ptr<float>() : nullptr,
       q_invrg ? q_invrg->data_ptr<float>() : nullptr, cur_stream());
   auto mean_n = torch::bmm(Ks, alpha).squeeze(-1);  // (m, P)
-  return y_mean.unsqueeze(1).addcmul(y_std.unsqueeze(1), mean_n).transpose(0, 1);
+  // CONTIGUOUS (P, m) result: returning the bare transpose view made every
+  // downstream consumer (cat, .contiguous(), elementwise) pay a strided
+  // copy per generation (torch.profiler: ~50 us/gen of direct_copy at
+  // pop=1600)
+  return y_mean.unsqueeze(1)
+      .addcmul(y_std.unsqueeze(1), mean_n)
+      .transpose(0, 1)
+      .contiguous();
 }
 
 // Fused SCE-UA CCE stage: candidate proposal and acceptance+resort, one
@@ -408,7 +415,9 @@ torch::Tensor pareto_rank(torch::Tensor Y, int64_t stop = -1) {
     auto fmask = torch::empty({3 * W}, opts_i);  // triple-buffered front mask
     auto n_dom = torch::empty({N}, opts_i);
     auto ctrl = torch::empty({2}, opts_i);
-    auto rank = torch::zeros({N}, opts_i);
+    // empty, not zeros: the peel kernels write every rank entry (the fill
+    // kernel cost ~24 us/call at N=3200)
+    auto rank = torch::empty({N}, opts_i);
     if (launch_coop_peel(Yc.data_ptr<float>(),
                          (unsigned int*)Dbits.data_ptr<int>(),
                          (unsigned int*)fmask.data_ptr<int>(),
