@@ -284,7 +284,7 @@ class _GPRBase:
                     )
                 return _hipops.gp_predict_mean(
                     xr.float().contiguous(), pa[0], pa[1], pa[2], pa[3],
-                    pa[4], pa[5], pa[6], q_lb=cache[1], q_invrg=cache[2],
+                    pa[4], pa[5], pa[6], cache[1], cache[2],
                 )
         xq = self.normalize_query(xr)
         mean, _ = self._fitted.predict(xq, return_var=self.return_mean_variance)

@@ -913,3 +913,38 @@ def test_smpso_gpu_e2e_uses_kernel(dev):
         opt.update(xg, torch.as_tensor(yg, dtype=torch.float32, device=dev), st)
     px, py = opt.population_objectives
     assert torch.isfinite(torch.as_tensor(np.asarray(py) if not isinstance(py, torch.Tensor) else py.cpu().numpy())).all()
+
+
+def test_generation_spawn_matches_split_composition(dev):
+    """generation_spawn (one binding call) must equal tournament_pool +
+    variation_events composed manually with the same seeds — locks the
+    fused production route against the pieces it fuses."""
+    from dmosopt_amd import _hipops
+
+    torch.manual_seed(11)
+    N, d, poolsize, C, M = 300, 12, 150, 80, 20
+    population = torch.rand(N, d, device=dev)
+    rank = torch.sort(torch.randint(0, 9, (N,), device=dev)).values.long()
+    total = 2 * C + M
+    perm = torch.randperm(total, device=dev)
+    ci = perm[: 2 * C].contiguous()
+    mi = perm[2 * C :].contiguous()
+    g = torch.Generator().manual_seed(5)
+    i1 = torch.randint(0, poolsize, (C,), generator=g).to(dev)
+    i2 = (i1 + 1 + torch.randint(0, poolsize - 1, (C,), generator=g).to(dev)) % poolsize
+    im = torch.randint(0, poolsize, (M,), generator=g).to(dev)
+    di_c = torch.full((d,), 1.0, device=dev)
+    di_m = torch.full((d,), 20.0, device=dev)
+    lo = torch.zeros(d, device=dev)
+    hi = torch.ones(d, device=dev)
+    st, s1, s2 = 777, 1234, 9876
+
+    pool, _ = _hipops.tournament_pool(population.contiguous(), rank.contiguous(),
+                                      poolsize, 0.5, st)
+    want = _hipops.variation_events(pool.contiguous(), ci, mi, i1, i2, im,
+                                    di_c, di_m, lo, hi, 0.1, s1, s2)
+    got = _hipops.generation_spawn(population.contiguous(), rank.contiguous(),
+                                   poolsize, 0.5, st, ci, mi, i1, i2, im,
+                                   di_c, di_m, lo, hi, 0.1, s1, s2,
+                                   rank_sorted=True)
+    assert torch.equal(got, want)
