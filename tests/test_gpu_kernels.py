@@ -948,3 +948,21 @@ def test_generation_spawn_matches_split_composition(dev):
                                    di_c, di_m, lo, hi, 0.1, s1, s2,
                                    rank_sorted=True)
     assert torch.equal(got, want)
+
+
+def test_large_pop_epoch_bitwise_deterministic(dev):
+    """pop=1600 epoch (the 8-GPU weak-scaling regime) twice with the same
+    seed: final front must be BIT-identical — locks the large-N kernels
+    (cooperative peel atomics, wide-block bitonics, event-decoded
+    variation) to run-to-run determinism, the replicated-control-flow
+    invariant."""
+    import bench as B
+
+    X, Y = B.make_archive(seed=9)
+    outs = []
+    for _ in range(2):
+        x_res, y_res, hv = B.one_epoch(X, Y, 1600, 0, 1, dev, seed=77, n_gen=30)
+        outs.append((x_res.cpu().clone(), y_res.cpu().clone(), float(hv)))
+    assert torch.equal(outs[0][0], outs[1][0])
+    assert torch.equal(outs[0][1], outs[1][1])
+    assert outs[0][2] == outs[1][2]
