@@ -140,12 +140,17 @@ class NSGA2Optimizer(MOEA):
             # tournament + event-decoded variation chained in ONE binding
             # call (the loop is host-dispatch-bound; the pool tensor never
             # surfaces to python)
-            from dmosopt_amd.moea.variation import spawn_generation_native
+            from dmosopt_amd.moea.variation import (
+                _SpawnPrefetch, spawn_generation_native,
+            )
 
+            pf = getattr(self, "_spawn_prefetch", None)
+            if pf is None:
+                pf = self._spawn_prefetch = _SpawnPrefetch()
             res = spawn_generation_native(
                 population, rank, poolsize, 0.5, rng, popsize,
                 p.crossover_prob, p.mutation_prob, p.mutation_rate,
-                di_c, di_m, xlb_f, xub_f,
+                di_c, di_m, xlb_f, xub_f, prefetch=pf,
             )
             if res is not None:
                 x_gen, crossover_indices, mutation_indices = res

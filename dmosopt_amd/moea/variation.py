@@ -95,6 +95,49 @@ def _assemble_event_indices(rng, popsize, poolsize, crossover_prob, mutation_pro
     return combined, C, M, seed_sbx, seed_mut
 
 
+class _SpawnPrefetch:
+    """Chunked prefetch of the per-generation host randomness.
+
+    The event stream, parent indices and slot layout depend ONLY on the
+    rng (never on results), so drawing CHUNK generations ahead — in the
+    SAME rng call order the sequential loop would use — yields a
+    bit-identical stream while amortizing the numpy assembly and the slow
+    pinned H2D (~1 GB/s on this platform, profiles/README.md) over one
+    transfer instead of CHUNK."""
+
+    CHUNK = 16
+
+    def __init__(self):
+        self.key = None
+        self.items = []
+
+    def refill(self, rng, popsize, poolsize, pc, pm, device):
+        metas = []
+        parts = []
+        for _ in range(self.CHUNK):
+            seed_t = int(rng.integers(0, 2**62))
+            combined, C, M, s1, s2 = _assemble_event_indices(
+                rng, popsize, poolsize, pc, pm
+            )
+            metas.append((seed_t, C, M, s1, s2, combined.shape[0]))
+            parts.append(combined)
+        dev = _to_device_pinned(np.concatenate(parts), device)
+        items = []
+        base = 0
+        for seed_t, C, M, s1, s2, ln in metas:
+            items.append((seed_t, C, M, s1, s2, dev[base : base + ln]))
+            base += ln
+        items.reverse()  # pop() yields generation order
+        self.items = items
+
+    def next(self, rng, popsize, poolsize, pc, pm, device):
+        key = (popsize, poolsize, pc, pm, str(device))
+        if key != self.key or not self.items:
+            self.key = key
+            self.refill(rng, popsize, poolsize, pc, pm, device)
+        return self.items.pop()
+
+
 def spawn_generation_native(
     population: torch.Tensor,
     rank: torch.Tensor,
@@ -109,6 +152,7 @@ def spawn_generation_native(
     di_mutation: torch.Tensor,
     xlb: torch.Tensor,
     xub: torch.Tensor,
+    prefetch: "_SpawnPrefetch" = None,
 ):
     """Tournament selection + whole-generation variation in ONE extension
     call (generation_spawn binding): the host-dispatch-bound loop pays one
@@ -117,11 +161,17 @@ def spawn_generation_native(
     tournament kernel refuses the shape (caller uses the split path)."""
     from dmosopt_amd import _hipops
 
-    seed_t = int(rng.integers(0, 2**62))
-    combined, C, M, seed_sbx, seed_mut = _assemble_event_indices(
-        rng, popsize, poolsize, crossover_prob, mutation_prob
-    )
-    dev = _to_device_pinned(combined, population.device)
+    if prefetch is not None:
+        seed_t, C, M, seed_sbx, seed_mut, dev = prefetch.next(
+            rng, popsize, poolsize, crossover_prob, mutation_prob,
+            population.device,
+        )
+    else:
+        seed_t = int(rng.integers(0, 2**62))
+        combined, C, M, seed_sbx, seed_mut = _assemble_event_indices(
+            rng, popsize, poolsize, crossover_prob, mutation_prob
+        )
+        dev = _to_device_pinned(combined, population.device)
     o = 0
     i1_t = dev[o : o + C]; o += C
     i2_t = dev[o : o + C]; o += C
