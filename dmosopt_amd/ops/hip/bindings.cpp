@@ -12,6 +12,7 @@
 
 extern "C" {
 void launch_matern_assemble_affine(const float*, const float*, const float*, float*, int, int, int, int, int, float, int, int, int, const float*, const float*, hipStream_t);
+void launch_gp_mean_gemv(const float*, const float*, const float*, const float*, float*, int, int, int, hipStream_t);
 void launch_matern_assemble(const float*, const float*, const float*, float*,
                             int, int, int, int, int, float, int, int, int,
                             hipStream_t);
@@ -189,15 +190,13 @@ torch::Tensor gp_predict_mean(torch::Tensor Xq, torch::Tensor X,
       Ks.data_ptr<float>(), B, P, N, D, theta.size(1), 0.0f, nu_code(nu),
       aniso ? 1 : 0, 0, q_lb ? q_lb->data_ptr<float>() : nullptr,
       q_invrg ? q_invrg->data_ptr<float>() : nullptr, cur_stream());
-  auto mean_n = torch::bmm(Ks, alpha).squeeze(-1);  // (m, P)
-  // CONTIGUOUS (P, m) result: returning the bare transpose view made every
-  // downstream consumer (cat, .contiguous(), elementwise) pay a strided
-  // copy per generation (torch.profiler: ~50 us/gen of direct_copy at
-  // pop=1600)
-  return y_mean.unsqueeze(1)
-      .addcmul(y_std.unsqueeze(1), mean_n)
-      .transpose(0, 1)
-      .contiguous();
+  // fused matvec + de-standardization into the contiguous (P, B) result:
+  // one kernel instead of bmm + addcmul + transpose-contiguous
+  auto out = torch::empty({P, (long)B}, X.options());
+  launch_gp_mean_gemv(Ks.data_ptr<float>(), alpha.data_ptr<float>(),
+                      y_mean.data_ptr<float>(), y_std.data_ptr<float>(),
+                      out.data_ptr<float>(), B, P, N, cur_stream());
+  return out;
 }
 
 // Fused SCE-UA CCE stage: candidate proposal and acceptance+resort, one
