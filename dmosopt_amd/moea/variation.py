@@ -69,6 +69,29 @@ def _draw_event_stream(rng, popsize: int, crossover_prob: float, mutation_prob: 
     return np.concatenate(c_parts), np.concatenate(m_parts)
 
 
+def _draw_event_stream_tail(rng, remaining_target, crossover_prob, mutation_prob):
+    """Continue an under-shot event stream until `remaining_target` more
+    children exist (rare batched-prefetch overflow path)."""
+    if remaining_target <= 0:
+        return (np.zeros(0, dtype=bool), np.zeros(0, dtype=bool))
+    c_parts, m_parts = [], []
+    cum = 0
+    while True:
+        u = rng.random((64, 2))
+        c = u[:, 0] < crossover_prob
+        m = u[:, 1] < mutation_prob
+        cs = cum + np.cumsum(2 * c.astype(np.int64) + m.astype(np.int64))
+        j0 = int(np.searchsorted(cs, remaining_target, side="left"))
+        if j0 < 64:
+            c_parts.append(c[: j0 + 1])
+            m_parts.append(m[: j0 + 1])
+            break
+        c_parts.append(c)
+        m_parts.append(m)
+        cum = int(cs[-1])
+    return np.concatenate(c_parts), np.concatenate(m_parts)
+
+
 def _assemble_event_indices(rng, popsize, poolsize, crossover_prob, mutation_prob):
     """Draw + assemble the generation's event indices (host numpy side).
 
@@ -112,14 +135,60 @@ class _SpawnPrefetch:
         self.items = []
 
     def refill(self, rng, popsize, poolsize, pc, pm, device):
+        """Batched assembly: each numpy Generator call costs ~6 us of host
+        time and each numpy op a few more, so the CHUNK generations' draws
+        are made with ONE call per component (event uniforms, parent
+        uniforms, seeds) and the per-generation work is reduced to the
+        ragged slot-layout math."""
+        G = self.CHUNK
+        target = popsize - 1
+        chunk = max(64, popsize)
+        U = rng.random((G, chunk, 2))
+        c_all = U[:, :, 0] < pc
+        m_all = U[:, :, 1] < pm
+        inc = 2 * c_all + m_all
+        cs = np.cumsum(inc, axis=1)
+        reached = cs >= target
+        ok = reached[:, -1]
+        j0 = np.argmax(reached, axis=1)  # first iteration reaching target
+        events = []
+        for g in range(G):
+            if ok[g]:
+                events.append((c_all[g, : j0[g] + 1], m_all[g, : j0[g] + 1]))
+            else:
+                # rare overflow (the G*chunk draw under-shot): finish this
+                # generation's stream with the sequential path
+                extra_c, extra_m = _draw_event_stream_tail(
+                    rng, target - int(cs[g, -1]), pc, pm
+                )
+                events.append((np.concatenate([c_all[g], extra_c]),
+                               np.concatenate([m_all[g], extra_m])))
+        Cs = [int(c.sum()) for c, _ in events]
+        Ms = [int(m.sum()) for _, m in events]
+        u_par = rng.random(2 * sum(Cs) + sum(Ms))
+        seeds = rng.integers(0, 2**62, (G, 3))
         metas = []
         parts = []
-        for _ in range(self.CHUNK):
-            seed_t = int(rng.integers(0, 2**62))
-            combined, C, M, s1, s2 = _assemble_event_indices(
-                rng, popsize, poolsize, pc, pm
-            )
-            metas.append((seed_t, C, M, s1, s2, combined.shape[0]))
+        up = 0
+        for g in range(G):
+            c_ev, m_ev = events[g]
+            C, M = Cs[g], Ms[g]
+            u = u_par[up : up + 2 * C + M]; up += 2 * C + M
+            if C:
+                i1 = (u[:C] * poolsize).astype(np.int64)
+                i2 = (u[C : 2 * C] * (poolsize - 1)).astype(np.int64)
+                i2 = i2 + (i2 >= i1)
+            else:
+                i1 = i2 = np.empty(0, dtype=np.int64)
+            im = (u[2 * C :] * poolsize).astype(np.int64) if M else np.empty(0, dtype=np.int64)
+            it_sizes = 2 * c_ev + m_ev
+            it_starts = np.cumsum(it_sizes) - it_sizes
+            ci = np.repeat(it_starts[c_ev], 2)
+            ci[1::2] += 1
+            mi = (it_starts + 2 * c_ev)[m_ev]
+            combined = np.concatenate([i1, i2, im, ci, mi])
+            metas.append((int(seeds[g, 0]), C, M, int(seeds[g, 1]),
+                          int(seeds[g, 2]), combined.shape[0]))
             parts.append(combined)
         dev = _to_device_pinned(np.concatenate(parts), device)
         items = []

@@ -69,3 +69,45 @@ def test_cpu_scatter_matches_slot_semantics():
     # every slot was written (no empty rows from a placement bug):
     # children are inside bounds and vary (not default-initialized zeros)
     assert x_gen.abs().sum() > 0
+
+
+def test_prefetch_batch_matches_invariants():
+    """The chunked-prefetch batched assembly must satisfy the same slot
+    invariants as the single-generation path, for every entry of a chunk."""
+    import torch
+    from dmosopt_amd.moea.variation import _SpawnPrefetch
+
+    rng = np.random.default_rng(11)
+    pf = _SpawnPrefetch()
+    dev = torch.device("cpu")
+    for _ in range(20):  # > one chunk: exercises refill twice
+        seed_t, C, M, s1, s2, buf = pf.next(rng, 120, 60, 0.9, 0.1, dev)
+        total = 2 * C + M
+        arr = buf.numpy()
+        i1, i2 = arr[:C], arr[C:2 * C]
+        im = arr[2 * C:2 * C + M]
+        ci = arr[2 * C + M:4 * C + M]
+        mi = arr[4 * C + M:]
+        slots = np.concatenate([ci, mi])
+        assert np.array_equal(np.sort(slots), np.arange(total))
+        assert np.array_equal(ci[1::2], ci[0::2] + 1)
+        assert ((i1 >= 0) & (i1 < 60)).all() and (i1 != i2).all()
+        assert ((im >= 0) & (im < 60)).all()
+        assert total >= 119
+
+
+def test_prefetch_overflow_path():
+    """mutation-only stream (pm tiny) forces many iterations per child —
+    exercises the rare batched-undershoot tail path."""
+    import torch
+    from dmosopt_amd.moea.variation import _SpawnPrefetch
+
+    rng = np.random.default_rng(2)
+    pf = _SpawnPrefetch()
+    seed_t, C, M, s1, s2, buf = pf.next(rng, 100, 50, 0.02, 0.05, torch.device("cpu"))
+    total = 2 * C + M
+    assert total >= 99
+    arr = buf.numpy()
+    ci = arr[2 * C + M:4 * C + M]
+    mi = arr[4 * C + M:]
+    assert np.array_equal(np.sort(np.concatenate([ci, mi])), np.arange(total))
