@@ -352,3 +352,41 @@ extern "C" void launch_smpso_velocity(const float* position,
                      stream, position, velocity, leader1, leader2, xlb, xub,
                      out, n, d, w, a1, a2, chi);
 }
+
+// Row-wise concat of two blocks and a column sum: trivial kernels that
+// replace at::cat / at::sum dispatches inside nsga2_select — each ATen
+// call costs ~7-10 us of host time in the host-dispatch-bound generation
+// loop; a raw launch costs ~2 us.
+__global__ void cat_rows_kernel(const float* __restrict__ a,
+                                const float* __restrict__ b,
+                                float* __restrict__ out, long long na_elems,
+                                long long total_elems) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total_elems) return;
+  out[i] = (i < na_elems) ? a[i] : b[i - na_elems];
+}
+
+extern "C" void launch_cat_rows(const float* a, const float* b, float* out,
+                                long long na_elems, long long total_elems,
+                                hipStream_t stream) {
+  hipLaunchKernelGGL(cat_rows_kernel,
+                     dim3((int)((total_elems + 255) / 256)), dim3(256), 0,
+                     stream, a, b, out, na_elems, total_elems);
+}
+
+// column sum of an (m, N) matrix -> (N,), fixed accumulation order over m
+// (deterministic; m is the objective count, <= ~10)
+__global__ void colsum_kernel(const float* __restrict__ per_dim,
+                              float* __restrict__ out, int m, int N) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= N) return;
+  float acc = 0.f;
+  for (int j = 0; j < m; ++j) acc += per_dim[(long long)j * N + i];
+  out[i] = acc;
+}
+
+extern "C" void launch_colsum(const float* per_dim, float* out, int m, int N,
+                              hipStream_t stream) {
+  hipLaunchKernelGGL(colsum_kernel, dim3((N + 255) / 256), dim3(256), 0,
+                     stream, per_dim, out, m, N);
+}
