@@ -163,39 +163,52 @@ class _SpawnPrefetch:
                 )
                 events.append((np.concatenate([c_all[g], extra_c]),
                                np.concatenate([m_all[g], extra_m])))
-        Cs = [int(c.sum()) for c, _ in events]
-        Ms = [int(m.sum()) for _, m in events]
-        u_par = rng.random(2 * sum(Cs) + sum(Ms))
+        # pad the ragged event streams to (G, Lmax) and do ALL remaining
+        # index math with flat masks — zero per-generation numpy ops
+        # (row-major flattening preserves generation order)
+        Lmax = max(len(c) for c, _ in events)
+        c_pad = np.zeros((G, Lmax), dtype=bool)
+        m_pad = np.zeros((G, Lmax), dtype=bool)
+        for g, (c_ev, m_ev) in enumerate(events):
+            c_pad[g, : len(c_ev)] = c_ev
+            m_pad[g, : len(m_ev)] = m_ev
+        Cs = c_pad.sum(axis=1)
+        Ms = m_pad.sum(axis=1)
+        nC, nM = int(Cs.sum()), int(Ms.sum())
+        u_par = rng.random(2 * nC + nM)
         seeds = rng.integers(0, 2**62, (G, 3))
-        metas = []
-        parts = []
-        up = 0
-        for g in range(G):
-            c_ev, m_ev = events[g]
-            C, M = Cs[g], Ms[g]
-            u = u_par[up : up + 2 * C + M]; up += 2 * C + M
-            if C:
-                i1 = (u[:C] * poolsize).astype(np.int64)
-                i2 = (u[C : 2 * C] * (poolsize - 1)).astype(np.int64)
-                i2 = i2 + (i2 >= i1)
-            else:
-                i1 = i2 = np.empty(0, dtype=np.int64)
-            im = (u[2 * C :] * poolsize).astype(np.int64) if M else np.empty(0, dtype=np.int64)
-            it_sizes = 2 * c_ev + m_ev
-            it_starts = np.cumsum(it_sizes) - it_sizes
-            ci = np.repeat(it_starts[c_ev], 2)
-            ci[1::2] += 1
-            mi = (it_starts + 2 * c_ev)[m_ev]
-            combined = np.concatenate([i1, i2, im, ci, mi])
-            metas.append((int(seeds[g, 0]), C, M, int(seeds[g, 1]),
-                          int(seeds[g, 2]), combined.shape[0]))
-            parts.append(combined)
-        dev = _to_device_pinned(np.concatenate(parts), device)
+        i1 = (u_par[:nC] * poolsize).astype(np.int64)
+        i2 = (u_par[nC : 2 * nC] * (poolsize - 1)).astype(np.int64)
+        i2 = i2 + (i2 >= i1)
+        im = (u_par[2 * nC :] * poolsize).astype(np.int64)
+        it_sizes = 2 * c_pad + m_pad
+        it_starts = np.cumsum(it_sizes, axis=1) - it_sizes
+        cstart = it_starts[c_pad]              # all gens' crossover starts
+        ci = np.repeat(cstart, 2)
+        ci[1::2] += 1
+        mi = (it_starts + 2 * c_pad)[m_pad]    # all gens' mutation slots
+        dev = _to_device_pinned(
+            np.concatenate([i1, i2, im, ci, mi]), device
+        )
+        # component group offsets within dev
+        o_i2 = nC
+        o_im = 2 * nC
+        o_ci = 2 * nC + nM
+        o_mi = o_ci + 2 * nC
+        coff = np.concatenate([[0], np.cumsum(Cs)])
+        moff = np.concatenate([[0], np.cumsum(Ms)])
         items = []
-        base = 0
-        for seed_t, C, M, s1, s2, ln in metas:
-            items.append((seed_t, C, M, s1, s2, dev[base : base + ln]))
-            base += ln
+        for g in range(G):
+            C0, C1 = int(coff[g]), int(coff[g + 1])
+            M0, M1 = int(moff[g]), int(moff[g + 1])
+            items.append((
+                int(seeds[g, 0]), C1 - C0, M1 - M0, int(seeds[g, 1]),
+                int(seeds[g, 2]),
+                (dev[C0:C1], dev[o_i2 + C0 : o_i2 + C1],
+                 dev[o_im + M0 : o_im + M1],
+                 dev[o_ci + 2 * C0 : o_ci + 2 * C1],
+                 dev[o_mi + M0 : o_mi + M1]),
+            ))
         items.reverse()  # pop() yields generation order
         self.items = items
 
@@ -231,22 +244,23 @@ def spawn_generation_native(
     from dmosopt_amd import _hipops
 
     if prefetch is not None:
-        seed_t, C, M, seed_sbx, seed_mut, dev = prefetch.next(
+        seed_t, C, M, seed_sbx, seed_mut, views = prefetch.next(
             rng, popsize, poolsize, crossover_prob, mutation_prob,
             population.device,
         )
+        i1_t, i2_t, im_t, c_idx_t, m_idx_t = views
     else:
         seed_t = int(rng.integers(0, 2**62))
         combined, C, M, seed_sbx, seed_mut = _assemble_event_indices(
             rng, popsize, poolsize, crossover_prob, mutation_prob
         )
         dev = _to_device_pinned(combined, population.device)
-    o = 0
-    i1_t = dev[o : o + C]; o += C
-    i2_t = dev[o : o + C]; o += C
-    im_t = dev[o : o + M]; o += M
-    c_idx_t = dev[o : o + 2 * C]; o += 2 * C
-    m_idx_t = dev[o : o + M]
+        o = 0
+        i1_t = dev[o : o + C]; o += C
+        i2_t = dev[o : o + C]; o += C
+        im_t = dev[o : o + M]; o += M
+        c_idx_t = dev[o : o + 2 * C]; o += 2 * C
+        m_idx_t = dev[o : o + M]
     x_gen = _hipops.generation_spawn(
         population.float().contiguous(), rank.long().contiguous(), poolsize,
         float(p_sel), seed_t, c_idx_t, m_idx_t, i1_t, i2_t, im_t,

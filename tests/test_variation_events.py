@@ -81,13 +81,10 @@ def test_prefetch_batch_matches_invariants():
     pf = _SpawnPrefetch()
     dev = torch.device("cpu")
     for _ in range(20):  # > one chunk: exercises refill twice
-        seed_t, C, M, s1, s2, buf = pf.next(rng, 120, 60, 0.9, 0.1, dev)
+        seed_t, C, M, s1, s2, views = pf.next(rng, 120, 60, 0.9, 0.1, dev)
         total = 2 * C + M
-        arr = buf.numpy()
-        i1, i2 = arr[:C], arr[C:2 * C]
-        im = arr[2 * C:2 * C + M]
-        ci = arr[2 * C + M:4 * C + M]
-        mi = arr[4 * C + M:]
+        i1, i2, im, ci, mi = (v.numpy() for v in views)
+        assert len(i1) == C and len(i2) == C and len(im) == M
         slots = np.concatenate([ci, mi])
         assert np.array_equal(np.sort(slots), np.arange(total))
         assert np.array_equal(ci[1::2], ci[0::2] + 1)
@@ -104,10 +101,8 @@ def test_prefetch_overflow_path():
 
     rng = np.random.default_rng(2)
     pf = _SpawnPrefetch()
-    seed_t, C, M, s1, s2, buf = pf.next(rng, 100, 50, 0.02, 0.05, torch.device("cpu"))
+    seed_t, C, M, s1, s2, views = pf.next(rng, 100, 50, 0.02, 0.05, torch.device("cpu"))
     total = 2 * C + M
     assert total >= 99
-    arr = buf.numpy()
-    ci = arr[2 * C + M:4 * C + M]
-    mi = arr[4 * C + M:]
+    ci, mi = views[3].numpy(), views[4].numpy()
     assert np.array_equal(np.sort(np.concatenate([ci, mi])), np.arange(total))
