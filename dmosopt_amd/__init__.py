@@ -7,6 +7,19 @@ for the hot population/GP math, RCCL collectives over xGMI for distribution.
 
 __version__ = "0.1.0"
 
+import os as _os
+
+# Run-to-run bit determinism on the CPU path: torch's CPU backend is MKL,
+# and MKL_DYNAMIC (default TRUE) lets MKL SHRINK its thread team when the
+# machine is loaded — a different team size changes GEMM/reduction
+# partitioning, the CPU GP fit's floats move by ULPs, SCE-UA accept
+# decisions flip, and two same-seed runs diverge (observed ~8% of
+# processes under load). The replicated-control-flow multi-rank scheme
+# and the seeded-reproducibility contract both require fixed teams.
+# setdefault: a user who explicitly set these keeps their choice.
+_os.environ.setdefault("MKL_DYNAMIC", "FALSE")
+_os.environ.setdefault("OMP_DYNAMIC", "FALSE")
+
 from dmosopt_amd.datatypes import ParameterSpace  # noqa: F401
 
 sopt_dict = {}

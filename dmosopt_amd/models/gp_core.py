@@ -16,10 +16,34 @@ log marginal likelihood (reference model.py:1182-1275, 1419-1753).
 
 from __future__ import annotations
 
+import contextlib
 import math
 from typing import Optional, Tuple
 
 import torch
+
+
+@contextlib.contextmanager
+def _single_threaded_cpu_math(enabled: bool):
+    """Pin torch to ONE intra-op thread for a CPU numerics region.
+
+    MKL's threaded GEMM/LAPACK paths are not bit-reproducible run-to-run
+    when the effective team or partitioning shifts (observed: same-seed
+    same-process GP fits diverging at ULP level under machine load, which
+    flips SCE-UA accept decisions and desynchronizes the replicated
+    multi-rank control flow). Sequential execution is deterministic; the
+    CPU fit trades some throughput for the reproducibility contract. The
+    GPU path is unaffected (device kernels are deterministic by
+    construction)."""
+    if not enabled:
+        yield
+        return
+    n0 = torch.get_num_threads()
+    torch.set_num_threads(1)
+    try:
+        yield
+    finally:
+        torch.set_num_threads(n0)
 
 SQRT5 = math.sqrt(5.0)
 SQRT3 = math.sqrt(3.0)
@@ -184,57 +208,58 @@ def batched_nmll(
     points). ``differentiable=True`` forces the autograd-capable torch path
     (Adam optimizer); the default dispatches to the fused gfx950 kernels.
     """
-    N = X.shape[0]
-    B = theta.shape[0]
+    with _single_threaded_cpu_math(not X.is_cuda):
+        N = X.shape[0]
+        B = theta.shape[0]
 
-    def _yb():
-        return (
-            y[None, :, None].expand(B, N, 1) if y.dim() == 1 else y[:, :, None]
-        ).contiguous()
+        def _yb():
+            return (
+                y[None, :, None].expand(B, N, 1) if y.dim() == 1 else y[:, :, None]
+            ).contiguous()
 
-    if differentiable:
-        yb = _yb()
-        K = build_kernel_torch(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
-        L, info = torch.linalg.cholesky_ex(K)
-        alpha = torch.cholesky_solve(yb, L)
-        quad = (yb * alpha).sum(dim=(1, 2))
-        logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(dim=-1)
-    else:
-        from dmosopt_amd import ops
+        if differentiable:
+            yb = _yb()
+            K = build_kernel_torch(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
+            L, info = torch.linalg.cholesky_ex(K)
+            alpha = torch.cholesky_solve(yb, L)
+            quad = (yb * alpha).sum(dim=(1, 2))
+            logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(dim=-1)
+        else:
+            from dmosopt_amd import ops
 
-        if (
-            X.is_cuda
-            and X.dtype == torch.float32
-            and ops.native_available()
-            and _nmll_graph_enabled()
-        ):
-            key = (
-                B, N, X.shape[1], theta.shape[1], y.dim(), nu, anisotropic,
-                float(jitter),
-            )
-            g = _nmll_graphs.get(key)
-            if g is None:
-                if len(_nmll_graphs) >= 8:  # archives grow: drop stale shapes
-                    _nmll_graphs.clear()
-                g = _nmll_graphs[key] = _NmllGraph(
-                    X.float(), y.float(), theta.float(), nu, anisotropic, jitter
+            if (
+                X.is_cuda
+                and X.dtype == torch.float32
+                and ops.native_available()
+                and _nmll_graph_enabled()
+            ):
+                key = (
+                    B, N, X.shape[1], theta.shape[1], y.dim(), nu, anisotropic,
+                    float(jitter),
                 )
-            return g.run(X.float(), y.float(), theta.float())
-        fused = ops.gp_nmll_fused(
-            X, theta, y if y.dim() == 2 else y, nu, anisotropic, jitter
-        )
-        if fused is not None:
-            return fused
-        K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
-        L, half_logdet, info = ops.chol_factor_batched(K)
-        # both paths return sum(log diag L)
-        logdet = 2.0 * half_logdet
-        z = ops.tri_solve_forward(L, _yb())  # L z = y
-        quad = (z * z).sum(dim=(1, 2))
-    nmll = 0.5 * quad + 0.5 * logdet + 0.5 * N * LOG2PI
-    nmll = torch.where(info != 0, torch.full_like(nmll, float("inf")), nmll)
-    nmll = torch.where(torch.isfinite(nmll), nmll, torch.full_like(nmll, float("inf")))
-    return nmll
+                g = _nmll_graphs.get(key)
+                if g is None:
+                    if len(_nmll_graphs) >= 8:  # archives grow: drop stale shapes
+                        _nmll_graphs.clear()
+                    g = _nmll_graphs[key] = _NmllGraph(
+                        X.float(), y.float(), theta.float(), nu, anisotropic, jitter
+                    )
+                return g.run(X.float(), y.float(), theta.float())
+            fused = ops.gp_nmll_fused(
+                X, theta, y if y.dim() == 2 else y, nu, anisotropic, jitter
+            )
+            if fused is not None:
+                return fused
+            K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
+            L, half_logdet, info = ops.chol_factor_batched(K)
+            # both paths return sum(log diag L)
+            logdet = 2.0 * half_logdet
+            z = ops.tri_solve_forward(L, _yb())  # L z = y
+            quad = (z * z).sum(dim=(1, 2))
+        nmll = 0.5 * quad + 0.5 * logdet + 0.5 * N * LOG2PI
+        nmll = torch.where(info != 0, torch.full_like(nmll, float("inf")), nmll)
+        nmll = torch.where(torch.isfinite(nmll), nmll, torch.full_like(nmll, float("inf")))
+        return nmll
 
 
 class FittedGP:
@@ -268,6 +293,12 @@ class FittedGP:
         from dmosopt_amd import ops
 
         m, N = theta.shape[0], X.shape[0]
+        with _single_threaded_cpu_math(not X.is_cuda):
+            self._build_posterior(X, Y, theta, y_mean, y_std, nu, anisotropic,
+                                  jitter, ops, m, N)
+
+    def _build_posterior(self, X, Y, theta, y_mean, y_std, nu, anisotropic,
+                         jitter, ops, m, N):
         K = build_kernel(X, None, theta, nu=nu, anisotropic=anisotropic, jitter=jitter)
         if self.compute == "bf16":
             self.L, _, info = ops.chol_factor_batched_bf16(K)
@@ -318,12 +349,28 @@ class FittedGP:
             Ks = ops.matern_cross_bf16_kernel(
                 Xq.float(), self.X, self.theta, self.nu, self.anisotropic
             )
+        elif not Xq.is_cuda:
+            # deterministic CPU math (thread-team-independent): the
+            # replicated-rank scheme compares these values across processes
+            with _single_threaded_cpu_math(True):
+                return self._predict_torch(Xq, return_var)
         else:
             Ks = build_kernel(Xq, self.X, self.theta, nu=self.nu, anisotropic=self.anisotropic)
         # (m, P, N)
         mean_n = torch.bmm(Ks, self.alpha)[:, :, 0]  # (m, P)
         if not return_var:
             return self.y_mean[None, :] + self.y_std[None, :] * mean_n.T, None
+        return self._finish_var(Xq, Ks, mean_n)
+
+    def _predict_torch(self, Xq, return_var):
+        Ks = build_kernel(Xq, self.X, self.theta, nu=self.nu,
+                          anisotropic=self.anisotropic)
+        mean_n = torch.bmm(Ks, self.alpha)[:, :, 0]  # (m, P)
+        if not return_var:
+            return self.y_mean[None, :] + self.y_std[None, :] * mean_n.T, None
+        return self._finish_var(Xq, Ks, mean_n)
+
+    def _finish_var(self, Xq, Ks, mean_n):
         sf2 = torch.exp(self.theta[:, 0])
         noise = torch.exp(self.theta[:, -1])
         kss = (sf2 + noise)[:, None]  # (m, 1): k(x,x) = sf2*1 + noise
