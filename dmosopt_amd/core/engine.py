@@ -455,7 +455,10 @@ def _cross_duplicates(X: torch.Tensor, Y: torch.Tensor, eps: float = 1e-16) -> n
     cross-matrices is not reproduced)."""
     if X.numel() == 0 or Y.numel() == 0:
         return np.zeros(X.shape[0], dtype=bool)
-    D = torch.cdist(X.double(), Y.double())
+    # direct-difference mode: exact-duplicate detection at eps=1e-16 must
+    # not depend on GEMM ULP noise (see ops/torch_ref.get_duplicates)
+    D = torch.cdist(X.double(), Y.double(),
+                    compute_mode="donot_use_mm_for_euclid_dist")
     D = torch.nan_to_num(D, nan=float("inf"))
     return (D <= eps).any(dim=1).cpu().numpy()
 

@@ -721,10 +721,13 @@ int64_t hv_fpras_hits(torch::Tensor P, torch::Tensor ref, torch::Tensor cdf,
 
 torch::Tensor get_duplicates(torch::Tensor X, double eps) {
   CHECK_GPU(X);
-  // distance-based duplicate mask on device via torch primitives (cdist is
-  // a rocBLAS GEMM under the hood); semantics of MOEA.py:426-436
+  // distance-based duplicate mask on device via torch primitives;
+  // semantics of MOEA.py:426-436. compute_mode 1 = direct differences:
+  // the GEMM (x^2-2xy+y^2) formulation leaves exact duplicates at
+  // norm-vs-dot rounding residuals that dwarf the eps=1e-16 threshold
+  // in fp32 — the difference path is exact at zero distance.
   const int n = X.size(0);
-  auto D = torch::cdist(X, X);
+  auto D = torch::cdist(X, X, 2.0, 1);
   auto iu = torch::triu_indices(n, n, 0, torch::TensorOptions()
                                               .dtype(torch::kLong)
                                               .device(X.device()));

@@ -274,7 +274,13 @@ def get_duplicates(X: Tensor, eps: float = 1e-16) -> Tensor:
     n = X.shape[0]
     if n == 0:
         return torch.zeros(0, dtype=torch.bool, device=X.device)
-    D = torch.cdist(X.double(), X.double())
+    # donot_use_mm: the GEMM (x^2-2xy+y^2) formulation leaves exact
+    # duplicates at ~eps-sized residuals whose ULPs depend on the BLAS
+    # threading/partitioning of the moment — the eps=1e-16 threshold then
+    # flips run to run (observed: same-seed archives diverging). The
+    # direct-difference path is deterministic and exact at zero distance.
+    D = torch.cdist(X.double(), X.double(),
+                    compute_mode="donot_use_mm_for_euclid_dist")
     iu = torch.triu_indices(n, n, offset=0, device=X.device)
     D[iu[0], iu[1]] = float("inf")
     D = torch.nan_to_num(D, nan=float("inf"))
