@@ -349,10 +349,22 @@ def run_epoch(
                 logger.warning(f"Unable to fit feasibility model: {e}")
 
     if surrogate_method_name is not None and mdl.objective is None:
+        smk = dict(surrogate_method_kwargs or {})
+        if smk.get("seed") is None and local_random is not None:
+            # the surrogate's OWN hyperparameter search (SCE-UA / Adam init)
+            # must inherit determinism from the run seed: the reference
+            # leaves it unseeded (model.py:1192 seed=None default -> OS
+            # entropy), which silently breaks same-seed reproducibility —
+            # two same-seed runs fit slightly different thetas and ~10% of
+            # the time the resample selection flips. Drawing the seed from
+            # local_random is replicated identically on every rank (all
+            # ranks execute this line), so the theta-broadcast scheme and
+            # the per-epoch archive hash guard stay consistent.
+            smk["seed"] = int(local_random.integers(0, 2**31 - 1))
         mdl.objective = train(
             nInput, nOutput, xlb, xub, Xinit, Yinit, C,
             surrogate_method_name=surrogate_method_name,
-            surrogate_method_kwargs=surrogate_method_kwargs,
+            surrogate_method_kwargs=smk,
             surrogate_return_mean_variance=optimize_mean_variance,
             logger=logger, file_path=file_path, device=device,
         )
